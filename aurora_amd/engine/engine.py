@@ -1,0 +1,493 @@
+"""The in-process LLM engine: continuous batching over a paged KV cache.
+
+This replaces the reference's provider-API boundary (reference:
+server/chat/backend/agent/llm.py:192 LLMManager.invoke — an HTTP call to a
+rented model) with an engine the agent loop steps directly.  Concurrency
+that the reference buys with Celery worker processes (celery_config.py:66-148)
+is in-engine here: every concurrent investigation is a sequence in one
+continuously-batched decode step.
+
+Design (MI355X-first):
+* packed varlen everywhere — no [B, S] padding;
+* chunked prefill: new sequences stream into the batch without stalling
+  running decodes; append-prefill reuses the session's cached KV pages
+  (the in-engine analog of the reference's prompt-prefix cache,
+  utils/prefix_cache.py:39-158);
+* decode steps batch every running sequence into one kernel launch set;
+* sampling supports grammar-guided tool-JSON decoding (engine/grammar.py).
+"""
+from __future__ import annotations
+
+import time
+from collections import OrderedDict, deque
+from dataclasses import dataclass, field
+from typing import Deque, Dict, List, Optional, Tuple
+
+import torch
+
+from ..models.config import ModelSpec, get_spec
+from ..models.llama import AttnMetadata, LlamaForCausalLM
+from ..parallel import dist as pdist
+from .kv_cache import PagedKVCache
+from .sampling_params import SamplingParams
+from .tokenizer import Tokenizer, get_tokenizer
+
+WAITING, RUNNING, FINISHED = "waiting", "running", "finished"
+
+
+@dataclass
+class EngineConfig:
+    model: str = "tiny-llama"
+    dtype: str = "bfloat16"
+    device: str = "auto"  # "auto" -> cuda:local_rank if available else cpu
+    page_size: int = 0  # 0 -> 256 on GPU, 16 on CPU
+    kv_cache_pages: int = 0  # 0 -> sized from kv_memory_fraction (GPU) / 512 pages (CPU)
+    kv_memory_fraction: float = 0.55
+    max_num_seqs: int = 64
+    max_batched_tokens: int = 8192
+    max_model_len: int = 32768
+    tp_size: int = 1
+    tokenizer_path: Optional[str] = None
+    seed: int = 0
+    grammar_topk: int = 32  # candidates tried per step under a grammar
+
+
+@dataclass
+class StepOutput:
+    request_id: str
+    new_token_ids: List[int]
+    new_text: str
+    finished: bool
+    finish_reason: Optional[str] = None
+
+
+class Sequence:
+    def __init__(self, req_id: str, token_ids: List[int], params: SamplingParams,
+                 session_id: Optional[str] = None):
+        self.req_id = req_id
+        self.session_id = session_id
+        self.token_ids: List[int] = list(token_ids)
+        self.num_prompt = len(token_ids)
+        self.num_cached = 0  # tokens whose KV is already in cache pages
+        self.pages: List[int] = []
+        self.params = params
+        self.status = WAITING
+        self.finish_reason: Optional[str] = None
+        self.out_bytes = bytearray()
+        self.grammar = params.grammar.fork() if params.grammar is not None else None
+        self.arrival = time.monotonic()
+        self.first_token_time: Optional[float] = None
+
+    @property
+    def num_tokens(self) -> int:
+        return len(self.token_ids)
+
+    @property
+    def num_output(self) -> int:
+        return len(self.token_ids) - self.num_prompt
+
+    @property
+    def prefill_target(self) -> int:
+        """Tokens whose KV must be cached before this seq can decode.
+        A preempted seq that already generated tokens re-prefills everything
+        but its last token (which becomes the next decode input — no
+        re-sampling)."""
+        return self.num_prompt if self.num_output == 0 else self.num_tokens - 1
+
+
+class LLMEngine:
+    def __init__(self, config: EngineConfig, model: Optional[LlamaForCausalLM] = None,
+                 tokenizer: Optional[Tokenizer] = None):
+        self.config = config
+        self.spec: ModelSpec = get_spec(config.model)
+        use_cuda = torch.cuda.is_available() and config.device != "cpu"
+        if config.device in ("auto", "cuda") and use_cuda:
+            self.device = f"cuda:{torch.cuda.current_device()}"
+        else:
+            self.device = "cpu"
+        self.dtype = getattr(torch, config.dtype) if self.device != "cpu" else torch.float32
+        pdist.init_distributed(tp_size=config.tp_size)
+        self.tokenizer = tokenizer or get_tokenizer(config.tokenizer_path)
+
+        if model is None:
+            model = LlamaForCausalLM(self.spec, dtype=self.dtype, device=self.device)
+            model.init_random(config.seed)
+        self.model = model.eval()
+
+        page_size = config.page_size or (256 if self.device != "cpu" else 16)
+        tp = pdist.tp_world_size()
+        kv_heads_local = max(self.spec.n_kv_heads // tp, 1)
+        if config.kv_cache_pages:
+            pages = config.kv_cache_pages
+        elif self.device != "cpu":
+            free, total = torch.cuda.mem_get_info()
+            budget = int(free * config.kv_memory_fraction)
+            pages = PagedKVCache.pages_for_memory(
+                budget, self.spec.n_layers, kv_heads_local, self.spec.head_dim,
+                page_size, 2 if self.dtype != torch.float32 else 4,
+            )
+        else:
+            pages = 512
+        self.kv = PagedKVCache(
+            self.spec.n_layers, kv_heads_local, self.spec.head_dim,
+            page_size=page_size, num_pages=pages, device=self.device,
+            dtype=self.dtype,
+        )
+        self.max_model_len = min(config.max_model_len, self.spec.max_position)
+
+        self.waiting: Deque[Sequence] = deque()
+        self.running: List[Sequence] = []
+        self.seqs: Dict[str, Sequence] = {}
+        # session KV reuse: session_id -> finished Sequence kept warm in cache
+        self.sessions: "OrderedDict[str, Sequence]" = OrderedDict()
+        self.gen = torch.Generator(device="cpu").manual_seed(config.seed)
+        # counters (engine observability; surfaced by service.metrics)
+        self.counters = {
+            "prompt_tokens": 0, "generated_tokens": 0, "prefill_steps": 0,
+            "decode_steps": 0, "preemptions": 0, "grammar_fallbacks": 0,
+        }
+
+    # ------------------------------------------------------------------ API
+
+    def add_request(self, request_id: str, prompt_token_ids: List[int],
+                    params: SamplingParams, session_id: Optional[str] = None) -> None:
+        if request_id in self.seqs:
+            raise ValueError(f"duplicate request id {request_id}")
+        if len(prompt_token_ids) >= self.max_model_len:
+            prompt_token_ids = prompt_token_ids[-(self.max_model_len - params.max_tokens - 1):]
+        seq = Sequence(request_id, prompt_token_ids, params, session_id)
+        if session_id is not None and session_id in self.sessions:
+            cached = self.sessions.pop(session_id)
+            match = 0
+            limit = min(cached.num_cached, seq.num_prompt - 1)
+            while match < limit and cached.token_ids[match] == seq.token_ids[match]:
+                match += 1
+            # adopt whole pages covered by the matching prefix
+            keep_pages = match // self.kv.page_size
+            if keep_pages > 0:
+                seq.pages = cached.pages[:keep_pages]
+                self.kv.free(cached.pages[keep_pages:])
+                seq.num_cached = keep_pages * self.kv.page_size
+            else:
+                self.kv.free(cached.pages)
+        self.seqs[request_id] = seq
+        self.waiting.append(seq)
+
+    def abort_request(self, request_id: str) -> None:
+        seq = self.seqs.pop(request_id, None)
+        if seq is None:
+            return
+        if seq in self.waiting:
+            self.waiting.remove(seq)
+        if seq in self.running:
+            self.running.remove(seq)
+        self.kv.free(seq.pages)
+        seq.pages = []
+        seq.status = FINISHED
+        seq.finish_reason = "aborted"
+
+    def has_work(self) -> bool:
+        return bool(self.waiting or self.running)
+
+    def num_running(self) -> int:
+        return len(self.running)
+
+    # ------------------------------------------------------------- schedule
+
+    def _evict_sessions(self, need_pages: int) -> None:
+        while self.kv.num_free() < need_pages and self.sessions:
+            _, old = self.sessions.popitem(last=False)
+            self.kv.free(old.pages)
+            old.pages = []
+
+    def _try_reserve(self, seq: Sequence, up_to_tokens: int) -> bool:
+        need = self.kv.pages_needed(up_to_tokens) - len(seq.pages)
+        if need <= 0:
+            return True
+        self._evict_sessions(need)
+        if self.kv.num_free() < need:
+            return False
+        seq.pages.extend(self.kv.alloc(need))
+        return True
+
+    def _preempt_lowest(self, exclude=()) -> bool:
+        """Evict the most recently admitted running seq (not in ``exclude``)
+        back to waiting, freeing its pages for recompute later."""
+        for j in range(len(self.running) - 1, -1, -1):
+            victim = self.running[j]
+            if victim in exclude:
+                continue
+            self.running.pop(j)
+            self.kv.free(victim.pages)
+            victim.pages = []
+            victim.num_cached = 0
+            victim.status = WAITING
+            self.waiting.appendleft(victim)
+            self.counters["preemptions"] += 1
+            return True
+        return False
+
+    def _schedule(self) -> Tuple[str, List[Sequence], List[int]]:
+        """Returns (mode, seqs, chunk_lens). Prefill takes priority so new
+        investigations reach first token fast; decode batches all running."""
+        budget = self.config.max_batched_tokens
+        prefill: List[Sequence] = []
+        chunks: List[int] = []
+        while self.waiting and len(self.running) + len(prefill) < self.config.max_num_seqs:
+            seq = self.waiting.popleft()
+            remaining = seq.prefill_target - seq.num_cached
+            chunk = min(remaining, budget)
+            if chunk <= 0:
+                self.waiting.appendleft(seq)
+                break
+            # Admission never preempts running seqs (that thrashes); decode
+            # growth below does the preempting. If nothing is running at all
+            # and the seq still does not fit, it never will.
+            reserved = self._try_reserve(seq, seq.num_cached + chunk)
+            if not reserved:
+                self.waiting.appendleft(seq)
+                if not prefill and not self.running:
+                    raise MemoryError(
+                        f"sequence {seq.req_id} needs more KV pages than the whole cache "
+                        f"({self.kv.num_pages} pages x {self.kv.page_size} tokens)"
+                    )
+                break
+            prefill.append(seq)
+            chunks.append(chunk)
+            budget -= chunk
+            if budget <= 0:
+                break
+        if prefill:
+            return "prefill", prefill, chunks
+        # decode: every running seq advances one token
+        decode: List[Sequence] = []
+        for seq in list(self.running):
+            if seq not in self.running:  # already preempted as a victim
+                continue
+            ok = self._try_reserve(seq, seq.num_tokens + 1)
+            while not ok:
+                # evict others first; as a last resort evict this seq itself
+                if self._preempt_lowest(exclude=set(decode) | {seq}):
+                    ok = self._try_reserve(seq, seq.num_tokens + 1)
+                else:
+                    self._preempt_lowest(exclude=set(decode))
+                    break
+            if ok:
+                decode.append(seq)
+        return "decode", decode, []
+
+    # -------------------------------------------------------------- tensors
+
+    def _slot(self, seq: Sequence, pos: int) -> int:
+        page = seq.pages[pos // self.kv.page_size]
+        return page * self.kv.page_size + pos % self.kv.page_size
+
+    def _block_table_tensor(self, seqs: List[Sequence]) -> torch.Tensor:
+        maxp = max(len(s.pages) for s in seqs)
+        bt = torch.zeros(len(seqs), maxp, dtype=torch.int32)
+        for i, s in enumerate(seqs):
+            if s.pages:
+                bt[i, : len(s.pages)] = torch.tensor(s.pages, dtype=torch.int32)
+        return bt.to(self.device)
+
+    # ------------------------------------------------------------------ step
+
+    @torch.inference_mode()
+    def step(self) -> List[StepOutput]:
+        mode, seqs, chunks = self._schedule()
+        if not seqs:
+            return []
+        dev = self.device
+        if mode == "prefill":
+            self.counters["prefill_steps"] += 1
+            ids: List[int] = []
+            positions: List[int] = []
+            slots: List[int] = []
+            cu = [0]
+            ctx_lens: List[int] = []
+            for seq, chunk in zip(seqs, chunks):
+                a = seq.num_cached
+                b = a + chunk
+                ids.extend(seq.token_ids[a:b])
+                positions.extend(range(a, b))
+                slots.extend(self._slot(seq, p) for p in range(a, b))
+                cu.append(cu[-1] + chunk)
+                ctx_lens.append(b)
+            meta = AttnMetadata(
+                mode="prefill",
+                positions=torch.tensor(positions, dtype=torch.int32, device=dev),
+                slot_mapping=torch.tensor(slots, dtype=torch.int32, device=dev),
+                block_tables=self._block_table_tensor(seqs),
+                context_lens=torch.tensor(ctx_lens, dtype=torch.int32, device=dev),
+                cu_seqlens_q=torch.tensor(cu, dtype=torch.int32, device=dev),
+            )
+            input_ids = torch.tensor(ids, dtype=torch.long, device=dev)
+            hidden = self.model(input_ids, self.kv.caches, meta)
+            outputs: List[StepOutput] = []
+            sample_rows = []
+            sample_seqs = []
+            for i, (seq, chunk) in enumerate(zip(seqs, chunks)):
+                had_output = seq.num_output > 0
+                seq.num_cached += chunk
+                self.counters["prompt_tokens"] += chunk
+                if seq.num_cached >= seq.prefill_target:
+                    seq.status = RUNNING
+                    if seq not in self.running:
+                        self.running.append(seq)
+                    if not had_output:  # preempted seqs resume decode, no re-sample
+                        sample_rows.append(cu[i + 1] - 1)
+                        sample_seqs.append(seq)
+                else:
+                    self.waiting.appendleft(seq)  # more chunks to go
+            if sample_seqs:
+                rows = hidden[torch.tensor(sample_rows, dtype=torch.long, device=dev)]
+                logits = self.model.compute_logits(rows)
+                outputs.extend(self._sample_and_append(sample_seqs, logits))
+            return outputs
+
+        # decode
+        self.counters["decode_steps"] += 1
+        ids = [s.token_ids[-1] for s in seqs]
+        positions = [s.num_tokens - 1 for s in seqs]
+        # note: the last token was appended by sampling but its KV is not yet
+        # written; it is this step's input token.
+        slots = [self._slot(s, s.num_tokens - 1) for s in seqs]
+        ctx_lens = [s.num_tokens for s in seqs]
+        meta = AttnMetadata(
+            mode="decode",
+            positions=torch.tensor(positions, dtype=torch.int32, device=dev),
+            slot_mapping=torch.tensor(slots, dtype=torch.int32, device=dev),
+            block_tables=self._block_table_tensor(seqs),
+            context_lens=torch.tensor(ctx_lens, dtype=torch.int32, device=dev),
+        )
+        input_ids = torch.tensor(ids, dtype=torch.long, device=dev)
+        hidden = self.model(input_ids, self.kv.caches, meta)
+        logits = self.model.compute_logits(hidden)
+        for s in seqs:
+            s.num_cached = s.num_tokens
+        return self._sample_and_append(seqs, logits)
+
+    # --------------------------------------------------------------- sample
+
+    def _sample_row(self, seq: Sequence, row: torch.Tensor) -> int:
+        p = seq.params
+        if seq.grammar is None:
+            return self._plain_sample(row, p)
+        # grammar-guided: walk candidates in probability order
+        k = min(self.config.grammar_topk, row.numel())
+        if p.temperature > 0:
+            probs = (row / p.temperature).softmax(-1)
+            cand = torch.multinomial(
+                probs.cpu(), num_samples=k, replacement=False, generator=self.gen
+            ).tolist()
+        else:
+            cand = row.topk(k).indices.tolist()
+        for tok in cand:
+            b = self.tokenizer.token_bytes(tok)
+            if tok == self.tokenizer.eos_id and seq.grammar.is_complete():
+                return tok
+            g = seq.grammar.fork()
+            if b and g.advance(b):
+                seq.grammar = g
+                return tok
+            if not b:
+                continue
+        # fallback: full-vocab scan in logit order
+        self.counters["grammar_fallbacks"] += 1
+        order = row.argsort(descending=True).tolist()
+        for tok in order:
+            b = self.tokenizer.token_bytes(tok)
+            if not b:
+                continue
+            g = seq.grammar.fork()
+            if g.advance(b):
+                seq.grammar = g
+                return tok
+        return self.tokenizer.eos_id
+
+    def _plain_sample(self, row: torch.Tensor, p: SamplingParams) -> int:
+        if p.temperature <= 0:
+            return int(row.argmax())
+        r = row / p.temperature
+        V = r.numel()
+        if p.top_k and p.top_k < V:
+            kth = r.topk(p.top_k).values[-1]
+            r = r.masked_fill(r < kth, float("-inf"))
+        if p.top_p < 1.0:
+            sr, si = r.sort(descending=True)
+            probs = sr.softmax(-1)
+            cum = probs.cumsum(-1)
+            cut = cum - probs >= p.top_p
+            sr = sr.masked_fill(cut, float("-inf"))
+            r = torch.full_like(r, float("-inf")).scatter(0, si, sr)
+        probs = r.softmax(-1)
+        return int(torch.multinomial(probs.cpu(), 1, generator=self.gen))
+
+    def _sample_and_append(self, seqs: List[Sequence], logits: torch.Tensor) -> List[StepOutput]:
+        outputs = []
+        lf = logits.float().cpu()
+        now = time.monotonic()
+        for i, seq in enumerate(seqs):
+            tok = self._sample_row(seq, lf[i])
+            seq.token_ids.append(tok)
+            if seq.first_token_time is None:
+                seq.first_token_time = now
+            self.counters["generated_tokens"] += 1
+            tb = self.tokenizer.token_bytes(tok)
+            seq.out_bytes.extend(tb)
+            new_text = tb.decode("utf-8", errors="ignore")
+            finish = None
+            p = seq.params
+            if tok == self.tokenizer.eos_id and not p.ignore_eos:
+                finish = "eos"
+            elif tok in p.stop_token_ids:
+                finish = "stop_token"
+            elif seq.grammar is not None and seq.grammar.is_complete():
+                finish = "grammar"
+            elif seq.num_output >= p.max_tokens:
+                finish = "max_tokens"
+            elif p.stop:
+                tail = seq.out_bytes[-256:].decode("utf-8", errors="ignore")
+                for s in p.stop:
+                    if s in tail:
+                        finish = "stop"
+                        break
+            elif seq.num_tokens >= self.max_model_len:
+                finish = "max_len"
+            if finish:
+                self._finish(seq, finish)
+            outputs.append(StepOutput(seq.req_id, [tok], new_text, bool(finish), finish))
+        return outputs
+
+    def _finish(self, seq: Sequence, reason: str) -> None:
+        seq.status = FINISHED
+        seq.finish_reason = reason
+        if seq in self.running:
+            self.running.remove(seq)
+        self.seqs.pop(seq.req_id, None)
+        if seq.session_id is not None:
+            seq.num_cached = seq.num_tokens - 1  # last token's KV not written
+            old = self.sessions.pop(seq.session_id, None)
+            if old is not None:
+                self.kv.free(old.pages)
+            self.sessions[seq.session_id] = seq
+        else:
+            self.kv.free(seq.pages)
+            seq.pages = []
+
+    # --------------------------------------------------------- convenience
+
+    def generate(self, prompts: List[List[int]], params: SamplingParams,
+                 max_steps: int = 100000) -> List[List[int]]:
+        """Blocking batch generate (tests / bench)."""
+        for i, p in enumerate(prompts):
+            self.add_request(f"gen-{id(prompts)}-{i}", p, params)
+        done: Dict[str, List[int]] = {}
+        outs: Dict[str, List[int]] = {f"gen-{id(prompts)}-{i}": [] for i in range(len(prompts))}
+        steps = 0
+        while self.has_work() and steps < max_steps:
+            for out in self.step():
+                if out.request_id in outs:
+                    outs[out.request_id].extend(out.new_token_ids)
+            steps += 1
+        return [outs[f"gen-{id(prompts)}-{i}"] for i in range(len(prompts))]

@@ -1,0 +1,233 @@
+"""Byte-level constrained decoding for tool-call JSON.
+
+The reference gets well-formed tool calls for free from provider APIs
+(reference: server/chat/backend/agent/agent.py:871-1047 — the SDK enforces
+the schema); an in-process engine must enforce the grammar during sampling
+(SURVEY.md §7 hard part 2).  This module is an incremental byte-level JSON
+validator; the sampler uses it as a filter: propose top-k candidates, keep
+the first whose bytes keep the FSM valid (guided rejection sampling — exact
+for a byte tokenizer, a high-probability guide for BPE).
+"""
+from __future__ import annotations
+
+import copy
+from typing import List, Optional
+
+# parser states
+_VALUE = 0        # expecting the start of a value
+_STRING = 1
+_STR_ESC = 2
+_STR_U = 3        # \uXXXX, counts hex digits via _u_left
+_NUM_INT = 4      # in integer part
+_NUM_FRAC = 5
+_NUM_EXP = 6
+_LITERAL = 7      # inside true/false/null
+_OBJ_KEY_OR_END = 8
+_OBJ_KEY = 9      # key string started (shares _STRING via _ctx)
+_OBJ_COLON = 10
+_OBJ_COMMA_OR_END = 11
+_ARR_COMMA_OR_END = 12
+_DONE = 13
+_OBJ_KEY_REQUIRED = 14  # after ',' in an object: '}' illegal
+
+_WS = b" \t\n\r"
+_DIGITS = b"0123456789"
+
+
+class JsonGrammar:
+    """Incremental JSON validator; max_depth bounds container nesting."""
+
+    def __init__(self, max_depth: int = 16, require_object: bool = True):
+        self.max_depth = max_depth
+        self.stack: List[int] = []  # b'{' or b'[' markers (as ints)
+        self.state = _VALUE
+        self.lit: bytes = b""
+        self.lit_pos = 0
+        self.u_left = 0
+        self.in_key = False
+        self.require_object = require_object
+        self.started = False
+        self.num_had_digit = False
+        self.arr_fresh = False  # directly after '[' (so ']' is legal, ',]' is not)
+
+    def fork(self) -> "JsonGrammar":
+        g = JsonGrammar.__new__(JsonGrammar)
+        g.__dict__.update(self.__dict__)
+        g.stack = list(self.stack)
+        return g
+
+    def is_complete(self) -> bool:
+        return self.state == _DONE or (
+            self.started and not self.stack and self.state in (_NUM_INT, _NUM_FRAC, _NUM_EXP)
+            and self.num_had_digit
+        )
+
+    def advance(self, data: bytes) -> bool:
+        for b in data:
+            if not self._advance_byte(b):
+                return False
+        return True
+
+    # -- single byte -------------------------------------------------------
+
+    def _pop_value_done(self) -> bool:
+        if not self.stack:
+            self.state = _DONE
+        elif self.stack[-1] == 0x7B:  # {
+            self.state = _OBJ_COMMA_OR_END
+        else:
+            self.state = _ARR_COMMA_OR_END
+        return True
+
+    def _start_value(self, b: int) -> bool:
+        if b in _WS:
+            return True
+        self.started = True
+        if b == 0x5D and self.arr_fresh and self.stack and self.stack[-1] == 0x5B:
+            # empty array "[]" (']' legal only directly after '[')
+            self.arr_fresh = False
+            self.stack.pop()
+            return self._pop_value_done()
+        self.arr_fresh = False
+        if b == 0x22:  # "
+            self.state = _STRING
+            self.in_key = False
+            return True
+        if b == 0x7B:  # {
+            if len(self.stack) >= self.max_depth:
+                return False
+            self.stack.append(0x7B)
+            self.state = _OBJ_KEY_OR_END
+            return True
+        if b == 0x5B:  # [
+            if len(self.stack) >= self.max_depth:
+                return False
+            self.stack.append(0x5B)
+            self.state = _VALUE
+            self.arr_fresh = True
+            return True
+        if b in _DIGITS or b == 0x2D:  # digit or -
+            self.state = _NUM_INT
+            self.num_had_digit = b != 0x2D
+            return True
+        for lit in (b"true", b"false", b"null"):
+            if b == lit[0]:
+                self.lit, self.lit_pos, self.state = lit, 1, _LITERAL
+                return True
+        return False
+
+    def _advance_byte(self, b: int) -> bool:
+        s = self.state
+        if s == _DONE:
+            return b in _WS
+        if s == _VALUE:
+            if self.require_object and not self.started and b not in _WS:
+                if b != 0x7B:
+                    return False
+            return self._start_value(b)
+        if s in (_STRING,):
+            if b == 0x22:
+                if self.in_key:
+                    self.state = _OBJ_COLON
+                else:
+                    return self._pop_value_done()
+                return True
+            if b == 0x5C:  # backslash
+                self.state = _STR_ESC
+                return True
+            return b >= 0x20  # control chars illegal; raw utf-8 bytes ok
+        if s == _STR_ESC:
+            if b == 0x75:  # u
+                self.state = _STR_U
+                self.u_left = 4
+                return True
+            if b in b'"\\/bfnrt':
+                self.state = _STRING
+                return True
+            return False
+        if s == _STR_U:
+            if b in b"0123456789abcdefABCDEF":
+                self.u_left -= 1
+                if self.u_left == 0:
+                    self.state = _STRING
+                return True
+            return False
+        if s in (_NUM_INT, _NUM_FRAC, _NUM_EXP):
+            if b in _DIGITS:
+                self.num_had_digit = True
+                return True
+            if s == _NUM_INT and b == 0x2E and self.num_had_digit:  # .
+                self.state = _NUM_FRAC
+                self.num_had_digit = False
+                return True
+            if s in (_NUM_INT, _NUM_FRAC) and b in b"eE" and self.num_had_digit:
+                self.state = _NUM_EXP
+                self.num_had_digit = False
+                return True
+            if s == _NUM_EXP and b in b"+-" and not self.num_had_digit:
+                return True
+            if not self.num_had_digit:
+                return False
+            # number terminated by this byte: re-dispatch in the parent state
+            self._pop_value_done()
+            return self._advance_byte(b)
+        if s == _LITERAL:
+            if self.lit_pos < len(self.lit) and b == self.lit[self.lit_pos]:
+                self.lit_pos += 1
+                if self.lit_pos == len(self.lit):
+                    return self._pop_value_done()
+                return True
+            return False
+        if s == _OBJ_KEY_OR_END:
+            if b in _WS:
+                return True
+            if b == 0x7D:  # }
+                self.stack.pop()
+                return self._pop_value_done()
+            if b == 0x22:
+                self.state = _STRING
+                self.in_key = True
+                return True
+            return False
+        if s == _OBJ_COLON:
+            if b in _WS:
+                return True
+            if b == 0x3A:  # :
+                self.state = _VALUE
+                self.in_key = False
+                return True
+            return False
+        if s == _OBJ_KEY_REQUIRED:
+            if b in _WS:
+                return True
+            if b == 0x22:
+                self.state = _STRING
+                self.in_key = True
+                return True
+            return False
+        if s == _OBJ_COMMA_OR_END:
+            if b in _WS:
+                return True
+            if b == 0x2C:  # ,
+                self.state = _OBJ_KEY_REQUIRED
+                return True
+            if b == 0x7D:
+                self.stack.pop()
+                return self._pop_value_done()
+            return False
+        if s == _ARR_COMMA_OR_END:
+            if b in _WS:
+                return True
+            if b == 0x2C:
+                self.state = _VALUE
+                return True
+            if b == 0x5D:
+                self.stack.pop()
+                return self._pop_value_done()
+            return False
+        return False
+
+
+def validate_json_bytes(data: bytes) -> bool:
+    g = JsonGrammar(require_object=False)
+    return g.advance(data) and g.is_complete()

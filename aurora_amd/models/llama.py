@@ -1,0 +1,208 @@
+"""Llama-3-family decoder (GQA, RoPE, SiLU-MLP) built on the aurora_amd op set.
+
+Every hot op routes through aurora_amd.ops (HIP/CDNA4 kernels on GPU, fp32
+torch reference on CPU); plain projections stay on hipBLASLt via F.linear.
+The forward is engine-shaped: packed varlen tokens + paged KV metadata, no
+[B, S] padding anywhere.
+
+Replaces the reference's rented chat models (reference:
+server/chat/backend/agent/llm.py:39-67, providers/) with an in-process
+model the engine steps directly.
+"""
+from __future__ import annotations
+
+from dataclasses import dataclass
+from typing import List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from .. import ops
+from ..parallel import dist as pdist
+from ..parallel.tp import (
+    ColumnParallelLinear,
+    MergedColumnParallelLinear,
+    QKVParallelLinear,
+    RowParallelLinear,
+)
+from .config import ModelSpec
+
+
+@dataclass
+class AttnMetadata:
+    """Per-step attention metadata shared by all layers.
+
+    mode "prefill": q tokens are packed varlen chunks (cu_seqlens_q); their
+    K/V have already been assigned cache slots via slot_mapping and the
+    kernel attends against the paged cache (covers fresh and append prefill).
+    mode "decode": one q token per sequence.
+    """
+
+    mode: str  # "prefill" | "decode"
+    positions: torch.Tensor  # [T] int32 absolute positions
+    slot_mapping: torch.Tensor  # [T] int32 cache slots for the new K/V
+    block_tables: torch.Tensor  # [num_seqs, max_pages] int32
+    context_lens: torch.Tensor  # [num_seqs] int32 (incl. new tokens)
+    cu_seqlens_q: Optional[torch.Tensor] = None  # [num_seqs+1] int32, prefill only
+
+
+class LlamaAttention(nn.Module):
+    def __init__(self, spec: ModelSpec, dtype):
+        super().__init__()
+        self.spec = spec
+        tp = pdist.tp_world_size()
+        self.n_heads = spec.n_heads // tp
+        self.n_kv_heads = spec.n_kv_heads // tp
+        self.head_dim = spec.head_dim
+        self.scale = spec.head_dim ** -0.5
+        self.qkv_proj = QKVParallelLinear(
+            spec.hidden_size, spec.n_heads, spec.n_kv_heads, spec.head_dim, dtype=dtype
+        )
+        self.o_proj = RowParallelLinear(spec.q_size, spec.hidden_size, dtype=dtype)
+
+    def forward(
+        self,
+        x: torch.Tensor,  # [T, H]
+        cos_sin: torch.Tensor,
+        kv_cache: Tuple[torch.Tensor, torch.Tensor],
+        meta: AttnMetadata,
+    ) -> torch.Tensor:
+        T = x.shape[0]
+        qkv = self.qkv_proj(x)
+        q, k, v = self.qkv_proj.split_qkv(qkv)
+        q = q.view(T, self.n_heads, self.head_dim)
+        k = k.view(T, self.n_kv_heads, self.head_dim)
+        v = v.view(T, self.n_kv_heads, self.head_dim)
+        q, k = ops.rope_apply(q, k, meta.positions, cos_sin)
+        k_cache, v_cache = kv_cache
+        ops.kv_cache_write(k, v, k_cache, v_cache, meta.slot_mapping)
+        if meta.mode == "prefill":
+            o = ops.attention_prefill_paged(
+                q, k_cache, v_cache, meta.block_tables,
+                meta.cu_seqlens_q, meta.context_lens, self.scale,
+            )
+        else:
+            o = ops.paged_attention_decode(
+                q, k_cache, v_cache, meta.block_tables, meta.context_lens, self.scale
+            )
+        return self.o_proj(o.reshape(T, self.n_heads * self.head_dim))
+
+
+class LlamaMLP(nn.Module):
+    def __init__(self, spec: ModelSpec, dtype):
+        super().__init__()
+        self.gate_up = MergedColumnParallelLinear(spec.hidden_size, spec.intermediate_size, dtype=dtype)
+        self.down = RowParallelLinear(spec.intermediate_size, spec.hidden_size, dtype=dtype)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.down(ops.silu_mul(self.gate_up(x)))
+
+
+class MoEMLP(nn.Module):
+    """Mixtral-style sparse MoE block (top-k routing over dense experts).
+
+    Experts are TP-sharded like the dense MLP (each expert's gate_up column-
+    sharded, down row-sharded) so the expert compute stays local and the
+    row-parallel all-reduce doubles as the TP combine.  Token->expert
+    dispatch is a host-side gather (grouped GEMM fusion is a later round).
+    """
+
+    def __init__(self, spec: ModelSpec, dtype):
+        super().__init__()
+        self.n_experts = spec.n_experts
+        self.top_k = spec.experts_per_token
+        self.router = nn.Linear(spec.hidden_size, spec.n_experts, bias=False, dtype=dtype)
+        self.router.weight.requires_grad_(False)
+        self.experts = nn.ModuleList(
+            [LlamaMLP(spec, dtype) for _ in range(spec.n_experts)]
+        )
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        T, H = x.shape
+        logits = self.router(x).float()  # [T, E]
+        weights, sel = logits.topk(self.top_k, dim=-1)
+        weights = weights.softmax(-1).to(x.dtype)  # [T, k]
+        out = torch.zeros_like(x)
+        for e in range(self.n_experts):
+            tok, slot = (sel == e).nonzero(as_tuple=True)
+            if tok.numel() == 0:
+                continue
+            out.index_add_(0, tok, self.experts[e](x[tok]) * weights[tok, slot, None])
+        return out
+
+
+class LlamaDecoderLayer(nn.Module):
+    def __init__(self, spec: ModelSpec, dtype):
+        super().__init__()
+        self.input_norm = nn.Parameter(torch.empty(spec.hidden_size, dtype=dtype), requires_grad=False)
+        self.post_attn_norm = nn.Parameter(torch.empty(spec.hidden_size, dtype=dtype), requires_grad=False)
+        self.eps = spec.rms_eps
+        self.attn = LlamaAttention(spec, dtype)
+        self.mlp = MoEMLP(spec, dtype) if spec.n_experts > 0 else LlamaMLP(spec, dtype)
+
+    def forward(self, x, residual, cos_sin, kv_cache, meta):
+        if residual is None:
+            residual = x
+            h = ops.rmsnorm(x, self.input_norm, self.eps)
+        else:
+            h, residual = ops.rmsnorm_residual(x, residual, self.input_norm, self.eps)
+        a = self.attn(h, cos_sin, kv_cache, meta)
+        h, residual = ops.rmsnorm_residual(a, residual, self.post_attn_norm, self.eps)
+        m = self.mlp(h)
+        return m, residual
+
+
+class LlamaForCausalLM(nn.Module):
+    def __init__(self, spec: ModelSpec, dtype=torch.bfloat16, device=None):
+        super().__init__()
+        self.spec = spec
+        self.dtype = dtype
+        self.embed = nn.Embedding(spec.vocab_size, spec.hidden_size, dtype=dtype)
+        self.embed.weight.requires_grad_(False)
+        self.layers = nn.ModuleList(
+            [LlamaDecoderLayer(spec, dtype) for _ in range(spec.n_layers)]
+        )
+        self.final_norm = nn.Parameter(torch.empty(spec.hidden_size, dtype=dtype), requires_grad=False)
+        if spec.tie_embeddings:
+            self.lm_head = None
+        else:
+            self.lm_head = nn.Linear(spec.hidden_size, spec.vocab_size, bias=False, dtype=dtype)
+            self.lm_head.weight.requires_grad_(False)
+        cos_sin = ops.torch_ref.rope_cos_sin_cache(
+            spec.head_dim, spec.max_position, spec.rope_theta, dtype=torch.float32
+        )
+        self.register_buffer("cos_sin", cos_sin, persistent=False)
+        if device is not None:
+            self.to(device)
+
+    @torch.inference_mode()
+    def forward(
+        self,
+        input_ids: torch.Tensor,  # [T]
+        kv_caches: List[Tuple[torch.Tensor, torch.Tensor]],
+        meta: AttnMetadata,
+    ) -> torch.Tensor:
+        x = self.embed(input_ids)
+        residual = None
+        for layer, cache in zip(self.layers, kv_caches):
+            x, residual = layer(x, residual, self.cos_sin, cache, meta)
+        x = ops.rmsnorm_residual(x, residual, self.final_norm, self.spec.rms_eps)[0]
+        return x  # hidden states [T, H]
+
+    @torch.inference_mode()
+    def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
+        w = self.embed.weight if self.lm_head is None else self.lm_head.weight
+        return F.linear(hidden, w).float()
+
+    @torch.inference_mode()
+    def init_random(self, seed: int = 0) -> None:
+        """Random-init all weights (synthetic-bench environment: no checkpoints)."""
+        gen = torch.Generator(device="cpu").manual_seed(seed)
+        for name, p in self.named_parameters():
+            if p.dim() >= 2:
+                fan_in = p.shape[-1]
+                std = fan_in ** -0.5
+                p.copy_(torch.randn(p.shape, generator=gen, dtype=torch.float32).mul_(std).to(p.dtype))
+            else:
+                p.copy_(torch.ones(p.shape, dtype=p.dtype))

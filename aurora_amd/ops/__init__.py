@@ -1,0 +1,158 @@
+"""Op dispatch layer.
+
+On CUDA (= ROCm/HIP) tensors every op routes to the in-tree gfx950 HIP
+extension ``aurora_amd._hip_ops`` and FAILS LOUDLY if it is not built —
+there is deliberately no silent eager fallback on the GPU path (the whole
+point is the CDNA4 kernels; a quiet PyTorch fallback would fake GPU tests).
+On CPU tensors ops route to the fp32 reference (``torch_ref``) so the full
+pipeline runs and is testable in CPU-only CI.
+
+Build the extension with ``python setup.py build_ext --inplace`` (hipcc,
+``--offload-arch=gfx950``) — see ``aurora_amd/ops/csrc/``.
+"""
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from . import torch_ref
+
+_hip = None
+_hip_import_error: Optional[BaseException] = None
+try:  # the in-tree extension: aurora_amd/_hip_ops*.so
+    from aurora_amd import _hip_ops as _hip  # type: ignore
+except Exception as e:  # pragma: no cover - exercised only when unbuilt
+    _hip_import_error = e
+
+
+def hip_available() -> bool:
+    return _hip is not None
+
+
+def _require_hip(op: str):
+    if _hip is None:
+        raise RuntimeError(
+            f"aurora_amd HIP extension is not built but op '{op}' was called on a GPU "
+            f"tensor. Build it in-tree with `python setup.py build_ext --inplace` "
+            f"(PYTORCH_ROCM_ARCH=gfx950). Import error: {_hip_import_error!r}"
+        )
+    return _hip
+
+
+def rmsnorm(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    if x.is_cuda:
+        return _require_hip("rmsnorm").rmsnorm(x.contiguous(), weight, eps)
+    return torch_ref.rmsnorm(x, weight, eps)
+
+
+def rmsnorm_residual(
+    x: torch.Tensor, residual: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    if x.is_cuda:
+        m = _require_hip("rmsnorm_residual")
+        return m.rmsnorm_residual(x.contiguous(), residual, weight, eps)
+    return torch_ref.rmsnorm_residual(x, residual, weight, eps)
+
+
+def rope_apply(
+    q: torch.Tensor, k: torch.Tensor, positions: torch.Tensor, cos_sin: torch.Tensor
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    if q.is_cuda:
+        m = _require_hip("rope_apply")
+        q = q.contiguous()
+        k = k.contiguous()
+        m.rope_apply_(q, k, positions.to(torch.int32), cos_sin)  # in-place
+        return q, k
+    return torch_ref.rope_apply(q, k, positions, cos_sin)
+
+
+def silu_mul(gate_up: torch.Tensor) -> torch.Tensor:
+    if gate_up.is_cuda:
+        return _require_hip("silu_mul").silu_mul(gate_up.contiguous())
+    return torch_ref.silu_mul(gate_up)
+
+
+def attention_prefill(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    cu_seqlens: torch.Tensor,
+    scale: float,
+    causal: bool = True,
+) -> torch.Tensor:
+    if q.is_cuda:
+        m = _require_hip("attention_prefill")
+        return m.attention_prefill(
+            q.contiguous(), k.contiguous(), v.contiguous(),
+            cu_seqlens.to(torch.int32), float(scale), bool(causal),
+        )
+    return torch_ref.attention_prefill(q, k, v, cu_seqlens, scale, causal)
+
+
+def attention_prefill_paged(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,
+    cu_seqlens_q: torch.Tensor,
+    context_lens: torch.Tensor,
+    scale: float,
+) -> torch.Tensor:
+    if q.is_cuda:
+        m = _require_hip("attention_prefill_paged")
+        return m.attention_prefill_paged(
+            q.contiguous(), k_cache, v_cache, block_tables.to(torch.int32),
+            cu_seqlens_q.to(torch.int32), context_lens.to(torch.int32), float(scale),
+        )
+    return torch_ref.attention_prefill_paged(
+        q, k_cache, v_cache, block_tables, cu_seqlens_q, context_lens, scale
+    )
+
+
+def kv_cache_write(
+    k: torch.Tensor, v: torch.Tensor,
+    k_cache: torch.Tensor, v_cache: torch.Tensor,
+    slot_mapping: torch.Tensor,
+) -> None:
+    if k.is_cuda:
+        m = _require_hip("kv_cache_write")
+        m.kv_cache_write(k.contiguous(), v.contiguous(), k_cache, v_cache,
+                         slot_mapping.to(torch.int32))
+        return
+    torch_ref.kv_cache_write(k, v, k_cache, v_cache, slot_mapping)
+
+
+def paged_attention_decode(
+    q: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    block_tables: torch.Tensor,
+    context_lens: torch.Tensor,
+    scale: float,
+) -> torch.Tensor:
+    if q.is_cuda:
+        m = _require_hip("paged_attention_decode")
+        return m.paged_attention_decode(
+            q.contiguous(), k_cache, v_cache,
+            block_tables.to(torch.int32), context_lens.to(torch.int32), float(scale),
+        )
+    return torch_ref.paged_attention_decode(q, k_cache, v_cache, block_tables, context_lens, scale)
+
+
+def embed_pool(hidden: torch.Tensor, cu_seqlens: torch.Tensor) -> torch.Tensor:
+    if hidden.is_cuda:
+        return _require_hip("embed_pool").embed_pool(hidden.contiguous(), cu_seqlens.to(torch.int32))
+    return torch_ref.embed_pool(hidden, cu_seqlens)
+
+
+def knn_topk(query: torch.Tensor, docs: torch.Tensor, k: int):
+    if query.is_cuda:
+        return _require_hip("knn_topk").knn_topk(query.contiguous(), docs, int(k))
+    return torch_ref.knn_topk(query, docs, k)
+
+
+def softmax_sample(logits, temperature, top_p, top_k, generator=None):
+    # Sampling is [B, V] with B small; keep it in torch on both devices for now
+    # (the GPU cost is one softmax over V — negligible next to the forward).
+    return torch_ref.softmax_sample(logits, temperature, top_p, top_k, generator)

@@ -1,0 +1,124 @@
+"""Distributed runtime: one process per GPU, torch.distributed over RCCL.
+
+The reference's scaling unit is a Celery worker process over Redis
+(reference: server/celery_config.py:37, CELERY_CONCURRENCY=4); here the
+scaling unit is a rank: DP shards concurrent investigations across engines
+(one per GPU), TP shards the model inside a group, and the collectives ride
+RCCL over xGMI (backend "nccl" IS RCCL on ROCm).  On CPU-only hosts (CI)
+the same code runs on gloo.
+"""
+from __future__ import annotations
+
+import datetime
+import os
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+_TP_GROUP: Optional[dist.ProcessGroup] = None
+_TP_RANK = 0
+_TP_WORLD = 1
+
+
+def env_rank() -> int:
+    return int(os.environ.get("RANK", "0"))
+
+
+def env_world_size() -> int:
+    return int(os.environ.get("WORLD_SIZE", "1"))
+
+
+def env_local_rank() -> int:
+    return int(os.environ.get("LOCAL_RANK", os.environ.get("RANK", "0")))
+
+
+def init_distributed(tp_size: int = 1, backend: Optional[str] = None,
+                     timeout_s: float = 600.0) -> None:
+    """Initialise the default process group and the TP subgroup.
+
+    World layout: ranks are grouped into contiguous TP groups of size
+    ``tp_size``; ranks in different groups are DP replicas.  xGMI note: TP
+    all-reduce is per-link bound (7 p2p links/GPU), so keep tp_size at 1, 4
+    or 8 and let RCCL pick the one-shot algorithm for decode-sized messages.
+    """
+    global _TP_GROUP, _TP_RANK, _TP_WORLD
+    world = env_world_size()
+    if world == 1 and not dist.is_initialized():
+        _TP_GROUP, _TP_RANK, _TP_WORLD = None, 0, 1
+        return
+    if not dist.is_initialized():
+        if backend is None:
+            backend = "nccl" if torch.cuda.is_available() else "gloo"
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29517")
+        dist.init_process_group(
+            backend=backend, rank=env_rank(), world_size=world,
+            timeout=datetime.timedelta(seconds=timeout_s),
+        )
+    if torch.cuda.is_available():
+        torch.cuda.set_device(env_local_rank() % max(torch.cuda.device_count(), 1))
+    assert world % tp_size == 0, f"world {world} not divisible by tp {tp_size}"
+    _TP_WORLD = tp_size
+    if tp_size == 1:
+        _TP_GROUP = None
+        _TP_RANK = 0
+        return
+    rank = dist.get_rank()
+    for g0 in range(0, world, tp_size):
+        ranks = list(range(g0, g0 + tp_size))
+        grp = dist.new_group(ranks=ranks)
+        if rank in ranks:
+            _TP_GROUP = grp
+            _TP_RANK = rank - g0
+
+
+def tp_world_size() -> int:
+    return _TP_WORLD
+
+
+def tp_rank() -> int:
+    return _TP_RANK
+
+
+def tp_group():
+    return _TP_GROUP
+
+
+def tp_all_reduce(x: torch.Tensor) -> torch.Tensor:
+    if _TP_WORLD == 1:
+        return x
+    dist.all_reduce(x, group=_TP_GROUP)
+    return x
+
+
+def tp_all_gather(x: torch.Tensor, dim: int = -1) -> torch.Tensor:
+    if _TP_WORLD == 1:
+        return x
+    parts = [torch.empty_like(x) for _ in range(_TP_WORLD)]
+    dist.all_gather(parts, x.contiguous(), group=_TP_GROUP)
+    return torch.cat(parts, dim=dim)
+
+
+def barrier() -> None:
+    if dist.is_initialized():
+        dist.barrier()
+
+
+def dp_rank() -> int:
+    if not dist.is_initialized():
+        return 0
+    return dist.get_rank() // _TP_WORLD
+
+
+def dp_world_size() -> int:
+    if not dist.is_initialized():
+        return 1
+    return dist.get_world_size() // _TP_WORLD
+
+
+def destroy() -> None:
+    global _TP_GROUP, _TP_RANK, _TP_WORLD
+    if dist.is_initialized():
+        dist.destroy_process_group()
+    _TP_GROUP, _TP_RANK, _TP_WORLD = None, 0, 1
