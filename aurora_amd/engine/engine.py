@@ -369,30 +369,20 @@ class LLMEngine:
 
     # --------------------------------------------------------------- sample
 
-    def _sample_row(self, seq: Sequence, row: torch.Tensor) -> int:
-        p = seq.params
-        if seq.grammar is None:
-            return self._plain_sample(row, p)
-        # grammar-guided: walk candidates in probability order
-        k = min(self.config.grammar_topk, row.numel())
-        if p.temperature > 0:
-            probs = (row / p.temperature).softmax(-1)
-            cand = torch.multinomial(
-                probs.cpu(), num_samples=k, replacement=False, generator=self.gen
-            ).tolist()
-        else:
-            cand = row.topk(k).indices.tolist()
+    def _grammar_pick(self, seq: Sequence, cand: List[int], row: torch.Tensor) -> int:
+        """Walk candidate tokens in probability order; keep the first whose
+        bytes keep the JSON FSM valid."""
         for tok in cand:
             b = self.tokenizer.token_bytes(tok)
             if tok == self.tokenizer.eos_id and seq.grammar.is_complete():
                 return tok
-            g = seq.grammar.fork()
-            if b and g.advance(b):
-                seq.grammar = g
-                return tok
             if not b:
                 continue
-        # fallback: full-vocab scan in logit order
+            g = seq.grammar.fork()
+            if g.advance(b):
+                seq.grammar = g
+                return tok
+        # fallback: full-vocab scan in logit order (rare)
         self.counters["grammar_fallbacks"] += 1
         order = row.argsort(descending=True).tolist()
         for tok in order:
@@ -405,30 +395,67 @@ class LLMEngine:
                 return tok
         return self.tokenizer.eos_id
 
-    def _plain_sample(self, row: torch.Tensor, p: SamplingParams) -> int:
-        if p.temperature <= 0:
-            return int(row.argmax())
-        r = row / p.temperature
-        V = r.numel()
-        if p.top_k and p.top_k < V:
-            kth = r.topk(p.top_k).values[-1]
-            r = r.masked_fill(r < kth, float("-inf"))
-        if p.top_p < 1.0:
-            sr, si = r.sort(descending=True)
-            probs = sr.softmax(-1)
-            cum = probs.cumsum(-1)
-            cut = cum - probs >= p.top_p
-            sr = sr.masked_fill(cut, float("-inf"))
-            r = torch.full_like(r, float("-inf")).scatter(0, si, sr)
-        probs = r.softmax(-1)
-        return int(torch.multinomial(probs.cpu(), 1, generator=self.gen))
+    def _batch_plain_sample(self, logits: torch.Tensor, seqs: List[Sequence]) -> torch.Tensor:
+        """Vectorized temperature/top-k/top-p sampling on the logits' device
+        (gumbel-max for the stochastic rows); returns token ids on CPU."""
+        B, V = logits.shape
+        temps = torch.tensor([s.params.temperature for s in seqs], device=logits.device)
+        greedy = temps <= 0
+        r = logits / temps.clamp(min=1e-6).unsqueeze(1)
+        for i, s in enumerate(seqs):
+            p = s.params
+            if p.temperature <= 0:
+                continue
+            row = r[i]
+            if p.top_k and p.top_k < V:
+                kth = row.topk(p.top_k).values[-1]
+                row.masked_fill_(row < kth, float("-inf"))
+            if p.top_p < 1.0:
+                sr, si = row.sort(descending=True)
+                probs = sr.softmax(-1)
+                cut = probs.cumsum(-1) - probs >= p.top_p
+                sr = sr.masked_fill(cut, float("-inf"))
+                row.copy_(torch.full_like(row, float("-inf")).scatter(0, si, sr))
+        if bool(greedy.all()):
+            return logits.argmax(-1).cpu()
+        u = torch.rand(B, V, device="cpu", generator=self.gen).to(logits.device)
+        gumbel = -torch.log(-torch.log(u.clamp(min=1e-20)).clamp(min=1e-20))
+        noisy = torch.where(greedy.unsqueeze(1), logits, r + gumbel)
+        return noisy.argmax(-1).cpu()
 
     def _sample_and_append(self, seqs: List[Sequence], logits: torch.Tensor) -> List[StepOutput]:
         outputs = []
-        lf = logits.float().cpu()
+        lf = logits.float()
         now = time.monotonic()
+        # plain rows: one batched device-side sample; grammar rows: top-k walk
+        plain_idx = [i for i, s in enumerate(seqs) if s.grammar is None]
+        gram_idx = [i for i, s in enumerate(seqs) if s.grammar is not None]
+        toks: Dict[int, int] = {}
+        if plain_idx:
+            sel = lf[plain_idx] if len(plain_idx) < len(seqs) else lf
+            ids = self._batch_plain_sample(sel, [seqs[i] for i in plain_idx])
+            for j, i in enumerate(plain_idx):
+                toks[i] = int(ids[j])
+        if gram_idx:
+            k = min(self.config.grammar_topk, lf.shape[1])
+            sub = lf[gram_idx]
+            seq_list = [seqs[i] for i in gram_idx]
+            cand: List[List[int]] = []
+            stochastic = any(s.params.temperature > 0 for s in seq_list)
+            if stochastic:
+                u = torch.rand(sub.shape, device="cpu", generator=self.gen).to(sub.device)
+                g = -torch.log(-torch.log(u.clamp(min=1e-20)).clamp(min=1e-20))
+                temps = torch.tensor(
+                    [max(s.params.temperature, 1e-6) for s in seq_list], device=sub.device
+                ).unsqueeze(1)
+                noisy = sub / temps + g
+                cand_t = noisy.topk(k, dim=-1).indices.cpu()
+            else:
+                cand_t = sub.topk(k, dim=-1).indices.cpu()
+            for j, i in enumerate(gram_idx):
+                toks[i] = self._grammar_pick(seqs[i], cand_t[j].tolist(), sub[j])
         for i, seq in enumerate(seqs):
-            tok = self._sample_row(seq, lf[i])
+            tok = toks[i]
             seq.token_ids.append(tok)
             if seq.first_token_time is None:
                 seq.first_token_time = now

@@ -37,7 +37,13 @@ _DIGITS = b"0123456789"
 class JsonGrammar:
     """Incremental JSON validator; max_depth bounds container nesting."""
 
-    def __init__(self, max_depth: int = 16, require_object: bool = True):
+    def __init__(self, max_depth: int = 16, require_object: bool = True,
+                 compact: bool = True):
+        # compact: reject whitespace outside strings — JSON allows unbounded
+        # inter-token whitespace, which lets a badly-steered model spin on
+        # '\r' forever without completing; generation uses compact=True,
+        # validation compact=False.
+        self.compact = compact
         self.max_depth = max_depth
         self.stack: List[int] = []  # b'{' or b'[' markers (as ints)
         self.state = _VALUE
@@ -118,6 +124,8 @@ class JsonGrammar:
 
     def _advance_byte(self, b: int) -> bool:
         s = self.state
+        if self.compact and b in _WS and s not in (_STRING, _STR_ESC, _STR_U):
+            return False
         if s == _DONE:
             return b in _WS
         if s == _VALUE:
@@ -229,5 +237,5 @@ class JsonGrammar:
 
 
 def validate_json_bytes(data: bytes) -> bool:
-    g = JsonGrammar(require_object=False)
+    g = JsonGrammar(require_object=False, compact=False)
     return g.advance(data) and g.is_complete()

@@ -103,3 +103,8 @@ register(ModelSpec(
     head_dim=16, intermediate_size=128, vocab_size=512, max_position=512,
     is_encoder=True,
 ))
+register(ModelSpec(  # GPU smoke/test model: real head_dim for the HIP kernels
+    name="smoke-llama", hidden_size=512, n_layers=4, n_heads=4, n_kv_heads=2,
+    head_dim=128, intermediate_size=1024, vocab_size=1024, max_position=8192,
+    tie_embeddings=True,
+))

@@ -197,12 +197,25 @@ class LlamaForCausalLM(nn.Module):
 
     @torch.inference_mode()
     def init_random(self, seed: int = 0) -> None:
-        """Random-init all weights (synthetic-bench environment: no checkpoints)."""
+        """Random-init all weights (synthetic-bench environment: no checkpoints).
+
+        Small params use a CPU generator so CPU and GPU instances of the same
+        spec get identical weights (engine equivalence tests); big params
+        (>16M elements, e.g. the 8B model's) are generated on-device for
+        speed — cross-device bit-equality is irrelevant at that scale.
+        """
         gen = torch.Generator(device="cpu").manual_seed(seed)
+        dev_gen = None
         for name, p in self.named_parameters():
             if p.dim() >= 2:
-                fan_in = p.shape[-1]
-                std = fan_in ** -0.5
-                p.copy_(torch.randn(p.shape, generator=gen, dtype=torch.float32).mul_(std).to(p.dtype))
+                std = p.shape[-1] ** -0.5
+                if p.numel() > (1 << 24) and p.device.type != "cpu":
+                    if dev_gen is None:
+                        dev_gen = torch.Generator(device=p.device).manual_seed(seed)
+                    r = torch.randn(p.shape, generator=dev_gen, device=p.device,
+                                    dtype=torch.float32)
+                else:
+                    r = torch.randn(p.shape, generator=gen, dtype=torch.float32)
+                p.copy_(r.mul_(std).to(p.dtype))
             else:
                 p.copy_(torch.ones(p.shape, dtype=p.dtype))

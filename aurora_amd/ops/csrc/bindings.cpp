@@ -1,0 +1,36 @@
+// Python bindings for the aurora_amd gfx950 kernel set.
+#include <torch/extension.h>
+
+torch::Tensor rmsnorm(torch::Tensor x, torch::Tensor weight, double eps);
+std::vector<torch::Tensor> rmsnorm_residual(torch::Tensor x, torch::Tensor res,
+                                            torch::Tensor weight, double eps);
+torch::Tensor silu_mul(torch::Tensor gu);
+void rope_apply_(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
+                 torch::Tensor cos_sin);
+void kv_cache_write(torch::Tensor k, torch::Tensor v, torch::Tensor kc,
+                    torch::Tensor vc, torch::Tensor slots);
+torch::Tensor paged_attention_decode(torch::Tensor q, torch::Tensor kc,
+                                     torch::Tensor vc, torch::Tensor bt,
+                                     torch::Tensor ctx, double scale);
+torch::Tensor attention_prefill_paged(torch::Tensor q, torch::Tensor kc,
+                                      torch::Tensor vc, torch::Tensor bt,
+                                      torch::Tensor cu_q, torch::Tensor ctx,
+                                      double scale);
+torch::Tensor attention_prefill(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                torch::Tensor cu, double scale, bool causal);
+torch::Tensor embed_pool(torch::Tensor hidden, torch::Tensor cu);
+std::vector<torch::Tensor> knn_topk(torch::Tensor query, torch::Tensor docs, long k);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &rmsnorm, "fused RMSNorm (bf16, gfx950)");
+  m.def("rmsnorm_residual", &rmsnorm_residual, "fused residual-add + RMSNorm");
+  m.def("silu_mul", &silu_mul, "fused SiLU(gate)*up");
+  m.def("rope_apply_", &rope_apply_, "in-place RoPE on q/k");
+  m.def("kv_cache_write", &kv_cache_write, "scatter K/V into paged cache");
+  m.def("paged_attention_decode", &paged_attention_decode, "paged GQA decode attention");
+  m.def("attention_prefill_paged", &attention_prefill_paged,
+        "MFMA flash prefill over paged KV");
+  m.def("attention_prefill", &attention_prefill, "varlen self-attention (encoder)");
+  m.def("embed_pool", &embed_pool, "mean-pool + L2 normalize");
+  m.def("knn_topk", &knn_topk, "brute-force kNN top-k");
+}

@@ -1,0 +1,75 @@
+// Common helpers for the aurora_amd gfx950 (CDNA4) kernels.
+// Wave width is 64 on CDNA4 — hard-coded per the platform guide.
+#pragma once
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+
+#define WAVE 64
+#define DEVINL __device__ __forceinline__
+
+typedef __attribute__((ext_vector_type(2))) float floatx2;
+typedef __attribute__((ext_vector_type(4))) float floatx4;
+typedef __attribute__((ext_vector_type(16))) float floatx16;
+typedef __attribute__((ext_vector_type(4))) short shortx4;
+typedef __attribute__((ext_vector_type(8))) short shortx8;
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
+typedef __attribute__((ext_vector_type(2))) __bf16 bf16x2;
+
+DEVINL float bf16_to_f32(unsigned short u) {
+  union { unsigned int i; float f; } v;
+  v.i = ((unsigned int)u) << 16;
+  return v.f;
+}
+
+DEVINL unsigned short f32_to_bf16(float f) {
+  union { float f; unsigned int i; } v;
+  v.f = f;
+  // round-to-nearest-even
+  unsigned int lsb = (v.i >> 16) & 1;
+  v.i += 0x7fff + lsb;
+  return (unsigned short)(v.i >> 16);
+}
+
+// dot product of 8 bf16 pairs (unpacked to f32 FMAs)
+DEVINL float dot8_bf16(const shortx8 a, const shortx8 b) {
+  float acc = 0.f;
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+    acc += bf16_to_f32((unsigned short)a[i]) * bf16_to_f32((unsigned short)b[i]);
+  return acc;
+}
+
+// full-wave (64-lane) reductions
+DEVINL float wave_sum(float x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) x += __shfl_xor(x, off, WAVE);
+  return x;
+}
+DEVINL float wave_max(float x) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) x = fmaxf(x, __shfl_xor(x, off, WAVE));
+  return x;
+}
+// reduction within a contiguous group of W lanes (W power of two <= 64)
+template <int W>
+DEVINL float group_sum(float x) {
+#pragma unroll
+  for (int off = W / 2; off > 0; off >>= 1) x += __shfl_xor(x, off, WAVE);
+  return x;
+}
+template <int W>
+DEVINL float group_max(float x) {
+#pragma unroll
+  for (int off = W / 2; off > 0; off >>= 1) x = fmaxf(x, __shfl_xor(x, off, WAVE));
+  return x;
+}
+
+#define HIP_CHECK_KERNEL()                                                     \
+  do {                                                                         \
+    hipError_t e = hipGetLastError();                                          \
+    if (e != hipSuccess)                                                       \
+      TORCH_CHECK(false, "HIP kernel launch failed: ", hipGetErrorString(e));  \
+  } while (0)

@@ -1,0 +1,175 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: incidents/hr end-to-end RCA (Llama-3-8B agent).
+
+Driver contract:
+    python bench.py --gpus N --steps K --warmup W
+For N>1 the driver launches this under torch.distributed.run with one rank
+per GPU (DP: each rank serves its own shard of concurrent investigations —
+weak scaling, the reference's Celery-worker model mapped onto GPUs).
+
+One STEP = one batch of --concurrency synthetic incident investigations
+processed end-to-end on this rank's engine: per investigation, --turns agent
+turns, each turn appending a synthetic alert/tool-output prompt chunk to the
+session (KV reused across turns) and decoding a fixed thought/tool-call
+budget.  Fixed token counts per step (no early exit) so the timed work is
+identical across steps and ranks.  Data: synthetic token streams,
+random-init weights (no network in this environment — BASELINE.md).
+
+Rank 0 prints ONE JSON line; value = whole-job incidents/hr over all ranks.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import statistics
+import time
+
+import torch
+
+
+def parse_args():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--model", default=os.environ.get("BENCH_MODEL", "llama3-8b"))
+    ap.add_argument("--concurrency", type=int,
+                    default=int(os.environ.get("BENCH_CONCURRENCY", "16")))
+    ap.add_argument("--turns", type=int, default=int(os.environ.get("BENCH_TURNS", "4")))
+    ap.add_argument("--first-prompt", type=int, default=768)
+    ap.add_argument("--turn-prompt", type=int, default=512)
+    ap.add_argument("--decode-per-turn", type=int, default=128)
+    ap.add_argument("--device", default="auto")
+    return ap.parse_args()
+
+
+def synth_tokens(rng: torch.Generator, n: int, vocab: int) -> list:
+    return torch.randint(0, min(vocab - 4, 250), (n,), generator=rng).tolist()
+
+
+def run_step(eng, args, rng, vocab, step_idx: int):
+    """One full investigation batch; returns per-incident latencies."""
+    from aurora_amd.engine.sampling_params import SamplingParams
+
+    B, turns = args.concurrency, args.turns
+    params = SamplingParams(temperature=0.7, top_p=0.95,
+                            max_tokens=args.decode_per_turn, ignore_eos=True)
+    ctx_tokens = {i: synth_tokens(rng, args.first_prompt, vocab) for i in range(B)}
+    turn_no = {i: 0 for i in range(B)}
+    t_start = {i: time.monotonic() for i in range(B)}
+    t_done = {}
+
+    def submit(i):
+        rid = f"s{step_idx}-inc{i}-t{turn_no[i]}"
+        eng.add_request(rid, list(ctx_tokens[i]), params, session_id=f"inc-{step_idx}-{i}")
+        return rid
+
+    live = {}
+    for i in range(B):
+        live[submit(i)] = i
+    gen_out = {i: [] for i in range(B)}
+    while live:
+        for out in eng.step():
+            i = live.get(out.request_id)
+            if i is None:
+                continue
+            gen_out[i].extend(out.new_token_ids)
+            if out.finished:
+                del live[out.request_id]
+                ctx_tokens[i].extend(gen_out[i][-args.decode_per_turn:])
+                turn_no[i] += 1
+                if turn_no[i] < turns:
+                    ctx_tokens[i].extend(synth_tokens(rng, args.turn_prompt, vocab))
+                    live[submit(i)] = i
+                else:
+                    t_done[i] = time.monotonic()
+    return [t_done[i] - t_start[i] for i in range(B)]
+
+
+def main():
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    use_cuda = torch.cuda.is_available() and args.device != "cpu"
+
+    from aurora_amd.engine.engine import EngineConfig, LLMEngine
+    from aurora_amd.models.config import get_spec
+    from aurora_amd.parallel import dist as pdist
+
+    if world > 1:
+        pdist.init_distributed(tp_size=1)
+
+    spec = get_spec(args.model)
+    cfg = EngineConfig(
+        model=args.model,
+        device="cpu" if not use_cuda else "auto",
+        max_num_seqs=max(args.concurrency + 2, 8),
+        max_batched_tokens=16384,
+        seed=1234 + rank,
+    )
+    eng = LLMEngine(cfg)
+    rng = torch.Generator().manual_seed(999 + rank)
+    vocab = spec.vocab_size
+
+    def sync():
+        if use_cuda:
+            torch.cuda.synchronize()
+        if world > 1:
+            pdist.barrier()
+            if use_cuda:
+                torch.cuda.synchronize()
+
+    for w in range(args.warmup):
+        run_step(eng, args, rng, vocab, -1 - w)
+    sync()
+    t0 = time.monotonic()
+    lat = []
+    for s in range(args.steps):
+        lat.extend(run_step(eng, args, rng, vocab, s))
+    sync()
+    elapsed = time.monotonic() - t0
+
+    # max over ranks (slowest rank defines the job)
+    if world > 1:
+        t = torch.tensor([elapsed])
+        import torch.distributed as dist
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t[0])
+
+    incidents = args.steps * args.concurrency * world
+    value = incidents / (elapsed / 3600.0)
+    p50 = statistics.median(lat)
+    tokens_per_incident = (args.first_prompt + (args.turns - 1) * args.turn_prompt
+                           + args.turns * args.decode_per_turn)
+    if rank == 0:
+        print(json.dumps({
+            "metric": "incidents/hr",
+            "value": round(value, 2),
+            "unit": "incidents/hr",
+            "n_gpus": world if use_cuda else 0,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000.0, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if use_cuda else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.concurrency * world,
+                "seq_len": tokens_per_incident,
+                "parallelism": f"dp{world}",
+                "concurrency_per_gpu": args.concurrency,
+                "turns": args.turns,
+                "decode_tokens_per_turn": args.decode_per_turn,
+                "p50_time_to_rca_s": round(p50, 3),
+                "prompt_tokens_per_incident": args.first_prompt
+                + (args.turns - 1) * args.turn_prompt,
+            },
+        }))
+
+
+if __name__ == "__main__":
+    main()

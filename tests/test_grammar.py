@@ -20,7 +20,7 @@ BAD = [
 
 @pytest.mark.parametrize("s", GOOD)
 def test_good(s):
-    g = JsonGrammar()
+    g = JsonGrammar(compact=False)
     assert g.advance(s.encode()), s
     assert g.is_complete(), s
     json.loads(s)  # sanity: python agrees
@@ -28,13 +28,13 @@ def test_good(s):
 
 @pytest.mark.parametrize("s", BAD)
 def test_bad(s):
-    g = JsonGrammar()
+    g = JsonGrammar(compact=False)
     ok = g.advance(s.encode()) and g.is_complete()
     assert not ok, s
 
 
 def test_incremental_byte_at_a_time():
-    s = b'{"tool": "kubectl", "args": {"cmd": "get pods", "n": 3.5}}'
+    s = b'{"tool":"kubectl","args":{"cmd":"get pods","n":3.5}}'
     g = JsonGrammar()
     for i in range(len(s)):
         assert g.advance(s[i:i+1]), s[:i+1]
@@ -44,7 +44,7 @@ def test_incremental_byte_at_a_time():
 
 
 def test_fork_independent():
-    g = JsonGrammar()
+    g = JsonGrammar(compact=False)
     assert g.advance(b'{"a"')
     f = g.fork()
     assert f.advance(b': 1}')
@@ -63,3 +63,13 @@ def test_validate_non_object_allowed():
 def test_utf8_in_strings():
     s = '{"k": "héllo ✓"}'.encode()
     assert validate_json_bytes(s)
+
+
+def test_compact_rejects_whitespace():
+    g = JsonGrammar()  # compact by default
+    assert not g.fork().advance(b' ')
+    assert not g.fork().advance(b'{ ')
+    assert g.fork().advance(b'{"a":1}')
+    g2 = JsonGrammar()
+    assert g2.advance(b'{"a b":" c "}')  # ws inside strings still fine
+    assert g2.is_complete()

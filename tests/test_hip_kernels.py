@@ -1,0 +1,256 @@
+"""GPU numerics tests: every gfx950 HIP kernel vs the plain-PyTorch fp32
+reference of the same op (aurora_amd.ops.torch_ref), on the same bf16 inputs.
+
+All marked @pytest.mark.gpu (run via gpurun on a real MI355X)."""
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+from aurora_amd import ops
+from aurora_amd.ops import torch_ref as R
+
+
+def _err(a, b):
+    a = a.float().cpu()
+    b = b.float().cpu()
+    denom = b.abs().max().clamp(min=1e-6)
+    return ((a - b).abs().max() / denom).item()
+
+
+@pytest.fixture(autouse=True)
+def _require_ext():
+    assert ops.hip_available(), "HIP extension must be built in-tree"
+
+
+def test_rmsnorm():
+    x = torch.randn(37, 4096, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(4096, device="cuda", dtype=torch.bfloat16)
+    y = ops.rmsnorm(x, w, 1e-5)
+    ref = R.rmsnorm(x.cpu(), w.cpu(), 1e-5)
+    assert _err(y, ref) < 2e-2
+
+
+def test_rmsnorm_residual():
+    x = torch.randn(16, 512, device="cuda", dtype=torch.bfloat16)
+    r = torch.randn(16, 512, device="cuda", dtype=torch.bfloat16)
+    w = torch.randn(512, device="cuda", dtype=torch.bfloat16)
+    y, nr = ops.rmsnorm_residual(x, r, w, 1e-5)
+    ry, rnr = R.rmsnorm_residual(x.cpu(), r.cpu(), w.cpu(), 1e-5)
+    assert _err(nr, rnr) < 1e-2
+    assert _err(y, ry) < 2e-2
+
+
+def test_rope():
+    T, Hq, Hkv, D = 33, 8, 2, 128
+    cs = R.rope_cos_sin_cache(D, 4096, 500000.0)
+    q = torch.randn(T, Hq, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(T, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    pos = torch.randint(0, 4096, (T,), dtype=torch.int32, device="cuda")
+    q0, k0 = q.cpu(), k.cpu()
+    q2, k2 = ops.rope_apply(q, k, pos, cs.cuda())
+    rq, rk = R.rope_apply(q0, k0, pos.cpu(), cs)
+    assert _err(q2, rq) < 2e-2
+    assert _err(k2, rk) < 2e-2
+
+
+def test_silu_mul():
+    x = torch.randn(64, 2048, device="cuda", dtype=torch.bfloat16)
+    y = ops.silu_mul(x)
+    assert _err(y, R.silu_mul(x.cpu())) < 2e-2
+
+
+def test_kv_cache_write():
+    T, Hkv, D, P, NP = 50, 4, 128, 64, 8
+    k = torch.randn(T, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(T, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    kc = torch.zeros(NP, Hkv, P, D, device="cuda", dtype=torch.bfloat16)
+    vc = torch.zeros_like(kc)
+    perm = torch.randperm(NP * P)[:T].to(torch.int32)
+    ops.kv_cache_write(k, v, kc, vc, perm.cuda())
+    kc_ref = torch.zeros(NP, Hkv, P, D)
+    vc_ref = torch.zeros_like(kc_ref)
+    R.kv_cache_write(k.cpu().float(), v.cpu().float(), kc_ref, vc_ref, perm)
+    assert _err(kc, kc_ref) < 1e-2
+    assert _err(vc, vc_ref) < 1e-2
+
+
+@pytest.mark.parametrize("D,G,Hkv", [(128, 4, 8), (128, 1, 4), (64, 8, 2), (128, 8, 1)])
+def test_paged_decode(D, G, Hkv):
+    torch.manual_seed(0)
+    B, P, NP = 5, 64, 40
+    Hq = G * Hkv
+    ctx = torch.tensor([17, 64, 130, 1, 301], dtype=torch.int32)
+    maxp = (int(ctx.max()) + P - 1) // P
+    bt = torch.zeros(B, maxp, dtype=torch.int32)
+    used = torch.randperm(NP)
+    u = 0
+    for b in range(B):
+        need = (int(ctx[b]) + P - 1) // P
+        bt[b, :need] = used[u:u + need].to(torch.int32)
+        u += need
+    kc = torch.randn(NP, Hkv, P, D, device="cuda", dtype=torch.bfloat16)
+    vc = torch.randn(NP, Hkv, P, D, device="cuda", dtype=torch.bfloat16)
+    q = torch.randn(B, Hq, D, device="cuda", dtype=torch.bfloat16)
+    scale = 1.0 / math.sqrt(D)
+    out = ops.paged_attention_decode(q, kc, vc, bt.cuda(), ctx.cuda(), scale)
+    ref = R.paged_attention_decode(q.cpu(), kc.cpu(), vc.cpu(), bt, ctx, scale)
+    assert _err(out, ref) < 3e-2
+
+
+@pytest.mark.parametrize("D,G,Hkv", [(128, 4, 2), (64, 1, 4)])
+def test_prefill_paged_fresh(D, G, Hkv):
+    """Fresh prefill (qlen == ctx), ragged lengths crossing 128-row blocks."""
+    torch.manual_seed(1)
+    P, NP = 256, 24
+    Hq = G * Hkv
+    lens = [5, 200, 513, 128]
+    T = sum(lens)
+    cu = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)), dtype=torch.int32)
+    ctx = torch.tensor(lens, dtype=torch.int32)
+    maxp = (max(lens) + P - 1) // P
+    bt = torch.zeros(len(lens), maxp, dtype=torch.int32)
+    used = torch.randperm(NP)
+    u = 0
+    slot_map = torch.empty(T, dtype=torch.int32)
+    for s, L in enumerate(lens):
+        need = (L + P - 1) // P
+        pages = used[u:u + need]
+        bt[s, :need] = pages.to(torch.int32)
+        u += need
+        a = int(cu[s])
+        for i in range(L):
+            slot_map[a + i] = int(pages[i // P]) * P + i % P
+    q = torch.randn(T, Hq, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(T, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(T, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    kc = torch.zeros(NP, Hkv, P, D, device="cuda", dtype=torch.bfloat16)
+    vc = torch.zeros_like(kc)
+    ops.kv_cache_write(k, v, kc, vc, slot_map.cuda())
+    scale = 1.0 / math.sqrt(D)
+    out = ops.attention_prefill_paged(q, kc, vc, bt.cuda(), cu.cuda(), ctx.cuda(), scale)
+    ref = R.attention_prefill_paged(q.cpu(), kc.cpu(), vc.cpu(), bt, cu, ctx, scale)
+    assert _err(out, ref) < 3e-2
+
+
+def test_prefill_paged_append():
+    """Append-prefill: 100 cached tokens + 37 new queries."""
+    torch.manual_seed(2)
+    D, Hq, Hkv, P, NP = 128, 8, 2, 256, 8
+    hist, new = 100, 37
+    L = hist + new
+    pages = [3, 1]
+    bt = torch.tensor([pages], dtype=torch.int32)
+    slot_map = torch.tensor([pages[i // P] * P + i % P for i in range(L)], dtype=torch.int32)
+    k = torch.randn(L, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(L, Hkv, D, device="cuda", dtype=torch.bfloat16)
+    kc = torch.zeros(NP, Hkv, P, D, device="cuda", dtype=torch.bfloat16)
+    vc = torch.zeros_like(kc)
+    ops.kv_cache_write(k, v, kc, vc, slot_map.cuda())
+    q = torch.randn(new, Hq, D, device="cuda", dtype=torch.bfloat16)
+    cu = torch.tensor([0, new], dtype=torch.int32)
+    ctx = torch.tensor([L], dtype=torch.int32)
+    scale = 1.0 / math.sqrt(D)
+    out = ops.attention_prefill_paged(q, kc, vc, bt.cuda(), cu.cuda(), ctx.cuda(), scale)
+    ref = R.attention_prefill_paged(q.cpu(), kc.cpu(), vc.cpu(), bt, cu, ctx, scale)
+    assert _err(out, ref) < 3e-2
+
+
+@pytest.mark.parametrize("causal", [False, True])
+def test_varlen_attention_encoder(causal):
+    torch.manual_seed(3)
+    D, H = 64, 12
+    lens = [7, 128, 33]
+    T = sum(lens)
+    cu = torch.tensor([0] + list(torch.tensor(lens).cumsum(0)), dtype=torch.int32)
+    q = torch.randn(T, H, D, device="cuda", dtype=torch.bfloat16)
+    k = torch.randn(T, H, D, device="cuda", dtype=torch.bfloat16)
+    v = torch.randn(T, H, D, device="cuda", dtype=torch.bfloat16)
+    scale = 1.0 / math.sqrt(D)
+    out = ops.attention_prefill(q, k, v, cu.cuda(), scale, causal)
+    ref = R.attention_prefill(q.cpu(), k.cpu(), v.cpu(), cu, scale, causal)
+    assert _err(out, ref) < 3e-2
+
+
+def test_embed_pool():
+    h = torch.randn(100, 768, device="cuda", dtype=torch.bfloat16)
+    cu = torch.tensor([0, 10, 100], dtype=torch.int32)
+    out = ops.embed_pool(h, cu.cuda())
+    ref = R.embed_pool(h.cpu(), cu)
+    assert _err(out, ref) < 2e-2
+
+
+def test_knn():
+    docs = torch.nn.functional.normalize(
+        torch.randn(5000, 768), dim=-1).to("cuda", torch.bfloat16)
+    qv = docs[123:125].float() + 0.001 * torch.randn(2, 768, device="cuda")
+    vals, idx = ops.knn_topk(qv.to(torch.bfloat16), docs, 10)
+    assert idx[0, 0].item() == 123 and idx[1, 0].item() == 124
+
+
+def test_engine_gpu_matches_cpu_reference():
+    """Whole-engine check: GPU bf16 forward vs CPU fp32 same-weight model."""
+    from aurora_amd.engine.engine import EngineConfig, LLMEngine
+    from aurora_amd.engine.sampling_params import SamplingParams
+
+    prompts = [list(range(10, 200)), list(range(300, 350)), [7, 8, 9]]
+    p = SamplingParams(temperature=0.0, max_tokens=16, ignore_eos=True)
+    gpu = LLMEngine(EngineConfig(model="smoke-llama", device="cuda", seed=11,
+                                 kv_cache_pages=64))
+    outs_gpu = gpu.generate(prompts, p)
+    cpu = LLMEngine(EngineConfig(model="smoke-llama", device="cpu", seed=11))
+    outs_cpu = cpu.generate(prompts, p)
+    # bf16 vs fp32 on random weights: token-level agreement is statistical;
+    # compare the prefill logits directly instead of token ids
+    import torch as _t
+    ids = _t.tensor(prompts[0], device="cuda")
+    from aurora_amd.models.llama import AttnMetadata
+    L = len(prompts[0])
+    meta = AttnMetadata(
+        mode="prefill",
+        positions=_t.arange(L, dtype=_t.int32, device="cuda"),
+        slot_mapping=_t.arange(L, dtype=_t.int32, device="cuda"),
+        block_tables=_t.arange(1, dtype=_t.int32, device="cuda").unsqueeze(0),
+        context_lens=_t.tensor([L], dtype=_t.int32, device="cuda"),
+        cu_seqlens_q=_t.tensor([0, L], dtype=_t.int32, device="cuda"),
+    )
+    kv = [(c[0].clone(), c[1].clone()) for c in gpu.kv.caches]
+    hidden = gpu.model(ids, kv, meta)
+    logits = gpu.model.compute_logits(hidden)
+    meta_cpu = AttnMetadata(
+        mode="prefill",
+        positions=_t.arange(L, dtype=_t.int32),
+        slot_mapping=_t.arange(L, dtype=_t.int32),
+        block_tables=_t.arange(16, dtype=_t.int32).unsqueeze(0),
+        context_lens=_t.tensor([L], dtype=_t.int32),
+        cu_seqlens_q=_t.tensor([0, L], dtype=_t.int32),
+    )
+    P = cpu.kv.page_size
+    npage = (L + P - 1) // P
+    meta_cpu.block_tables = _t.arange(npage, dtype=_t.int32).unsqueeze(0)
+    kv_cpu = [(c[0].clone(), c[1].clone()) for c in cpu.kv.caches]
+    hidden_cpu = cpu.model(_t.tensor(prompts[0]), kv_cpu, meta_cpu)
+    logits_cpu = cpu.model.compute_logits(hidden_cpu)
+    a = logits[-1].float().cpu()
+    b = logits_cpu[-1].float()
+    cos = torch.nn.functional.cosine_similarity(a, b, dim=0)
+    assert cos.item() > 0.98, cos.item()
+    # GPU generation must be deterministic under greedy
+    gpu2 = LLMEngine(EngineConfig(model="smoke-llama", device="cuda", seed=11,
+                                  kv_cache_pages=64))
+    assert gpu2.generate(prompts, p) == outs_gpu
+
+
+def test_engine_gpu_grammar_json():
+    from aurora_amd.engine.engine import EngineConfig, LLMEngine
+    from aurora_amd.engine.grammar import JsonGrammar, validate_json_bytes
+    from aurora_amd.engine.sampling_params import SamplingParams
+
+    eng = LLMEngine(EngineConfig(model="smoke-llama", device="cuda", seed=5,
+                                 kv_cache_pages=64))
+    p = SamplingParams(temperature=0.8, max_tokens=200, grammar=JsonGrammar())
+    out = eng.generate([[1, 2, 3]], p)[0]
+    data = bytes(t for t in out if t < 256)
+    assert validate_json_bytes(data), data
