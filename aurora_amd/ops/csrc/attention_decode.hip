@@ -113,8 +113,8 @@ __global__ __launch_bounds__(256) void paged_decode_kernel(
       Lsum += w * s_l[g][i];
       O += w * s_o[g][i][d];
     }
-    out[(((long)b * Hkv + kvh) * G + g) * D + d] =
-        __hip_bfloat16(f32_to_bf16(O / Lsum));
+    *(unsigned short*)(out + (((long)b * Hkv + kvh) * G + g) * D + d) =
+        f32_to_bf16(O / Lsum);
   }
 }
 

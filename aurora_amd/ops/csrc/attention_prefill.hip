@@ -51,6 +51,10 @@ __global__ __launch_bounds__(256, 1) void prefill_kernel(
   constexpr int NC = D / 16;        // Q^T fragments (MFMA K-chunks of QK^T)
   constexpr int NDT = D / 32;       // 32-wide d-tiles of O
   constexpr int VROW = KVB + 8;     // padded V^T row length (elements)
+  // K-tile XOR swizzle mask: a row has D/8 16-B slots, so the swizzle may
+  // only permute within that many slots (at D=64, (row&15)<<4 would escape
+  // the 128-B row)
+  constexpr int SWZ = (D / 8) - 1;
   const int nb = blockIdx.x;
   const int h = blockIdx.y;
   const int kvh = h / (Hq / Hkv);
@@ -125,7 +129,7 @@ __global__ __launch_bounds__(256, 1) void prefill_kernel(
         kv8 = *reinterpret_cast<const shortx8*>(kc + base);
         vv8 = *reinterpret_cast<const shortx8*>(vc + base);
       }
-      const int kb = (row * D * 2 + ((d0 * 2) ^ ((row & 15) << 4))) / 2;
+      const int kb = (row * D * 2 + ((d0 * 2) ^ ((row & SWZ) << 4))) / 2;
       *reinterpret_cast<shortx8*>(kbuf + kb) = kv8;
 #pragma unroll
       for (int j = 0; j < 8; ++j) vtbuf[(d0 + j) * VROW + row] = vv8[j];
@@ -146,7 +150,7 @@ __global__ __launch_bounds__(256, 1) void prefill_kernel(
 #pragma unroll
       for (int c = 0; c < NC; ++c) {
         const int kb =
-            (krow * D * 2 + (((c * 16 + half * 8) * 2) ^ ((krow & 15) << 4))) / 2;
+            (krow * D * 2 + (((c * 16 + half * 8) * 2) ^ ((krow & SWZ) << 4))) / 2;
         frag_u kf;
         const shortx8 kv = *reinterpret_cast<const shortx8*>(kbuf + kb);
         kf.u[0] = ((const unsigned int*)&kv)[0];

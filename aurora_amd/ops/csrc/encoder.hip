@@ -78,7 +78,7 @@ __global__ __launch_bounds__(256) void varlen_attn_kernel(
       L += w * s_l[wave][i];
       O += w * s_o[wave][i][d];
     }
-    out[((long)tq * Hq + h) * D + d] = __hip_bfloat16(f32_to_bf16(O / L));
+    *(unsigned short*)(out + ((long)tq * Hq + h) * D + d) = f32_to_bf16(O / L);
   }
 }
 
@@ -156,7 +156,7 @@ __global__ void embed_pool_kernel(const __hip_bfloat16* __restrict__ hidden,
   for (int c = 0; c < nchunk; ++c) {
     const int d = tid + c * blockDim.x;
     if (d < H)
-      out[(long)seq * H + d] = __hip_bfloat16(f32_to_bf16(acc[c] * s_inv));
+      *(unsigned short*)(out + (long)seq * H + d) = f32_to_bf16(acc[c] * s_inv);
   }
 }
 
