@@ -1,0 +1,81 @@
+"""LLMManager: the model registry + invoke surface for the agent stack.
+
+The reference's ModelConfig maps purposes (MAIN/RCA/ORCHESTRATOR/SUBAGENT/
+SUMMARIZATION/GUARDRAILS) to rented provider models (reference:
+llm.py:32-67); here every purpose resolves to a purpose-tag over the shared
+in-process engine (one engine per GPU; different purposes share weights,
+which is the right MI355X design: 288 GB holds the model once, concurrency
+comes from continuous batching, not from more model copies).  Usage is
+tracked per call into llm_usage (reference: llm.py:270-350).
+"""
+from __future__ import annotations
+
+import time
+from dataclasses import dataclass, field
+from typing import Any, Dict, Optional
+
+from ..engine.sampling_params import SamplingParams
+from ..engine.server import AsyncEngineServer, GenResult
+
+
+@dataclass
+class ModelConfig:
+    main_model: str = "tiny-llama"
+    rca_model: str = ""
+    orchestrator_model: str = ""
+    subagent_model: str = ""
+    summarization_model: str = ""
+    guardrails_model: str = ""
+    vision_model: str = ""
+
+    def resolve(self, purpose: str) -> str:
+        return getattr(self, f"{purpose}_model", "") or self.main_model
+
+
+class LLMManager:
+    def __init__(self, server: AsyncEngineServer, config: Optional[ModelConfig] = None,
+                 store=None):
+        self.server = server
+        self.config = config or ModelConfig()
+        self.store = store  # OrgStore for usage tracking (optional)
+
+    @property
+    def tokenizer(self):
+        return self.server.tokenizer()
+
+    def count_tokens(self, text: str) -> int:
+        return self.tokenizer.count(text)
+
+    async def invoke(self, prompt: str, params: Optional[SamplingParams] = None,
+                     purpose: str = "main", session_id: Optional[str] = None) -> GenResult:
+        params = params or SamplingParams(temperature=0.2, max_tokens=512)
+        t0 = time.monotonic()
+        res = await self.server.complete_text(prompt, params, session_id=session_id)
+        if self.store is not None:
+            self.store.track_usage(
+                model=self.config.resolve(purpose),
+                prompt_tokens=res.prompt_tokens,
+                completion_tokens=len(res.token_ids),
+                response_ms=(time.monotonic() - t0) * 1000.0,
+                purpose=purpose, session_id=session_id)
+        return res
+
+    async def complete_text(self, prompt: str, params: SamplingParams,
+                            session_id: Optional[str] = None) -> GenResult:
+        """Raw passthrough (guardrail judges call this shape)."""
+        return await self.invoke(prompt, params, purpose="guardrails",
+                                 session_id=session_id)
+
+    async def summarize(self, text: str, max_tokens: int = 256,
+                        instruction: str = "Summarize the following:") -> str:
+        """Tool-output summarisation with context capping
+        (reference: llm.py:356-445)."""
+        budget = 6000
+        if self.count_tokens(text) > budget:
+            # head+tail cap before summarising (reference: tool_output_cap.py:22)
+            text = text[: budget * 2] + "\n...[truncated]...\n" + text[-budget:]
+        res = await self.invoke(
+            f"{instruction}\n---\n{text}\n---\nSummary:",
+            SamplingParams(temperature=0.0, max_tokens=max_tokens, stop=["\n---"]),
+            purpose="summarization")
+        return res.text.strip()

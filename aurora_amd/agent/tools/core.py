@@ -1,0 +1,297 @@
+"""Built-in agent tools (reference: chat/backend/agent/tools/* — the core
+execution, introspection, knowledge and artifact surface rebuilt)."""
+from __future__ import annotations
+
+import asyncio
+import json
+import os
+import shlex
+import tempfile
+import time
+from typing import Any, Dict, Optional
+
+from ...guardrails.command_safety import evaluate_command
+from .registry import GLOBAL_REGISTRY as R
+from .registry import ToolContext
+
+# --------------------------------------------------------------- execution
+
+
+@R.add("terminal_exec",
+       "Run a shell command in the sandboxed diagnostics environment "
+       "(read-only infra commands; every command is safety-gated).",
+       {"command": "the shell command to run"}, read_only=False)
+async def terminal_exec(ctx: ToolContext, command: str, timeout: int = 60) -> str:
+    verdict = await evaluate_command(command, llm=ctx.llm, org_store=ctx.org_store,
+                                     user_id=ctx.user_id)
+    if not verdict.safe:
+        return (f"BLOCKED by guardrail ({verdict.source}"
+                f"{':' + verdict.rule_id if verdict.rule_id else ''}): {verdict.reason}")
+    # sandbox: clean env, temp cwd, hard timeout (the reference runs these in
+    # per-user K8s pods — terminal_pod_manager.py:68; locally we subprocess
+    # with a stripped environment)
+    env = {"PATH": "/usr/bin:/bin:/usr/sbin:/sbin", "HOME": tempfile.gettempdir(),
+           "LANG": "C.UTF-8"}
+    try:
+        proc = await asyncio.create_subprocess_shell(
+            command, stdout=asyncio.subprocess.PIPE, stderr=asyncio.subprocess.PIPE,
+            env=env, cwd=tempfile.gettempdir())
+        out, err = await asyncio.wait_for(proc.communicate(), timeout=timeout)
+        text = out.decode("utf-8", "replace")
+        if err:
+            text += "\n[stderr]\n" + err.decode("utf-8", "replace")
+        return text or f"(exit {proc.returncode}, no output)"
+    except asyncio.TimeoutError:
+        try:
+            proc.kill()
+        except ProcessLookupError:
+            pass
+        return f"ERROR: command timed out after {timeout}s"
+
+
+@R.add("kubectl",
+       "Run a kubectl command against the connected cluster (safety-gated).",
+       {"args": "kubectl arguments, e.g. 'get pods -n default'"}, read_only=False)
+async def kubectl(ctx: ToolContext, args: str) -> str:
+    return await terminal_exec(ctx, f"kubectl {args}")
+
+
+@R.add("cloud_exec",
+       "Run a cloud-provider CLI command (aws/az/gcloud...) with isolated "
+       "credentials (safety-gated).",
+       {"command": "full CLI command starting with the provider binary"},
+       read_only=False)
+async def cloud_exec(ctx: ToolContext, command: str) -> str:
+    binname = shlex.split(command)[0] if command.strip() else ""
+    if binname not in ("aws", "az", "gcloud", "scw", "ovhcli", "tailscale", "flyctl"):
+        return "ERROR: cloud_exec only accepts aws/az/gcloud/scw/ovhcli/tailscale/flyctl"
+    return await terminal_exec(ctx, command)
+
+
+# ------------------------------------------------------------ introspection
+# (reference: tools/introspection_tools.py:101-341)
+
+
+@R.add("list_incidents", "List recent incidents for this org.",
+       {"status": "optional filter: open|investigating|resolved"})
+async def list_incidents(ctx: ToolContext, status: str = "") -> str:
+    if ctx.org_store is None:
+        return "ERROR: no data store bound"
+    rows = ctx.org_store.list_incidents(status or None, limit=20)
+    return json.dumps([
+        {k: r[k] for k in ("id", "title", "source", "severity", "status",
+                           "aurora_status", "created_at")} for r in rows], indent=1)
+
+
+@R.add("get_incident", "Full record of one incident incl. summary and root cause.",
+       {"incident_id": "the incident id"})
+async def get_incident(ctx: ToolContext, incident_id: str) -> str:
+    row = ctx.org_store.get_incident(incident_id)
+    return json.dumps(row, indent=1) if row else f"ERROR: no incident {incident_id}"
+
+
+@R.add("get_incident_timeline", "Lifecycle events + tool steps of an incident.",
+       {"incident_id": "the incident id"})
+async def get_incident_timeline(ctx: ToolContext, incident_id: str) -> str:
+    ev = ctx.org_store.lifecycle_events(incident_id)
+    steps = ctx.org_store.steps(incident_id)
+    return json.dumps({"events": ev, "steps": [
+        {k: s[k] for k in ("tool", "status", "started_at", "finished_at")}
+        for s in steps]}, indent=1)
+
+
+@R.add("get_alert_field", "Read a field from the triggering alert payload.",
+       {"field": "dot-path into the alert payload, e.g. 'alert.severity'"})
+async def get_alert_field(ctx: ToolContext, field: str = "") -> str:
+    if not ctx.incident_id:
+        return "ERROR: no incident bound to this session"
+    inc = ctx.org_store.get_incident(ctx.incident_id)
+    if not inc:
+        return "ERROR: incident not found"
+    doc: Any = json.loads(inc.get("payload") or "{}")
+    if not field:
+        return json.dumps(doc, indent=1)[:8000]
+    for part in field.split("."):
+        if isinstance(doc, dict) and part in doc:
+            doc = doc[part]
+        else:
+            return f"ERROR: field '{field}' not present"
+    return json.dumps(doc) if isinstance(doc, (dict, list)) else str(doc)
+
+
+@R.add("list_services", "List services known to the infrastructure graph.")
+async def list_services(ctx: ToolContext) -> str:
+    if ctx.graph is None:
+        return "ERROR: no infrastructure graph available"
+    return json.dumps(ctx.graph.list_services(), indent=1)
+
+
+@R.add("get_service_impact",
+       "Downstream impact radius of a service from the dependency graph.",
+       {"service": "service name"})
+async def get_service_impact(ctx: ToolContext, service: str) -> str:
+    if ctx.graph is None:
+        return "ERROR: no infrastructure graph available"
+    return json.dumps(ctx.graph.impact_radius(service), indent=1)
+
+
+@R.add("get_infrastructure_context",
+       "Dependency-graph context around a service (up/downstream, SPOFs).",
+       {"service": "service name (empty = whole-graph overview)"})
+async def get_infrastructure_context(ctx: ToolContext, service: str = "") -> str:
+    if ctx.graph is None:
+        return "ERROR: no infrastructure graph available"
+    if not service:
+        return json.dumps(ctx.graph.overview(), indent=1)
+    return json.dumps({
+        "upstream": ctx.graph.upstream(service),
+        "downstream": ctx.graph.downstream(service),
+        "spof": ctx.graph.single_points_of_failure(),
+    }, indent=1)
+
+
+@R.add("save_infrastructure_context",
+       "Record a discovered service or dependency into the infra graph.",
+       {"service": "service name", "depends_on": "optional dependency name",
+        "kind": "node kind, e.g. service/database/queue"}, read_only=False)
+async def save_infrastructure_context(ctx: ToolContext, service: str,
+                                      depends_on: str = "", kind: str = "service") -> str:
+    if ctx.graph is None:
+        return "ERROR: no infrastructure graph available"
+    ctx.graph.upsert_service(service, kind=kind)
+    if depends_on:
+        ctx.graph.upsert_service(depends_on)
+        ctx.graph.add_dependency(service, depends_on)
+    return f"saved: {service}" + (f" -> {depends_on}" if depends_on else "")
+
+
+# ----------------------------------------------------------- knowledge base
+
+
+@R.add("knowledge_base_search",
+       "Hybrid (semantic + keyword) search over the org knowledge base / runbooks.",
+       {"query": "search query", "k": "number of results (default 5)"})
+async def knowledge_base_search(ctx: ToolContext, query: str, k: int = 5) -> str:
+    if ctx.rag is None:
+        return "ERROR: knowledge base not available"
+    hits = ctx.rag.search(query, k=int(k))
+    if not hits:
+        return "No matching knowledge-base content."
+    return "\n\n".join(
+        f"[{i+1}] (score {h['score']:.3f}) {h['doc_name']}:\n{h['text']}"
+        for i, h in enumerate(hits))
+
+
+@R.add("web_search", "Search the web for relevant context.",
+       {"query": "search query"})
+async def web_search(ctx: ToolContext, query: str) -> str:
+    # The reference proxies a self-hosted SearXNG (web_search_service.py:162);
+    # this environment has no egress, so fall back to the knowledge base and
+    # say so explicitly rather than fabricate results.
+    if ctx.rag is not None:
+        hits = ctx.rag.search(query, k=3)
+        if hits:
+            return ("(no internet egress; nearest knowledge-base matches)\n\n"
+                    + "\n\n".join(h["text"][:500] for h in hits))
+    return "ERROR: web search unavailable (no egress) and knowledge base empty"
+
+
+# -------------------------------------------------------------- RCA / cards
+
+
+@R.add("trigger_rca",
+       "Open a new incident and dispatch a background RCA investigation for it.",
+       {"title": "incident title", "details": "what happened"}, read_only=False)
+async def trigger_rca(ctx: ToolContext, title: str, details: str = "") -> str:
+    if ctx.depth > 0:
+        return "ERROR: nested RCA triggers are blocked"  # trigger_rca_tool.py:38
+    if ctx.org_store is None or ctx.queue is None:
+        return "ERROR: background pipeline unavailable"
+    iid = ctx.org_store.create_incident(
+        title=title, source="agent", payload={"details": details},
+        user_id=ctx.user_id)
+    ctx.queue.enqueue("run_background_chat",
+                      {"org_id": ctx.org_store.org_id, "incident_id": iid,
+                       "prompt": f"{title}\n{details}", "user_id": ctx.user_id},
+                      dedup_key=f"rca:{iid}")
+    return f"RCA dispatched for incident {iid}"
+
+
+@R.add("get_postmortem", "Fetch the postmortem document of an incident.",
+       {"incident_id": "the incident id"})
+async def get_postmortem(ctx: ToolContext, incident_id: str) -> str:
+    pm = ctx.org_store.get_postmortem(incident_id)
+    return pm["content"] if pm else f"No postmortem for {incident_id} yet."
+
+
+@R.add("save_postmortem", "Save/overwrite the postmortem document of an incident.",
+       {"incident_id": "the incident id", "content": "markdown postmortem"},
+       read_only=False)
+async def save_postmortem(ctx: ToolContext, incident_id: str, content: str) -> str:
+    pid = ctx.org_store.save_postmortem(incident_id, content)
+    return f"postmortem saved ({pid})"
+
+
+# ---------------------------------------------------------------- artifacts
+
+
+@R.add("artifact_list", "List persistent investigation documents.")
+async def artifact_list(ctx: ToolContext) -> str:
+    rows = ctx.org_store.artifacts(ctx.session_id)
+    return json.dumps([{"name": r["name"], "updated_at": r["updated_at"]}
+                       for r in rows], indent=1)
+
+
+@R.add("artifact_read", "Read a persistent investigation document.",
+       {"name": "artifact name"})
+async def artifact_read(ctx: ToolContext, name: str) -> str:
+    row = ctx.org_store.get_artifact(name)
+    return row["content"] if row else f"ERROR: no artifact '{name}'"
+
+
+@R.add("artifact_write", "Create or update a persistent investigation document.",
+       {"name": "artifact name", "content": "full document content"},
+       read_only=False)
+async def artifact_write(ctx: ToolContext, name: str, content: str) -> str:
+    ctx.org_store.save_artifact(name, content, session_id=ctx.session_id)
+    return f"artifact '{name}' saved ({len(content)} chars)"
+
+
+# -------------------------------------------------------------------- skills
+
+
+@R.add("load_skill", "Load an investigation skill/playbook by name.",
+       {"name": "skill name from the skill catalog"})
+async def load_skill(ctx: ToolContext, name: str) -> str:
+    from ..skills.registry import get_skill_registry
+    sk = get_skill_registry().get(name)
+    return sk.body if sk else f"ERROR: unknown skill '{name}'"
+
+
+# ----------------------------------------------------- connector-gated stubs
+# The provider API-client tools of the reference (datadog/sentry/splunk/...)
+# need live provider endpoints; the registration + gating contract is kept
+# and each tool degrades with an explicit error when unreachable.
+
+def _provider_tool(name: str, provider: str, desc: str):
+    @R.add(name, desc, {"query": "provider query"}, requires_connection=provider)
+    async def _tool(ctx: ToolContext, query: str, _p=provider) -> str:
+        return (f"ERROR: {_p} API unreachable from this deployment "
+                f"(no egress); connection is configured but the call failed")
+    return _tool
+
+
+_provider_tool("query_datadog", "datadog", "Query Datadog metrics/logs/monitors.")
+_provider_tool("query_newrelic", "newrelic", "Run a NRQL query against New Relic.")
+_provider_tool("query_sentry", "sentry", "Search Sentry issues/events.")
+_provider_tool("search_splunk", "splunk", "Run a Splunk search.")
+_provider_tool("query_dynatrace", "dynatrace", "Query Dynatrace problems/metrics.")
+_provider_tool("query_grafana", "grafana", "Query Grafana dashboards/datasources.")
+_provider_tool("query_cloudwatch", "aws", "Query CloudWatch metrics/logs.")
+_provider_tool("github_rca", "github", "Correlate incident window with recent commits/PRs.")
+_provider_tool("github_fix", "github", "Draft a fix suggestion PR for the incident.")
+_provider_tool("jenkins_rca", "jenkins", "Analyze recent Jenkins build failures.")
+_provider_tool("query_slack", "slack", "Search Slack channels/threads for context.")
+_provider_tool("query_jira", "jira", "Search Jira issues.")
+_provider_tool("notion_search", "notion", "Search Notion workspace pages.")
+_provider_tool("confluence_search", "confluence", "Search Confluence pages/runbooks.")

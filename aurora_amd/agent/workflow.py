@@ -1,0 +1,193 @@
+"""The agent workflow: guardrailed ReAct loop over the in-process engine.
+
+Rebuilds the reference's Workflow/Agent pair (reference: workflow.py:90,942
+Workflow.stream; agent.py:251 agentic_tool_flow): context load, concurrent
+input-rail check awaited before the first model call (fail closed), the
+Thought->Action(JSON)->Observation loop bounded by AGENT_RECURSION_LIMIT,
+execution capture, context persistence, and an event stream the chat/WS and
+background paths both consume.
+
+Event dicts: {"type": "thought"|"token"|"tool_call"|"tool_result"|"final"|
+"blocked"|"error"|"usage", ...}
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import time
+from typing import Any, AsyncIterator, Dict, List, Optional
+
+from ..engine.grammar import JsonGrammar
+from ..engine.sampling_params import SamplingParams
+from ..guardrails.input_rail import check_input
+from ..utils.config import settings as get_settings
+from .llm import LLMManager
+from .prompts import assemble_system_prompt, build_prompt_segments, render_history
+from .skills.registry import get_skill_registry
+from .state import State
+from .tools.registry import GLOBAL_REGISTRY, ToolContext, ToolRegistry
+
+
+class Workflow:
+    def __init__(self, llm: LLMManager, ctx: ToolContext,
+                 registry: Optional[ToolRegistry] = None,
+                 tool_subset: Optional[List[str]] = None,
+                 system_prompt_override: Optional[str] = None,
+                 max_turns: Optional[int] = None):
+        self.llm = llm
+        self.ctx = ctx
+        self.registry = registry or GLOBAL_REGISTRY
+        self.tool_subset = tool_subset
+        self.system_prompt_override = system_prompt_override
+        s = get_settings()
+        # recursion limit counts model+tool super-steps (reference
+        # AGENT_RECURSION_LIMIT=240 ~ 120 tool turns)
+        self.max_turns = max_turns if max_turns is not None else s.agent_recursion_limit // 2
+        self.settings = s
+
+    # ------------------------------------------------------------------ run
+
+    async def stream(self, state: State) -> AsyncIterator[Dict[str, Any]]:
+        store = self.ctx.org_store
+        t_start = time.monotonic()
+        # 1. load persisted context (reference: workflow.py:975-1002)
+        if store is not None and state.session_id and not state.messages:
+            state.messages = store.load_context_history(state.session_id)
+        state.messages.append({"role": "user", "content": state.question})
+        if store is not None and state.session_id:
+            store.append_session_messages(
+                state.session_id, [{"role": "user", "content": state.question}])
+
+        # 2. input rail fired concurrently, awaited before the first model
+        # call (reference: workflow.py:1076, agent.py:875-910); FAIL CLOSED
+        rail_task = asyncio.create_task(check_input(
+            state.question, llm=self.llm,
+            enabled=self.settings.guardrails_enabled))
+
+        # 3. prompt assembly concurrent with the rail
+        tools = self.registry.available(self.ctx, subset=self.tool_subset)
+        connected = []
+        if store is not None and state.user_id:
+            connected = store.connected_providers(state.user_id)
+        if self.system_prompt_override:
+            system = self.system_prompt_override
+        else:
+            skills_text = get_skill_registry().assemble_rca_prompt(
+                self.settings.rca_skills_token_budget,
+                self.llm.count_tokens, connected)
+            alert_ctx = ""
+            if store is not None and state.incident_id:
+                inc = store.get_incident(state.incident_id)
+                if inc:
+                    alert_ctx = json.dumps({
+                        "incident_id": inc["id"], "title": inc["title"],
+                        "source": inc["source"], "severity": inc["severity"],
+                        "payload": json.loads(inc.get("payload") or "{}"),
+                    }, indent=1)[:6000]
+            system = assemble_system_prompt(
+                build_prompt_segments(state, tools, skills_text, alert_ctx))
+
+        rail = await rail_task
+        if not rail.allowed:
+            state.guardrail_blocked = True
+            if store is not None:
+                store.audit("input_rail_block", json.dumps(
+                    {"classification": rail.classification, "reason": rail.reason}))
+            yield {"type": "blocked", "classification": rail.classification,
+                   "reason": rail.reason, "latency_ms": rail.latency_ms}
+            self._persist(state)
+            return
+
+        tool_names = {t.name for t in tools} | {"final_answer"}
+        final_text: Optional[str] = None
+        for turn in range(self.max_turns):
+            if (time.monotonic() - t_start) > self.settings.workflow_timeout_s:
+                yield {"type": "error", "error": "workflow timeout"}
+                break
+            prompt = (system + "\n\n# Investigation so far\n"
+                      + render_history(state.messages) + "\nThought:")
+            # --- thought (streamed) ---
+            thought_parts: List[str] = []
+            async for out in self.llm.server.stream(
+                    self.llm.tokenizer.encode(prompt, add_bos=True),
+                    SamplingParams(temperature=0.3, max_tokens=160,
+                                   stop=["\nAction", "\nObservation", "\nThought"],
+                                   ignore_eos=False),
+                    session_id=state.session_id):
+                if out.new_text:
+                    thought_parts.append(out.new_text)
+                    yield {"type": "token", "content": out.new_text}
+            thought = "".join(thought_parts).strip()
+            state.messages.append({"role": "thought", "content": thought})
+            yield {"type": "thought", "content": thought}
+            if store is not None and state.incident_id:
+                store.add_thought(state.incident_id, thought)
+
+            # --- action: grammar-constrained tool JSON ---
+            act_prompt = prompt + " " + thought + "\nAction:"
+            action_raw = ""
+            action: Optional[dict] = None
+            for attempt in range(3):  # retry malformed/unknown (agent.py:1031)
+                res = await self.llm.invoke(
+                    act_prompt,
+                    SamplingParams(temperature=0.1 + 0.3 * attempt, max_tokens=256,
+                                   grammar=JsonGrammar()),
+                    purpose="main", session_id=state.session_id)
+                action_raw = res.text.strip()
+                try:
+                    doc = json.loads(action_raw)
+                except json.JSONDecodeError:
+                    continue
+                if isinstance(doc, dict) and doc.get("tool") in tool_names:
+                    action = doc
+                    break
+                act_prompt += (f" {action_raw}\nObservation: ERROR unknown tool; "
+                               f"valid tools: {sorted(tool_names)}\nAction:")
+            if action is None:
+                state.messages.append({
+                    "role": "observation",
+                    "content": f"ERROR: could not produce a valid action "
+                               f"(last: {action_raw[:200]})"})
+                continue
+
+            name = action["tool"]
+            args = action.get("args") or {}
+            if not isinstance(args, dict):
+                args = {"value": args}
+            state.messages.append({"role": "action", "tool": name, "args": args})
+            if name == "final_answer":
+                final_text = str(args.get("answer", "")).strip() or thought
+                break
+            yield {"type": "tool_call", "tool": name, "args": args}
+            output = await self.registry.run(self.ctx, name, args)
+            state.messages.append({"role": "observation", "content": output})
+            yield {"type": "tool_result", "tool": name,
+                   "output": output[:2000]}
+        else:
+            yield {"type": "error", "error": "recursion limit reached"}
+
+        if final_text is None:
+            # best-effort summary when the loop ended without a final answer
+            final_text = await self._forced_final(system, state)
+        state.messages.append({"role": "assistant", "content": final_text})
+        yield {"type": "final", "content": final_text}
+        yield {"type": "usage", "counters": self.llm.server.counters(),
+               "elapsed_ms": (time.monotonic() - t_start) * 1000.0}
+        self._persist(state)
+
+    async def _forced_final(self, system: str, state: State) -> str:
+        res = await self.llm.invoke(
+            system + "\n\n# Investigation so far\n" + render_history(state.messages)
+            + "\nWrite the final RCA report now.\nFinal answer:",
+            SamplingParams(temperature=0.2, max_tokens=512),
+            purpose="rca", session_id=state.session_id)
+        return res.text.strip()
+
+    def _persist(self, state: State) -> None:
+        store = self.ctx.org_store
+        if store is None or not state.session_id:
+            return
+        store.save_context_history(state.session_id, state.messages)
+        ui = [m for m in state.messages
+              if m.get("role") in ("assistant", "thought", "action")]
+        store.append_session_messages(state.session_id, ui[-4:])

@@ -1,0 +1,119 @@
+"""Async front-end for the engine: one background stepping thread, many
+concurrent awaitable generations.
+
+This is the in-process replacement for the reference's provider HTTP
+boundary *and* its Celery worker concurrency: every concurrent
+investigation/chat turn is just another sequence in the same continuously
+batched engine (SURVEY.md §2.12 row 'Concurrent investigations')."""
+from __future__ import annotations
+
+import asyncio
+import queue as _queue
+import threading
+import time
+import uuid
+from dataclasses import dataclass, field
+from typing import AsyncIterator, Dict, List, Optional
+
+from .engine import EngineConfig, LLMEngine, StepOutput
+from .sampling_params import SamplingParams
+
+
+@dataclass
+class GenResult:
+    token_ids: List[int] = field(default_factory=list)
+    text: str = ""
+    finish_reason: Optional[str] = None
+    prompt_tokens: int = 0
+    elapsed_ms: float = 0.0
+
+
+class AsyncEngineServer:
+    def __init__(self, config: EngineConfig, engine: Optional[LLMEngine] = None):
+        self.engine = engine or LLMEngine(config)
+        self._subs: Dict[str, _queue.SimpleQueue] = {}
+        self._lock = threading.Lock()
+        self._stop = threading.Event()
+        self._wake = threading.Event()
+        self._thread = threading.Thread(target=self._run, daemon=True,
+                                        name="engine-step-loop")
+        self._thread.start()
+
+    # ------------------------------------------------------------- lifecycle
+
+    def _run(self) -> None:
+        while not self._stop.is_set():
+            with self._lock:
+                busy = self.engine.has_work()
+            if not busy:
+                self._wake.wait(timeout=0.02)
+                self._wake.clear()
+                continue
+            with self._lock:
+                outs = self.engine.step()
+            for out in outs:
+                q = self._subs.get(out.request_id)
+                if q is not None:
+                    q.put(out)
+
+    def shutdown(self) -> None:
+        self._stop.set()
+        self._wake.set()
+        self._thread.join(timeout=5)
+
+    # ------------------------------------------------------------------- API
+
+    def tokenizer(self):
+        return self.engine.tokenizer
+
+    async def stream(self, prompt_token_ids: List[int], params: SamplingParams,
+                     session_id: Optional[str] = None,
+                     request_id: Optional[str] = None) -> AsyncIterator[StepOutput]:
+        rid = request_id or f"req-{uuid.uuid4().hex[:12]}"
+        q: _queue.SimpleQueue = _queue.SimpleQueue()
+        self._subs[rid] = q
+        with self._lock:
+            self.engine.add_request(rid, prompt_token_ids, params, session_id=session_id)
+        self._wake.set()
+        loop = asyncio.get_running_loop()
+        finished = False
+        try:
+            while True:
+                out = await loop.run_in_executor(None, q.get)
+                yield out
+                if out.finished:
+                    finished = True
+                    return
+        finally:
+            self._subs.pop(rid, None)
+            if not finished:  # consumer bailed (cancel / WS drop): free the seq
+                with self._lock:
+                    self.engine.abort_request(rid)
+
+    async def generate(self, prompt_token_ids: List[int], params: SamplingParams,
+                       session_id: Optional[str] = None) -> GenResult:
+        t0 = time.monotonic()
+        res = GenResult(prompt_tokens=len(prompt_token_ids))
+        async for out in self.stream(prompt_token_ids, params, session_id):
+            res.token_ids.extend(out.new_token_ids)
+            res.text += out.new_text
+            if out.finished:
+                res.finish_reason = out.finish_reason
+        res.elapsed_ms = (time.monotonic() - t0) * 1000.0
+        return res
+
+    async def complete_text(self, prompt: str, params: SamplingParams,
+                            session_id: Optional[str] = None) -> GenResult:
+        tok = self.engine.tokenizer
+        ids = tok.encode(prompt, add_bos=True)
+        return await self.generate(ids, params, session_id=session_id)
+
+    def abort(self, request_id: str) -> None:
+        with self._lock:
+            self.engine.abort_request(request_id)
+        q = self._subs.pop(request_id, None)
+        if q is not None:
+            q.put(StepOutput(request_id, [], "", True, "aborted"))
+
+    def counters(self) -> Dict[str, int]:
+        return dict(self.engine.counters)

@@ -1,0 +1,132 @@
+"""Process runtime: wires store + queue + engine + per-org services.
+
+The composition root the API server, WS chatbot and background workers all
+share (the reference splits this across main_compute.py, main_chatbot.py
+and celery_config.py worker prewarm — reference: celery_config.py:272-304).
+"""
+from __future__ import annotations
+
+import asyncio
+import collections
+import time
+from typing import Any, Dict, List, Optional
+
+from ..agent.llm import LLMManager, ModelConfig
+from ..agent.tools.registry import GLOBAL_REGISTRY, ToolContext
+from ..engine.engine import EngineConfig
+from ..engine.server import AsyncEngineServer
+from ..graph.infra_graph import InfraGraph
+from ..rag.index import RagIndex
+from ..utils.config import Settings, settings as get_settings
+from .queue import TaskQueue
+from .store import AdminStore, Database, OrgStore
+
+# tool modules register themselves on import
+from ..agent.tools import core as _core_tools  # noqa: F401
+
+
+class EventBus:
+    """In-process pub/sub feeding the SSE stream and WS status pushes
+    (reference: incidents_sse.py:20,34)."""
+
+    def __init__(self):
+        self._subs: Dict[str, List[asyncio.Queue]] = collections.defaultdict(list)
+        self.history: Dict[str, List[dict]] = collections.defaultdict(list)
+
+    def publish(self, org_id: str, event: dict) -> None:
+        event = dict(event, ts=time.time())
+        hist = self.history[org_id]
+        hist.append(event)
+        del hist[:-200]
+        for q in list(self._subs.get(org_id, [])):
+            try:
+                q.put_nowait(event)
+            except asyncio.QueueFull:
+                pass
+
+    def subscribe(self, org_id: str) -> asyncio.Queue:
+        q: asyncio.Queue = asyncio.Queue(maxsize=500)
+        self._subs[org_id].append(q)
+        return q
+
+    def unsubscribe(self, org_id: str, q: asyncio.Queue) -> None:
+        try:
+            self._subs[org_id].remove(q)
+        except ValueError:
+            pass
+
+
+class Runtime:
+    def __init__(self, cfg: Optional[Settings] = None,
+                 engine_server: Optional[AsyncEngineServer] = None,
+                 engine_config: Optional[EngineConfig] = None):
+        self.settings = cfg or get_settings()
+        self.db = Database(self.settings.db_path)
+        self.admin = AdminStore(self.db)
+        self.queue = TaskQueue(self.db, concurrency=self.settings.worker_concurrency,
+                               default_time_limit_s=self.settings.workflow_timeout_s)
+        self.events = EventBus()
+        if engine_server is None:
+            ecfg = engine_config or EngineConfig(
+                model=self.settings.main_model,
+                device=self.settings.engine_device,
+                kv_memory_fraction=self.settings.kv_memory_fraction,
+                max_num_seqs=self.settings.max_num_seqs,
+                tp_size=self.settings.tp_size,
+            )
+            engine_server = AsyncEngineServer(ecfg)
+        self.engine = engine_server
+        self._rag: Dict[str, RagIndex] = {}
+        self._graph: Dict[str, InfraGraph] = {}
+        self._rca_rate: Dict[str, float] = {}  # user -> last background RCA ts
+        from . import worker as _worker  # registers task handlers
+        _worker.register_tasks(self)
+
+    # ---- per-org services ----
+    def org(self, org_id: str) -> OrgStore:
+        return self.admin.org_store(org_id)
+
+    def llm(self, org_id: str) -> LLMManager:
+        return LLMManager(self.engine, ModelConfig(
+            main_model=self.settings.main_model), store=self.org(org_id))
+
+    def rag(self, org_id: str) -> RagIndex:
+        if org_id not in self._rag:
+            self._rag[org_id] = RagIndex.from_store(
+                self.org(org_id), tokenizer=self.engine.tokenizer())
+        return self._rag[org_id]
+
+    def graph(self, org_id: str) -> InfraGraph:
+        if org_id not in self._graph:
+            self._graph[org_id] = InfraGraph(self.org(org_id))
+        return self._graph[org_id]
+
+    def tool_ctx(self, org_id: str, user_id: Optional[str] = None,
+                 session_id: Optional[str] = None,
+                 incident_id: Optional[str] = None, mode: str = "agent",
+                 notifier=None, depth: int = 0) -> ToolContext:
+        store = self.org(org_id)
+        return ToolContext(
+            org_store=store, user_id=user_id, session_id=session_id,
+            incident_id=incident_id, llm=self.llm(org_id), mode=mode,
+            rag=self.rag(org_id), graph=self.graph(org_id), queue=self.queue,
+            notifier=notifier, settings=self.settings, depth=depth)
+
+    # ---- RCA trigger rate limit (reference: task.py:336) ----
+    def rca_allowed(self, user_key: str) -> bool:
+        now = time.time()
+        last = self._rca_rate.get(user_key, 0.0)
+        if now - last < self.settings.rca_rate_window_s:
+            return False
+        self._rca_rate[user_key] = now
+        return True
+
+    def start(self) -> None:
+        """Start workers + beats (call from a running event loop)."""
+        # beat schedules (reference: celery_config.py:113-146)
+        self.queue.add_beat("cleanup_stale_investigations", 300.0, {})
+        self.queue.add_beat("action_scheduler", 60.0, {})
+        self.queue.start()
+
+    async def stop(self) -> None:
+        await self.queue.stop()

@@ -1,0 +1,195 @@
+"""Background investigation pipeline: the run_background_chat rebuild.
+
+Task flow mirrors the reference exactly (reference:
+chat/background/task.py:447-697,1322-1440 and SURVEY.md §3.2): dedup ->
+incident linking -> aurora_status='running' + investigation_started_at
+(the MTTD stamp) -> lifecycle event -> on_incident actions -> the headless
+agent workflow -> status update, severity, summary, recommendations,
+citations, postmortem, notifications, SSE.
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import time
+from typing import Any, Dict, Optional
+
+from ..agent.state import State
+from ..agent.workflow import Workflow
+from .postrca import build_postmortem, generate_recommendations, save_citations
+from .summarization import (alert_only_summary, determine_severity,
+                            generate_incident_summary)
+
+
+def build_rca_prompt(source: str, title: str, payload: dict,
+                     prior_rcas: Optional[list] = None) -> str:
+    """reference: rca_prompt_builder.py:336 — alert-field extraction +
+    'Aurora Learn' similar-good-RCA injection."""
+    lines = [
+        f"A {source} alert fired. Investigate it end-to-end and produce an RCA.",
+        f"Title: {title}",
+        f"Alert payload: {json.dumps(payload)[:4000]}",
+    ]
+    if prior_rcas:
+        lines.append("Similar past investigations (for reference):")
+        for r in prior_rcas[:2]:
+            lines.append(f"- {r[:500]}")
+    lines.append("Investigate now. Use tools to gather evidence; finish with "
+                 "final_answer containing the RCA.")
+    return "\n".join(lines)
+
+
+async def run_background_chat(rt, args: Dict[str, Any]) -> dict:
+    org_id = args["org_id"]
+    incident_id = args["incident_id"]
+    user_id = args.get("user_id")
+    store = rt.org(org_id)
+    inc = store.get_incident(incident_id)
+    if inc is None:
+        return {"error": "incident not found"}
+    if inc["aurora_status"] == "running":
+        return {"skipped": "already running"}  # dedup (task.py:488-499)
+
+    session_id = inc.get("aurora_chat_session_id") or store.create_session(
+        user_id=user_id, is_background=True, incident_id=incident_id)
+    t_pickup = time.time()
+    store.update_incident(
+        incident_id, aurora_status="running", status="investigating",
+        aurora_chat_session_id=session_id,
+        investigation_started_at=t_pickup)  # MTTD stamp (task.py:620-626)
+    store.lifecycle(incident_id, "rca_started", {"session_id": session_id})
+    rt.events.publish(org_id, {"type": "incident_update", "incident_id": incident_id,
+                               "aurora_status": "running"})
+    # on_incident actions (reference: services/actions/executor.py:111)
+    rt.queue.enqueue("dispatch_on_incident_actions",
+                     {"org_id": org_id, "incident_id": incident_id})
+
+    payload = json.loads(inc.get("payload") or "{}")
+    prompt = args.get("prompt") or build_rca_prompt(
+        inc["source"], inc["title"], payload)
+    llm = rt.llm(org_id)
+    ctx = rt.tool_ctx(org_id, user_id=user_id, session_id=session_id,
+                      incident_id=incident_id, depth=1)  # depth=1 blocks nested RCA
+    wf = Workflow(llm, ctx, max_turns=int(args.get("max_turns", 0)) or None)
+    state = State(question=prompt, user_id=user_id, org_id=org_id,
+                  session_id=session_id, incident_id=incident_id,
+                  incident_start_time=inc["created_at"], is_background=True)
+    final = ""
+    error = None
+    try:
+        async for ev in wf.stream(state):
+            if ev["type"] == "final":
+                final = ev["content"]
+            elif ev["type"] == "blocked":
+                error = f"guardrail blocked: {ev['reason']}"
+            elif ev["type"] == "error":
+                error = ev["error"]
+    except Exception as e:  # noqa: BLE001
+        error = f"workflow failure: {e}"
+
+    # ---- post-RCA stage ----
+    if error and not final:
+        summary = alert_only_summary(inc["title"], payload)
+        store.update_incident(incident_id, aurora_status="failed", summary=summary)
+        store.lifecycle(incident_id, "rca_failed", {"error": error})
+        rt.events.publish(org_id, {"type": "incident_update",
+                                   "incident_id": incident_id,
+                                   "aurora_status": "failed"})
+        return {"error": error}
+
+    try:
+        summary = await generate_incident_summary(llm, state.messages, session_id)
+    except Exception:
+        summary = alert_only_summary(inc["title"], payload)
+    severity = determine_severity(summary, inc.get("severity"))
+    n_cits = save_citations(store, incident_id, state.messages)
+    suggestions = await generate_recommendations(llm, store, incident_id,
+                                                 state.messages)
+    store.update_incident(incident_id, aurora_status="completed",
+                          summary=summary, severity=severity,
+                          root_cause=_extract_root_cause(summary))
+    store.lifecycle(incident_id, "rca_completed",
+                    {"citations": n_cits, "suggestions": len(suggestions)})
+    # auto postmortem (system action)
+    pm = build_postmortem(store.get_incident(incident_id), summary,
+                          store.citations(incident_id), suggestions,
+                          store.lifecycle_events(incident_id))
+    store.save_postmortem(incident_id, pm)
+    rt.events.publish(org_id, {"type": "incident_update", "incident_id": incident_id,
+                               "aurora_status": "completed",
+                               "severity": severity})
+    return {"ok": True, "summary_chars": len(summary), "citations": n_cits,
+            "time_to_rca_s": time.time() - inc["created_at"]}
+
+
+def _extract_root_cause(summary: str) -> str:
+    for line in summary.splitlines():
+        low = line.strip().lower()
+        if low.startswith("root cause"):
+            return line.strip()[:500]
+    return ""
+
+
+async def trigger_delayed_rca(rt, args: Dict[str, Any]) -> dict:
+    """The second Celery hop of the webhook path (reference:
+    routes/pagerduty/tasks.py:235-431): dedup via the session link, rate
+    limit, build prompt, create session, dispatch run_background_chat."""
+    org_id = args["org_id"]
+    incident_id = args["incident_id"]
+    store = rt.org(org_id)
+    inc = store.get_incident(incident_id)
+    if inc is None:
+        return {"error": "incident gone"}
+    if inc.get("aurora_chat_session_id"):
+        return {"skipped": "rca already dispatched"}
+    user_key = args.get("user_id") or org_id
+    if not rt.rca_allowed(f"bg:{user_key}"):
+        store.lifecycle(incident_id, "rca_rate_limited", {})
+        return {"skipped": "rate limited"}
+    tid = rt.queue.enqueue(
+        "run_background_chat",
+        {"org_id": org_id, "incident_id": incident_id,
+         "user_id": args.get("user_id"),
+         "max_turns": args.get("max_turns", 0)},
+        dedup_key=f"rca:{incident_id}")
+    if tid:
+        store.update_incident(incident_id, rca_task_id=tid)
+    return {"task_id": tid}
+
+
+async def cleanup_stale_investigations(rt, args: Dict[str, Any]) -> dict:
+    """Beat: mark investigations whose heartbeat (incidents.updated_at,
+    touched on every thought save) went stale as failed (reference:
+    task.py:2390,2442,2350)."""
+    stale = rt.admin.stale_running_incidents(older_than_s=900.0)
+    for inc in stale:
+        store = rt.org(inc["org_id"])
+        store.update_incident(inc["id"], aurora_status="failed")
+        store.lifecycle(inc["id"], "rca_reaped", {"reason": "stale heartbeat"})
+        rt.events.publish(inc["org_id"], {
+            "type": "incident_update", "incident_id": inc["id"],
+            "aurora_status": "failed"})
+    return {"reaped": len(stale)}
+
+
+async def dispatch_on_incident_actions(rt, args: Dict[str, Any]) -> dict:
+    """reference: services/actions/executor.py:16,111"""
+    from .actions import run_incident_actions
+    return await run_incident_actions(rt, args["org_id"], args["incident_id"])
+
+
+async def action_scheduler(rt, args: Dict[str, Any]) -> dict:
+    """Per-minute scheduled-action beat (reference: scheduler.py:38)."""
+    from .actions import run_scheduled_actions
+    return await run_scheduled_actions(rt)
+
+
+def register_tasks(rt) -> None:
+    q = rt.queue
+    q.register("run_background_chat", lambda a: run_background_chat(rt, a))
+    q.register("trigger_delayed_rca", lambda a: trigger_delayed_rca(rt, a))
+    q.register("cleanup_stale_investigations",
+               lambda a: cleanup_stale_investigations(rt, a))
+    q.register("dispatch_on_incident_actions",
+               lambda a: dispatch_on_incident_actions(rt, a))
+    q.register("action_scheduler", lambda a: action_scheduler(rt, a))
