@@ -39,7 +39,7 @@ class Workflow:
         self.registry = registry or GLOBAL_REGISTRY
         self.tool_subset = tool_subset
         self.system_prompt_override = system_prompt_override
-        s = get_settings()
+        s = getattr(ctx, "settings", None) or get_settings()
         # recursion limit counts model+tool super-steps (reference
         # AGENT_RECURSION_LIMIT=240 ~ 120 tool turns)
         self.max_turns = max_turns if max_turns is not None else s.agent_recursion_limit // 2

@@ -49,8 +49,23 @@ class AsyncEngineServer:
                 self._wake.wait(timeout=0.02)
                 self._wake.clear()
                 continue
-            with self._lock:
-                outs = self.engine.step()
+            try:
+                with self._lock:
+                    outs = self.engine.step()
+            except Exception as e:  # noqa: BLE001 - a step failure must not
+                # kill the loop: fail every in-flight request loudly instead
+                import traceback
+                err = f"engine step failed: {e}"
+                traceback.print_exc()
+                with self._lock:
+                    for rid in list(self._subs):
+                        try:
+                            self.engine.abort_request(rid)
+                        except Exception:  # noqa: BLE001
+                            pass
+                for rid, q in list(self._subs.items()):
+                    q.put(StepOutput(rid, [], "", True, err))
+                continue
             for out in outs:
                 q = self._subs.get(out.request_id)
                 if q is not None:
@@ -77,15 +92,26 @@ class AsyncEngineServer:
         self._wake.set()
         loop = asyncio.get_running_loop()
         finished = False
+
+        def _get():  # bounded wait so executor threads never leak
+            while True:
+                try:
+                    return q.get(timeout=0.5)
+                except _queue.Empty:
+                    if self._stop.is_set():
+                        return StepOutput(rid, [], "", True, "engine shutdown")
+
         try:
             while True:
-                out = await loop.run_in_executor(None, q.get)
+                out = await loop.run_in_executor(None, _get)
                 yield out
                 if out.finished:
                     finished = True
                     return
         finally:
             self._subs.pop(rid, None)
+            # release any executor thread still blocked in _get
+            q.put(StepOutput(rid, [], "", True, "stream closed"))
             if not finished:  # consumer bailed (cancel / WS drop): free the seq
                 with self._lock:
                     self.engine.abort_request(rid)

@@ -1,0 +1,315 @@
+"""HTTP API (FastAPI): the reference's Flask surface rebuilt.
+
+Route groups mirror reference blueprints (reference: main_compute.py:343-587
+registrations; routes/*): webhooks per source, incidents CRUD + SSE,
+monitor (waterfall/fleet), metrics (MTTD/pickup), postmortems, artifacts,
+findings, llm usage, audit, command policies, knowledge base, graph,
+connectors, MCP-dispatchable tool surface, health.
+
+Auth model: X-User-Id + X-Org-Id headers resolved/verified against the user
+table (the reference's stateless-auth + user<->org binding middleware,
+main_compute.py:299); INTERNAL_API_SECRET header for service-to-service
+calls (AGENTS.md:96).
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import time
+from typing import Optional
+
+from fastapi import Depends, FastAPI, Header, HTTPException, Request
+from fastapi.responses import JSONResponse, StreamingResponse
+
+from .runtime import Runtime
+from .store import OrgStore
+from .webhooks import process_webhook, sources, verify_signature
+
+
+def create_app(rt: Runtime) -> FastAPI:
+    app = FastAPI(title="aurora-mi355x", version="0.1.0")
+    app.state.rt = rt
+
+    @app.on_event("startup")
+    async def _startup():
+        rt.start()
+
+    @app.on_event("shutdown")
+    async def _shutdown():
+        await rt.stop()
+
+    # ---------------------------------------------------------------- auth
+
+    def auth(x_user_id: Optional[str] = Header(None),
+             x_org_id: Optional[str] = Header(None),
+             x_internal_api_secret: Optional[str] = Header(None)) -> OrgStore:
+        s = rt.settings
+        if s.internal_api_secret and x_internal_api_secret == s.internal_api_secret:
+            if x_org_id:
+                return rt.org(x_org_id)
+            raise HTTPException(400, "internal call requires X-Org-Id")
+        if not x_user_id:
+            raise HTTPException(401, "X-User-Id required")
+        user = rt.admin.get_user(x_user_id)
+        if user is None:
+            raise HTTPException(401, "unknown user")
+        # user<->org binding enforcement (main_compute.py:299): a client-sent
+        # org id that mismatches the user's org is rejected, not trusted
+        if x_org_id and x_org_id != user["org_id"]:
+            raise HTTPException(403, "user is not a member of that org")
+        return rt.org(user["org_id"])
+
+    # ------------------------------------------------------------- webhooks
+
+    @app.post("/webhook/{source}/{org_id}")
+    async def webhook(source: str, org_id: str, request: Request,
+                      user_id: Optional[str] = None):
+        if source not in sources():
+            raise HTTPException(404, f"unknown source; known: {sources()}")
+        body = await request.body()
+        store = rt.org(org_id)
+        secret = store.get_token_ref("system", f"webhook_secret:{source}")
+        if not verify_signature(source, body, dict(request.headers), secret):
+            store.audit("webhook_bad_signature", source)
+            raise HTTPException(403, "bad signature")
+        try:
+            payload = json.loads(body or b"{}")
+        except json.JSONDecodeError:
+            raise HTTPException(400, "invalid JSON") from None
+        iid, created = process_webhook(rt, org_id, source, payload, user_id=user_id)
+        return {"incident_id": iid, "created": created}
+
+    # ------------------------------------------------------------ incidents
+
+    @app.get("/incidents")
+    async def list_incidents(status: Optional[str] = None,
+                             store: OrgStore = Depends(auth)):
+        return store.list_incidents(status)
+
+    @app.get("/incidents/{iid}")
+    async def get_incident(iid: str, store: OrgStore = Depends(auth)):
+        inc = store.get_incident(iid)
+        if inc is None:
+            raise HTTPException(404, "no such incident")
+        return inc
+
+    @app.post("/incidents")
+    async def create_incident(body: dict, store: OrgStore = Depends(auth)):
+        iid = store.create_incident(
+            title=body.get("title", "manual incident"),
+            source=body.get("source", "manual"),
+            payload=body.get("payload", {}))
+        return {"incident_id": iid}
+
+    @app.post("/incidents/{iid}/investigate")
+    async def investigate(iid: str, store: OrgStore = Depends(auth)):
+        if store.get_incident(iid) is None:
+            raise HTTPException(404, "no such incident")
+        tid = rt.queue.enqueue("run_background_chat",
+                               {"org_id": store.org_id, "incident_id": iid},
+                               dedup_key=f"rca:{iid}:{int(time.time())}")
+        return {"task_id": tid}
+
+    @app.post("/incidents/{iid}/cancel")
+    async def cancel_rca(iid: str, store: OrgStore = Depends(auth)):
+        inc = store.get_incident(iid)
+        if inc is None:
+            raise HTTPException(404, "no such incident")
+        ok = rt.queue.cancel(inc.get("rca_task_id") or "")
+        return {"cancelled": ok}
+
+    @app.post("/incidents/{iid}/resolve")
+    async def resolve(iid: str, store: OrgStore = Depends(auth)):
+        store.update_incident(iid, status="resolved", resolved_at=time.time())
+        store.lifecycle(iid, "resolved", {})
+        return {"ok": True}
+
+    @app.get("/incidents/{iid}/thoughts")
+    async def thoughts(iid: str, store: OrgStore = Depends(auth)):
+        return store.thoughts(iid)
+
+    @app.get("/incidents/{iid}/citations")
+    async def citations(iid: str, store: OrgStore = Depends(auth)):
+        return store.citations(iid)
+
+    @app.get("/incidents/{iid}/suggestions")
+    async def suggestions(iid: str, store: OrgStore = Depends(auth)):
+        return store.suggestions(iid)
+
+    @app.get("/incidents/{iid}/findings")
+    async def findings(iid: str, store: OrgStore = Depends(auth)):
+        return store.findings(iid)
+
+    @app.get("/incidents/{iid}/timeline")
+    async def timeline(iid: str, store: OrgStore = Depends(auth)):
+        return store.lifecycle_events(iid)
+
+    # ---- SSE stream (reference: incidents_sse.py:34) ----
+    @app.get("/incidents/stream")
+    async def sse(store: OrgStore = Depends(auth)):
+        org_id = store.org_id
+
+        async def gen():
+            q = rt.events.subscribe(org_id)
+            try:
+                yield "retry: 3000\n\n"
+                while True:
+                    try:
+                        ev = await asyncio.wait_for(q.get(), timeout=15.0)
+                        yield f"data: {json.dumps(ev)}\n\n"
+                    except asyncio.TimeoutError:
+                        yield ": keepalive\n\n"
+            finally:
+                rt.events.unsubscribe(org_id, q)
+
+        return StreamingResponse(gen(), media_type="text/event-stream")
+
+    # ---- monitor: waterfall + fleet (reference: monitor/waterfall_routes.py) ----
+    @app.get("/monitor/waterfall/{iid}")
+    async def waterfall(iid: str, store: OrgStore = Depends(auth)):
+        steps = store.steps(iid)
+        return {"incident_id": iid, "steps": steps,
+                "total_ms": sum(((s["finished_at"] or s["started_at"])
+                                 - s["started_at"]) * 1000 for s in steps)}
+
+    @app.get("/monitor/tool-stats")
+    async def tool_stats(store: OrgStore = Depends(auth)):
+        rows = rt.db.query(
+            "SELECT tool, COUNT(*) n, AVG(COALESCE(finished_at,started_at)-started_at) "
+            "avg_s, SUM(status='error') errors FROM execution_steps WHERE org_id=? "
+            "GROUP BY tool", [store.org_id])
+        return [dict(r) for r in rows]
+
+    @app.get("/monitor/fleet")
+    async def fleet(store: OrgStore = Depends(auth)):
+        running = store.list_incidents()
+        return {"running": [i for i in running if i["aurora_status"] == "running"],
+                "queue_depth": rt.queue.pending_count(),
+                "engine": rt.engine.counters()}
+
+    # ---- metrics (reference: metrics_routes.py: MTTD / pickup latency) ----
+    @app.get("/metrics")
+    async def metrics(store: OrgStore = Depends(auth)):
+        rows = store.list_incidents(limit=500)
+        pickup = [r["investigation_started_at"] - r["created_at"]
+                  for r in rows if r.get("investigation_started_at")]
+        done = [r for r in rows if r["aurora_status"] == "completed"]
+        ttr = [r["updated_at"] - r["created_at"] for r in done]
+        sev = {}
+        for r in rows:
+            sev[r.get("severity") or "unknown"] = sev.get(r.get("severity") or "unknown", 0) + 1
+        med = (lambda xs: sorted(xs)[len(xs) // 2] if xs else None)
+        return {
+            "incidents_total": len(rows),
+            "incidents_completed": len(done),
+            "severity_mix": sev,
+            "pickup_latency_p50_s": med(pickup),
+            "time_to_rca_p50_s": med(ttr),
+            "engine_counters": rt.engine.counters(),
+        }
+
+    # ---- postmortems / artifacts ----
+    @app.get("/postmortems/{iid}")
+    async def get_pm(iid: str, store: OrgStore = Depends(auth)):
+        pm = store.get_postmortem(iid)
+        if pm is None:
+            raise HTTPException(404, "no postmortem")
+        return pm
+
+    @app.put("/postmortems/{iid}")
+    async def put_pm(iid: str, body: dict, store: OrgStore = Depends(auth)):
+        return {"id": store.save_postmortem(iid, body.get("content", ""))}
+
+    @app.get("/artifacts")
+    async def list_artifacts(session_id: Optional[str] = None,
+                             store: OrgStore = Depends(auth)):
+        return store.artifacts(session_id)
+
+    # ---- llm usage / audit ----
+    @app.get("/llm-usage")
+    async def llm_usage(store: OrgStore = Depends(auth)):
+        return store.usage_summary()
+
+    @app.get("/audit")
+    async def audit_log(store: OrgStore = Depends(auth)):
+        return store.audit_events()
+
+    # ---- command policies ----
+    @app.get("/command-policies")
+    async def get_policies(store: OrgStore = Depends(auth)):
+        return store.command_policies()
+
+    @app.post("/command-policies")
+    async def add_policy(body: dict, store: OrgStore = Depends(auth)):
+        if body.get("action") not in ("allow", "deny"):
+            raise HTTPException(400, "action must be allow|deny")
+        return {"id": store.add_command_policy(body["pattern"], body["action"])}
+
+    @app.delete("/command-policies/{pid}")
+    async def del_policy(pid: str, store: OrgStore = Depends(auth)):
+        return {"deleted": store.delete_command_policy(pid)}
+
+    # ---- knowledge base ----
+    @app.post("/knowledge-base/documents")
+    async def kb_upload(body: dict, store: OrgStore = Depends(auth)):
+        name = body.get("name", "untitled")
+        content = body.get("content", "")
+        did = store.add_kb_document(name, content)
+        from ..rag.index import chunk_text
+        for i, ch in enumerate(chunk_text(content)):
+            store.add_kb_chunk(did, i, ch)
+        rt.rag(store.org_id).add_document(did, name, content)
+        return {"document_id": did}
+
+    @app.get("/knowledge-base/documents")
+    async def kb_list(store: OrgStore = Depends(auth)):
+        return [{k: d[k] for k in ("id", "name", "ts")} for d in store.kb_documents()]
+
+    @app.get("/knowledge-base/search")
+    async def kb_search(q: str, k: int = 5, store: OrgStore = Depends(auth)):
+        return rt.rag(store.org_id).search(q, k=k)
+
+    # ---- infra graph (reference: graph_routes.py) ----
+    @app.get("/graph/services")
+    async def graph_services(store: OrgStore = Depends(auth)):
+        return rt.graph(store.org_id).list_services()
+
+    @app.get("/graph/impact/{service}")
+    async def graph_impact(service: str, store: OrgStore = Depends(auth)):
+        return rt.graph(store.org_id).impact_radius(service)
+
+    @app.get("/graph/overview")
+    async def graph_overview(store: OrgStore = Depends(auth)):
+        return rt.graph(store.org_id).overview()
+
+    @app.post("/graph/services")
+    async def graph_upsert(body: dict, store: OrgStore = Depends(auth)):
+        g = rt.graph(store.org_id)
+        g.upsert_service(body["name"], kind=body.get("kind", "service"))
+        for dep in body.get("depends_on", []):
+            g.upsert_service(dep)
+            g.add_dependency(body["name"], dep)
+        return {"ok": True}
+
+    # ---- connectors ----
+    @app.get("/connectors")
+    async def connectors(store: OrgStore = Depends(auth),
+                         x_user_id: Optional[str] = Header(None)):
+        from ..connectors.registry import connector_catalog
+        connected = store.connected_providers(x_user_id or "")
+        return [dict(c, connected=c["name"] in connected)
+                for c in connector_catalog()]
+
+    @app.post("/connectors/{provider}/connect")
+    async def connect(provider: str, body: dict, store: OrgStore = Depends(auth),
+                      x_user_id: Optional[str] = Header(None)):
+        from ..connectors.registry import connect_provider
+        return connect_provider(rt, store, x_user_id or "", provider, body)
+
+    # ---- health ----
+    @app.get("/health")
+    async def health():
+        return {"status": "ok", "engine": rt.engine.counters(),
+                "queue_depth": rt.queue.pending_count()}
+
+    return app

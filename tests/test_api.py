@@ -1,0 +1,186 @@
+"""HTTP API + WebSocket chat tests (CPU, tiny model)."""
+import asyncio
+import json
+
+import pytest
+from fastapi.testclient import TestClient
+
+from aurora_amd.engine.engine import EngineConfig
+from aurora_amd.engine.server import AsyncEngineServer
+from aurora_amd.service.app import create_app
+from aurora_amd.service.runtime import Runtime
+from aurora_amd.utils.config import Settings
+
+
+@pytest.fixture(scope="module")
+def env():
+    cfg = Settings(db_path=":memory:", secrets_path="/tmp/test_secrets.json")
+    server = AsyncEngineServer(EngineConfig(model="tiny-llama", device="cpu", seed=2))
+    rt = Runtime(cfg, engine_server=server)
+    oid = rt.admin.create_org("acme")
+    uid = rt.admin.create_user(oid, "a@acme.io", user_id="user-a")
+    app = create_app(rt)
+    with TestClient(app) as client:
+        yield rt, client, oid, uid
+    server.shutdown()
+
+
+H = {"X-User-Id": "user-a"}
+
+
+def test_health(env):
+    rt, client, oid, uid = env
+    r = client.get("/health")
+    assert r.status_code == 200 and r.json()["status"] == "ok"
+
+
+def test_auth_required(env):
+    rt, client, *_ = env
+    assert client.get("/incidents").status_code == 401
+    assert client.get("/incidents", headers={"X-User-Id": "ghost"}).status_code == 401
+    # org binding enforcement: claiming another org is rejected
+    r = client.get("/incidents", headers={"X-User-Id": "user-a", "X-Org-Id": "other"})
+    assert r.status_code == 403
+
+
+def test_webhook_to_incident_api(env):
+    rt, client, oid, uid = env
+    r = client.post(f"/webhook/pagerduty/{oid}", json={
+        "event": {"data": {"id": "X1", "title": "DB down",
+                           "service": {"summary": "db"}}}})
+    assert r.status_code == 200
+    iid = r.json()["incident_id"]
+    r = client.get(f"/incidents/{iid}", headers=H)
+    assert r.status_code == 200 and r.json()["title"] == "DB down"
+    r = client.get("/incidents", headers=H)
+    assert any(i["id"] == iid for i in r.json())
+    # timeline + waterfall + metrics all answer
+    assert client.get(f"/incidents/{iid}/timeline", headers=H).status_code == 200
+    assert client.get(f"/monitor/waterfall/{iid}", headers=H).status_code == 200
+    m = client.get("/metrics", headers=H).json()
+    assert m["incidents_total"] >= 1
+
+
+def test_unknown_source_404(env):
+    rt, client, oid, uid = env
+    assert client.post(f"/webhook/nosuch/{oid}", json={}).status_code == 404
+
+
+def test_command_policies_api(env):
+    rt, client, *_ = env
+    r = client.post("/command-policies", headers=H,
+                    json={"pattern": "rm -rf", "action": "deny"})
+    pid = r.json()["id"]
+    assert any(p["id"] == pid for p in client.get("/command-policies", headers=H).json())
+    assert client.delete(f"/command-policies/{pid}", headers=H).json()["deleted"] == 1
+
+
+def test_knowledge_base_api(env):
+    rt, client, *_ = env
+    r = client.post("/knowledge-base/documents", headers=H, json={
+        "name": "runbook-db",
+        "content": "When the database connection pool is exhausted, restart "
+                   "pgbouncer and check for long transactions.\n\nAlso check "
+                   "replication lag on the standby."})
+    assert r.status_code == 200
+    hits = client.get("/knowledge-base/search", headers=H,
+                      params={"q": "connection pool exhausted"}).json()
+    assert hits and "pgbouncer" in hits[0]["text"]
+
+
+def test_graph_api(env):
+    rt, client, *_ = env
+    client.post("/graph/services", headers=H,
+                json={"name": "checkout", "depends_on": ["db", "cache"]})
+    ov = client.get("/graph/overview", headers=H).json()
+    assert ov["services"] >= 3
+    imp = client.get("/graph/impact/db", headers=H).json()
+    assert "checkout" in imp["impacted_dependents"]
+
+
+def test_connectors_api(env):
+    rt, client, *_ = env
+    cat = client.get("/connectors", headers=H).json()
+    assert len(cat) >= 25
+    r = client.post("/connectors/github/connect", headers=H,
+                    json={"credentials": {"token": "ghp_abc"}})
+    assert r.json()["connected"]
+    cat = client.get("/connectors", headers=H).json()
+    gh = next(c for c in cat if c["name"] == "github")
+    assert gh["connected"]
+    # stored as a ref, not the raw token
+    store = rt.org(rt.admin.get_user("user-a")["org_id"])
+    ref = store.get_token_ref("user-a", "github")
+    assert ref and "ghp_abc" not in ref
+    from aurora_amd.connectors.registry import get_credentials
+    assert get_credentials(store, "user-a", "github")["token"] == "ghp_abc"
+
+
+def test_postmortem_api(env):
+    rt, client, oid, uid = env
+    r = client.post("/incidents", headers=H, json={"title": "manual"})
+    iid = r.json()["incident_id"]
+    client.put(f"/postmortems/{iid}", headers=H, json={"content": "# PM"})
+    assert client.get(f"/postmortems/{iid}", headers=H).json()["content"] == "# PM"
+
+
+def test_websocket_chat():
+    from aurora_amd.service.ws import ChatServer, mask_frame, read_frame
+
+    cfg = Settings(db_path=":memory:")
+    cfg.agent_recursion_limit = 4  # tiny random model: keep the loop short
+    server = AsyncEngineServer(EngineConfig(model="tiny-llama", device="cpu", seed=4))
+    rt = Runtime(cfg, engine_server=server)
+    oid = rt.admin.create_org("ws-org")
+    rt.admin.create_user(oid, "w@x.io", user_id="wsu")
+
+    async def main():
+        cs = ChatServer(rt)
+        port = await cs.start()
+        reader, writer = await asyncio.open_connection("127.0.0.1", port)
+        writer.write((
+            f"GET /?token=wsu.n1 HTTP/1.1\r\nHost: x\r\nUpgrade: websocket\r\n"
+            f"Connection: Upgrade\r\nSec-WebSocket-Key: dGhlIHNhbXBsZSBub25jZQ==\r\n"
+            f"Sec-WebSocket-Version: 13\r\n\r\n").encode())
+        await writer.drain()
+        resp = await reader.readuntil(b"\r\n\r\n")
+        assert b"101" in resp.split(b"\r\n")[0]
+        opcode, data = await read_frame(reader)
+        assert json.loads(data)["type"] == "ready"
+        writer.write(mask_frame(json.dumps({"type": "init"}).encode()))
+        await writer.drain()
+        opcode, data = await read_frame(reader)
+        msg = json.loads(data)
+        assert msg["type"] == "initialized" and msg["session_id"]
+        # a query streams events and ends with final+usage
+        writer.write(mask_frame(json.dumps(
+            {"type": "query", "content": "hello"}).encode()))
+        await writer.drain()
+        types = []
+        for _ in range(400):
+            opcode, data = await asyncio.wait_for(read_frame(reader), timeout=120)
+            ev = json.loads(data)
+            types.append(ev["type"])
+            if ev["type"] in ("final", "error"):
+                break
+        assert "final" in types or "error" in types
+        writer.close()
+        await cs.stop()
+
+        # bad token rejected
+        cs2 = ChatServer(rt)
+        port2 = await cs2.start()
+        r2, w2 = await asyncio.open_connection("127.0.0.1", port2)
+        w2.write((
+            "GET /?token=ghost HTTP/1.1\r\nHost: x\r\nUpgrade: websocket\r\n"
+            "Connection: Upgrade\r\nSec-WebSocket-Key: dGhlIHNhbXBsZSBub25jZQ==\r\n"
+            "\r\n").encode())
+        await w2.drain()
+        line = await r2.readline()
+        assert b"401" in line
+        await cs2.stop()
+
+    try:
+        asyncio.run(main())
+    finally:
+        server.shutdown()
