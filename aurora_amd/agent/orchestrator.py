@@ -1,0 +1,207 @@
+"""Multi-agent orchestrator: triage -> dispatch -> parallel sub-agents ->
+synthesis (reference: chat/backend/agent/orchestrator/ — triage.py:60-314,
+dispatcher.py:39,220-288, sub_agent.py:81,241,268, synthesis.py:47-556,
+findings_writer.py, roles/*.md).
+
+The reference fans out with LangGraph Send and merges findings through a
+state reducer; here the fan-out is asyncio tasks whose sub-agent decode
+streams batch INTO THE SAME ENGINE (continuous batching — the MI355X-native
+replacement for process-level parallelism), and findings
+merge through the findings table.
+"""
+from __future__ import annotations
+
+import asyncio
+import json
+import time
+from dataclasses import dataclass, field
+from typing import Any, Dict, List, Optional
+
+from ..engine.grammar import JsonGrammar
+from ..engine.sampling_params import SamplingParams
+from .state import State
+from .workflow import Workflow
+
+# role briefs (reference: orchestrator/roles/*.md)
+ROLES: Dict[str, Dict[str, Any]] = {
+    "general_investigator": {
+        "brief": "You broadly investigate the incident: read the alert, check "
+                 "service health, and identify the most suspicious subsystem.",
+        "tools": ["get_alert_field", "get_infrastructure_context",
+                  "list_incidents", "terminal_exec", "kubectl"],
+    },
+    "error_signal_investigator": {
+        "brief": "You focus on error signals: logs, error rates, stack traces "
+                 "around the incident start time.",
+        "tools": ["terminal_exec", "kubectl", "get_alert_field"],
+    },
+    "recent_change_investigator": {
+        "brief": "You focus on recent changes: deploys, config changes, "
+                 "migrations in the 60 minutes before the incident.",
+        "tools": ["get_incident_timeline", "list_incidents", "github_rca",
+                  "get_alert_field"],
+    },
+    "runtime_state_investigator": {
+        "brief": "You inspect live runtime state: resource saturation, "
+                 "process health, connectivity.",
+        "tools": ["terminal_exec", "kubectl", "get_service_impact"],
+    },
+    "runbook_lookup": {
+        "brief": "You find and summarise the most relevant runbooks and past "
+                 "incidents for this alert.",
+        "tools": ["knowledge_base_search", "list_incidents", "get_postmortem"],
+    },
+    "ticket_history": {
+        "brief": "You mine past incident history for recurrences of this "
+                 "failure signature.",
+        "tools": ["list_incidents", "get_incident", "get_postmortem"],
+    },
+}
+
+TRIAGE_PROMPT = """You are the RCA orchestrator. Given the alert, pick the
+investigation plan as a JSON object:
+{{"mode": "direct" | "multi_agent", "roles": ["<role>", ...], "waves": 1}}
+Valid roles: {roles}. Use "direct" for simple/low-severity alerts; pick 2-4
+roles for complex ones.
+Alert: {alert}
+JSON:"""
+
+SYNTH_PROMPT = """You are the synthesis agent. Merge the sub-agent findings
+below into either a follow-up plan or a final RCA. Reply a JSON object:
+{{"decision": "end" | "dispatch", "roles": [...], "rca": "<final RCA text if
+decision is end>"}}
+Findings:
+{findings}
+JSON:"""
+
+
+@dataclass
+class OrchestratorResult:
+    final: str
+    waves: int
+    findings: List[dict] = field(default_factory=list)
+    triage: Optional[dict] = None
+
+
+class Orchestrator:
+    def __init__(self, llm, ctx, max_waves: int = 2, sub_agent_turns: int = 6,
+                 sub_agent_timeout_s: float = 600.0, max_roles_per_wave: int = 4):
+        self.llm = llm
+        self.ctx = ctx
+        self.max_waves = max_waves
+        self.sub_agent_turns = sub_agent_turns
+        self.sub_agent_timeout_s = sub_agent_timeout_s
+        self.max_roles = max_roles_per_wave
+
+    async def run(self, state: State) -> OrchestratorResult:
+        store = self.ctx.org_store
+        alert = state.question[:3000]
+        # ---- triage (reference: triage.py:54-71) ----
+        triage = await self._json_call(
+            TRIAGE_PROMPT.format(roles=sorted(ROLES), alert=alert))
+        roles = self._filter_roles(triage.get("roles", []))
+        state.triage_decision = triage
+        if triage.get("mode") != "multi_agent" or not roles:
+            # direct ReAct path (reference: workflow.py:171-205 direct_react)
+            wf = Workflow(self.llm, self.ctx, max_turns=self.sub_agent_turns * 2)
+            final = ""
+            async for ev in wf.stream(state):
+                if ev["type"] == "final":
+                    final = ev["content"]
+            return OrchestratorResult(final, 0, triage=triage)
+
+        all_findings: List[dict] = []
+        final = ""
+        for wave in range(self.max_waves):
+            state.synthesis_wave = wave
+            # ---- dispatch: pre-emit finding rows then fan out
+            # (reference: dispatcher.py:220-288) ----
+            fids = {}
+            for role in roles:
+                fids[role] = store.upsert_finding(
+                    state.incident_id or "-", role, wave, "running") \
+                    if store is not None else None
+            tasks = [self._sub_agent(state, role, wave, fids[role])
+                     for role in roles]
+            results = await asyncio.gather(*tasks, return_exceptions=True)
+            wave_findings = []
+            for role, res in zip(roles, results):
+                if isinstance(res, Exception):
+                    res = f"(sub-agent {role} failed: {res})"
+                    if store is not None and fids[role]:
+                        store.upsert_finding(state.incident_id or "-", role,
+                                             wave, "failed", str(res),
+                                             finding_id=fids[role])
+                wave_findings.append({"role": role, "wave": wave,
+                                      "content": str(res)[:4000]})
+            all_findings.extend(wave_findings)
+            state.finding_refs.extend(f["role"] for f in wave_findings)
+            # ---- synthesis (reference: synthesis.py:47-61,556) ----
+            synth = await self._json_call(SYNTH_PROMPT.format(
+                findings=json.dumps(all_findings)[:8000]))
+            state.synthesis_history.append(json.dumps(synth)[:500])
+            if synth.get("decision") == "dispatch" and wave + 1 < self.max_waves:
+                roles = self._filter_roles(synth.get("roles", [])) or roles
+                continue
+            final = str(synth.get("rca", "")).strip()
+            break
+        if not final:
+            final = "Root cause undetermined. Sub-agent findings:\n" + "\n".join(
+                f"- [{f['role']}] {f['content'][:300]}" for f in all_findings)
+        return OrchestratorResult(final, state.synthesis_wave + 1,
+                                  findings=all_findings, triage=triage)
+
+    # ------------------------------------------------------------- helpers
+
+    def _filter_roles(self, roles: List[str]) -> List[str]:
+        """known-role filter + dedupe + cap (reference: dispatcher.py:69-122)."""
+        out: List[str] = []
+        for r in roles:
+            if r in ROLES and r not in out:
+                out.append(r)
+        return out[: self.max_roles]
+
+    async def _json_call(self, prompt: str) -> dict:
+        try:
+            res = await self.llm.invoke(
+                prompt, SamplingParams(temperature=0.1, max_tokens=220,
+                                       grammar=JsonGrammar()),
+                purpose="orchestrator")
+            doc = json.loads(res.text)
+            return doc if isinstance(doc, dict) else {}
+        except Exception:  # noqa: BLE001
+            return {}
+
+    async def _sub_agent(self, state: State, role: str, wave: int,
+                         finding_id: Optional[str]) -> str:
+        """Role-briefed ReAct run with a tool subset + timeout (reference:
+        sub_agent.py:81,241,268); the finding is persisted to the findings
+        table (findings_writer.py)."""
+        brief = ROLES[role]["brief"]
+        subset = ROLES[role]["tools"] + ["final_answer"]
+        sub_state = State(
+            question=f"[role: {role}] {brief}\nIncident: {state.question}",
+            user_id=state.user_id, org_id=state.org_id,
+            session_id=None,  # sub-agents keep their own scratch context
+            incident_id=state.incident_id, is_background=True)
+        wf = Workflow(self.llm, self.ctx, tool_subset=subset,
+                      system_prompt_override=(
+                          f"You are the {role} sub-agent of an incident "
+                          f"investigation.\n{brief}\nFinish with final_answer "
+                          f"containing your findings."),
+                      max_turns=self.sub_agent_turns)
+        final = ""
+        try:
+            async def _run():
+                nonlocal final
+                async for ev in wf.stream(sub_state):
+                    if ev["type"] == "final":
+                        final = ev["content"]
+            await asyncio.wait_for(_run(), timeout=self.sub_agent_timeout_s)
+        except asyncio.TimeoutError:
+            final = final or "(sub-agent timed out)"
+        store = self.ctx.org_store
+        if store is not None and finding_id:
+            store.upsert_finding(state.incident_id or "-", role, wave,
+                                 "completed", final, finding_id=finding_id)
+        return final

@@ -70,20 +70,29 @@ async def run_background_chat(rt, args: Dict[str, Any]) -> dict:
     llm = rt.llm(org_id)
     ctx = rt.tool_ctx(org_id, user_id=user_id, session_id=session_id,
                       incident_id=incident_id, depth=1)  # depth=1 blocks nested RCA
-    wf = Workflow(llm, ctx, max_turns=int(args.get("max_turns", 0)) or None)
     state = State(question=prompt, user_id=user_id, org_id=org_id,
                   session_id=session_id, incident_id=incident_id,
                   incident_start_time=inc["created_at"], is_background=True)
     final = ""
     error = None
     try:
-        async for ev in wf.stream(state):
-            if ev["type"] == "final":
-                final = ev["content"]
-            elif ev["type"] == "blocked":
-                error = f"guardrail blocked: {ev['reason']}"
-            elif ev["type"] == "error":
-                error = ev["error"]
+        if rt.settings.orchestrator_enabled:
+            # multi-agent path (reference: workflow.py:171-205 with
+            # ORCHESTRATOR_ENABLED=true)
+            from ..agent.orchestrator import Orchestrator
+            orch = Orchestrator(llm, ctx,
+                                sub_agent_turns=int(args.get("max_turns", 0)) or 6)
+            result = await orch.run(state)
+            final = result.final
+        else:
+            wf = Workflow(llm, ctx, max_turns=int(args.get("max_turns", 0)) or None)
+            async for ev in wf.stream(state):
+                if ev["type"] == "final":
+                    final = ev["content"]
+                elif ev["type"] == "blocked":
+                    error = f"guardrail blocked: {ev['reason']}"
+                elif ev["type"] == "error":
+                    error = ev["error"]
     except Exception as e:  # noqa: BLE001
         error = f"workflow failure: {e}"
 

@@ -1,0 +1,151 @@
+"""Orchestrator (multi-agent) + MCP server tests with a scripted LLM."""
+import asyncio
+import json
+
+import pytest
+
+from aurora_amd.service.store import AdminStore, Database
+
+
+class FakeResult:
+    def __init__(self, text):
+        self.text = text
+        self.token_ids = [0]
+        self.prompt_tokens = 1
+        self.elapsed_ms = 1.0
+        self.finish_reason = "eos"
+
+
+class ScriptedServer:
+    """Returns scripted grammar responses; plain text elsewhere."""
+
+    def __init__(self, script):
+        self.script = list(script)
+
+    def tokenizer(self):
+        from aurora_amd.engine.tokenizer import ByteTokenizer
+        return ByteTokenizer()
+
+    def counters(self):
+        return {}
+
+    async def stream(self, ids, params, session_id=None, request_id=None):
+        from aurora_amd.engine.engine import StepOutput
+        yield StepOutput("x", [1], "thinking...", True, "stop")
+
+    async def complete_text(self, prompt, params, session_id=None):
+        if params.grammar is not None and self.script:
+            return FakeResult(self.script.pop(0))
+        return FakeResult("summary text")
+
+
+@pytest.fixture()
+def env():
+    db = Database(":memory:")
+    admin = AdminStore(db)
+    oid = admin.create_org("t")
+    store = admin.org_store(oid)
+    iid = store.create_incident("checkout 5xx", "pagerduty", {"service": "checkout"})
+    return store, iid
+
+
+def _mk(store, iid, script):
+    from aurora_amd.agent.llm import LLMManager
+    from aurora_amd.agent.tools.registry import ToolContext
+    import aurora_amd.agent.tools.core  # noqa: F401
+    server = ScriptedServer(script)
+    llm = LLMManager(server, store=store)
+    ctx = ToolContext(org_store=store, incident_id=iid, llm=llm)
+    return llm, ctx
+
+
+def test_orchestrator_multi_agent_flow(env):
+    store, iid = env
+    from aurora_amd.agent.orchestrator import Orchestrator
+    from aurora_amd.agent.state import State
+    script = [
+        # triage: pick two roles
+        json.dumps({"mode": "multi_agent",
+                    "roles": ["runbook_lookup", "bogus_role",
+                              "recent_change_investigator"], "waves": 1}),
+        # each sub-agent immediately finishes (grammar action calls)
+        json.dumps({"tool": "final_answer", "args": {"answer": "runbook says X"}}),
+        json.dumps({"tool": "final_answer", "args": {"answer": "deploy at 14:00"}}),
+        # synthesis: end with final RCA
+        json.dumps({"decision": "end",
+                    "rca": "Root cause: the 14:00 deploy broke checkout."}),
+    ]
+    llm, ctx = _mk(store, iid, script)
+    orch = Orchestrator(llm, ctx, sub_agent_turns=2)
+    state = State(question="checkout 5xx alert", incident_id=iid)
+
+    res = asyncio.run(orch.run(state))
+    assert "Root cause" in res.final
+    # unknown role filtered (dispatcher._filter_known_roles analog)
+    assert state.triage_decision["roles"][1] == "bogus_role"
+    roles_run = {f["role"] for f in res.findings}
+    assert roles_run == {"runbook_lookup", "recent_change_investigator"}
+    # findings persisted with status transitions
+    rows = store.findings(iid)
+    assert len(rows) == 2 and all(r["status"] == "completed" for r in rows)
+
+
+def test_orchestrator_direct_path(env):
+    store, iid = env
+    from aurora_amd.agent.orchestrator import Orchestrator
+    from aurora_amd.agent.state import State
+    script = [
+        json.dumps({"mode": "direct", "roles": []}),
+        json.dumps({"tool": "final_answer", "args": {"answer": "simple fix"}}),
+    ]
+    llm, ctx = _mk(store, iid, script)
+    orch = Orchestrator(llm, ctx)
+    res = asyncio.run(orch.run(State(question="minor alert", incident_id=iid)))
+    assert res.final == "simple fix" and res.waves == 0
+
+
+def test_mcp_server(env):
+    store, iid = env
+    from aurora_amd.service.mcp_server import MCPServer, search_dispatch_entries
+
+    class RT:  # minimal runtime facade
+        def __init__(self, store):
+            self._store = store
+            self.admin = type("A", (), {"get_user": staticmethod(
+                lambda t: {"id": "u", "org_id": store.org_id} if t == "tok" else None)})()
+
+        def org(self, org_id):
+            return self._store
+
+        def rag(self, org_id):
+            from aurora_amd.rag.index import RagIndex
+            return RagIndex()
+
+        def graph(self, org_id):
+            from aurora_amd.graph.infra_graph import InfraGraph
+            return InfraGraph(self._store)
+
+        class queue:  # noqa: N801
+            @staticmethod
+            def enqueue(*a, **k):
+                return "task-1"
+
+    srv = MCPServer(RT(store))
+
+    async def main():
+        r = await srv.handle({"jsonrpc": "2.0", "id": 1, "method": "initialize"}, "tok")
+        assert r["result"]["serverInfo"]["name"] == "aurora-mi355x"
+        r = await srv.handle({"jsonrpc": "2.0", "id": 2, "method": "tools/list"}, "tok")
+        names = [t["name"] for t in r["result"]["tools"]]
+        assert "aurora_list_incidents" in names
+        assert not any(n in names for n in ("terminal_exec", "kubectl"))
+        r = await srv.handle({"jsonrpc": "2.0", "id": 3, "method": "tools/call",
+                              "params": {"name": "aurora_get_incident",
+                                         "arguments": {"incident_id": iid}}}, "tok")
+        body = json.loads(r["result"]["content"][0]["text"])
+        assert body["title"] == "checkout 5xx"
+        # auth required
+        r = await srv.handle({"jsonrpc": "2.0", "id": 4, "method": "tools/list"}, "bad")
+        assert "error" in r
+    asyncio.run(main())
+    assert search_dispatch_entries("postmortem")
