@@ -50,6 +50,7 @@ class EngineConfig:
     tokenizer_path: Optional[str] = None
     seed: int = 0
     grammar_topk: int = 32  # candidates tried per step under a grammar
+    use_hip_graphs: bool = True  # capture the decode forward per batch bucket
 
 
 @dataclass
@@ -141,6 +142,18 @@ class LLMEngine:
         # session KV reuse: session_id -> finished Sequence kept warm in cache
         self.sessions: "OrderedDict[str, Sequence]" = OrderedDict()
         self.gen = torch.Generator(device="cpu").manual_seed(config.seed)
+        self._graph_cache: Dict[int, tuple] = {}
+        self._graph_pool = None
+        if self._graphs_enabled():
+            # reserve the idle-row scratch page + pinned host staging now,
+            # while the cache is guaranteed non-full
+            mp = (self.max_model_len + self.kv.page_size - 1) // self.kv.page_size
+            self._scratch_page = self.kv.alloc(1)[0]
+            self._scratch_slot = self._scratch_page * self.kv.page_size
+            maxb = self._BUCKETS[-1]
+            self._host_i64 = torch.empty(maxb, dtype=torch.int64, pin_memory=True)
+            self._host_i32 = torch.empty(maxb, 3, dtype=torch.int32, pin_memory=True)
+            self._host_bt = torch.empty(maxb, mp, dtype=torch.int32, pin_memory=True)
         # counters (engine observability; surfaced by service.metrics)
         self.counters = {
             "prompt_tokens": 0, "generated_tokens": 0, "prefill_steps": 0,
@@ -353,19 +366,103 @@ class LLMEngine:
         # written; it is this step's input token.
         slots = [self._slot(s, s.num_tokens - 1) for s in seqs]
         ctx_lens = [s.num_tokens for s in seqs]
-        meta = AttnMetadata(
-            mode="decode",
-            positions=torch.tensor(positions, dtype=torch.int32, device=dev),
-            slot_mapping=torch.tensor(slots, dtype=torch.int32, device=dev),
-            block_tables=self._block_table_tensor(seqs),
-            context_lens=torch.tensor(ctx_lens, dtype=torch.int32, device=dev),
-        )
-        input_ids = torch.tensor(ids, dtype=torch.long, device=dev)
-        hidden = self.model(input_ids, self.kv.caches, meta)
-        logits = self.model.compute_logits(hidden)
+        if self._graphs_enabled() and len(seqs) <= self._BUCKETS[-1]:
+            logits = self._decode_with_graph(seqs, ids, positions, slots, ctx_lens)
+        else:
+            meta = AttnMetadata(
+                mode="decode",
+                positions=torch.tensor(positions, dtype=torch.int32, device=dev),
+                slot_mapping=torch.tensor(slots, dtype=torch.int32, device=dev),
+                block_tables=self._block_table_tensor(seqs),
+                context_lens=torch.tensor(ctx_lens, dtype=torch.int32, device=dev),
+            )
+            input_ids = torch.tensor(ids, dtype=torch.long, device=dev)
+            hidden = self.model(input_ids, self.kv.caches, meta)
+            logits = self.model.compute_logits(hidden)
         for s in seqs:
             s.num_cached = s.num_tokens
         return self._sample_and_append(seqs, logits)
+
+    # ----------------------------------------------------- hipGraph decode
+
+    _BUCKETS = (1, 2, 4, 8, 16, 32, 64)
+
+    def _graphs_enabled(self) -> bool:
+        return (self.config.use_hip_graphs and self.device != "cpu"
+                and self.config.tp_size == 1)
+
+    def _decode_with_graph(self, seqs, ids, positions, slots, ctx_lens):
+        """Replay a hipGraph-captured decode forward (per batch-size bucket,
+        fixed-shape static buffers; idle rows decode a reserved scratch page
+        and are ignored).  Removes the per-layer kernel-launch host time
+        that dominates small-batch decode."""
+        B = len(seqs)
+        bucket = next(b for b in self._BUCKETS if b >= B)
+        g = self._graph_cache.get(bucket)
+        if g is None:
+            g = self._capture_decode_graph(bucket)
+            self._graph_cache[bucket] = g
+        (graph, in_ids, in_pos, in_slot, in_bt, in_ctx, out_logits) = g
+        mp = in_bt.shape[1]
+        # fill the static buffers (idle rows -> scratch page, ctx 1)
+        ids_t = self._host_i64
+        ids_t[:B] = torch.tensor(ids, dtype=torch.int64)
+        pos = self._host_i32
+        pos[:B, 0] = torch.tensor(positions, dtype=torch.int32)
+        pos[B:bucket, 0] = 0
+        pos[:B, 1] = torch.tensor(slots, dtype=torch.int32)
+        pos[B:bucket, 1] = self._scratch_slot
+        pos[:B, 2] = torch.tensor(ctx_lens, dtype=torch.int32)
+        pos[B:bucket, 2] = 1
+        bt_host = self._host_bt
+        bt_host[:bucket].fill_(self._scratch_page)
+        for i, s in enumerate(seqs):
+            n = min(len(s.pages), mp)
+            bt_host[i, :n] = torch.tensor(s.pages[:n], dtype=torch.int32)
+        in_ids.copy_(ids_t[:bucket], non_blocking=True)
+        in_pos.copy_(pos[:bucket, 0], non_blocking=True)
+        in_slot.copy_(pos[:bucket, 1], non_blocking=True)
+        in_ctx.copy_(pos[:bucket, 2], non_blocking=True)
+        in_bt.copy_(bt_host[:bucket], non_blocking=True)
+        graph.replay()
+        return out_logits[:B]
+
+    def _capture_decode_graph(self, bucket: int):
+        dev = self.device
+        mp = (self.max_model_len + self.kv.page_size - 1) // self.kv.page_size
+        if not hasattr(self, "_scratch_page"):
+            self._scratch_page = self.kv.alloc(1)[0]
+            self._scratch_slot = self._scratch_page * self.kv.page_size
+            maxb = self._BUCKETS[-1]
+            self._host_i64 = torch.empty(maxb, dtype=torch.int64, pin_memory=True)
+            self._host_i32 = torch.empty(maxb, 3, dtype=torch.int32, pin_memory=True)
+            self._host_bt = torch.empty(maxb, mp, dtype=torch.int32, pin_memory=True)
+        in_ids = torch.zeros(bucket, dtype=torch.int64, device=dev)
+        in_pos = torch.zeros(bucket, dtype=torch.int32, device=dev)
+        in_slot = torch.full((bucket,), self._scratch_slot, dtype=torch.int32,
+                             device=dev)
+        in_bt = torch.full((bucket, mp), self._scratch_page, dtype=torch.int32,
+                           device=dev)
+        in_ctx = torch.ones(bucket, dtype=torch.int32, device=dev)
+        meta = AttnMetadata(mode="decode", positions=in_pos, slot_mapping=in_slot,
+                            block_tables=in_bt, context_lens=in_ctx)
+
+        def fwd():
+            hidden = self.model(in_ids, self.kv.caches, meta)
+            return self.model.compute_logits(hidden)
+
+        for _ in range(2):  # warmup outside capture
+            fwd()
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()
+        if self._graph_pool is None:
+            with torch.cuda.graph(graph):
+                out_logits = fwd()
+            self._graph_pool = graph.pool()
+        else:
+            with torch.cuda.graph(graph, pool=self._graph_pool):
+                out_logits = fwd()
+        return (graph, in_ids, in_pos, in_slot, in_bt, in_ctx, out_logits)
 
     # --------------------------------------------------------------- sample
 
@@ -395,30 +492,37 @@ class LLMEngine:
                 return tok
         return self.tokenizer.eos_id
 
+    def _dev_gen(self, device) -> torch.Generator:
+        if device.type == "cpu":
+            return self.gen
+        if getattr(self, "_cuda_gen", None) is None:
+            self._cuda_gen = torch.Generator(device=device).manual_seed(
+                self.config.seed)
+        return self._cuda_gen
+
     def _batch_plain_sample(self, logits: torch.Tensor, seqs: List[Sequence]) -> torch.Tensor:
-        """Vectorized temperature/top-k/top-p sampling on the logits' device
-        (gumbel-max for the stochastic rows); returns token ids on CPU."""
+        """Fully-batched temperature/top-k/top-p sampling on the logits'
+        device (gumbel-max for stochastic rows) — one sort + one argmax for
+        the whole batch, no per-row python work, no host randoms."""
         B, V = logits.shape
-        temps = torch.tensor([s.params.temperature for s in seqs], device=logits.device)
+        dev = logits.device
+        temps = torch.tensor([s.params.temperature for s in seqs], device=dev)
         greedy = temps <= 0
-        r = logits / temps.clamp(min=1e-6).unsqueeze(1)
-        for i, s in enumerate(seqs):
-            p = s.params
-            if p.temperature <= 0:
-                continue
-            row = r[i]
-            if p.top_k and p.top_k < V:
-                kth = row.topk(p.top_k).values[-1]
-                row.masked_fill_(row < kth, float("-inf"))
-            if p.top_p < 1.0:
-                sr, si = row.sort(descending=True)
-                probs = sr.softmax(-1)
-                cut = probs.cumsum(-1) - probs >= p.top_p
-                sr = sr.masked_fill(cut, float("-inf"))
-                row.copy_(torch.full_like(row, float("-inf")).scatter(0, si, sr))
         if bool(greedy.all()):
             return logits.argmax(-1).cpu()
-        u = torch.rand(B, V, device="cpu", generator=self.gen).to(logits.device)
+        r = logits / temps.clamp(min=1e-6).unsqueeze(1)
+        top_ks = torch.tensor([s.params.top_k or V for s in seqs], device=dev)
+        top_ps = torch.tensor([s.params.top_p for s in seqs], device=dev)
+        if bool((top_ks < V).any()) or bool((top_ps < 1.0).any()):
+            sr, si = r.sort(dim=-1, descending=True)
+            probs = sr.softmax(-1)
+            cum = probs.cumsum(-1) - probs
+            ranks = torch.arange(V, device=dev).unsqueeze(0)
+            cut = (cum >= top_ps.unsqueeze(1)) | (ranks >= top_ks.unsqueeze(1))
+            cut[:, 0] = False  # always keep the top token
+            sr = sr.masked_fill(cut, float("-inf"))
+            r = torch.full_like(r, float("-inf")).scatter_(1, si, sr)
+        u = torch.rand(B, V, device=dev, generator=self._dev_gen(dev))
         gumbel = -torch.log(-torch.log(u.clamp(min=1e-20)).clamp(min=1e-20))
         noisy = torch.where(greedy.unsqueeze(1), logits, r + gumbel)
         return noisy.argmax(-1).cpu()

@@ -193,7 +193,7 @@ class LlamaForCausalLM(nn.Module):
     @torch.inference_mode()
     def compute_logits(self, hidden: torch.Tensor) -> torch.Tensor:
         w = self.embed.weight if self.lm_head is None else self.lm_head.weight
-        return F.linear(hidden, w).float()
+        return ops.linear(hidden, w).float()
 
     @torch.inference_mode()
     def init_random(self, seed: int = 0) -> None:

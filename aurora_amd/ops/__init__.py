@@ -152,6 +152,19 @@ def knn_topk(query: torch.Tensor, docs: torch.Tensor, k: int):
     return torch_ref.knn_topk(query, docs, k)
 
 
+def linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
+    """Bias-free linear: custom skinny GEMM on the GPU decode path (M<=64),
+    hipBLASLt otherwise.  Dispatch thresholds are measured
+    (profiles/kernel_perf_r01_*.json): hipBLASLt wins at very large N
+    (lm-head shapes), the skinny kernel at mid-size per-layer shapes."""
+    import torch.nn.functional as F
+    if (x.is_cuda and x.dim() == 2 and 0 < x.shape[0] <= 64
+            and x.shape[1] % 64 == 0 and weight.shape[0] <= 32768
+            and x.dtype == torch.bfloat16 and _hip is not None):
+        return _hip.skinny_gemm(x.contiguous(), weight)
+    return F.linear(x, weight)
+
+
 def softmax_sample(logits, temperature, top_p, top_k, generator=None):
     # Sampling is [B, V] with B small; keep it in torch on both devices for now
     # (the GPU cost is one softmax over V — negligible next to the forward).

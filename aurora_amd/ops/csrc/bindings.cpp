@@ -20,6 +20,7 @@ torch::Tensor attention_prefill(torch::Tensor q, torch::Tensor k, torch::Tensor 
                                 torch::Tensor cu, double scale, bool causal);
 torch::Tensor embed_pool(torch::Tensor hidden, torch::Tensor cu);
 std::vector<torch::Tensor> knn_topk(torch::Tensor query, torch::Tensor docs, long k);
+torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "fused RMSNorm (bf16, gfx950)");
@@ -33,4 +34,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attention_prefill", &attention_prefill, "varlen self-attention (encoder)");
   m.def("embed_pool", &embed_pool, "mean-pool + L2 normalize");
   m.def("knn_topk", &knn_topk, "brute-force kNN top-k");
+  m.def("skinny_gemm", &skinny_gemm, "decode-path skinny GEMM (M<=64)");
 }

@@ -12,6 +12,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+from .. import ops
 from . import dist as pdist
 
 
@@ -31,7 +32,7 @@ class ColumnParallelLinear(nn.Module):
         )
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        y = F.linear(x, self.weight)
+        y = ops.linear(x, self.weight)
         if self.gather_output:
             y = pdist.tp_all_gather(y, dim=-1)
         return y
@@ -52,7 +53,7 @@ class RowParallelLinear(nn.Module):
         )
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        y = F.linear(x, self.weight)
+        y = ops.linear(x, self.weight)
         return pdist.tp_all_reduce(y)
 
 

@@ -18,6 +18,7 @@ SRC = [
     "aurora_amd/ops/csrc/attention_decode.hip",
     "aurora_amd/ops/csrc/attention_prefill.hip",
     "aurora_amd/ops/csrc/encoder.hip",
+    "aurora_amd/ops/csrc/skinny_gemm.hip",
 ]
 
 setup(

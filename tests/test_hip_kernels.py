@@ -254,3 +254,17 @@ def test_engine_gpu_grammar_json():
     out = eng.generate([[1, 2, 3]], p)[0]
     data = bytes(t for t in out if t < 256)
     assert validate_json_bytes(data), data
+
+
+@pytest.mark.parametrize("M,N,K", [(1, 6144, 4096), (16, 6144, 4096),
+                                   (16, 128256, 4096), (33, 4096, 14336),
+                                   (64, 28672, 4096), (5, 512, 512)])
+def test_skinny_gemm(M, N, K):
+    from aurora_amd import _hip_ops
+    torch.manual_seed(0)
+    x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16) / (K ** 0.25)
+    w = torch.randn(N, K, device="cuda", dtype=torch.bfloat16) / (K ** 0.25)
+    out = _hip_ops.skinny_gemm(x, w)
+    ref = (x.float() @ w.float().t())
+    err = (out.float() - ref).abs().max() / ref.abs().max().clamp(min=1e-6)
+    assert err.item() < 3e-2, err.item()

@@ -115,8 +115,32 @@ def main():
     # decode-shaped GEMV batch: [16, 4096] x [4096, 6144]
     a = torch.randn(16, 4096, device=dev, dtype=torch.bfloat16)
     t = timeit(lambda: a @ b.t())
-    out["gemv_16x6144x4096"] = {"us": t * 1e6,
-                                "GBps": 6144 * 4096 * 2 / t / 1e9}
+    out["gemv_16x6144x4096_blaslt"] = {"us": t * 1e6,
+                                       "GBps": 6144 * 4096 * 2 / t / 1e9}
+    from aurora_amd import _hip_ops
+    t = timeit(lambda: _hip_ops.skinny_gemm(a, b))
+    out["gemv_16x6144x4096_skinny"] = {"us": t * 1e6,
+                                       "GBps": 6144 * 4096 * 2 / t / 1e9}
+    bd = torch.randn(4096, 14336, device=dev, dtype=torch.bfloat16)
+    t = timeit(lambda: _hip_ops.skinny_gemm(a[:, :4096], bd.t().contiguous()))
+    t2 = timeit(lambda: _hip_ops.skinny_gemm(torch.randn(16, 14336, device=dev, dtype=torch.bfloat16), bd))
+    out["gemv_16x4096x14336_skinny_down"] = {"us": t2 * 1e6,
+                                             "GBps": 14336 * 4096 * 2 / t2 / 1e9}
+    for (nn, kk, tag) in [(4096, 4096, "o"), (28672, 4096, "gateup"),
+                          (4096, 14336, "down")]:
+        ww = torch.randn(nn, kk, device=dev, dtype=torch.bfloat16)
+        aa = torch.randn(16, kk, device=dev, dtype=torch.bfloat16)
+        tb = timeit(lambda: aa @ ww.t())
+        ts = timeit(lambda: _hip_ops.skinny_gemm(aa, ww))
+        out[f"gemv_16_{tag}"] = {"blaslt_GBps": nn * kk * 2 / tb / 1e9,
+                                 "skinny_GBps": nn * kk * 2 / ts / 1e9}
+    blm = torch.randn(128256, 4096, device=dev, dtype=torch.bfloat16)
+    t = timeit(lambda: _hip_ops.skinny_gemm(a, blm), iters=10)
+    out["gemv_16x128256x4096_skinny_lmhead"] = {"us": t * 1e6,
+                                                "GBps": 128256 * 4096 * 2 / t / 1e9}
+    t = timeit(lambda: a @ blm.t(), iters=10)
+    out["gemv_16x128256x4096_blaslt_lmhead"] = {"us": t * 1e6,
+                                                "GBps": 128256 * 4096 * 2 / t / 1e9}
 
     print(json.dumps(out, indent=1))
 
