@@ -159,6 +159,9 @@ class LLMEngine:
             "prompt_tokens": 0, "generated_tokens": 0, "prefill_steps": 0,
             "decode_steps": 0, "preemptions": 0, "grammar_fallbacks": 0,
         }
+        # phase timers (seconds, cumulative) for perf triage
+        self.timers = {"schedule": 0.0, "build": 0.0, "forward": 0.0,
+                       "sample": 0.0, "prefill_forward": 0.0}
 
     # ------------------------------------------------------------------ API
 
@@ -307,7 +310,9 @@ class LLMEngine:
 
     @torch.inference_mode()
     def step(self) -> List[StepOutput]:
+        _t0 = time.monotonic()
         mode, seqs, chunks = self._schedule()
+        self.timers["schedule"] += time.monotonic() - _t0
         if not seqs:
             return []
         dev = self.device
@@ -335,7 +340,11 @@ class LLMEngine:
                 cu_seqlens_q=torch.tensor(cu, dtype=torch.int32, device=dev),
             )
             input_ids = torch.tensor(ids, dtype=torch.long, device=dev)
+            _tp = time.monotonic()
             hidden = self.model(input_ids, self.kv.caches, meta)
+            if dev != "cpu":
+                torch.cuda.synchronize()
+            self.timers["prefill_forward"] += time.monotonic() - _tp
             outputs: List[StepOutput] = []
             sample_rows = []
             sample_seqs = []
@@ -366,6 +375,7 @@ class LLMEngine:
         # written; it is this step's input token.
         slots = [self._slot(s, s.num_tokens - 1) for s in seqs]
         ctx_lens = [s.num_tokens for s in seqs]
+        _t1 = time.monotonic()
         if self._graphs_enabled() and len(seqs) <= self._BUCKETS[-1]:
             logits = self._decode_with_graph(seqs, ids, positions, slots, ctx_lens)
         else:
@@ -379,9 +389,15 @@ class LLMEngine:
             input_ids = torch.tensor(ids, dtype=torch.long, device=dev)
             hidden = self.model(input_ids, self.kv.caches, meta)
             logits = self.model.compute_logits(hidden)
+        if dev != "cpu":
+            torch.cuda.synchronize()
+        _t2 = time.monotonic()
+        self.timers["forward"] += _t2 - _t1
         for s in seqs:
             s.num_cached = s.num_tokens
-        return self._sample_and_append(seqs, logits)
+        outs = self._sample_and_append(seqs, logits)
+        self.timers["sample"] += time.monotonic() - _t2
+        return outs
 
     # ----------------------------------------------------- hipGraph decode
 

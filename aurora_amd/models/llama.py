@@ -70,13 +70,11 @@ class LlamaAttention(nn.Module):
     ) -> torch.Tensor:
         T = x.shape[0]
         qkv = self.qkv_proj(x)
-        q, k, v = self.qkv_proj.split_qkv(qkv)
-        q = q.view(T, self.n_heads, self.head_dim)
-        k = k.view(T, self.n_kv_heads, self.head_dim)
-        v = v.view(T, self.n_kv_heads, self.head_dim)
-        q, k = ops.rope_apply(q, k, meta.positions, cos_sin)
         k_cache, v_cache = kv_cache
-        ops.kv_cache_write(k, v, k_cache, v_cache, meta.slot_mapping)
+        # fused split + RoPE + paged K/V scatter (one kernel on GPU)
+        q = ops.fused_rope_kv_write(qkv, meta.positions, meta.slot_mapping,
+                                    cos_sin, k_cache, v_cache,
+                                    self.n_heads, self.n_kv_heads)
         if meta.mode == "prefill":
             o = ops.attention_prefill_paged(
                 q, k_cache, v_cache, meta.block_tables,

@@ -140,6 +140,26 @@ def paged_attention_decode(
     return torch_ref.paged_attention_decode(q, k_cache, v_cache, block_tables, context_lens, scale)
 
 
+def fused_rope_kv_write(qkv, positions, slots, cos_sin, k_cache, v_cache,
+                        n_heads: int, n_kv_heads: int):
+    """Split fused qkv output, RoPE q/k, scatter k/v into the paged cache;
+    returns contiguous roped q [T, Hq, D].  One kernel on GPU; composed
+    reference ops on CPU."""
+    if qkv.is_cuda:
+        m = _require_hip("fused_rope_kv_write")
+        return m.fused_rope_kv_write(qkv.contiguous(), positions.to(torch.int32),
+                                     slots.to(torch.int32), cos_sin,
+                                     k_cache, v_cache, n_heads, n_kv_heads)
+    T = qkv.shape[0]
+    D = k_cache.shape[3]
+    q = qkv[:, : n_heads * D].reshape(T, n_heads, D)
+    k = qkv[:, n_heads * D: (n_heads + n_kv_heads) * D].reshape(T, n_kv_heads, D)
+    v = qkv[:, (n_heads + n_kv_heads) * D:].reshape(T, n_kv_heads, D)
+    q, k = torch_ref.rope_apply(q, k, positions, cos_sin)
+    torch_ref.kv_cache_write(k, v, k_cache, v_cache, slots)
+    return q
+
+
 def embed_pool(hidden: torch.Tensor, cu_seqlens: torch.Tensor) -> torch.Tensor:
     if hidden.is_cuda:
         return _require_hip("embed_pool").embed_pool(hidden.contiguous(), cu_seqlens.to(torch.int32))
@@ -159,9 +179,16 @@ def linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     (lm-head shapes), the skinny kernel at mid-size per-layer shapes."""
     import torch.nn.functional as F
     if (x.is_cuda and x.dim() == 2 and 0 < x.shape[0] <= 64
-            and x.shape[1] % 64 == 0 and weight.shape[0] <= 32768
-            and x.dtype == torch.bfloat16 and _hip is not None):
-        return _hip.skinny_gemm(x.contiguous(), weight)
+            and x.shape[1] % 64 == 0 and x.dtype == torch.bfloat16
+            and _hip is not None):
+        w_bytes = weight.shape[0] * weight.shape[1] * 2
+        # measured crossover (tools/sweep_skinny.py, profiles/): hipBLASLt is
+        # near-streaming-peak above ~100 MB of W (gate_up/down/lm-head) but
+        # 1.7-2.5 TB/s on the small qkv/o shapes where this kernel reaches
+        # 2.8-2.9 TB/s with a short split-K
+        if w_bytes <= 96 * 1024 * 1024:
+            s_target = 256 if w_bytes <= 40 * 1024 * 1024 else 512
+            return _hip.skinny_gemm_ex(x.contiguous(), weight, s_target, 2)
     return F.linear(x, weight)
 
 

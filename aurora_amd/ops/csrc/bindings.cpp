@@ -21,6 +21,11 @@ torch::Tensor attention_prefill(torch::Tensor q, torch::Tensor k, torch::Tensor 
 torch::Tensor embed_pool(torch::Tensor hidden, torch::Tensor cu);
 std::vector<torch::Tensor> knn_topk(torch::Tensor query, torch::Tensor docs, long k);
 torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w);
+torch::Tensor fused_rope_kv_write(torch::Tensor qkv, torch::Tensor positions,
+                                  torch::Tensor slots, torch::Tensor cos_sin,
+                                  torch::Tensor kc, torch::Tensor vc,
+                                  long Hq, long Hkv);
+torch::Tensor skinny_gemm_ex(torch::Tensor x, torch::Tensor w, long s_target, long unroll);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "fused RMSNorm (bf16, gfx950)");
@@ -35,4 +40,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("embed_pool", &embed_pool, "mean-pool + L2 normalize");
   m.def("knn_topk", &knn_topk, "brute-force kNN top-k");
   m.def("skinny_gemm", &skinny_gemm, "decode-path skinny GEMM (M<=64)");
+  m.def("fused_rope_kv_write", &fused_rope_kv_write,
+        "fused qkv split + RoPE + paged KV scatter");
+  m.def("skinny_gemm_ex", &skinny_gemm_ex, "skinny GEMM with tuning knobs");
 }

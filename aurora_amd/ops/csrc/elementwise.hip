@@ -248,3 +248,78 @@ void kv_cache_write(torch::Tensor k, torch::Tensor v, torch::Tensor kc,
                      slots.data_ptr<int>(), T, Hkv, P, D);
   HIP_CHECK_KERNEL();
 }
+
+// ------------------------------------------------- fused qkv post-process
+// One kernel replaces: q/k/v slice copies (the qkv GEMM output is sliced
+// non-contiguously), RoPE on q and k, and the paged-cache K/V scatter —
+// rocprof showed those as ~5 launches + 3 copies per layer of pure launch
+// floor in the decode step (profiles/decode_kernel_stats_r01.csv).
+// qkv: [T, (Hq+2*Hkv)*D] -> writes q_out [T, Hq, D] (roped), kc/vc slots.
+__global__ void fused_rope_kv_write_kernel(
+    const __hip_bfloat16* __restrict__ qkv,
+    __hip_bfloat16* __restrict__ q_out,
+    __hip_bfloat16* __restrict__ kc, __hip_bfloat16* __restrict__ vc,
+    const int* __restrict__ positions, const int* __restrict__ slots,
+    const float* __restrict__ cos_sin,
+    int T, int Hq, int Hkv, int D, int P) {
+  const int half = D / 2;
+  const int HT = Hq + 2 * Hkv;
+  const int waves_per_block = blockDim.x / WAVE;
+  const long wave_id = (long)blockIdx.x * waves_per_block + threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  if (wave_id >= (long)T * HT) return;
+  const int t = wave_id / HT;
+  const int h = wave_id % HT;
+  const __hip_bfloat16* src = qkv + ((long)t * HT + h) * D;
+  if (h < Hq + Hkv) {
+    // roped write (q -> q_out, k -> cache)
+    const int pos = positions[t];
+    __hip_bfloat16* dst;
+    if (h < Hq) {
+      dst = q_out + ((long)t * Hq + h) * D;
+    } else {
+      const int slot = slots[t];
+      if (slot < 0) return;
+      dst = kc + ((((long)(slot / P) * Hkv + (h - Hq)) * P + slot % P)) * D;
+    }
+    for (int i = lane; i < half; i += WAVE) {
+      const float c = cos_sin[(long)pos * D + i];
+      const float s = cos_sin[(long)pos * D + half + i];
+      const float x1 = bf16_to_f32(*(const unsigned short*)(src + i));
+      const float x2 = bf16_to_f32(*(const unsigned short*)(src + half + i));
+      *(unsigned short*)(dst + i) = f32_to_bf16(x1 * c - x2 * s);
+      *(unsigned short*)(dst + half + i) = f32_to_bf16(x2 * c + x1 * s);
+    }
+  } else {
+    const int slot = slots[t];
+    if (slot < 0) return;
+    __hip_bfloat16* dst =
+        vc + ((((long)(slot / P) * Hkv + (h - Hq - Hkv)) * P + slot % P)) * D;
+    for (int i = lane * 8; i < D; i += WAVE * 8)
+      *reinterpret_cast<shortx8*>(dst + i) =
+          *reinterpret_cast<const shortx8*>(src + i);
+  }
+}
+
+torch::Tensor fused_rope_kv_write(torch::Tensor qkv, torch::Tensor positions,
+                                  torch::Tensor slots, torch::Tensor cos_sin,
+                                  torch::Tensor kc, torch::Tensor vc,
+                                  long Hq, long Hkv) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.dtype() == torch::kBFloat16, "bf16 GPU only");
+  const int T = qkv.size(0);
+  const int P = kc.size(2), D = kc.size(3);
+  TORCH_CHECK(qkv.size(1) == (Hq + 2 * Hkv) * D, "qkv width mismatch");
+  auto q_out = torch::empty({(long)T, Hq, (long)D}, qkv.options());
+  if (T == 0) return q_out;
+  const long waves = (long)T * (Hq + 2 * Hkv);
+  const int wpb = 4;
+  hipLaunchKernelGGL(fused_rope_kv_write_kernel,
+                     dim3((waves + wpb - 1) / wpb), dim3(wpb * WAVE), 0,
+                     CUR_STREAM, (const __hip_bfloat16*)qkv.data_ptr(),
+                     (__hip_bfloat16*)q_out.data_ptr(),
+                     (__hip_bfloat16*)kc.data_ptr(), (__hip_bfloat16*)vc.data_ptr(),
+                     positions.data_ptr<int>(), slots.data_ptr<int>(),
+                     cos_sin.data_ptr<float>(), T, (int)Hq, (int)Hkv, D, P);
+  HIP_CHECK_KERNEL();
+  return q_out;
+}

@@ -25,7 +25,7 @@ union sfrag {
 };
 
 // MT = number of 16-row M tiles (M <= 16*MT); SPLIT: write f32 partials.
-template <int MT, bool SPLIT>
+template <int MT, bool SPLIT, int UNR>
 __global__ __launch_bounds__(256) void skinny_gemm_kernel(
     const __hip_bfloat16* __restrict__ x,  // [M, K]
     const __hip_bfloat16* __restrict__ w,  // [N, K]
@@ -61,15 +61,15 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   for (int t = 0; t < MT; ++t)
     xbase[t] = (long)min(t * 16 + col, M - 1) * K;  // clamped (masked at store)
 
-  // 2 k-steps per iteration: 2 W loads (+2*MT x loads) in flight
-  for (int k = k_begin; k < k_end; k += 64) {
-    shortx8 wv[2];
-    sfrag bf[2], af[2][MT];
+  // UNR k-steps per iteration: UNR W loads (+UNR*MT x loads) in flight
+  for (int k = k_begin; k < k_end; k += 32 * UNR) {
+    shortx8 wv[UNR];
+    sfrag bf[UNR], af[UNR][MT];
 #pragma unroll
-    for (int u = 0; u < 2; ++u)
+    for (int u = 0; u < UNR; ++u)
       wv[u] = *reinterpret_cast<const shortx8*>(w + wbase + k + u * 32 + krow);
 #pragma unroll
-    for (int u = 0; u < 2; ++u)
+    for (int u = 0; u < UNR; ++u)
 #pragma unroll
       for (int t = 0; t < MT; ++t) {
         const shortx8 xv =
@@ -77,7 +77,7 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
         __builtin_memcpy(&af[u][t], &xv, 16);
       }
 #pragma unroll
-    for (int u = 0; u < 2; ++u) {
+    for (int u = 0; u < UNR; ++u) {
       __builtin_memcpy(&bf[u], &wv[u], 16);
 #pragma unroll
       for (int t = 0; t < MT; ++t)
@@ -114,7 +114,7 @@ __global__ void skinny_combine_kernel(const float* __restrict__ part,
   }
 }
 
-torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w) {
+torch::Tensor skinny_gemm_ex(torch::Tensor x, torch::Tensor w, long s_target, long unroll) {
   TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16, "bf16 GPU only");
   TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.size(1) == w.size(1));
   const int M = x.size(0), K = x.size(1), N = w.size(0);
@@ -123,9 +123,12 @@ torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w) {
   auto out = torch::empty({M, N}, x.options());
   if (M == 0 || N == 0) return out;
   const int nblocks = (N + 63) / 64;
-  int S = std::max(1, std::min(16, 512 / std::max(nblocks, 1)));
-  int KS = ((K / S + 63) / 64) * 64;
+  const int target = s_target > 0 ? (int)s_target : 1024;
+  const int kquant = 32 * (int)unroll;
+  int S = std::max(1, std::min(32, target / std::max(nblocks, 1)));
+  int KS = ((K / S + kquant - 1) / kquant) * kquant;
   S = (K + KS - 1) / KS;  // actual split count after rounding
+  TORCH_CHECK(K % kquant == 0, "K must be a multiple of ", kquant);
   const int MT = (M + 15) / 16;
   auto stream = CUR_STREAM;
   torch::Tensor part;
@@ -133,7 +136,14 @@ torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w) {
     part = torch::empty({S, M, N}, x.options().dtype(torch::kFloat32));
   dim3 grid(nblocks, S), block(256);
 #define SG_LAUNCH(MTV, SPLITV)                                               \
-  hipLaunchKernelGGL((skinny_gemm_kernel<MTV, SPLITV>), grid, block, 0,      \
+  if (unroll == 4)                                                           \
+    hipLaunchKernelGGL((skinny_gemm_kernel<MTV, SPLITV, 4>), grid, block, 0, \
+                     stream, (const __hip_bfloat16*)x.data_ptr(),            \
+                     (const __hip_bfloat16*)w.data_ptr(),                    \
+                     (__hip_bfloat16*)out.data_ptr(),                        \
+                     SPLITV ? part.data_ptr<float>() : nullptr, M, N, K, KS);\
+  else                                                                       \
+    hipLaunchKernelGGL((skinny_gemm_kernel<MTV, SPLITV, 2>), grid, block, 0, \
                      stream, (const __hip_bfloat16*)x.data_ptr(),            \
                      (const __hip_bfloat16*)w.data_ptr(),                    \
                      (__hip_bfloat16*)out.data_ptr(),                        \
@@ -161,4 +171,9 @@ torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w) {
 #undef SG_LAUNCH
   HIP_CHECK_KERNEL();
   return out;
+}
+
+
+torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w) {
+  return skinny_gemm_ex(x, w, 0, 2);
 }

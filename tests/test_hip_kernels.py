@@ -268,3 +268,23 @@ def test_skinny_gemm(M, N, K):
     ref = (x.float() @ w.float().t())
     err = (out.float() - ref).abs().max() / ref.abs().max().clamp(min=1e-6)
     assert err.item() < 3e-2, err.item()
+
+
+def test_fused_rope_kv_write():
+    torch.manual_seed(4)
+    T, Hq, Hkv, D, P, NP = 29, 8, 2, 128, 64, 4
+    cs = R.rope_cos_sin_cache(D, 2048, 500000.0).cuda()
+    qkv = torch.randn(T, (Hq + 2 * Hkv) * D, device="cuda", dtype=torch.bfloat16)
+    kc = torch.zeros(NP, Hkv, P, D, device="cuda", dtype=torch.bfloat16)
+    vc = torch.zeros_like(kc)
+    pos = torch.randint(0, 2048, (T,), dtype=torch.int32, device="cuda")
+    slots = torch.randperm(NP * P)[:T].to(torch.int32).cuda()
+    q = ops.fused_rope_kv_write(qkv, pos, slots, cs, kc, vc, Hq, Hkv)
+    # CPU composed reference on the same inputs
+    kc_ref = torch.zeros(NP, Hkv, P, D)
+    vc_ref = torch.zeros_like(kc_ref)
+    q_ref = ops.fused_rope_kv_write(qkv.cpu().float(), pos.cpu(), slots.cpu(),
+                                    cs.cpu(), kc_ref, vc_ref, Hq, Hkv)
+    assert _err(q, q_ref) < 2e-2
+    assert _err(kc, kc_ref) < 2e-2
+    assert _err(vc, vc_ref) < 2e-2

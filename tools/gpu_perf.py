@@ -62,6 +62,13 @@ def main():
     kv_bytes = B * Hkv * ctxv * D * 2 * 2
     out[f"paged_decode_B{B}_ctx{ctxv}"] = {"us": t * 1e6, "KV_GBps": kv_bytes / t / 1e9}
 
+    B3 = 2
+    q3 = torch.randn(B3, Hq, D, device=dev, dtype=torch.bfloat16)
+    ctx3 = torch.full((B3,), ctxv, dtype=torch.int32, device=dev)
+    t = timeit(lambda: ops.paged_attention_decode(q3, kc, vc, bt[:B3], ctx3, scale))
+    out[f"paged_decode_B{B3}_ctx{ctxv}"] = {
+        "us": t * 1e6, "KV_GBps": B3 * Hkv * ctxv * D * 2 * 2 / t / 1e9}
+
     B2 = 64
     bt2 = bt.repeat(4, 1)
     q2 = torch.randn(B2, Hq, D, device=dev, dtype=torch.bfloat16)
