@@ -1,0 +1,131 @@
+"""Multi-process distributed tests (gloo backend, world_size=2, CPU).
+
+The driver's 8-GPU scaling runs are out of our hands; these make the
+distributed path correct by construction: TP=2 forward must equal the
+single-process full model, and the DP bench contract must aggregate."""
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+import torch
+import torch.multiprocessing as mp
+
+ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _tp_worker(rank: int, world: int, port: int, q):
+    os.environ.update(RANK=str(rank), WORLD_SIZE=str(world),
+                      LOCAL_RANK=str(rank), MASTER_ADDR="127.0.0.1",
+                      MASTER_PORT=str(port))
+    sys.path.insert(0, ROOT)
+    import torch
+    from aurora_amd.models.config import get_spec
+    from aurora_amd.models.llama import AttnMetadata, LlamaForCausalLM
+    from aurora_amd.parallel import dist as pdist
+
+    # phase 1: TP=1 so the full (unsharded) model builds correctly
+    pdist.init_distributed(tp_size=1, backend="gloo")
+    spec = get_spec("tiny-llama")
+    torch.manual_seed(0)
+    full = LlamaForCausalLM(spec, dtype=torch.float32)
+    full.init_random(7)
+    D = spec.head_dim
+    hq, hkv = spec.n_heads // world, spec.n_kv_heads // world
+    inter = spec.intermediate_size // world
+
+    def build_tp(full, spec, rank, world):
+        from aurora_amd.models.llama import LlamaForCausalLM
+        tp = LlamaForCausalLM(spec, dtype=torch.float32)
+        _copy_shards(tp, full, spec, rank, world)
+        return tp
+
+    def _copy_shards(tp, full, spec, rank, world):
+      with torch.no_grad():
+            tp.embed.weight.copy_(full.embed.weight)
+            tp.final_norm.copy_(full.final_norm)
+            for lf, lt in zip(full.layers, tp.layers):
+                lt.input_norm.copy_(lf.input_norm)
+                lt.post_attn_norm.copy_(lf.post_attn_norm)
+                wf = lf.attn.qkv_proj.weight  # [(Hq+2Hkv)*D, H] full
+                q_w = wf[: spec.n_heads * D]
+                k_w = wf[spec.n_heads * D: (spec.n_heads + spec.n_kv_heads) * D]
+                v_w = wf[(spec.n_heads + spec.n_kv_heads) * D:]
+                lt.attn.qkv_proj.weight.copy_(torch.cat([
+                    q_w[rank * hq * D: (rank + 1) * hq * D],
+                    k_w[rank * hkv * D: (rank + 1) * hkv * D],
+                    v_w[rank * hkv * D: (rank + 1) * hkv * D]]))
+                # o_proj: row-parallel (input sharded by head)
+                lt.attn.o_proj.weight.copy_(
+                    lf.attn.o_proj.weight[:, rank * hq * D: (rank + 1) * hq * D])
+                gf = lf.mlp.gate_up.weight  # [2I, H]
+                lt.mlp.gate_up.weight.copy_(torch.cat([
+                    gf[rank * inter: (rank + 1) * inter],
+                    gf[spec.intermediate_size + rank * inter:
+                       spec.intermediate_size + (rank + 1) * inter]]))
+                lt.mlp.down.weight.copy_(
+                    lf.mlp.down.weight[:, rank * inter: (rank + 1) * inter])
+
+    L = 12
+    P = 16
+    ids = torch.arange(5, 5 + L)
+
+    def run(model, kvh):
+        caches = [(torch.zeros(2, kvh, P, D), torch.zeros(2, kvh, P, D))
+                  for _ in range(spec.n_layers)]
+        meta = AttnMetadata(
+            mode="prefill",
+            positions=torch.arange(L, dtype=torch.int32),
+            slot_mapping=torch.arange(L, dtype=torch.int32),
+            block_tables=torch.arange(2, dtype=torch.int32).unsqueeze(0),
+            context_lens=torch.tensor([L], dtype=torch.int32),
+            cu_seqlens_q=torch.tensor([0, L], dtype=torch.int32),
+        )
+        h = model(ids, caches, meta)
+        return model.compute_logits(h)
+
+    full_logits = run(full, spec.n_kv_heads)
+    # phase 2: switch to TP=2, build the sharded model, carve shards
+    pdist.init_distributed(tp_size=world, backend="gloo")
+    tp = build_tp(full, spec, rank, world)
+    tp_logits = run(tp, hkv)
+    err = (tp_logits - full_logits).abs().max().item()
+    q.put((rank, err))
+
+
+@pytest.mark.timeout(300)
+def test_tp2_matches_single_process():
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29611
+    procs = [ctx.Process(target=_tp_worker, args=(r, 2, port, q)) for r in range(2)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=240) for _ in range(2)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, err in results:
+        assert err < 1e-3, (rank, err)
+
+
+@pytest.mark.timeout(300)
+def test_bench_dp2_cpu():
+    """The driver's multi-rank launch shape: torch.distributed.run, 2 CPU
+    ranks, gloo; rank 0 prints one JSON line with whole-job aggregation."""
+    env = dict(os.environ, BENCH_MODEL="tiny-llama", BENCH_CONCURRENCY="2",
+               BENCH_TURNS="2", MASTER_ADDR="127.0.0.1")
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29612", "bench.py", "--gpus", "2", "--steps", "1",
+         "--warmup", "0", "--device", "cpu", "--first-prompt", "32",
+         "--turn-prompt", "16", "--decode-per-turn", "8"],
+        cwd=ROOT, env=env, capture_output=True, text=True, timeout=240)
+    lines = [ln for ln in out.stdout.splitlines() if ln.startswith("{")]
+    assert lines, out.stdout + out.stderr
+    doc = json.loads(lines[-1])
+    assert doc["metric"] == "incidents/hr"
+    assert doc["config"]["global_batch"] == 4  # 2 ranks x 2 incidents
+    assert doc["config"]["parallelism"] == "dp2"
+    assert doc["value"] > 0
