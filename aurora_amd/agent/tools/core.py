@@ -68,6 +68,31 @@ async def cloud_exec(ctx: ToolContext, command: str) -> str:
     return await terminal_exec(ctx, command)
 
 
+@R.add("on_prem_kubectl",
+       "Run kubectl on a connected on-prem cluster through its outbound "
+       "agent tunnel (safety-gated).",
+       {"cluster": "cluster id (empty = the only connected cluster)",
+        "args": "kubectl arguments"}, read_only=False)
+async def on_prem_kubectl(ctx: ToolContext, args: str, cluster: str = "") -> str:
+    agents = ctx.extra.get("kubectl_agents")
+    if agents is None and ctx.queue is not None:
+        agents = getattr(ctx.extra.get("runtime"), "kubectl_agents", None)
+    if not agents:
+        return "ERROR: no on-prem kubectl agent connected"
+    if not cluster:
+        if len(agents) != 1:
+            return f"ERROR: specify cluster; connected: {sorted(agents)}"
+        cluster = next(iter(agents))
+    conn = agents.get(cluster)
+    if conn is None:
+        return f"ERROR: cluster '{cluster}' not connected"
+    verdict = await evaluate_command(f"kubectl {args}", llm=ctx.llm,
+                                     org_store=ctx.org_store, user_id=ctx.user_id)
+    if not verdict.safe:
+        return f"BLOCKED by guardrail ({verdict.source}): {verdict.reason}"
+    return await conn.exec(args)
+
+
 # ------------------------------------------------------------ introspection
 # (reference: tools/introspection_tools.py:101-341)
 

@@ -104,6 +104,7 @@ class Workflow:
             if (time.monotonic() - t_start) > self.settings.workflow_timeout_s:
                 yield {"type": "error", "error": "workflow timeout"}
                 break
+            await self._maybe_compress(state)
             prompt = (system + "\n\n# Investigation so far\n"
                       + render_history(state.messages) + "\nThought:")
             # --- thought (streamed) ---
@@ -175,13 +176,36 @@ class Workflow:
                "elapsed_ms": (time.monotonic() - t_start) * 1000.0}
         self._persist(state)
 
+    async def _maybe_compress(self, state: State) -> None:
+        """Preflight conversation compression when the context outgrows the
+        engine's real KV budget (reference: chat_context_manager.py:109
+        should_summarize_context + agent.py:785-801 preflight compression;
+        retuned against our own max_model_len per SURVEY.md §5.7)."""
+        budget = int(self.llm.server.engine.max_model_len * 0.5)
+        hist = render_history(state.messages, max_messages=10**6)
+        if self.llm.count_tokens(hist) <= budget or len(state.messages) <= 6:
+            return
+        keep = state.messages[-4:]
+        old = render_history(state.messages[:-4], max_messages=10**6,
+                             tool_output_chars=800)
+        summary = await self.llm.summarize(
+            old, max_tokens=256,
+            instruction="Compress this investigation history, keeping every "
+                        "confirmed fact, ruled-out cause and open hypothesis:")
+        state.messages = (
+            [state.messages[0]] if state.messages[0].get("role") == "user" else []
+        ) + [{"role": "observation",
+              "content": f"[compressed earlier investigation]\n{summary}"}] + keep
+
     async def _forced_final(self, system: str, state: State) -> str:
         res = await self.llm.invoke(
             system + "\n\n# Investigation so far\n" + render_history(state.messages)
             + "\nWrite the final RCA report now.\nFinal answer:",
             SamplingParams(temperature=0.2, max_tokens=512),
             purpose="rca", session_id=state.session_id)
-        return res.text.strip()
+        return res.text.strip() or (
+            "Root cause undetermined. The investigation loop ended without "
+            "a conclusive final answer; see the execution trace.")
 
     def _persist(self, state: State) -> None:
         store = self.ctx.org_store

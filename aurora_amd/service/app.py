@@ -79,6 +79,22 @@ def create_app(rt: Runtime) -> FastAPI:
         iid, created = process_webhook(rt, org_id, source, payload, user_id=user_id)
         return {"incident_id": iid, "created": created}
 
+    @app.post("/webhook/pr/{provider}/{org_id}")
+    async def pr_webhook(provider: str, org_id: str, request: Request):
+        # change gating (reference: tasks/github_webhook_tasks.py:1090)
+        if provider not in ("github", "bitbucket"):
+            raise HTTPException(404, "provider must be github|bitbucket")
+        body = await request.body()
+        store = rt.org(org_id)
+        secret = store.get_token_ref("system", f"webhook_secret:{provider}")
+        from .webhooks import verify_signature as _vs
+        if not _vs(provider, body, dict(request.headers), secret):
+            raise HTTPException(403, "bad signature")
+        from .change_gating import process_pr_webhook
+        tid = process_pr_webhook(rt, org_id, provider,
+                                 json.loads(body or b"{}"))
+        return {"task_id": tid}
+
     # ------------------------------------------------------------ incidents
 
     @app.get("/incidents")
@@ -228,7 +244,8 @@ def create_app(rt: Runtime) -> FastAPI:
     # ---- llm usage / audit ----
     @app.get("/llm-usage")
     async def llm_usage(store: OrgStore = Depends(auth)):
-        return store.usage_summary()
+        from ..utils.billing import usage_with_cost
+        return usage_with_cost(store.usage_summary())
 
     @app.get("/audit")
     async def audit_log(store: OrgStore = Depends(auth)):

@@ -79,6 +79,7 @@ class Runtime:
         self._rag: Dict[str, RagIndex] = {}
         self._graph: Dict[str, InfraGraph] = {}
         self._rca_rate: Dict[str, float] = {}  # user -> last background RCA ts
+        self.kubectl_agents: Dict[str, Any] = {}  # cluster_id -> AgentConn
         from . import worker as _worker  # registers task handlers
         _worker.register_tasks(self)
 
@@ -110,7 +111,8 @@ class Runtime:
             org_store=store, user_id=user_id, session_id=session_id,
             incident_id=incident_id, llm=self.llm(org_id), mode=mode,
             rag=self.rag(org_id), graph=self.graph(org_id), queue=self.queue,
-            notifier=notifier, settings=self.settings, depth=depth)
+            notifier=notifier, settings=self.settings, depth=depth,
+            extra={"kubectl_agents": self.kubectl_agents, "runtime": self})
 
     # ---- RCA trigger rate limit (reference: task.py:336) ----
     def rca_allowed(self, user_key: str) -> bool:
@@ -126,6 +128,7 @@ class Runtime:
         # beat schedules (reference: celery_config.py:113-146)
         self.queue.add_beat("cleanup_stale_investigations", 300.0, {})
         self.queue.add_beat("action_scheduler", 60.0, {})
+        self.queue.add_beat("run_prediscovery", 3600.0, {})  # hourly (reference: celery_config.py:126-129)
         self.queue.start()
 
     async def stop(self) -> None:

@@ -22,14 +22,18 @@ from .summarization import (alert_only_summary, determine_severity,
 
 
 def build_rca_prompt(source: str, title: str, payload: dict,
-                     prior_rcas: Optional[list] = None) -> str:
+                     prior_rcas: Optional[list] = None,
+                     prediscovery: str = "") -> str:
     """reference: rca_prompt_builder.py:336 — alert-field extraction +
-    'Aurora Learn' similar-good-RCA injection."""
+    prediscovery context injection + 'Aurora Learn' similar-good-RCA
+    injection."""
     lines = [
         f"A {source} alert fired. Investigate it end-to-end and produce an RCA.",
         f"Title: {title}",
         f"Alert payload: {json.dumps(payload)[:4000]}",
     ]
+    if prediscovery:
+        lines.append(f"Known infrastructure (prediscovery): {prediscovery[:2000]}")
     if prior_rcas:
         lines.append("Similar past investigations (for reference):")
         for r in prior_rcas[:2]:
@@ -60,13 +64,28 @@ async def run_background_chat(rt, args: Dict[str, Any]) -> dict:
     store.lifecycle(incident_id, "rca_started", {"session_id": session_id})
     rt.events.publish(org_id, {"type": "incident_update", "incident_id": incident_id,
                                "aurora_status": "running"})
+    # billing/entitlement hook gate (reference: task.py:677-697)
+    from ..utils.hooks import fire
+    if not await fire("before_llm_call", org_id=org_id, user_id=user_id,
+                      purpose="background_rca", default=True):
+        store.update_incident(incident_id, aurora_status="failed")
+        store.lifecycle(incident_id, "rca_denied", {"reason": "hook gate"})
+        return {"error": "before_llm_call hook denied"}
     # on_incident actions (reference: services/actions/executor.py:111)
     rt.queue.enqueue("dispatch_on_incident_actions",
                      {"org_id": org_id, "incident_id": incident_id})
+    from .notifications import notify_investigation_started
+    await notify_investigation_started(rt, org_id, incident_id)
 
     payload = json.loads(inc.get("payload") or "{}")
+    from .prediscovery import prediscovery_context
+    # "Aurora Learn": inject summaries of similar completed RCAs
+    prior = [i["summary"] for i in store.list_incidents(limit=30)
+             if i["id"] != incident_id and i.get("summary")
+             and i["aurora_status"] == "completed"][:2]
     prompt = args.get("prompt") or build_rca_prompt(
-        inc["source"], inc["title"], payload)
+        inc["source"], inc["title"], payload, prior_rcas=prior,
+        prediscovery=prediscovery_context(store))
     llm = rt.llm(org_id)
     ctx = rt.tool_ctx(org_id, user_id=user_id, session_id=session_id,
                       incident_id=incident_id, depth=1)  # depth=1 blocks nested RCA
@@ -127,6 +146,9 @@ async def run_background_chat(rt, args: Dict[str, Any]) -> dict:
     rt.events.publish(org_id, {"type": "incident_update", "incident_id": incident_id,
                                "aurora_status": "completed",
                                "severity": severity})
+    from .notifications import notify_investigation_completed
+    await notify_investigation_completed(rt, org_id, incident_id)
+    await fire("report_usage", org_id=org_id, incident_id=incident_id, default=None)
     return {"ok": True, "summary_chars": len(summary), "citations": n_cits,
             "time_to_rca_s": time.time() - inc["created_at"]}
 
@@ -202,3 +224,7 @@ def register_tasks(rt) -> None:
     q.register("dispatch_on_incident_actions",
                lambda a: dispatch_on_incident_actions(rt, a))
     q.register("action_scheduler", lambda a: action_scheduler(rt, a))
+    from .change_gating import investigate_pr
+    from .prediscovery import run_prediscovery
+    q.register("investigate_pr", lambda a: investigate_pr(rt, a))
+    q.register("run_prediscovery", lambda a: run_prediscovery(rt, a))

@@ -24,6 +24,36 @@ from ..agent.workflow import Workflow
 _GUID = "258EAFA5-E914-47DA-95CA-C5AB0DC85B11"
 
 
+class AgentConn:
+    """Server-side handle of a paired kubectl-agent (reference:
+    utils/kubectl/agent_ws_handler.py)."""
+
+    def __init__(self, writer):
+        self.writer = writer
+        self.pending: Dict[str, asyncio.Future] = {}
+        self._n = 0
+
+    async def exec(self, args: str, timeout_s: float = 90.0) -> str:
+        import uuid as _uuid
+        rid = _uuid.uuid4().hex[:12]
+        fut = asyncio.get_running_loop().create_future()
+        self.pending[rid] = fut
+        try:
+            self.writer.write(encode_frame(json.dumps(
+                {"type": "exec", "id": rid, "args": args}).encode()))
+            await self.writer.drain()
+            return await asyncio.wait_for(fut, timeout=timeout_s)
+        except asyncio.TimeoutError:
+            return "ERROR: on-prem agent did not respond in time"
+        finally:
+            self.pending.pop(rid, None)
+
+    def resolve(self, rid: str, output: str) -> None:
+        fut = self.pending.get(rid)
+        if fut is not None and not fut.done():
+            fut.set_result(output)
+
+
 # ------------------------------------------------------------ frame layer
 
 def _accept_key(key: str) -> str:
@@ -165,6 +195,10 @@ class ChatServer:
             await writer.drain()
             return
         query = dict(urllib.parse.parse_qsl(urllib.parse.urlsplit(path).query))
+        if "agent_token" in query:  # kubectl-agent pairing (reference:
+            # main_chatbot.py:911-914 routes agents off the same socket)
+            await self._agent_session(reader, writer, key, query["agent_token"])
+            return
         user = self._auth(query)
         if user is None:
             writer.write(b"HTTP/1.1 401 Unauthorized\r\n\r\n")
@@ -176,6 +210,44 @@ class ChatServer:
             + _accept_key(key).encode() + b"\r\n\r\n")
         await writer.drain()
         await self._session(reader, writer, user)
+
+    async def _agent_session(self, reader, writer, key: str, token: str) -> None:
+        from ..kubectl_agent.agent import cluster_id_from_token
+        from .kubectl_registry import resolve_agent_token
+        org_id = resolve_agent_token(self.rt, token)
+        if org_id is None:
+            writer.write(b"HTTP/1.1 401 Unauthorized\r\n\r\n")
+            await writer.drain()
+            return
+        writer.write(
+            b"HTTP/1.1 101 Switching Protocols\r\nUpgrade: websocket\r\n"
+            b"Connection: Upgrade\r\nSec-WebSocket-Accept: "
+            + _accept_key(key).encode() + b"\r\n\r\n")
+        await writer.drain()
+        cid = cluster_id_from_token(token)
+        conn = AgentConn(writer)
+        self.rt.kubectl_agents[cid] = conn
+        self.rt.org(org_id).audit("kubectl_agent_connected", cid)
+        try:
+            while True:
+                opcode, data = await read_frame(reader)
+                if opcode == 0x8:
+                    return
+                if opcode == 0x9:
+                    writer.write(encode_frame(data, 0xA))
+                    await writer.drain()
+                    continue
+                if opcode not in (0x1, 0x2):
+                    continue
+                try:
+                    msg = json.loads(data)
+                except json.JSONDecodeError:
+                    continue
+                if msg.get("type") == "result":
+                    conn.resolve(msg.get("id"), msg.get("output", ""))
+        finally:
+            if self.rt.kubectl_agents.get(cid) is conn:
+                del self.rt.kubectl_agents[cid]
 
     async def _send(self, writer, obj: dict) -> None:
         writer.write(encode_frame(json.dumps(obj).encode()))

@@ -184,3 +184,61 @@ def test_websocket_chat():
         asyncio.run(main())
     finally:
         server.shutdown()
+
+
+def test_kubectl_agent_tunnel():
+    """End-to-end on-prem kubectl: agent dials in over the WS port, the
+    tool proxies a command through the tunnel."""
+    import asyncio as aio
+    from aurora_amd.kubectl_agent.agent import KubectlAgent
+    from aurora_amd.service.kubectl_registry import register_kubectl_agent
+    from aurora_amd.service.ws import ChatServer
+
+    cfg = Settings(db_path=":memory:", secrets_path="/tmp/test_ka_secrets.json")
+    server = AsyncEngineServer(EngineConfig(model="tiny-llama", device="cpu", seed=6))
+    rt = Runtime(cfg, engine_server=server)
+    oid = rt.admin.create_org("ka-org")
+    rt.admin.create_user(oid, "k@x.io", user_id="kau")
+    token = "cluster-secret-token-1"
+    cid = register_kubectl_agent(rt, oid, token)
+
+    async def main():
+        cs = ChatServer(rt)
+        port = await cs.start()
+
+        def fake_runner(args):
+            return f"FAKE-KUBECTL[{args}]"
+
+        agent = KubectlAgent("127.0.0.1", port, token, runner=fake_runner,
+                             reconnect_s=0.2)
+        task = aio.create_task(agent.run_forever())
+        for _ in range(100):
+            if cid in rt.kubectl_agents:
+                break
+            await aio.sleep(0.05)
+        assert cid in rt.kubectl_agents
+        from aurora_amd.agent.tools.registry import GLOBAL_REGISTRY
+        ctx = rt.tool_ctx(oid, user_id="kau")
+        out = await GLOBAL_REGISTRY.run(ctx, "on_prem_kubectl",
+                                        {"args": "get pods -n default"})
+        assert "FAKE-KUBECTL[get pods -n default]" in out
+        # guardrail still applies through the tunnel
+        out = await GLOBAL_REGISTRY.run(ctx, "on_prem_kubectl",
+                                        {"args": "delete pods --all-namespaces"})
+        assert "BLOCKED" in out
+        # bad token rejected at pairing
+        bad = KubectlAgent("127.0.0.1", port, "wrong-token", runner=fake_runner,
+                           reconnect_s=5)
+        bad_task = aio.create_task(bad.run_forever())
+        await aio.sleep(0.3)
+        assert len(rt.kubectl_agents) == 1
+        agent.stop()
+        bad.stop()
+        task.cancel()
+        bad_task.cancel()
+        await cs.stop()
+
+    try:
+        asyncio.run(main())
+    finally:
+        server.shutdown()

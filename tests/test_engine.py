@@ -126,3 +126,32 @@ def test_abort():
     eng.abort_request("a")
     assert not eng.has_work()
     assert eng.kv.num_free() == eng.kv.num_pages
+
+
+def test_moe_model_generates():
+    """Mixtral-style MoE path through the whole engine (tiny-moe, CPU)."""
+    eng = LLMEngine(EngineConfig(model="tiny-moe", device="cpu", seed=5))
+    out = eng.generate([[3, 1, 4, 1, 5]],
+                       SamplingParams(temperature=0.0, max_tokens=6, ignore_eos=True))
+    assert len(out[0]) == 6
+    # deterministic
+    eng2 = LLMEngine(EngineConfig(model="tiny-moe", device="cpu", seed=5))
+    assert eng2.generate([[3, 1, 4, 1, 5]],
+                         SamplingParams(temperature=0.0, max_tokens=6,
+                                        ignore_eos=True)) == out
+
+
+def test_moe_routing_uses_multiple_experts():
+    import torch
+    from aurora_amd.models.config import get_spec
+    from aurora_amd.models.llama import MoEMLP
+    spec = get_spec("tiny-moe")
+    m = MoEMLP(spec, torch.float32)
+    for p in m.parameters():
+        torch.nn.init.normal_(p, std=0.1)
+    x = torch.randn(40, spec.hidden_size)
+    logits = m.router(x).float()
+    sel = logits.topk(spec.experts_per_token, dim=-1).indices
+    assert sel.unique().numel() > 1  # routing spreads across experts
+    y = m(x)
+    assert y.shape == x.shape and torch.isfinite(y).all()
