@@ -181,7 +181,8 @@ class Workflow:
         engine's real KV budget (reference: chat_context_manager.py:109
         should_summarize_context + agent.py:785-801 preflight compression;
         retuned against our own max_model_len per SURVEY.md §5.7)."""
-        budget = int(self.llm.server.engine.max_model_len * 0.5)
+        eng = getattr(self.llm.server, "engine", None)
+        budget = int(getattr(eng, "max_model_len", 32768) * 0.5)
         hist = render_history(state.messages, max_messages=10**6)
         if self.llm.count_tokens(hist) <= budget or len(state.messages) <= 6:
             return
