@@ -21,7 +21,9 @@
 
 #define CUR_STREAM c10::hip::getCurrentHIPStream().stream()
 
-#define TUNR 4  // tokens prefetched per group iteration
+#ifndef TUNR
+#define TUNR 4  // tokens prefetched per group iteration (8 measured neutral)
+#endif
 
 template <int D, int G>
 __global__ __launch_bounds__(256) void paged_decode_kernel(

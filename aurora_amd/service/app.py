@@ -95,6 +95,27 @@ def create_app(rt: Runtime) -> FastAPI:
                                  json.loads(body or b"{}"))
         return {"task_id": tid}
 
+    @app.post("/chat-events/{platform}/{org_id}")
+    async def chat_events(platform: str, org_id: str, request: Request):
+        # Slack / Google Chat two-way bridge
+        if platform not in ("slack", "google_chat"):
+            raise HTTPException(404, "platform must be slack|google_chat")
+        from .chat_integrations import normalize_chat_event
+        payload = await request.json()
+        ev = normalize_chat_event(platform, payload)
+        if ev is None:
+            return {"ok": True}
+        if "challenge" in ev:  # Slack URL verification handshake
+            return {"challenge": ev["challenge"]}
+        tid = rt.queue.enqueue("handle_chat_command",
+                               {"org_id": org_id, "event": ev})
+        return {"task_id": tid}
+
+    @app.get("/incidents/{iid}/visualizations")
+    async def visualizations(iid: str, store: OrgStore = Depends(auth)):
+        row = store.get_artifact(f"visualizations:{iid}")
+        return json.loads(row["content"]) if row else []
+
     # ------------------------------------------------------------ incidents
 
     @app.get("/incidents")

@@ -146,6 +146,8 @@ async def run_background_chat(rt, args: Dict[str, Any]) -> dict:
     rt.events.publish(org_id, {"type": "incident_update", "incident_id": incident_id,
                                "aurora_status": "completed",
                                "severity": severity})
+    from .visualization import publish_visualizations
+    publish_visualizations(rt, org_id, incident_id, state.messages)
     from .notifications import notify_investigation_completed
     await notify_investigation_completed(rt, org_id, incident_id)
     await fire("report_usage", org_id=org_id, incident_id=incident_id, default=None)
@@ -226,5 +228,7 @@ def register_tasks(rt) -> None:
     q.register("action_scheduler", lambda a: action_scheduler(rt, a))
     from .change_gating import investigate_pr
     from .prediscovery import run_prediscovery
+    from .chat_integrations import handle_chat_command
     q.register("investigate_pr", lambda a: investigate_pr(rt, a))
     q.register("run_prediscovery", lambda a: run_prediscovery(rt, a))
+    q.register("handle_chat_command", lambda a: handle_chat_command(rt, a))

@@ -167,6 +167,8 @@ def main():
                 "p50_time_to_rca_s": round(p50, 3),
                 "prompt_tokens_per_incident": args.first_prompt
                 + (args.turns - 1) * args.turn_prompt,
+                "engine_timers_s": {k: round(v, 2) for k, v in eng.timers.items()},
+                "engine_counters": eng.counters,
             },
         }))
 
