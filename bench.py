@@ -41,6 +41,9 @@ def parse_args():
     ap.add_argument("--turn-prompt", type=int, default=512)
     ap.add_argument("--decode-per-turn", type=int, default=128)
     ap.add_argument("--device", default="auto")
+    ap.add_argument("--rag", type=int, default=int(os.environ.get("BENCH_RAG", "1")),
+                    help="per-turn on-GPU RAG retrieval + input-rail check (1=on)")
+    ap.add_argument("--rag-docs", type=int, default=10000)
     return ap.parse_args()
 
 
@@ -48,7 +51,46 @@ def synth_tokens(rng: torch.Generator, n: int, vocab: int) -> list:
     return torch.randint(0, min(vocab - 4, 250), (n,), generator=rng).tolist()
 
 
-def run_step(eng, args, rng, vocab, step_idx: int):
+class RagPath:
+    """BASELINE config 2/3: per-turn retrieval over an HBM-resident corpus
+    (on-GPU sentence encoder + kNN) + the input-rail heuristic check."""
+
+    def __init__(self, args, device: str):
+        import torch as _t
+        from aurora_amd import ops
+        from aurora_amd.models.encoder import SentenceEncoder
+        self.ops = ops
+        self.device = device
+        self.encoder = SentenceEncoder(
+            "bge-base", dtype=_t.bfloat16 if device != "cpu" else _t.float32,
+            device=device)
+        self.encoder.init_random(3)
+        rng = _t.Generator().manual_seed(17)
+        # synthetic corpus: embed args.rag_docs chunks once at setup
+        chunks = [_t.randint(1, 250, (96,), generator=rng).tolist()
+                  for _ in range(args.rag_docs)]
+        embs = []
+        for i in range(0, len(chunks), 512):
+            embs.append(self.encoder.encode_ids(chunks[i:i + 512]))
+        self.docs = _t.cat(embs)
+        self.rail_patterns = None
+
+    def rail_check(self, tokens) -> bool:
+        from aurora_amd.guardrails.input_rail import _INJECTION_PATTERNS
+        text = bytes(t % 256 for t in tokens[:512]).decode("latin-1")
+        return not any(p.search(text) for p in _INJECTION_PATTERNS)
+
+    def retrieve(self, query_tokens) -> list:
+        qv = self.encoder.encode_ids([query_tokens[:96]])
+        vals, idx = self.ops.knn_topk(qv, self.docs, 4)
+        # deterministic "retrieved chunk" tokens derived from the hit ids
+        out = []
+        for i in idx[0].tolist():
+            out.extend(((i * 131 + j * 7) % 250 for j in range(32)))
+        return out
+
+
+def run_step(eng, args, rng, vocab, step_idx: int, rag=None):
     """One full investigation batch; returns per-incident latencies."""
     from aurora_amd.engine.sampling_params import SamplingParams
 
@@ -62,6 +104,12 @@ def run_step(eng, args, rng, vocab, step_idx: int):
 
     def submit(i):
         rid = f"s{step_idx}-inc{i}-t{turn_no[i]}"
+        if rag is not None:
+            # input rail on the webhook/tool-authored turn text + RAG
+            # retrieval over the HBM corpus, inside the timed path
+            rag.rail_check(ctx_tokens[i][-args.turn_prompt:])
+            retrieved = rag.retrieve(ctx_tokens[i][-96:])
+            ctx_tokens[i][-len(retrieved):] = retrieved  # fixed token count
         eng.add_request(rid, list(ctx_tokens[i]), params, session_id=f"inc-{step_idx}-{i}")
         return rid
 
@@ -111,6 +159,7 @@ def main():
     eng = LLMEngine(cfg)
     rng = torch.Generator().manual_seed(999 + rank)
     vocab = spec.vocab_size
+    rag = RagPath(args, eng.device) if args.rag else None
 
     def sync():
         if use_cuda:
@@ -121,12 +170,12 @@ def main():
                 torch.cuda.synchronize()
 
     for w in range(args.warmup):
-        run_step(eng, args, rng, vocab, -1 - w)
+        run_step(eng, args, rng, vocab, -1 - w, rag)
     sync()
     t0 = time.monotonic()
     lat = []
     for s in range(args.steps):
-        lat.extend(run_step(eng, args, rng, vocab, s))
+        lat.extend(run_step(eng, args, rng, vocab, s, rag))
     sync()
     elapsed = time.monotonic() - t0
 
@@ -164,6 +213,8 @@ def main():
                 "concurrency_per_gpu": args.concurrency,
                 "turns": args.turns,
                 "decode_tokens_per_turn": args.decode_per_turn,
+                "rag": bool(rag), "rag_docs": args.rag_docs if rag else 0,
+                "guardrail_per_turn": bool(rag),
                 "p50_time_to_rca_s": round(p50, 3),
                 "prompt_tokens_per_incident": args.first_prompt
                 + (args.turns - 1) * args.turn_prompt,

@@ -114,7 +114,7 @@ def test_bench_dp2_cpu():
     """The driver's multi-rank launch shape: torch.distributed.run, 2 CPU
     ranks, gloo; rank 0 prints one JSON line with whole-job aggregation."""
     env = dict(os.environ, BENCH_MODEL="tiny-llama", BENCH_CONCURRENCY="2",
-               BENCH_TURNS="2", MASTER_ADDR="127.0.0.1")
+               BENCH_TURNS="2", BENCH_RAG="0", MASTER_ADDR="127.0.0.1")
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
