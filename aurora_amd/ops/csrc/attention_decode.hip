@@ -90,15 +90,21 @@ __global__ __launch_bounds__(256) void paged_decode_kernel(
       float s[G];
 #pragma unroll
       for (int g = 0; g < G; ++g) {
-        s[g] = group_sum<GL>(dot8_bf16(kf[u], qf[g])) * scale;
-        const float mn = fmaxf(m[g], s[g]);
-        const float alpha = __expf(m[g] - mn);
-        const float pv = __expf(s[g] - mn);
-        m[g] = mn;
-        l[g] = l[g] * alpha + pv;
+        s[g] = group_sum<GL>(dot8_bf16_fast(kf[u], qf[g])) * scale;
+        // defer-style rescale: the running max rarely grows, so the O/l
+        // rescale (8*G VALU per token) is branch-skipped when it does not
+        if (s[g] > m[g]) {
+          const float alpha = __expf(m[g] - s[g]);
+          m[g] = s[g];
+          l[g] *= alpha;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) o[g][j] *= alpha;
+        }
+        const float pv = __expf(s[g] - m[g]);
+        l[g] += pv;
 #pragma unroll
         for (int j = 0; j < 8; ++j)
-          o[g][j] = o[g][j] * alpha + pv * bf16_to_f32((unsigned short)vf[u][j]);
+          o[g][j] += pv * bf16_to_f32((unsigned short)vf[u][j]);
       }
     }
   }
