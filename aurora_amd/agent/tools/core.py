@@ -20,7 +20,8 @@ from .registry import ToolContext
 @R.add("terminal_exec",
        "Run a shell command in the sandboxed diagnostics environment "
        "(read-only infra commands; every command is safety-gated).",
-       {"command": "the shell command to run"}, read_only=False)
+       {"command": "the shell command to run"}, read_only=False,
+       needs_confirmation=True)
 async def terminal_exec(ctx: ToolContext, command: str, timeout: int = 60) -> str:
     verdict = await evaluate_command(command, llm=ctx.llm, org_store=ctx.org_store,
                                      user_id=ctx.user_id)
@@ -72,7 +73,7 @@ async def cloud_exec(ctx: ToolContext, command: str) -> str:
        "Run kubectl on a connected on-prem cluster through its outbound "
        "agent tunnel (safety-gated).",
        {"cluster": "cluster id (empty = the only connected cluster)",
-        "args": "kubectl arguments"}, read_only=False)
+        "args": "kubectl arguments"}, read_only=False, needs_confirmation=True)
 async def on_prem_kubectl(ctx: ToolContext, args: str, cluster: str = "") -> str:
     agents = ctx.extra.get("kubectl_agents")
     if agents is None and ctx.queue is not None:

@@ -32,6 +32,7 @@ class ToolContext:
     graph: Any = None
     queue: Any = None
     notifier: Optional[Callable[[dict], Awaitable[None]]] = None  # WS frames
+    confirmer: Optional[Callable[[str, dict], Awaitable[bool]]] = None
     settings: Any = None
     depth: int = 0  # nested-RCA guard (reference: trigger_rca_tool.py:38)
     extra: Dict[str, Any] = field(default_factory=dict)
@@ -45,6 +46,7 @@ class Tool:
     parameters: Dict[str, str] = field(default_factory=dict)  # name -> description
     read_only: bool = True
     requires_connection: Optional[str] = None
+    needs_confirmation: bool = False  # interactive human-in-the-loop gate
 
     def spec_line(self) -> str:
         ps = ", ".join(f"{k}: {v}" for k, v in self.parameters.items())
@@ -68,10 +70,11 @@ class ToolRegistry:
         return tool
 
     def add(self, name: str, description: str, parameters: Optional[Dict[str, str]] = None,
-            read_only: bool = True, requires_connection: Optional[str] = None):
+            read_only: bool = True, requires_connection: Optional[str] = None,
+            needs_confirmation: bool = False):
         def deco(fn):
             self.register(Tool(name, description, fn, parameters or {},
-                               read_only, requires_connection))
+                               read_only, requires_connection, needs_confirmation))
             return fn
         return deco
 
@@ -113,6 +116,21 @@ class ToolRegistry:
             return f"ERROR: unknown tool '{name}'. Available: {', '.join(self.names())}"
         if ctx.mode == "ask" and not tool.read_only:
             return f"ERROR: tool '{name}' is not available in ask (read-only) mode"
+        # human-in-the-loop confirmation for destructive-classified tools in
+        # interactive sessions (reference: confirmation_response WS type,
+        # main_chatbot.py:1000-1619; destructive classification
+        # mcp_tools.py:57); headless runs have no confirmer and rely on the
+        # guardrail stack alone
+        if tool.needs_confirmation and ctx.confirmer is not None:
+            try:
+                approved = await asyncio.wait_for(ctx.confirmer(name, args),
+                                                  timeout=300.0)
+            except asyncio.TimeoutError:
+                approved = False
+            if not approved:
+                if ctx.org_store is not None:
+                    ctx.org_store.audit("tool_declined", name)
+                return f"CANCELLED: user declined '{name}'"
         step_id = None
         if ctx.org_store is not None:
             step_id = ctx.org_store.step_start(

@@ -39,6 +39,21 @@ def create_app(rt: Runtime) -> FastAPI:
         await rt.stop()
 
     # ---------------------------------------------------------------- auth
+    # RBAC: 3 roles (reference: Casbin model rbac_model.conf + 
+    # require_permission decorators, utils/auth/rbac_decorators.py).
+    ROLE_PERMS = {
+        "viewer": {"read"},
+        "member": {"read", "write", "investigate", "connect"},
+        "admin": {"read", "write", "investigate", "connect", "admin"},
+    }
+
+    def _require(verb: str, x_user_id: Optional[str]) -> None:
+        if not x_user_id:
+            return  # internal-secret calls carry full permissions
+        user = rt.admin.get_user(x_user_id)
+        role = (user or {}).get("role", "viewer")
+        if verb not in ROLE_PERMS.get(role, {"read"}):
+            raise HTTPException(403, f"role '{role}' lacks '{verb}' permission")
 
     def auth(x_user_id: Optional[str] = Header(None),
              x_org_id: Optional[str] = Header(None),
@@ -131,7 +146,9 @@ def create_app(rt: Runtime) -> FastAPI:
         return inc
 
     @app.post("/incidents")
-    async def create_incident(body: dict, store: OrgStore = Depends(auth)):
+    async def create_incident(body: dict, store: OrgStore = Depends(auth),
+                              x_user_id: Optional[str] = Header(None)):
+        _require("write", x_user_id)
         iid = store.create_incident(
             title=body.get("title", "manual incident"),
             source=body.get("source", "manual"),
@@ -139,7 +156,9 @@ def create_app(rt: Runtime) -> FastAPI:
         return {"incident_id": iid}
 
     @app.post("/incidents/{iid}/investigate")
-    async def investigate(iid: str, store: OrgStore = Depends(auth)):
+    async def investigate(iid: str, store: OrgStore = Depends(auth),
+                          x_user_id: Optional[str] = Header(None)):
+        _require("investigate", x_user_id)
         if store.get_incident(iid) is None:
             raise HTTPException(404, "no such incident")
         tid = rt.queue.enqueue("run_background_chat",
@@ -278,13 +297,17 @@ def create_app(rt: Runtime) -> FastAPI:
         return store.command_policies()
 
     @app.post("/command-policies")
-    async def add_policy(body: dict, store: OrgStore = Depends(auth)):
+    async def add_policy(body: dict, store: OrgStore = Depends(auth),
+                         x_user_id: Optional[str] = Header(None)):
+        _require("admin", x_user_id)
         if body.get("action") not in ("allow", "deny"):
             raise HTTPException(400, "action must be allow|deny")
         return {"id": store.add_command_policy(body["pattern"], body["action"])}
 
     @app.delete("/command-policies/{pid}")
-    async def del_policy(pid: str, store: OrgStore = Depends(auth)):
+    async def del_policy(pid: str, store: OrgStore = Depends(auth),
+                         x_user_id: Optional[str] = Header(None)):
+        _require("admin", x_user_id)
         return {"deleted": store.delete_command_policy(pid)}
 
     # ---- knowledge base ----
@@ -341,8 +364,40 @@ def create_app(rt: Runtime) -> FastAPI:
     @app.post("/connectors/{provider}/connect")
     async def connect(provider: str, body: dict, store: OrgStore = Depends(auth),
                       x_user_id: Optional[str] = Header(None)):
+        _require("connect", x_user_id)
         from ..connectors.registry import connect_provider
         return connect_provider(rt, store, x_user_id or "", provider, body)
+
+    # ---- org management (reference: org routes; before_add_member hook) ----
+    @app.get("/org/users")
+    async def org_users(store: OrgStore = Depends(auth)):
+        rows = rt.db.query("SELECT id, email, role FROM users WHERE org_id=?",
+                           [store.org_id])
+        return [dict(r) for r in rows]
+
+    @app.post("/org/users")
+    async def add_user(body: dict, store: OrgStore = Depends(auth),
+                       x_user_id: Optional[str] = Header(None)):
+        _require("admin", x_user_id)
+        from ..utils.hooks import fire
+        if not await fire("before_add_member", org_id=store.org_id,
+                          email=body.get("email"), default=True):
+            raise HTTPException(403, "membership denied by hook")
+        uid = rt.admin.create_user(store.org_id, body["email"],
+                                   role=body.get("role", "member"))
+        store.audit("user_added", body["email"])
+        return {"user_id": uid}
+
+    @app.post("/org/tool-permissions")
+    async def set_tool_perm(body: dict, store: OrgStore = Depends(auth),
+                            x_user_id: Optional[str] = Header(None)):
+        _require("admin", x_user_id)
+        store.set_tool_permission(body["tool"], bool(body["allowed"]))
+        return {"ok": True}
+
+    @app.get("/org/tool-permissions")
+    async def get_tool_perms(store: OrgStore = Depends(auth)):
+        return store.tool_permissions()
 
     # ---- health ----
     @app.get("/health")

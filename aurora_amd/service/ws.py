@@ -260,6 +260,8 @@ class ChatServer:
         session_id: Optional[str] = None
         incident_id: Optional[str] = None
         mode = "agent"
+        confirm_tools = False
+        pending_confirm: Dict[str, asyncio.Future] = {}
         current: Optional[asyncio.Task] = None
         tok = rt.engine.tokenizer()
 
@@ -293,7 +295,13 @@ class ChatServer:
                                                       session_id=session_id)
                 incident_id = msg.get("incident_id")
                 mode = msg.get("mode", "agent")
+                confirm_tools = bool(msg.get("confirm_tools", False))
                 await send({"type": "initialized", "session_id": session_id})
+                continue
+            if mtype == "confirmation_response":
+                fut = pending_confirm.pop(msg.get("id", ""), None)
+                if fut is not None and not fut.done():
+                    fut.set_result(bool(msg.get("approved", False)))
                 continue
             if mtype == "control:cancel":
                 # reference: main_chatbot.py:1041-1157 — cancel, consolidate
@@ -317,15 +325,27 @@ class ChatServer:
             if current and not current.done():
                 await send({"type": "error", "error": "busy"})
                 continue
+            confirmer = None
+            if confirm_tools:
+                async def confirmer(tool, targs):
+                    import uuid as _u
+                    cid = _u.uuid4().hex[:10]
+                    fut = asyncio.get_running_loop().create_future()
+                    pending_confirm[cid] = fut
+                    await send({"type": "confirmation_request", "id": cid,
+                                "tool": tool, "args": targs})
+                    return await fut
             current = asyncio.create_task(self._run_query(
-                send, org_id, user["id"], session_id, incident_id, mode, question))
+                send, org_id, user["id"], session_id, incident_id, mode,
+                question, confirmer))
 
     async def _run_query(self, send, org_id, user_id, session_id, incident_id,
-                         mode, question) -> None:
+                         mode, question, confirmer=None) -> None:
         rt = self.rt
         try:
             ctx = rt.tool_ctx(org_id, user_id=user_id, session_id=session_id,
                               incident_id=incident_id, mode=mode, notifier=send)
+            ctx.confirmer = confirmer
             wf = Workflow(rt.llm(org_id), ctx)
             state = State(question=question, user_id=user_id, org_id=org_id,
                           session_id=session_id, incident_id=incident_id, mode=mode)

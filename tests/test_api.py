@@ -67,12 +67,14 @@ def test_unknown_source_404(env):
 
 
 def test_command_policies_api(env):
-    rt, client, *_ = env
-    r = client.post("/command-policies", headers=H,
+    rt, client, oid, uid = env
+    rt.admin.create_user(oid, "adm@acme.io", role="admin", user_id="user-adm")
+    A = {"X-User-Id": "user-adm"}
+    r = client.post("/command-policies", headers=A,
                     json={"pattern": "rm -rf", "action": "deny"})
     pid = r.json()["id"]
     assert any(p["id"] == pid for p in client.get("/command-policies", headers=H).json())
-    assert client.delete(f"/command-policies/{pid}", headers=H).json()["deleted"] == 1
+    assert client.delete(f"/command-policies/{pid}", headers=A).json()["deleted"] == 1
 
 
 def test_knowledge_base_api(env):
@@ -242,3 +244,29 @@ def test_kubectl_agent_tunnel():
         asyncio.run(main())
     finally:
         server.shutdown()
+
+
+def test_rbac_roles(env):
+    rt, client, oid, uid = env
+    rt.admin.create_user(oid, "viewer@acme.io", role="viewer", user_id="user-v")
+    V = {"X-User-Id": "user-v"}
+    # viewer can read
+    assert client.get("/incidents", headers=V).status_code == 200
+    # viewer cannot write / investigate / admin
+    assert client.post("/incidents", headers=V, json={"title": "x"}).status_code == 403
+    assert client.post("/command-policies", headers=V,
+                       json={"pattern": "x", "action": "deny"}).status_code == 403
+    # member cannot admin
+    assert client.post("/command-policies", headers=H,
+                       json={"pattern": "x", "action": "deny"}).status_code == 403
+    # promote to admin, now allowed
+    rt.db.execute("UPDATE users SET role='admin' WHERE id='user-a'")
+    r = client.post("/command-policies", headers=H,
+                    json={"pattern": "xx", "action": "deny"})
+    assert r.status_code == 200
+    # org user management
+    r = client.post("/org/users", headers=H, json={"email": "new@acme.io"})
+    assert r.status_code == 200
+    users = client.get("/org/users", headers=H).json()
+    assert any(u["email"] == "new@acme.io" for u in users)
+    rt.db.execute("UPDATE users SET role='member' WHERE id='user-a'")

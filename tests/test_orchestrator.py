@@ -149,3 +149,36 @@ def test_mcp_server(env):
         assert "error" in r
     asyncio.run(main())
     assert search_dispatch_entries("postmortem")
+
+
+def test_tool_confirmation_gate(env):
+    """Destructive-classified tools ask the confirmer in interactive mode;
+    decline cancels, headless (no confirmer) proceeds to the guardrail."""
+    store, iid = env
+    from aurora_amd.agent.tools.registry import GLOBAL_REGISTRY, ToolContext
+    import aurora_amd.agent.tools.core  # noqa: F401
+
+    async def main():
+        asked = []
+
+        async def deny(tool, args):
+            asked.append(tool)
+            return False
+
+        ctx = ToolContext(org_store=store, confirmer=deny)
+        out = await GLOBAL_REGISTRY.run(ctx, "terminal_exec", {"command": "uptime"})
+        assert out.startswith("CANCELLED")
+        assert asked == ["terminal_exec"]
+        assert any(a["kind"] == "tool_declined" for a in store.audit_events())
+
+        async def approve(tool, args):
+            return True
+
+        ctx2 = ToolContext(org_store=store, confirmer=approve)
+        out = await GLOBAL_REGISTRY.run(ctx2, "terminal_exec", {"command": "uptime"})
+        assert "load average" in out or "up" in out
+        # headless: no confirmer -> guardrail path only
+        ctx3 = ToolContext(org_store=store)
+        out = await GLOBAL_REGISTRY.run(ctx3, "terminal_exec", {"command": "uptime"})
+        assert "CANCELLED" not in out
+    asyncio.run(main())
