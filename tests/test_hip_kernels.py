@@ -288,3 +288,17 @@ def test_fused_rope_kv_write():
     assert _err(q, q_ref) < 2e-2
     assert _err(kc, kc_ref) < 2e-2
     assert _err(vc, vc_ref) < 2e-2
+
+
+def test_engine_gpu_moe():
+    """Mixtral-style MoE through the GPU engine (HIP kernels + routing)."""
+    from aurora_amd.engine.engine import EngineConfig, LLMEngine
+    from aurora_amd.engine.sampling_params import SamplingParams
+    eng = LLMEngine(EngineConfig(model="smoke-moe", device="cuda", seed=9,
+                                 kv_cache_pages=64))
+    p = SamplingParams(temperature=0.0, max_tokens=8, ignore_eos=True)
+    outs = eng.generate([[5, 6, 7, 8] * 10, [9] * 17], p)
+    assert all(len(o) == 8 for o in outs)
+    eng2 = LLMEngine(EngineConfig(model="smoke-moe", device="cuda", seed=9,
+                                  kv_cache_pages=64))
+    assert eng2.generate([[5, 6, 7, 8] * 10, [9] * 17], p) == outs
