@@ -48,6 +48,7 @@ class EngineConfig:
     max_model_len: int = 32768
     tp_size: int = 1
     tokenizer_path: Optional[str] = None
+    checkpoint_path: Optional[str] = None  # HF-layout safetensors dir
     seed: int = 0
     grammar_topk: int = 32  # candidates tried per step under a grammar
     use_hip_graphs: bool = True  # capture the decode forward per batch bucket
@@ -112,7 +113,11 @@ class LLMEngine:
 
         if model is None:
             model = LlamaForCausalLM(self.spec, dtype=self.dtype, device=self.device)
-            model.init_random(config.seed)
+            if config.checkpoint_path:
+                from ..models.loader import load_llama_safetensors
+                load_llama_safetensors(model, config.checkpoint_path)
+            else:
+                model.init_random(config.seed)
         self.model = model.eval()
 
         page_size = config.page_size or (256 if self.device != "cpu" else 16)
