@@ -53,6 +53,10 @@ DEVINL float dot8_bf16_fast(const shortx8 a, const shortx8 b, float acc = 0.f) {
   return acc;
 }
 
+DEVINL unsigned int pack2_bf16d(float lo, float hi) {
+  return ((unsigned int)f32_to_bf16(hi) << 16) | f32_to_bf16(lo);
+}
+
 // full-wave (64-lane) reductions
 DEVINL float wave_sum(float x) {
 #pragma unroll
