@@ -188,7 +188,9 @@ torch::Tensor paged_attention_decode(torch::Tensor q, torch::Tensor kc,
   auto out = torch::empty_like(q);
   if (B == 0) return out;
   // context splits: fill the chip at small batch (>=512 WGs target)
-  int CS = std::max(1, std::min(8, 512 / std::max(B * Hkv, 1)));
+  // fill target ~1024 WGs: at B=64 the LDS-bound residency is only
+  // 2 WGs/CU, so deeper context splits buy memory-level parallelism
+  int CS = std::max(1, std::min(8, 1024 / std::max(B * Hkv, 1)));
   torch::Tensor part;
   if (CS > 1)
     part = torch::empty({CS, (long)B * Hq, D + 2},
