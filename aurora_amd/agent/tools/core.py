@@ -321,3 +321,109 @@ _provider_tool("query_slack", "slack", "Search Slack channels/threads for contex
 _provider_tool("query_jira", "jira", "Search Jira issues.")
 _provider_tool("notion_search", "notion", "Search Notion workspace pages.")
 _provider_tool("confluence_search", "confluence", "Search Confluence pages/runbooks.")
+
+
+# ------------------------------------------------------- files / artefacts
+
+
+@R.add("analyze_zip_file",
+       "List and summarise the contents of an uploaded zip archive.",
+       {"artifact": "name of the zip artifact (base64 content)"})
+async def analyze_zip_file(ctx: ToolContext, artifact: str) -> str:
+    import base64
+    import io
+    import zipfile
+    row = ctx.org_store.get_artifact(artifact)
+    if not row:
+        return f"ERROR: no artifact '{artifact}'"
+    try:
+        data = base64.b64decode(row["content"])
+        zf = zipfile.ZipFile(io.BytesIO(data))
+    except Exception as e:  # noqa: BLE001
+        return f"ERROR: not a readable zip: {e}"
+    lines = []
+    for info in zf.infolist()[:200]:
+        lines.append(f"{info.filename} ({info.file_size} B)")
+    return "Archive contents:\n" + "\n".join(lines)
+
+
+@R.add("rag_index_zip",
+       "Index the text files of an uploaded zip archive into the knowledge base.",
+       {"artifact": "name of the zip artifact (base64 content)"}, read_only=False)
+async def rag_index_zip(ctx: ToolContext, artifact: str) -> str:
+    import base64
+    import io
+    import zipfile
+    row = ctx.org_store.get_artifact(artifact)
+    if not row:
+        return f"ERROR: no artifact '{artifact}'"
+    try:
+        zf = zipfile.ZipFile(io.BytesIO(base64.b64decode(row["content"])))
+    except Exception as e:  # noqa: BLE001
+        return f"ERROR: not a readable zip: {e}"
+    n = 0
+    for info in zf.infolist():
+        if info.file_size > 2_000_000 or info.is_dir():
+            continue
+        if not info.filename.endswith((".txt", ".md", ".log", ".yaml", ".yml",
+                                       ".json", ".conf", ".cfg", ".py", ".sh")):
+            continue
+        try:
+            text = zf.read(info).decode("utf-8", "replace")
+        except Exception:  # noqa: BLE001
+            continue
+        did = ctx.org_store.add_kb_document(f"{artifact}:{info.filename}", text)
+        if ctx.rag is not None:
+            ctx.rag.add_document(did, info.filename, text)
+        n += 1
+    return f"indexed {n} files from {artifact} into the knowledge base"
+
+
+# -------------------------------------------------------------- automation
+
+
+@R.add("iac_tool",
+       "Write and plan Terraform for a proposed infrastructure change "
+       "(plan only; apply requires human confirmation).",
+       {"name": "change name", "hcl": "terraform configuration"},
+       read_only=False, needs_confirmation=True)
+async def iac_tool(ctx: ToolContext, name: str, hcl: str) -> str:
+    import shutil
+    import tempfile
+    ctx.org_store.save_artifact(f"iac:{name}.tf", hcl, session_id=ctx.session_id)
+    if shutil.which("terraform") is None:
+        return (f"terraform configuration saved as artifact 'iac:{name}.tf'. "
+                f"(terraform binary unavailable in this deployment; plan/apply "
+                f"must run from a workstation)")
+    with tempfile.TemporaryDirectory() as d:
+        with open(f"{d}/main.tf", "w") as f:
+            f.write(hcl)
+        out = await terminal_exec(ctx, f"cd {d} && terraform init -backend=false "
+                                       f"-input=false && terraform plan -input=false")
+        return out
+
+
+@R.add("rightsizing_analysis",
+       "Analyze pod resource usage vs limits and propose HPA/VPA right-sizing.",
+       {"usage": "kubectl top / metrics output (name cpu memory per line)",
+        "limits": "current requests/limits description"})
+async def rightsizing_analysis(ctx: ToolContext, usage: str, limits: str = "") -> str:
+    # deterministic analysis core of the reference's hpa_vpa right-sizing
+    # cards (reference: services/actions/hpa_vpa_* 958 LoC)
+    import re
+    rows = []
+    for line in usage.splitlines():
+        m = re.match(r"\s*(\S+)\s+(\d+)m?\s+(\d+)(Mi|Gi)", line)
+        if m:
+            mem = int(m.group(3)) * (1024 if m.group(4) == "Gi" else 1)
+            rows.append((m.group(1), int(m.group(2)), mem))
+    if not rows:
+        return "ERROR: no parsable usage rows (expected 'name cpu mem' lines)"
+    recs = []
+    for name, cpu_m, mem_mi in rows:
+        rec_cpu = max(int(cpu_m * 1.4), 10)
+        rec_mem = max(int(mem_mi * 1.3), 32)
+        recs.append(f"- {name}: request cpu={rec_cpu}m memory={rec_mem}Mi "
+                    f"(observed {cpu_m}m / {mem_mi}Mi + headroom)")
+    return ("Right-sizing recommendations (1.4x cpu / 1.3x memory headroom "
+            "over observed usage):\n" + "\n".join(recs))

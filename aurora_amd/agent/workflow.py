@@ -105,6 +105,7 @@ class Workflow:
                 yield {"type": "error", "error": "workflow timeout"}
                 break
             await self._maybe_compress(state)
+            self._inject_context_updates(state)
             prompt = (system + "\n\n# Investigation so far\n"
                       + render_history(state.messages) + "\nThought:")
             # --- thought (streamed) ---
@@ -175,6 +176,24 @@ class Workflow:
         yield {"type": "usage", "counters": self.llm.server.counters(),
                "elapsed_ms": (time.monotonic() - t_start) * 1000.0}
         self._persist(state)
+
+    def _inject_context_updates(self, state: State) -> None:
+        """Correlated alerts that arrived mid-investigation become
+        observations in the running loop (reference: context_updates.py)."""
+        store = self.ctx.org_store
+        if store is None or not state.session_id:
+            return
+        row = store.get_artifact(f"context_updates:{state.session_id}")
+        if not row:
+            return
+        updates = json.loads(row["content"])
+        if not updates:
+            return
+        store.save_artifact(f"context_updates:{state.session_id}", "[]")
+        state.messages.append({
+            "role": "observation",
+            "content": "[context update] additional correlated alerts arrived "
+                       "during this investigation: " + json.dumps(updates)[:1500]})
 
     async def _maybe_compress(self, state: State) -> None:
         """Preflight conversation compression when the context outgrows the

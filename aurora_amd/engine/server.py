@@ -35,6 +35,8 @@ class AsyncEngineServer:
         self._lock = threading.Lock()
         self._stop = threading.Event()
         self._wake = threading.Event()
+        self._last_progress = time.monotonic()
+        self.watchdog_timeout_s = 300.0
         self._thread = threading.Thread(target=self._run, daemon=True,
                                         name="engine-step-loop")
         self._thread.start()
@@ -52,6 +54,7 @@ class AsyncEngineServer:
             try:
                 with self._lock:
                     outs = self.engine.step()
+                self._last_progress = time.monotonic()
             except Exception as e:  # noqa: BLE001 - a step failure must not
                 # kill the loop: fail every in-flight request loudly instead
                 import traceback
@@ -143,3 +146,11 @@ class AsyncEngineServer:
 
     def counters(self) -> Dict[str, int]:
         return dict(self.engine.counters)
+
+    def healthy(self) -> bool:
+        """Engine watchdog (SURVEY.md §5.3 'engine watchdog'): unhealthy when
+        a step has been stuck (hung kernel / collective) past the timeout."""
+        with self._lock:
+            busy = self.engine.has_work()
+        return (not busy) or (time.monotonic() - self._last_progress
+                              < self.watchdog_timeout_s)

@@ -402,7 +402,9 @@ def create_app(rt: Runtime) -> FastAPI:
     # ---- health ----
     @app.get("/health")
     async def health():
-        return {"status": "ok", "engine": rt.engine.counters(),
+        ok = rt.engine.healthy() if hasattr(rt.engine, "healthy") else True
+        return {"status": "ok" if ok else "engine_stuck",
+                "engine": rt.engine.counters(),
                 "queue_depth": rt.queue.pending_count()}
 
     return app

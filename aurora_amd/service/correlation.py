@@ -93,6 +93,17 @@ class AlertCorrelator:
             self.store.lifecycle(res.incident_id, "alert_correlated",
                                  {"strategy": res.strategy, "score": res.score})
             self.store.touch_incident(res.incident_id)
+            # queue a context update for a RUNNING investigation (reference:
+            # chat/background/context_updates.py:145,343)
+            inc = self.store.get_incident(res.incident_id)
+            sid = (inc or {}).get("aurora_chat_session_id")
+            if sid:
+                row = self.store.get_artifact(f"context_updates:{sid}")
+                updates = json.loads(row["content"]) if row else []
+                updates.append({"source": source, "title": title,
+                                "ts": time.time()})
+                self.store.save_artifact(f"context_updates:{sid}",
+                                         json.dumps(updates[-10:]))
             return res.incident_id, False
         iid = self.store.create_incident(
             title=title, source=source, payload=payload, user_id=user_id,
