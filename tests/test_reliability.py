@@ -1,0 +1,110 @@
+"""Failure-detection / recovery tests (SURVEY.md §5.3 semantics kept:
+orphan reaper, cancellation, watchdog, queue crash retry)."""
+import asyncio
+import time
+
+import pytest
+
+from aurora_amd.engine.engine import EngineConfig
+from aurora_amd.engine.server import AsyncEngineServer
+from aurora_amd.service.runtime import Runtime
+from aurora_amd.utils.config import Settings
+
+
+@pytest.fixture()
+def rt():
+    cfg = Settings(db_path=":memory:")
+    server = AsyncEngineServer(EngineConfig(model="tiny-llama", device="cpu", seed=1))
+    r = Runtime(cfg, engine_server=server)
+    yield r
+    server.shutdown()
+
+
+def test_stale_investigation_reaper(rt):
+    """A crashed investigation (stale heartbeat) is marked failed by the
+    reaper beat (reference: task.py:2390,2442,2350)."""
+    async def main():
+        org = rt.admin.create_org("r")
+        store = rt.org(org)
+        iid = store.create_incident("stuck", "manual", {})
+        store.update_incident(iid, aurora_status="running")
+        # age the heartbeat past the threshold
+        rt.db.execute("UPDATE incidents SET updated_at = ? WHERE id = ?",
+                      [time.time() - 3600, iid])
+        from aurora_amd.service.worker import cleanup_stale_investigations
+        res = await cleanup_stale_investigations(rt, {})
+        assert res["reaped"] == 1
+        assert store.get_incident(iid)["aurora_status"] == "failed"
+        events = [e["event"] for e in store.lifecycle_events(iid)]
+        assert "rca_reaped" in events
+        # a LIVE investigation (fresh heartbeat) is untouched
+        iid2 = store.create_incident("alive", "manual", {})
+        store.update_incident(iid2, aurora_status="running")
+        res = await cleanup_stale_investigations(rt, {})
+        assert res["reaped"] == 0
+    asyncio.run(main())
+
+
+def test_rca_cancellation(rt):
+    """Pending RCA tasks cancel via the stored task id (reference:
+    cancel_rca_for_incident task.py:83)."""
+    org = rt.admin.create_org("c")
+    store = rt.org(org)
+    iid = store.create_incident("x", "manual", {})
+    tid = rt.queue.enqueue("run_background_chat",
+                           {"org_id": org, "incident_id": iid})
+    store.update_incident(iid, rca_task_id=tid)
+    # queue not started -> task still pending -> cancellable
+    assert rt.queue.cancel(tid)
+    assert rt.queue.get_task(tid)["status"] == "cancelled"
+    assert not rt.queue.cancel(tid)  # second cancel is a no-op
+
+
+def test_engine_watchdog_healthy(rt):
+    assert rt.engine.healthy()
+    # simulate a stuck step: mark busy + stale progress
+    rt.engine._last_progress = time.monotonic() - 10_000
+    from aurora_amd.engine.sampling_params import SamplingParams
+    with rt.engine._lock:
+        rt.engine.engine.add_request("stuck", [1, 2, 3],
+                                     SamplingParams(max_tokens=1))
+        busy = rt.engine.engine.has_work()
+    assert busy and not rt.engine.healthy()
+    rt.engine._wake.set()
+    for _ in range(100):
+        if rt.engine.healthy():
+            break
+        time.sleep(0.05)
+    assert rt.engine.healthy()  # stepping resumed => healthy again
+
+
+def test_session_cache_eviction_under_pressure():
+    """Session KV pages are evicted LRU when new work needs pages."""
+    from aurora_amd.engine.engine import EngineConfig, LLMEngine
+    from aurora_amd.engine.sampling_params import SamplingParams
+    eng = LLMEngine(EngineConfig(model="tiny-llama", device="cpu", seed=2,
+                                 kv_cache_pages=10))
+    p = SamplingParams(temperature=0.0, max_tokens=4, ignore_eos=True)
+    for i in range(4):  # each session holds ~3 pages; 4 sessions > 12 pages
+        eng.add_request(f"r{i}", list(range(1, 40)), p, session_id=f"s{i}")
+        while eng.has_work():
+            eng.step()
+    assert len(eng.sessions) < 4  # oldest sessions were evicted
+    assert eng.kv.num_free() >= 0
+
+
+def test_engine_step_failure_fails_requests_loudly():
+    from aurora_amd.engine.sampling_params import SamplingParams
+    server = AsyncEngineServer(EngineConfig(model="tiny-llama", device="cpu"))
+
+    def boom():
+        raise RuntimeError("injected fault")
+    server.engine.step = boom  # fault injection
+
+    async def main():
+        res = await server.generate([1, 2, 3], SamplingParams(max_tokens=4))
+        assert "engine step failed" in (res.finish_reason or "")
+    try:
+        asyncio.run(main())
+    finally:
+        server.shutdown()
