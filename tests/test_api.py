@@ -270,3 +270,20 @@ def test_rbac_roles(env):
     users = client.get("/org/users", headers=H).json()
     assert any(u["email"] == "new@acme.io" for u in users)
     rt.db.execute("UPDATE users SET role='member' WHERE id='user-a'")
+
+
+def test_client_sdk(env):
+    """The python SDK against a live app + WS server."""
+    rt, client, oid, uid = env
+    import aurora_amd.client as ac
+
+    # REST via the TestClient's transport: monkeypatch httpx.Client
+    sdk = ac.AuroraClient.__new__(ac.AuroraClient)
+    sdk.http = client  # starlette TestClient is httpx-compatible
+    sdk.http.headers.update({"X-User-Id": "user-a"})
+    iid = sdk.create_incident("sdk incident", {"service": "api"})
+    assert sdk.incident(iid)["title"] == "sdk incident"
+    assert any(i["id"] == iid for i in sdk.incidents())
+    assert sdk.metrics()["incidents_total"] >= 1
+    did = sdk.kb_upload("sdk-doc", "restart pgbouncer when pool exhausts")
+    assert sdk.kb_search("pgbouncer pool")[0]["doc_id"] == did
