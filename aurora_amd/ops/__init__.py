@@ -186,8 +186,15 @@ def linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
         # near-streaming-peak above ~100 MB of W (gate_up/down/lm-head) but
         # 1.7-2.5 TB/s on the small qkv/o shapes where this kernel reaches
         # 2.8-2.9 TB/s with a short split-K
-        if w_bytes <= 96 * 1024 * 1024:
-            s_target = 256 if w_bytes <= 40 * 1024 * 1024 else 512
+        import os
+        max_mb = int(os.environ.get("AURORA_SKINNY_MAX_MB", "96"))
+        if w_bytes <= max_mb * 1024 * 1024:
+            if w_bytes <= 40 * 1024 * 1024:
+                s_target = 256
+            elif w_bytes <= 96 * 1024 * 1024:
+                s_target = 512
+            else:
+                s_target = 2048
             return _hip.skinny_gemm_ex(x.contiguous(), weight, s_target, 2)
     return F.linear(x, weight)
 
