@@ -14,8 +14,11 @@
 //  * O accumulates in registers (col d = lane&31 per d-tile); the epilogue
 //    transposes through LDS so global stores are coalesced 16-B rows.
 //
-// Geometry: 4 waves/WG; each wave owns 32 q rows (QBLK = 128/WG); KV tiles
-// of 64 keys staged cooperatively, computed as two 32-key MFMA steps.
+// Geometry: 8 waves/WG (512 threads); the WG covers HG = min(G, 4) query
+// heads of ONE kv head x (8/HG)*32 query rows, so every staged K/V tile
+// feeds 8 waves of MFMA (GQA heads share the stage — staging bandwidth was
+// the 156-TF bound of the 4-wave/1-head structure).  KV tiles of 64 keys,
+// two 32-key MFMA steps each.
 // Covers fresh prefill and append-prefill (chunked agent turns) since K/V
 // are read from the paged cache with an absolute-position causal bound.
 #include "common.h"
@@ -35,8 +38,8 @@ union frag_u {
   bf16x8 v;
 };
 
-template <int D>
-__global__ __launch_bounds__(256, 1) void prefill_kernel(
+template <int D, int HG>  // HG = query heads per workgroup
+__global__ __launch_bounds__(512, 1) void prefill_kernel(
     const __hip_bfloat16* __restrict__ q,   // [T, Hq, D]
     const __hip_bfloat16* __restrict__ kc,  // [NP, Hkv, P, D]
     const __hip_bfloat16* __restrict__ vc,
@@ -55,9 +58,8 @@ __global__ __launch_bounds__(256, 1) void prefill_kernel(
   // only permute within that many slots (at D=64, (row&15)<<4 would escape
   // the 128-B row)
   constexpr int SWZ = (D / 8) - 1;
+  constexpr int QS = 8 / HG;          // 32-row q sub-blocks per WG
   const int nb = blockIdx.x;
-  const int h = blockIdx.y;
-  const int kvh = h / (Hq / Hkv);
   const int seq = block_seq[nb];
   const int qoff = block_qoff[nb];
   const int q_start = cu_q[seq];
@@ -70,15 +72,20 @@ __global__ __launch_bounds__(256, 1) void prefill_kernel(
   const int lane = tid % WAVE;
   const int col = lane & 31;   // q column owned by this lane (within wave's 32)
   const int half = lane >> 5;  // half-wave
+  // wave -> (query head, q sub-block): heads vary fastest so the HG heads
+  // of one sub-block sit on consecutive waves
+  const int h = blockIdx.y * HG + (wave % HG);
+  const int kvh = h / (Hq / Hkv);
+  const int qsub = wave / HG;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   short* kbuf = (short*)smem;                       // [KVB][D] XOR-swizzled
   short* vtbuf = (short*)(smem + KVB * D * 2);      // [D][VROW] transposed
-  float* scratch = (float*)(smem + KVB * D * 2 + D * VROW * 2);  // [4][32]
+  float* scratch = (float*)(smem + KVB * D * 2 + D * VROW * 2);  // [8][32]
   float* wsc = scratch + wave * 32;
 
   // ---- Q fragments (B-operand of swapped QK^T): held for the whole block
-  const int qrow = qoff + wave * 32 + col;  // local q row in this seq
+  const int qrow = qoff + qsub * 32 + col;  // local q row in this seq
   const bool qvalid = qrow < qlen;
   const long qbase = ((long)(q_start + (qvalid ? qrow : 0)) * Hq + h) * D;
   frag_u qf[NC];
@@ -105,19 +112,19 @@ __global__ __launch_bounds__(256, 1) void prefill_kernel(
     for (int r = 0; r < 16; ++r) o[dt][r] = 0.f;
 
   // keys needed by this q block: up to the last row's absolute position
-  const int blk_rows = min(qlen - qoff, 128);
+  const int blk_rows = min(qlen - qoff, QS * 32);
   const int kend = min(ctx, ctx - qlen + qoff + blk_rows);
   const int ntiles = (kend + KVB - 1) / KVB;
   // absolute position of the wave's first (smallest) q row
-  const int q_abs_min = ctx - qlen + qoff + wave * 32;
+  const int q_abs_min = ctx - qlen + qoff + qsub * 32;
 
   for (int tile = 0; tile < ntiles; ++tile) {
     const int kbase = tile * KVB;
-    // ---- stage K (swizzled) and V^T cooperatively: 256 threads x 4 chunks
+    // ---- stage K (swizzled) and V^T cooperatively: 512 threads
     __syncthreads();
 #pragma unroll
-    for (int i = 0; i < KVB * D / (256 * 8); ++i) {
-      const int c = tid + i * 256;
+    for (int i = 0; i < KVB * D / (512 * 8); ++i) {
+      const int c = tid + i * 512;
       const int row = c / (D / 8);
       const int d0 = (c % (D / 8)) * 8;
       const int t = kbase + row;
@@ -250,28 +257,31 @@ __global__ __launch_bounds__(256, 1) void prefill_kernel(
 #pragma unroll
     for (int j = 0; j < 4; ++j) linv[g][j] = 1.f / lv[j];
   }
-  __syncthreads();  // everyone done with K/V buffers; reuse as O staging
-  // wave's staging area: 32 rows x D bf16
-  short* obuf = (short*)smem + wave * 32 * D;
+  // everyone done with K/V buffers; reuse as O staging in two rounds of
+  // 4 waves (8 waves' staging would not fit the K+V area)
 #pragma unroll
-  for (int dt = 0; dt < NDT; ++dt) {
-    const int d = dt * 32 + col;
+  for (int round = 0; round < 2; ++round) {
+    __syncthreads();
+    if ((wave >> 2) != round) continue;
+    short* obuf = (short*)smem + (wave & 3) * 32 * D;
 #pragma unroll
-    for (int g = 0; g < 4; ++g) {
+    for (int dt = 0; dt < NDT; ++dt) {
+      const int d = dt * 32 + col;
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        const int r = (j) + 8 * g + 4 * half;  // q row within the wave's 32
-        obuf[r * D + d] = (short)f32_to_bf16(o[dt][4 * g + j] * linv[g][j]);
+      for (int g = 0; g < 4; ++g) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const int r = (j) + 8 * g + 4 * half;  // q row within the wave's 32
+          obuf[r * D + d] = (short)f32_to_bf16(o[dt][4 * g + j] * linv[g][j]);
+        }
       }
     }
-  }
-  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-  __builtin_amdgcn_sched_barrier(0);
-  // each lane stores half a row (D/2 elems = D bytes) with 16-B stores
-  {
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
+    // each lane stores half a row (D/2 elems = D bytes) with 16-B stores
     const int r = lane >> 1;            // 0..31
     const int seg = (lane & 1) * (D / 2);
-    const int qr = qoff + wave * 32 + r;
+    const int qr = qoff + qsub * 32 + r;
     if (qr < qlen) {
       const long ob = ((long)(q_start + qr) * Hq + h) * D + seg;
 #pragma unroll
@@ -294,6 +304,10 @@ torch::Tensor attention_prefill_paged(torch::Tensor q, torch::Tensor kc,
   TORCH_CHECK(D == 128 || D == 64, "head_dim must be 64 or 128, got ", D);
   auto out = torch::empty_like(q);
   if (T == 0) return out;
+  // heads per workgroup: the GQA group shares each staged K/V tile
+  const int G = Hq / Hkv;
+  const int HG = G >= 4 ? 4 : G;  // 1, 2 or 4
+  const int QROWS = 32 * (8 / HG);  // q rows covered per workgroup
   // host-side q-block map (cu_q is tiny; engine built it on CPU this step)
   auto cu_cpu = cu_q.to(torch::kCPU);
   const int* cu = cu_cpu.data_ptr<int>();
@@ -301,7 +315,7 @@ torch::Tensor attention_prefill_paged(torch::Tensor q, torch::Tensor kc,
   std::vector<int> bseq, boff;
   for (int s = 0; s < S; ++s) {
     const int qlen = cu[s + 1] - cu[s];
-    for (int off = 0; off < qlen; off += 128) {
+    for (int off = 0; off < qlen; off += QROWS) {
       bseq.push_back(s);
       boff.push_back(off);
     }
@@ -312,11 +326,11 @@ torch::Tensor attention_prefill_paged(torch::Tensor q, torch::Tensor kc,
   auto boff_t = torch::from_blob(boff.data(), {NB}, torch::kInt32).to(opts.device());
 
   const int VROW = 64 + 8;
-  const size_t lds = 64 * D * 2 + D * VROW * 2 + 4 * 32 * 4;
-  dim3 grid(NB, Hq), block(256);
+  const size_t lds = 64 * D * 2 + D * VROW * 2 + 8 * 32 * 4;
+  dim3 grid(NB, Hq / HG), block(512);
   auto stream = CUR_STREAM;
-#define PLAUNCH(DD)                                                           \
-  hipLaunchKernelGGL((prefill_kernel<DD>), grid, block, lds, stream,          \
+#define PLAUNCH(DD, HGV)                                                      \
+  hipLaunchKernelGGL((prefill_kernel<DD, HGV>), grid, block, lds, stream,     \
                      (const __hip_bfloat16*)q.data_ptr(),                     \
                      (const __hip_bfloat16*)kc.data_ptr(),                    \
                      (const __hip_bfloat16*)vc.data_ptr(),                    \
@@ -324,8 +338,18 @@ torch::Tensor attention_prefill_paged(torch::Tensor q, torch::Tensor kc,
                      bseq_t.data_ptr<int>(), boff_t.data_ptr<int>(),          \
                      cu_q.data_ptr<int>(), bt.data_ptr<int>(),                \
                      ctx.data_ptr<int>(), Hq, Hkv, P, MP, (float)scale)
-  if (D == 128) PLAUNCH(128);
-  else PLAUNCH(64);
+#define PLAUNCH_D(DD)                                                         \
+  switch (HG) {                                                               \
+    case 1: PLAUNCH(DD, 1); break;                                            \
+    case 2: PLAUNCH(DD, 2); break;                                            \
+    default: PLAUNCH(DD, 4); break;                                           \
+  }
+  if (D == 128) {
+    PLAUNCH_D(128)
+  } else {
+    PLAUNCH_D(64)
+  }
+#undef PLAUNCH_D
 #undef PLAUNCH
   HIP_CHECK_KERNEL();
   return out;
