@@ -11,8 +11,8 @@ tracked per call into llm_usage (reference: llm.py:270-350).
 from __future__ import annotations
 
 import time
-from dataclasses import dataclass, field
-from typing import Any, Dict, Optional
+from dataclasses import dataclass
+from typing import Optional
 
 from ..engine.sampling_params import SamplingParams
 from ..engine.server import AsyncEngineServer, GenResult

@@ -6,7 +6,7 @@ utils/prefix_cache.py:39-158)."""
 from __future__ import annotations
 
 import json
-from typing import List, Optional
+from typing import List
 
 from .state import State
 from .tools.registry import Tool

@@ -4,11 +4,9 @@ from __future__ import annotations
 
 import asyncio
 import json
-import os
 import shlex
 import tempfile
-import time
-from typing import Any, Dict, Optional
+from typing import Any
 
 from ...guardrails.command_safety import evaluate_command
 from .registry import GLOBAL_REGISTRY as R

@@ -9,7 +9,6 @@ and tool_start/tool_end frames pushed to the chat transport.
 from __future__ import annotations
 
 import asyncio
-import inspect
 import json
 import time
 from dataclasses import dataclass, field

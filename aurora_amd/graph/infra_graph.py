@@ -14,7 +14,7 @@ from __future__ import annotations
 import json
 import time
 from collections import defaultdict, deque
-from typing import Dict, List, Optional, Set
+from typing import Dict, List, Set
 
 
 class InfraGraph:

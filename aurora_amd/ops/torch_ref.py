@@ -15,7 +15,6 @@ KV cache layout (chosen for the MI355X decode kernel's streaming reads):
 """
 from __future__ import annotations
 
-import math
 from typing import Optional, Tuple
 
 import torch

@@ -71,7 +71,7 @@ def test_engine_watchdog_healthy(rt):
         busy = rt.engine.engine.has_work()
     assert busy and not rt.engine.healthy()
     rt.engine._wake.set()
-    for _ in range(100):
+    for _ in range(400):
         if rt.engine.healthy():
             break
         time.sleep(0.05)
