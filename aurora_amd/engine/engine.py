@@ -278,23 +278,28 @@ class LLMEngine:
             budget -= chunk
             if budget <= 0:
                 break
-        if prefill:
-            return "prefill", prefill, chunks
-        # decode: every running seq advances one token
+        # decode: every running seq (not being prefilled) advances one token.
+        # Prefill does NOT stall decodes: both regimes batch into one mixed
+        # forward (projections share the packed batch; attention splits).
+        in_prefill = set(prefill)
         decode: List[Sequence] = []
         for seq in list(self.running):
-            if seq not in self.running:  # already preempted as a victim
+            if seq not in self.running or seq in in_prefill:
                 continue
             ok = self._try_reserve(seq, seq.num_tokens + 1)
             while not ok:
                 # evict others first; as a last resort evict this seq itself
-                if self._preempt_lowest(exclude=set(decode) | {seq}):
+                if self._preempt_lowest(exclude=set(decode) | in_prefill | {seq}):
                     ok = self._try_reserve(seq, seq.num_tokens + 1)
                 else:
-                    self._preempt_lowest(exclude=set(decode))
+                    self._preempt_lowest(exclude=set(decode) | in_prefill)
                     break
             if ok:
                 decode.append(seq)
+        if prefill and decode:
+            return "mixed", (prefill, decode), chunks
+        if prefill:
+            return "prefill", prefill, chunks
         return "decode", decode, []
 
     # -------------------------------------------------------------- tensors
@@ -321,6 +326,8 @@ class LLMEngine:
         if not seqs:
             return []
         dev = self.device
+        if mode == "mixed":
+            return self._step_mixed(*seqs, chunks)
         if mode == "prefill":
             self.counters["prefill_steps"] += 1
             ids: List[int] = []
@@ -401,6 +408,79 @@ class LLMEngine:
         for s in seqs:
             s.num_cached = s.num_tokens
         outs = self._sample_and_append(seqs, logits)
+        self.timers["sample"] += time.monotonic() - _t2
+        return outs
+
+    @torch.inference_mode()
+    def _step_mixed(self, prefill: List[Sequence], decode: List[Sequence],
+                    chunks: List[int]) -> List[StepOutput]:
+        """One forward for prefill chunks + decode tokens: the projections
+        and norms run on the whole packed batch, attention splits per
+        regime — decode streams no longer stall while new investigations
+        prefill (this was ~30% of the bench step as a serial bubble)."""
+        dev = self.device
+        self.counters["prefill_steps"] += 1
+        self.counters["decode_steps"] += 1
+        ids: List[int] = []
+        positions: List[int] = []
+        slots: List[int] = []
+        cu = [0]
+        pre_ctx: List[int] = []
+        for seq, chunk in zip(prefill, chunks):
+            a = seq.num_cached
+            b = a + chunk
+            ids.extend(seq.token_ids[a:b])
+            positions.extend(range(a, b))
+            slots.extend(self._slot(seq, p) for p in range(a, b))
+            cu.append(cu[-1] + chunk)
+            pre_ctx.append(b)
+        Tp = len(ids)
+        ids.extend(s.token_ids[-1] for s in decode)
+        positions.extend(s.num_tokens - 1 for s in decode)
+        slots.extend(self._slot(s, s.num_tokens - 1) for s in decode)
+        dec_ctx = [s.num_tokens for s in decode]
+        meta = AttnMetadata(
+            mode="mixed",
+            positions=torch.tensor(positions, dtype=torch.int32, device=dev),
+            slot_mapping=torch.tensor(slots, dtype=torch.int32, device=dev),
+            block_tables=self._block_table_tensor(prefill),
+            context_lens=torch.tensor(pre_ctx, dtype=torch.int32, device=dev),
+            cu_seqlens_q=torch.tensor(cu, dtype=torch.int32, device=dev),
+            num_prefill_tokens=Tp,
+            dec_block_tables=self._block_table_tensor(decode),
+            dec_context_lens=torch.tensor(dec_ctx, dtype=torch.int32, device=dev),
+        )
+        input_ids = torch.tensor(ids, dtype=torch.long, device=dev)
+        _t1 = time.monotonic()
+        hidden = self.model(input_ids, self.kv.caches, meta)
+        # sample rows: completed prefills + every decode row
+        sample_rows: List[int] = []
+        sample_seqs: List[Sequence] = []
+        for i, (seq, chunk) in enumerate(zip(prefill, chunks)):
+            had_output = seq.num_output > 0
+            seq.num_cached += chunk
+            self.counters["prompt_tokens"] += chunk
+            if seq.num_cached >= seq.prefill_target:
+                seq.status = RUNNING
+                if seq not in self.running:
+                    self.running.append(seq)
+                if not had_output:
+                    sample_rows.append(cu[i + 1] - 1)
+                    sample_seqs.append(seq)
+            else:
+                self.waiting.appendleft(seq)
+        for j, seq in enumerate(decode):
+            sample_rows.append(Tp + j)
+            sample_seqs.append(seq)
+        rows = hidden[torch.tensor(sample_rows, dtype=torch.long, device=dev)]
+        logits = self.model.compute_logits(rows)
+        if dev != "cpu":
+            torch.cuda.synchronize()
+        self.timers["forward"] += time.monotonic() - _t1
+        for seq in decode:
+            seq.num_cached = seq.num_tokens
+        _t2 = time.monotonic()
+        outs = self._sample_and_append(sample_seqs, logits)
         self.timers["sample"] += time.monotonic() - _t2
         return outs
 

@@ -39,12 +39,17 @@ class AttnMetadata:
     mode "decode": one q token per sequence.
     """
 
-    mode: str  # "prefill" | "decode"
+    mode: str  # "prefill" | "decode" | "mixed"
     positions: torch.Tensor  # [T] int32 absolute positions
     slot_mapping: torch.Tensor  # [T] int32 cache slots for the new K/V
     block_tables: torch.Tensor  # [num_seqs, max_pages] int32
     context_lens: torch.Tensor  # [num_seqs] int32 (incl. new tokens)
     cu_seqlens_q: Optional[torch.Tensor] = None  # [num_seqs+1] int32, prefill only
+    # mixed mode: tokens are [prefill chunks..., decode tokens]; the fields
+    # above describe the prefill part, the ones below the decode part
+    num_prefill_tokens: int = 0
+    dec_block_tables: Optional[torch.Tensor] = None
+    dec_context_lens: Optional[torch.Tensor] = None
 
 
 class LlamaAttention(nn.Module):
@@ -80,10 +85,23 @@ class LlamaAttention(nn.Module):
                 q, k_cache, v_cache, meta.block_tables,
                 meta.cu_seqlens_q, meta.context_lens, self.scale,
             )
-        else:
+        elif meta.mode == "decode":
             o = ops.paged_attention_decode(
                 q, k_cache, v_cache, meta.block_tables, meta.context_lens, self.scale
             )
+        else:  # mixed: prefill chunks + decode tokens in one forward — the
+            # projections/norms run on the whole packed batch (bigger GEMMs),
+            # attention splits per regime
+            Tp = meta.num_prefill_tokens
+            o_pre = ops.attention_prefill_paged(
+                q[:Tp], k_cache, v_cache, meta.block_tables,
+                meta.cu_seqlens_q, meta.context_lens, self.scale,
+            ) if Tp else q[:0]
+            o_dec = ops.paged_attention_decode(
+                q[Tp:].contiguous(), k_cache, v_cache, meta.dec_block_tables,
+                meta.dec_context_lens, self.scale,
+            ) if T > Tp else q[:0]
+            o = torch.cat([o_pre, o_dec]) if (Tp and T > Tp) else (o_pre if Tp else o_dec)
         return self.o_proj(o.reshape(T, self.n_heads * self.head_dim))
 
 

@@ -155,3 +155,29 @@ def test_moe_routing_uses_multiple_experts():
     assert sel.unique().numel() > 1  # routing spreads across experts
     y = m(x)
     assert y.shape == x.shape and torch.isfinite(y).all()
+
+
+def test_mixed_batch_equivalence():
+    """A request arriving mid-decode (mixed prefill+decode steps) must not
+    change the in-flight sequence's greedy output."""
+    cfg = EngineConfig(model="tiny-llama", device="cpu", seed=7)
+    p = SamplingParams(temperature=0.0, max_tokens=14, ignore_eos=True)
+    a_prompt, b_prompt = list(range(5, 45)), list(range(60, 100))
+    # reference: run A alone
+    ref_a = LLMEngine(cfg).generate([a_prompt], p)[0]
+    ref_b = LLMEngine(cfg).generate([b_prompt], p)[0]
+    # mixed: start A, inject B after a few decode steps
+    eng = LLMEngine(cfg)
+    eng.add_request("A", a_prompt, p)
+    outs = {"A": [], "B": []}
+    steps = 0
+    while eng.has_work():
+        if steps == 4:
+            eng.add_request("B", b_prompt, p)
+        for o in eng.step():
+            outs[o.request_id].extend(o.new_token_ids)
+        steps += 1
+    assert outs["A"] == ref_a
+    assert outs["B"] == ref_b
+    # mixed steps actually happened (prefill of B overlapped A's decode)
+    assert eng.counters["prefill_steps"] >= 2
