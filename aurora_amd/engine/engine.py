@@ -489,10 +489,10 @@ class LLMEngine:
     _BUCKETS = (1, 2, 4, 8, 16, 32, 64)
 
     def _graphs_enabled(self) -> bool:
-        # MoE routing is data-dependent (token->expert nonzero gather):
-        # not graph-capturable; dense models only
+        # MoE decode uses the fixed-capacity batched dispatch (static
+        # shapes), so it is capturable too
         return (self.config.use_hip_graphs and self.device != "cpu"
-                and self.config.tp_size == 1 and self.spec.n_experts == 0)
+                and self.config.tp_size == 1)
 
     def _decode_with_graph(self, seqs, ids, positions, slots, ctx_lens):
         """Replay a hipGraph-captured decode forward (per batch-size bucket,

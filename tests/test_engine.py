@@ -181,3 +181,26 @@ def test_mixed_batch_equivalence():
     assert outs["B"] == ref_b
     # mixed steps actually happened (prefill of B overlapped A's decode)
     assert eng.counters["prefill_steps"] >= 2
+
+
+
+def test_moe_batched_matches_loop():
+    import torch
+    from aurora_amd.models.config import get_spec
+    from aurora_amd.models.llama import MoEMLP
+    spec = get_spec("tiny-moe")
+    m = MoEMLP(spec, torch.float32)
+    for p in m.parameters():
+        torch.nn.init.normal_(p, std=0.1)
+    x = torch.randn(33, spec.hidden_size)
+    logits = m.router(x).float()
+    w, sel = logits.topk(spec.experts_per_token, dim=-1)
+    w = w.softmax(-1)
+    batched = m._forward_batched(x, w, sel)
+    # loop reference
+    ref = torch.zeros_like(x)
+    for e in range(spec.n_experts):
+        tok, slot = (sel == e).nonzero(as_tuple=True)
+        if tok.numel():
+            ref.index_add_(0, tok, m.experts[e](x[tok]) * w[tok, slot, None])
+    assert torch.allclose(batched, ref, atol=1e-5), (batched - ref).abs().max()
