@@ -167,7 +167,10 @@ class MoEMLP(nn.Module):
         order = flat.argsort(stable=True)  # tokens grouped by expert
         tok_of = order // k  # token index of each sorted assignment
         # rank within the expert group = position - group start
-        counts = torch.bincount(flat, minlength=E)
+        # (bincount host-syncs to size its output -> not stream-capturable;
+        # a scatter_add histogram is)
+        counts = torch.zeros(E, dtype=torch.long, device=x.device).scatter_add_(
+            0, flat, torch.ones_like(flat))
         starts = torch.cumsum(counts, 0) - counts
         rank = torch.arange(T * k, device=x.device) - starts[flat[order]]
         slot_idx = flat[order] * C + rank  # flat [E*C] position (rank < C always)
