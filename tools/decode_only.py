@@ -3,11 +3,13 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 from aurora_amd.engine.engine import LLMEngine, EngineConfig
 from aurora_amd.engine.sampling_params import SamplingParams
-eng = LLMEngine(EngineConfig(model="llama3-8b", seed=0, kv_memory_fraction=0.3))
+B = int(os.environ.get("B", "16"))
+CTX = int(os.environ.get("CTX", "512"))
+eng = LLMEngine(EngineConfig(model="llama3-8b", seed=0, kv_memory_fraction=0.45))
 p = SamplingParams(temperature=0.0, max_tokens=400, ignore_eos=True)
-for i in range(16):
-    eng.add_request(f"r{i}", list(range(30, 542)), p)
-for _ in range(40):
+for i in range(B):
+    eng.add_request(f"r{i}", [30 + (j % 500) for j in range(CTX)], p)
+for _ in range(40 + CTX // 256):
     eng.step()
 torch.cuda.synchronize()
 for _ in range(100):
