@@ -178,24 +178,19 @@ def linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     (profiles/kernel_perf_r01_*.json): hipBLASLt wins at very large N
     (lm-head shapes), the skinny kernel at mid-size per-layer shapes."""
     import torch.nn.functional as F
-    if (x.is_cuda and x.dim() == 2 and 0 < x.shape[0] <= 64
+    if (x.is_cuda and x.dim() == 2 and 0 < x.shape[0] <= 16
             and x.shape[1] % 64 == 0 and x.dtype == torch.bfloat16
             and _hip is not None):
         w_bytes = weight.shape[0] * weight.shape[1] * 2
-        # measured crossover (tools/sweep_skinny.py, profiles/): hipBLASLt is
-        # near-streaming-peak above ~100 MB of W (gate_up/down/lm-head) but
-        # 1.7-2.5 TB/s on the small qkv/o shapes where this kernel reaches
-        # 2.8-2.9 TB/s with a short split-K
+        # measured crossover (tools/sweep_skinny2.py, COLD weights rotated
+        # past the 256 MB LLC — the in-engine regime): at M<=16 the direct
+        # MFMA-fragment kernel wins (2.7 TB/s vs blaslt 2.5 on qkv); at
+        # M=64 its per-lane 16 B row-gathers of A cap it at ~1.3 TB/s while
+        # hipBLASLt reaches 1.7-5.0 TB/s, so larger M goes to blaslt.
         import os
         max_mb = int(os.environ.get("AURORA_SKINNY_MAX_MB", "96"))
         if w_bytes <= max_mb * 1024 * 1024:
-            if w_bytes <= 40 * 1024 * 1024:
-                s_target = 256
-            elif w_bytes <= 96 * 1024 * 1024:
-                s_target = 512
-            else:
-                s_target = 2048
-            return _hip.skinny_gemm_ex(x.contiguous(), weight, s_target, 2)
+            return _hip.skinny_gemm_ex(x.contiguous(), weight, 1024, 4)
     return F.linear(x, weight)
 
 
