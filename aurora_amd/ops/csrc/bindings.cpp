@@ -26,6 +26,7 @@ torch::Tensor fused_rope_kv_write(torch::Tensor qkv, torch::Tensor positions,
                                   torch::Tensor kc, torch::Tensor vc,
                                   long Hq, long Hkv);
 torch::Tensor skinny_gemm_ex(torch::Tensor x, torch::Tensor w, long s_target, long unroll);
+torch::Tensor skinny_gemm2_ex(torch::Tensor x, torch::Tensor w, long s_target);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "fused RMSNorm (bf16, gfx950)");
@@ -43,4 +44,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("fused_rope_kv_write", &fused_rope_kv_write,
         "fused qkv split + RoPE + paged KV scatter");
   m.def("skinny_gemm_ex", &skinny_gemm_ex, "skinny GEMM with tuning knobs");
+  m.def("skinny_gemm2_ex", &skinny_gemm2_ex, "LDS-staged skinny GEMM (M<=64)");
 }

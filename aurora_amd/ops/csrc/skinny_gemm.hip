@@ -103,6 +103,135 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// v2: LDS-staged variant for 17 <= M <= 64.  v1's per-lane 16 B fragment
+// loads are row-gathers (consecutive lanes touch lines K*2 bytes apart);
+// at MT=4 that is 5 gathered loads per 32-k step and caps the kernel at
+// ~1.3 TB/s cold (tools/sweep_skinny2.py).  Here a WG cooperatively
+// streams W and x tiles with 64 B-per-thread coalesced loads into
+// XOR-swizzled LDS ([64 rows][128 k] bf16, unit16 ^= row&15 -- conflict-
+// free per 16-lane phase for both the b128 stores and the fragment
+// reads), double-buffered so the global stream overlaps the MFMAs.
+// ---------------------------------------------------------------------------
+#define SG2_BK 128                 // k elements per staged chunk
+#define SG2_TILE_B (64 * SG2_BK * 2)  // one tile: 16 KB
+
+__device__ __forceinline__ int sg2_addr(int row, int unit) {
+  return row * (SG2_BK * 2) + ((unit ^ (row & 15)) << 4);
+}
+
+template <bool SPLIT>
+__global__ __launch_bounds__(256, 2) void skinny_gemm2_kernel(
+    const __hip_bfloat16* __restrict__ x,  // [M, K]
+    const __hip_bfloat16* __restrict__ w,  // [N, K]
+    __hip_bfloat16* __restrict__ out,      // [M, N] (SPLIT=false)
+    float* __restrict__ part,              // [S, M, N] (SPLIT=true)
+    int M, int N, int K, int KS) {
+  __shared__ char lds[2][2 * SG2_TILE_B];  // [buf][W tile | X tile]
+  const int lane = threadIdx.x % WAVE;
+  const int wave = threadIdx.x / WAVE;
+  const int n0 = blockIdx.x * 64;
+  const int split = blockIdx.y;
+  const int k_begin = split * KS;
+  const int k_end = min(K, k_begin + KS);
+  const int nchunks = (k_end - k_begin + SG2_BK - 1) / SG2_BK;
+  if (nchunks <= 0 || n0 >= N) return;
+
+  // cooperative load mapping: thread -> (row = t>>2, 64B segment = t&3)
+  const int lrow = threadIdx.x >> 2;
+  const int lseg = threadIdx.x & 3;
+  const long wrow = (long)min(n0 + lrow, N - 1) * K;  // clamped
+  const long xrow = (long)min(lrow, M - 1) * K;       // clamped
+
+  float4 gw0, gw1, gw2, gw3, gx0, gx1, gx2, gx3;
+#define SG2_GLOAD(c)                                                        \
+  {                                                                         \
+    const long kc = k_begin + (long)(c)*SG2_BK + lseg * 32;                 \
+    gw0 = *reinterpret_cast<const float4*>(w + wrow + kc);                  \
+    gw1 = *reinterpret_cast<const float4*>(w + wrow + kc + 8);              \
+    gw2 = *reinterpret_cast<const float4*>(w + wrow + kc + 16);             \
+    gw3 = *reinterpret_cast<const float4*>(w + wrow + kc + 24);             \
+    gx0 = *reinterpret_cast<const float4*>(x + xrow + kc);                  \
+    gx1 = *reinterpret_cast<const float4*>(x + xrow + kc + 8);              \
+    gx2 = *reinterpret_cast<const float4*>(x + xrow + kc + 16);             \
+    gx3 = *reinterpret_cast<const float4*>(x + xrow + kc + 24);             \
+  }
+#define SG2_STORE(b)                                                        \
+  {                                                                         \
+    char* wt = lds[b];                                                      \
+    char* xt = lds[b] + SG2_TILE_B;                                         \
+    *reinterpret_cast<float4*>(wt + sg2_addr(lrow, lseg * 4 + 0)) = gw0;    \
+    *reinterpret_cast<float4*>(wt + sg2_addr(lrow, lseg * 4 + 1)) = gw1;    \
+    *reinterpret_cast<float4*>(wt + sg2_addr(lrow, lseg * 4 + 2)) = gw2;    \
+    *reinterpret_cast<float4*>(wt + sg2_addr(lrow, lseg * 4 + 3)) = gw3;    \
+    *reinterpret_cast<float4*>(xt + sg2_addr(lrow, lseg * 4 + 0)) = gx0;    \
+    *reinterpret_cast<float4*>(xt + sg2_addr(lrow, lseg * 4 + 1)) = gx1;    \
+    *reinterpret_cast<float4*>(xt + sg2_addr(lrow, lseg * 4 + 2)) = gx2;    \
+    *reinterpret_cast<float4*>(xt + sg2_addr(lrow, lseg * 4 + 3)) = gx3;    \
+  }
+
+  f32x4 acc[4];
+#pragma unroll
+  for (int t = 0; t < 4; ++t)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) acc[t][r] = 0.f;
+
+  const int bcol = lane & 15;    // W-tile row / x-tile row offset
+  const int kg = lane >> 4;      // unit offset of this lane's 16 B
+
+  SG2_GLOAD(0);
+  SG2_STORE(0);
+  if (nchunks > 1) SG2_GLOAD(1);
+  __syncthreads();
+
+  for (int c = 0; c < nchunks; ++c) {
+    const int buf = c & 1;
+    if (c + 1 < nchunks) {
+      SG2_STORE(!buf);                    // stage next chunk
+      if (c + 2 < nchunks) SG2_GLOAD(c + 2);  // start chunk after next
+    }
+    const char* wt = lds[buf];
+    const char* xt = lds[buf] + SG2_TILE_B;
+    const int klim = min(SG2_BK, k_end - k_begin - c * SG2_BK);
+#pragma unroll
+    for (int s = 0; s < SG2_BK / 32; ++s) {
+      if (s * 32 >= klim) break;
+      const shortx8 bv = *reinterpret_cast<const shortx8*>(
+          wt + sg2_addr(wave * 16 + bcol, s * 4 + kg));
+      sfrag bfr;
+      __builtin_memcpy(&bfr, &bv, 16);
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        const shortx8 av = *reinterpret_cast<const shortx8*>(
+            xt + sg2_addr(t * 16 + bcol, s * 4 + kg));
+        sfrag afr;
+        __builtin_memcpy(&afr, &av, 16);
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afr.v, bfr.v,
+                                                         acc[t], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+  const int n = n0 + wave * 16 + bcol;
+  if (n >= N) return;
+#pragma unroll
+  for (int t = 0; t < 4; ++t) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int m = t * 16 + (lane >> 4) * 4 + r;
+      if (m >= M) continue;
+      if (SPLIT) {
+        part[((long)split * M + m) * N + n] = acc[t][r];
+      } else {
+        *(unsigned short*)(out + (long)m * N + n) = f32_to_bf16(acc[t][r]);
+      }
+    }
+  }
+}
+#undef SG2_GLOAD
+#undef SG2_STORE
+
 __global__ void skinny_combine_kernel(const float* __restrict__ part,
                                       __hip_bfloat16* __restrict__ out,
                                       int S, long MN) {
@@ -176,4 +305,42 @@ torch::Tensor skinny_gemm_ex(torch::Tensor x, torch::Tensor w, long s_target, lo
 
 torch::Tensor skinny_gemm(torch::Tensor x, torch::Tensor w) {
   return skinny_gemm_ex(x, w, 0, 2);
+}
+
+torch::Tensor skinny_gemm2_ex(torch::Tensor x, torch::Tensor w, long s_target) {
+  TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16, "bf16 GPU only");
+  TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.size(1) == w.size(1));
+  const int M = x.size(0), K = x.size(1), N = w.size(0);
+  TORCH_CHECK(M <= 64, "skinny_gemm2 is for M <= 64");
+  TORCH_CHECK(K % SG2_BK == 0, "K must be a multiple of ", SG2_BK);
+  auto out = torch::empty({M, N}, x.options());
+  if (M == 0 || N == 0) return out;
+  const int nblocks = (N + 63) / 64;
+  const int target = s_target > 0 ? (int)s_target : 768;
+  int S = std::max(1, std::min(16, target / std::max(nblocks, 1)));
+  int KS = ((K / S + SG2_BK - 1) / SG2_BK) * SG2_BK;
+  S = (K + KS - 1) / KS;
+  auto stream = CUR_STREAM;
+  torch::Tensor part;
+  dim3 grid(nblocks, S), block(256);
+  if (S > 1) {
+    part = torch::empty({S, M, N}, x.options().dtype(torch::kFloat32));
+    hipLaunchKernelGGL((skinny_gemm2_kernel<true>), grid, block, 0, stream,
+                       (const __hip_bfloat16*)x.data_ptr(),
+                       (const __hip_bfloat16*)w.data_ptr(),
+                       (__hip_bfloat16*)out.data_ptr(),
+                       part.data_ptr<float>(), M, N, K, KS);
+    const long MN = (long)M * N;
+    int cblocks = (int)std::min<long>((MN + 255) / 256, 1024);
+    hipLaunchKernelGGL(skinny_combine_kernel, dim3(cblocks), dim3(256), 0,
+                       stream, part.data_ptr<float>(),
+                       (__hip_bfloat16*)out.data_ptr(), S, MN);
+  } else {
+    hipLaunchKernelGGL((skinny_gemm2_kernel<false>), grid, block, 0, stream,
+                       (const __hip_bfloat16*)x.data_ptr(),
+                       (const __hip_bfloat16*)w.data_ptr(),
+                       (__hip_bfloat16*)out.data_ptr(), nullptr, M, N, K, KS);
+  }
+  HIP_CHECK_KERNEL();
+  return out;
 }

@@ -37,6 +37,15 @@ for tag, M, K, N in SHAPES:
         for unr in (2, 4):
             dt = timeit(lambda i: hip.skinny_gemm_ex(x, ws[i % ncopy], st, unr), iters)
             res[f"sk_s{st}_u{unr}"] = dt
+    if K % 128 == 0:
+        for st in (256, 512, 768, 1024, 2048):
+            dt = timeit(lambda i: hip.skinny_gemm2_ex(x, ws[i % ncopy], st), iters)
+            res[f"v2_s{st}"] = dt
+        # numerics: v2 vs fp32 reference on the last W
+        ref = (x.float() @ ws[0].float().t())
+        got = hip.skinny_gemm2_ex(x, ws[0], 768).float()
+        err = (got - ref).abs().max().item() / max(ref.abs().max().item(), 1e-6)
+        assert err < 2e-2, f"v2 numerics off: rel {err}"
     res["blaslt"] = timeit(lambda i: torch.matmul(x, ws[i % ncopy].t()), iters)
     best = min(res, key=res.get)
     print(f"== {tag} M={M} K={K} N={N} W={wb/1e6:.0f}MB (x{ncopy} copies)")
