@@ -178,19 +178,20 @@ def linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     (profiles/kernel_perf_r01_*.json): hipBLASLt wins at very large N
     (lm-head shapes), the skinny kernel at mid-size per-layer shapes."""
     import torch.nn.functional as F
-    if (x.is_cuda and x.dim() == 2 and 0 < x.shape[0] <= 16
-            and x.shape[1] % 64 == 0 and x.dtype == torch.bfloat16
+    if (x.is_cuda and x.dim() == 2 and 0 < x.shape[0] <= 64
+            and x.shape[1] % 128 == 0 and x.dtype == torch.bfloat16
             and _hip is not None):
         w_bytes = weight.shape[0] * weight.shape[1] * 2
         # measured crossover (tools/sweep_skinny2.py, COLD weights rotated
-        # past the 256 MB LLC — the in-engine regime): at M<=16 the direct
-        # MFMA-fragment kernel wins (2.7 TB/s vs blaslt 2.5 on qkv); at
-        # M=64 its per-lane 16 B row-gathers of A cap it at ~1.3 TB/s while
-        # hipBLASLt reaches 1.7-5.0 TB/s, so larger M goes to blaslt.
+        # past the 256 MB LLC — the in-engine regime): the LDS-staged v2
+        # kernel beats hipBLASLt up to ~128 MB of W (o 15.0 vs 21.4 us,
+        # down 34.1 vs 42.2 us at M=64); above that (gate_up 235 MB,
+        # lm-head) blaslt's large-N kernels reach ~5 TB/s and win.
         import os
-        max_mb = int(os.environ.get("AURORA_SKINNY_MAX_MB", "96"))
+        max_mb = int(os.environ.get("AURORA_SKINNY_MAX_MB", "128"))
         if w_bytes <= max_mb * 1024 * 1024:
-            return _hip.skinny_gemm_ex(x.contiguous(), weight, 1024, 4)
+            s_target = 256 if w_bytes <= 40 * 1024 * 1024 else 512
+            return _hip.skinny_gemm2_ex(x.contiguous(), weight, s_target)
     return F.linear(x, weight)
 
 
