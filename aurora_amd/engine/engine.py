@@ -158,7 +158,9 @@ class LLMEngine:
             maxb = self._BUCKETS[-1]
             self._host_i64 = torch.empty(maxb, dtype=torch.int64, pin_memory=True)
             self._host_i32 = torch.empty(maxb, 3, dtype=torch.int32, pin_memory=True)
-            self._host_bt = torch.empty(maxb, mp, dtype=torch.int32, pin_memory=True)
+            self._host_bt = torch.full((maxb, mp), self._scratch_page,
+                                       dtype=torch.int32, pin_memory=True)
+            self._bt_staged = [None] * maxb
         # counters (engine observability; surfaced by service.metrics)
         self.counters = {
             "prompt_tokens": 0, "generated_tokens": 0, "prefill_steps": 0,
@@ -517,11 +519,21 @@ class LLMEngine:
         pos[B:bucket, 1] = self._scratch_slot
         pos[:B, 2] = torch.tensor(ctx_lens, dtype=torch.int32)
         pos[B:bucket, 2] = 1
+        # block-table rows change only on page append / batch-slot change:
+        # rewrite a staged row only when its (seq, npages) key moved
         bt_host = self._host_bt
-        bt_host[:bucket].fill_(self._scratch_page)
+        staged = self._bt_staged
         for i, s in enumerate(seqs):
-            n = min(len(s.pages), mp)
-            bt_host[i, :n] = torch.tensor(s.pages[:n], dtype=torch.int32)
+            key = (s.req_id, len(s.pages))
+            if staged[i] != key:
+                n = min(len(s.pages), mp)
+                bt_host[i, :n] = torch.tensor(s.pages[:n], dtype=torch.int32)
+                bt_host[i, n:] = self._scratch_page
+                staged[i] = key
+        for i in range(len(seqs), bucket):
+            if staged[i] is not None:
+                bt_host[i].fill_(self._scratch_page)
+                staged[i] = None
         in_ids.copy_(ids_t[:bucket], non_blocking=True)
         in_pos.copy_(pos[:bucket, 0], non_blocking=True)
         in_slot.copy_(pos[:bucket, 1], non_blocking=True)
@@ -539,7 +551,9 @@ class LLMEngine:
             maxb = self._BUCKETS[-1]
             self._host_i64 = torch.empty(maxb, dtype=torch.int64, pin_memory=True)
             self._host_i32 = torch.empty(maxb, 3, dtype=torch.int32, pin_memory=True)
-            self._host_bt = torch.empty(maxb, mp, dtype=torch.int32, pin_memory=True)
+            self._host_bt = torch.full((maxb, mp), self._scratch_page,
+                                       dtype=torch.int32, pin_memory=True)
+            self._bt_staged = [None] * maxb
         in_ids = torch.zeros(bucket, dtype=torch.int64, device=dev)
         in_pos = torch.zeros(bucket, dtype=torch.int32, device=dev)
         in_slot = torch.full((bucket,), self._scratch_slot, dtype=torch.int32,
@@ -609,10 +623,11 @@ class LLMEngine:
         the whole batch, no per-row python work, no host randoms."""
         B, V = logits.shape
         dev = logits.device
+        all_greedy = all(s.params.temperature <= 0 for s in seqs)
+        if all_greedy:
+            return logits.argmax(-1).cpu()
         temps = torch.tensor([s.params.temperature for s in seqs], device=dev)
         greedy = temps <= 0
-        if bool(greedy.all()):
-            return logits.argmax(-1).cpu()
         r = logits / temps.clamp(min=1e-6).unsqueeze(1)
         top_ks = torch.tensor([s.params.top_k or V for s in seqs], device=dev)
         top_ps = torch.tensor([s.params.top_p for s in seqs], device=dev)
@@ -653,7 +668,8 @@ class LLMEngine:
             cand: List[List[int]] = []
             stochastic = any(s.params.temperature > 0 for s in seq_list)
             if stochastic:
-                u = torch.rand(sub.shape, device="cpu", generator=self.gen).to(sub.device)
+                u = torch.rand(sub.shape, device=sub.device,
+                               generator=self._dev_gen(sub.device))
                 g = -torch.log(-torch.log(u.clamp(min=1e-20)).clamp(min=1e-20))
                 temps = torch.tensor(
                     [max(s.params.temperature, 1e-6) for s in seq_list], device=sub.device
