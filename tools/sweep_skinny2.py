@@ -7,7 +7,9 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 from aurora_amd import _hip_ops as hip
 
-SHAPES = [  # (tag, M, K, N)
+import json
+_env = os.environ.get("SHAPES")
+SHAPES = json.loads(_env) if _env else [  # (tag, M, K, N)
     ("qkv", 64, 4096, 6144),
     ("o", 64, 4096, 4096),
     ("gate_up", 64, 4096, 28672),
