@@ -488,7 +488,7 @@ class LLMEngine:
 
     # ----------------------------------------------------- hipGraph decode
 
-    _BUCKETS = (1, 2, 4, 8, 16, 32, 64, 96, 128)
+    _BUCKETS = (1, 2, 4, 8, 16, 32, 64, 96, 128, 192, 256)
 
     def _graphs_enabled(self) -> bool:
         # MoE decode uses the fixed-capacity batched dispatch (static
