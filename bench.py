@@ -35,7 +35,7 @@ def parse_args():
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--model", default=os.environ.get("BENCH_MODEL", "llama3-8b"))
     ap.add_argument("--concurrency", type=int,
-                    default=int(os.environ.get("BENCH_CONCURRENCY", "64")))
+                    default=int(os.environ.get("BENCH_CONCURRENCY", "256")))
     ap.add_argument("--turns", type=int, default=int(os.environ.get("BENCH_TURNS", "4")))
     ap.add_argument("--first-prompt", type=int, default=768)
     ap.add_argument("--turn-prompt", type=int, default=512)
