@@ -270,6 +270,21 @@ def test_skinny_gemm(M, N, K):
     assert err.item() < 3e-2, err.item()
 
 
+@pytest.mark.parametrize("M,N,K,st", [(1, 6144, 4096, 256), (17, 4096, 4096, 256),
+                                      (64, 6144, 4096, 512), (64, 4096, 14336, 512),
+                                      (64, 28672, 4096, 1), (33, 4000, 1152, 512)])
+def test_skinny_gemm2(M, N, K, st):
+    """LDS-staged v2 (split and no-split paths, ragged N) vs fp32 torch."""
+    from aurora_amd import _hip_ops
+    torch.manual_seed(1)
+    x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16) / (K ** 0.25)
+    w = torch.randn(N, K, device="cuda", dtype=torch.bfloat16) / (K ** 0.25)
+    out = _hip_ops.skinny_gemm2_ex(x, w, st)
+    ref = (x.float() @ w.float().t())
+    err = (out.float() - ref).abs().max() / ref.abs().max().clamp(min=1e-6)
+    assert err.item() < 3e-2, err.item()
+
+
 def test_fused_rope_kv_write():
     torch.manual_seed(4)
     T, Hq, Hkv, D, P, NP = 29, 8, 2, 128, 64, 4
