@@ -5,7 +5,8 @@ from aurora_amd.engine.engine import LLMEngine, EngineConfig
 from aurora_amd.engine.sampling_params import SamplingParams
 B = int(os.environ.get("B", "16"))
 CTX = int(os.environ.get("CTX", "512"))
-eng = LLMEngine(EngineConfig(model="llama3-8b", seed=0, kv_memory_fraction=0.45))
+eng = LLMEngine(EngineConfig(model="llama3-8b", seed=0, kv_memory_fraction=0.45,
+                             max_num_seqs=B + 2))
 p = SamplingParams(temperature=0.0, max_tokens=400, ignore_eos=True)
 for i in range(B):
     eng.add_request(f"r{i}", [30 + (j % 500) for j in range(CTX)], p)
