@@ -232,6 +232,119 @@ __global__ __launch_bounds__(256, 2) void skinny_gemm2_kernel(
 #undef SG2_GLOAD
 #undef SG2_STORE
 
+// ---------------------------------------------------------------------------
+// v2 mid-M variant (65 <= M <= 256, MT = 8 or 16 m-tiles): hipBLASLt's
+// skinny-M kernels measured 1.2-1.8 TB/s cold on the B=256 decode shapes
+// (tools/sweep_skinny2.py).  LDS cannot hold a double-buffered [MT*16|128]
+// X tile, so this variant is SINGLE-buffered and relies on 2 workgroups
+// per CU overlapping each other's load and compute phases; the global
+// prefetch for chunk c+1 is issued before the MFMA loop of chunk c.
+// ---------------------------------------------------------------------------
+template <int MT, bool SPLIT>
+__global__ __launch_bounds__(256, 2) void skinny_gemm2m_kernel(
+    const __hip_bfloat16* __restrict__ x,  // [M, K]
+    const __hip_bfloat16* __restrict__ w,  // [N, K]
+    __hip_bfloat16* __restrict__ out,      // [M, N] (SPLIT=false)
+    float* __restrict__ part,              // [S, M, N] (SPLIT=true)
+    int M, int N, int K, int KS) {
+  __shared__ char lds[SG2_TILE_B + MT * 16 * SG2_BK * 2];  // [W tile | X tile]
+  char* const wt = lds;
+  char* const xt = lds + SG2_TILE_B;
+  const int lane = threadIdx.x % WAVE;
+  const int wave = threadIdx.x / WAVE;
+  const int n0 = blockIdx.x * 64;
+  const int split = blockIdx.y;
+  const int k_begin = split * KS;
+  const int k_end = min(K, k_begin + KS);
+  const int nchunks = (k_end - k_begin + SG2_BK - 1) / SG2_BK;
+  if (nchunks <= 0 || n0 >= N) return;
+
+  // W tile staging: thread -> (row t>>2, 64B segment t&3)
+  const int wlrow = threadIdx.x >> 2;
+  const int wlseg = threadIdx.x & 3;
+  const long wrow = (long)min(n0 + wlrow, N - 1) * K;
+  // X tile staging: 16/MT threads per row, MT 16B units each
+  constexpr int TPR = 16 / MT;          // threads per X row
+  const int xlrow = threadIdx.x / TPR;
+  const int xuseg = (threadIdx.x % TPR) * MT;  // first unit of this thread
+  const long xrow = (long)min(xlrow, M - 1) * K;
+
+  // staging in batches of 4 float4s keeps live ranges register-sized
+  // (a held gx[MT] array spills to scratch); cross-chunk latency hiding
+  // comes from the second workgroup on the CU, not a register prefetch
+#define SG2M_STAGE(c)                                                       \
+  {                                                                         \
+    const long kcw = k_begin + (long)(c)*SG2_BK + wlseg * 32;               \
+    float4 w0 = *reinterpret_cast<const float4*>(w + wrow + kcw);           \
+    float4 w1 = *reinterpret_cast<const float4*>(w + wrow + kcw + 8);       \
+    float4 w2 = *reinterpret_cast<const float4*>(w + wrow + kcw + 16);      \
+    float4 w3 = *reinterpret_cast<const float4*>(w + wrow + kcw + 24);      \
+    const long kcx = k_begin + (long)(c)*SG2_BK + xuseg * 8;                \
+    _Pragma("unroll") for (int b = 0; b < MT / 4; ++b) {                    \
+      float4 x0 = *reinterpret_cast<const float4*>(x + xrow + kcx + b * 32);\
+      float4 x1 = *reinterpret_cast<const float4*>(x + xrow + kcx + b * 32 + 8);  \
+      float4 x2 = *reinterpret_cast<const float4*>(x + xrow + kcx + b * 32 + 16); \
+      float4 x3 = *reinterpret_cast<const float4*>(x + xrow + kcx + b * 32 + 24); \
+      *reinterpret_cast<float4*>(xt + sg2_addr(xlrow, xuseg + b * 4)) = x0;     \
+      *reinterpret_cast<float4*>(xt + sg2_addr(xlrow, xuseg + b * 4 + 1)) = x1; \
+      *reinterpret_cast<float4*>(xt + sg2_addr(xlrow, xuseg + b * 4 + 2)) = x2; \
+      *reinterpret_cast<float4*>(xt + sg2_addr(xlrow, xuseg + b * 4 + 3)) = x3; \
+    }                                                                       \
+    *reinterpret_cast<float4*>(wt + sg2_addr(wlrow, wlseg * 4 + 0)) = w0;   \
+    *reinterpret_cast<float4*>(wt + sg2_addr(wlrow, wlseg * 4 + 1)) = w1;   \
+    *reinterpret_cast<float4*>(wt + sg2_addr(wlrow, wlseg * 4 + 2)) = w2;   \
+    *reinterpret_cast<float4*>(wt + sg2_addr(wlrow, wlseg * 4 + 3)) = w3;   \
+  }
+
+  f32x4 acc[MT];
+#pragma unroll
+  for (int t = 0; t < MT; ++t)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) acc[t][r] = 0.f;
+
+  const int bcol = lane & 15;
+  const int kg = lane >> 4;
+
+  for (int c = 0; c < nchunks; ++c) {
+    __syncthreads();  // all waves done reading the tile of chunk c-1
+    SG2M_STAGE(c);
+    __syncthreads();  // tile of chunk c visible
+#pragma unroll
+    for (int s = 0; s < SG2_BK / 32; ++s) {
+      const shortx8 bv = *reinterpret_cast<const shortx8*>(
+          wt + sg2_addr(wave * 16 + bcol, s * 4 + kg));
+      sfrag bfr;
+      __builtin_memcpy(&bfr, &bv, 16);
+#pragma unroll
+      for (int t = 0; t < MT; ++t) {
+        const shortx8 av = *reinterpret_cast<const shortx8*>(
+            xt + sg2_addr(t * 16 + bcol, s * 4 + kg));
+        sfrag afr;
+        __builtin_memcpy(&afr, &av, 16);
+        acc[t] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afr.v, bfr.v,
+                                                         acc[t], 0, 0, 0);
+      }
+    }
+  }
+
+  const int n = n0 + wave * 16 + bcol;
+  if (n >= N) return;
+#pragma unroll
+  for (int t = 0; t < MT; ++t) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int m = t * 16 + (lane >> 4) * 4 + r;
+      if (m >= M) continue;
+      if (SPLIT) {
+        part[((long)split * M + m) * N + n] = acc[t][r];
+      } else {
+        *(unsigned short*)(out + (long)m * N + n) = f32_to_bf16(acc[t][r]);
+      }
+    }
+  }
+}
+#undef SG2M_STAGE
+
 __global__ void skinny_combine_kernel(const float* __restrict__ part,
                                       __hip_bfloat16* __restrict__ out,
                                       int S, long MN) {
@@ -311,7 +424,7 @@ torch::Tensor skinny_gemm2_ex(torch::Tensor x, torch::Tensor w, long s_target) {
   TORCH_CHECK(x.is_cuda() && x.dtype() == torch::kBFloat16, "bf16 GPU only");
   TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.size(1) == w.size(1));
   const int M = x.size(0), K = x.size(1), N = w.size(0);
-  TORCH_CHECK(M <= 64, "skinny_gemm2 is for M <= 64");
+  TORCH_CHECK(M <= 256, "skinny_gemm2 is for M <= 256");
   TORCH_CHECK(K % SG2_BK == 0, "K must be a multiple of ", SG2_BK);
   auto out = torch::empty({M, N}, x.options());
   if (M == 0 || N == 0) return out;
@@ -323,24 +436,38 @@ torch::Tensor skinny_gemm2_ex(torch::Tensor x, torch::Tensor w, long s_target) {
   auto stream = CUR_STREAM;
   torch::Tensor part;
   dim3 grid(nblocks, S), block(256);
+  float* pptr = nullptr;
   if (S > 1) {
     part = torch::empty({S, M, N}, x.options().dtype(torch::kFloat32));
-    hipLaunchKernelGGL((skinny_gemm2_kernel<true>), grid, block, 0, stream,
-                       (const __hip_bfloat16*)x.data_ptr(),
-                       (const __hip_bfloat16*)w.data_ptr(),
-                       (__hip_bfloat16*)out.data_ptr(),
-                       part.data_ptr<float>(), M, N, K, KS);
+    pptr = part.data_ptr<float>();
+  }
+#define SG2_DISPATCH(SPLITV)                                                 \
+  if (M <= 64)                                                               \
+    hipLaunchKernelGGL((skinny_gemm2_kernel<SPLITV>), grid, block, 0,        \
+                       stream, (const __hip_bfloat16*)x.data_ptr(),          \
+                       (const __hip_bfloat16*)w.data_ptr(),                  \
+                       (__hip_bfloat16*)out.data_ptr(), pptr, M, N, K, KS);  \
+  else if (M <= 128)                                                         \
+    hipLaunchKernelGGL((skinny_gemm2m_kernel<8, SPLITV>), grid, block, 0,    \
+                       stream, (const __hip_bfloat16*)x.data_ptr(),          \
+                       (const __hip_bfloat16*)w.data_ptr(),                  \
+                       (__hip_bfloat16*)out.data_ptr(), pptr, M, N, K, KS);  \
+  else                                                                       \
+    hipLaunchKernelGGL((skinny_gemm2m_kernel<16, SPLITV>), grid, block, 0,   \
+                       stream, (const __hip_bfloat16*)x.data_ptr(),          \
+                       (const __hip_bfloat16*)w.data_ptr(),                  \
+                       (__hip_bfloat16*)out.data_ptr(), pptr, M, N, K, KS)
+  if (S > 1) {
+    SG2_DISPATCH(true);
     const long MN = (long)M * N;
-    int cblocks = (int)std::min<long>((MN + 255) / 256, 1024);
+    int cblocks = (int)std::min<long>((MN + 255) / 256, 2048);
     hipLaunchKernelGGL(skinny_combine_kernel, dim3(cblocks), dim3(256), 0,
                        stream, part.data_ptr<float>(),
                        (__hip_bfloat16*)out.data_ptr(), S, MN);
   } else {
-    hipLaunchKernelGGL((skinny_gemm2_kernel<false>), grid, block, 0, stream,
-                       (const __hip_bfloat16*)x.data_ptr(),
-                       (const __hip_bfloat16*)w.data_ptr(),
-                       (__hip_bfloat16*)out.data_ptr(), nullptr, M, N, K, KS);
+    SG2_DISPATCH(false);
   }
+#undef SG2_DISPATCH
   HIP_CHECK_KERNEL();
   return out;
 }

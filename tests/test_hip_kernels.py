@@ -272,7 +272,10 @@ def test_skinny_gemm(M, N, K):
 
 @pytest.mark.parametrize("M,N,K,st", [(1, 6144, 4096, 256), (17, 4096, 4096, 256),
                                       (64, 6144, 4096, 512), (64, 4096, 14336, 512),
-                                      (64, 28672, 4096, 1), (33, 4000, 1152, 512)])
+                                      (64, 28672, 4096, 1), (33, 4000, 1152, 512),
+                                      (128, 6144, 4096, 512), (100, 4096, 4096, 512),
+                                      (256, 6144, 4096, 512), (256, 4096, 14336, 512),
+                                      (200, 28672, 4096, 1), (256, 4000, 1152, 512)])
 def test_skinny_gemm2(M, N, K, st):
     """LDS-staged v2 (split and no-split paths, ragged N) vs fp32 torch."""
     from aurora_amd import _hip_ops

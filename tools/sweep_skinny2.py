@@ -35,8 +35,8 @@ for tag, M, K, N in SHAPES:
     x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
     iters = max(20, min(60, 3 * ncopy))
     res = {}
-    only_blaslt = os.environ.get("BLASLT_ONLY") == "1" or M > 64
-    for st in () if only_blaslt else (256, 512, 1024, 2048):
+    only_blaslt = os.environ.get("BLASLT_ONLY") == "1"
+    for st in () if (only_blaslt or M > 64) else (256, 512, 1024, 2048):
         for unr in (2, 4):
             dt = timeit(lambda i: hip.skinny_gemm_ex(x, ws[i % ncopy], st, unr), iters)
             res[f"sk_s{st}_u{unr}"] = dt
