@@ -235,10 +235,13 @@ __global__ __launch_bounds__(256, 2) void skinny_gemm2_kernel(
 // ---------------------------------------------------------------------------
 // v2 mid-M variant (65 <= M <= 256, MT = 8 or 16 m-tiles): hipBLASLt's
 // skinny-M kernels measured 1.2-1.8 TB/s cold on the B=256 decode shapes
-// (tools/sweep_skinny2.py).  LDS cannot hold a double-buffered [MT*16|128]
-// X tile, so this variant is SINGLE-buffered and relies on 2 workgroups
-// per CU overlapping each other's load and compute phases; the global
-// prefetch for chunk c+1 is issued before the MFMA loop of chunk c.
+// (tools/sweep_skinny2.py).  Only the x tile is LDS-staged (it is L2-hot:
+// every workgroup re-reads the same [M,K] activations, so the staging
+// loads return in ~100 cycles); W streams straight into MFMA B-fragments
+// as 16 B per-lane row-gathers exactly like v1 -- at MT tiles per wave
+// each W fragment feeds MT MFMAs, so the gather instruction rate that
+// capped v1 at MT=4 is amortized 2-4x further.  Single-buffered; cross-
+// chunk latency hides behind the other 2-5 workgroups on the CU.
 // ---------------------------------------------------------------------------
 template <int MT, bool SPLIT>
 __global__ __launch_bounds__(256, 2) void skinny_gemm2m_kernel(
@@ -247,9 +250,7 @@ __global__ __launch_bounds__(256, 2) void skinny_gemm2m_kernel(
     __hip_bfloat16* __restrict__ out,      // [M, N] (SPLIT=false)
     float* __restrict__ part,              // [S, M, N] (SPLIT=true)
     int M, int N, int K, int KS) {
-  __shared__ char lds[SG2_TILE_B + MT * 16 * SG2_BK * 2];  // [W tile | X tile]
-  char* const wt = lds;
-  char* const xt = lds + SG2_TILE_B;
+  __shared__ char xt[MT * 16 * SG2_BK * 2];  // x tile, XOR-swizzled
   const int lane = threadIdx.x % WAVE;
   const int wave = threadIdx.x / WAVE;
   const int n0 = blockIdx.x * 64;
@@ -259,42 +260,16 @@ __global__ __launch_bounds__(256, 2) void skinny_gemm2m_kernel(
   const int nchunks = (k_end - k_begin + SG2_BK - 1) / SG2_BK;
   if (nchunks <= 0 || n0 >= N) return;
 
-  // W tile staging: thread -> (row t>>2, 64B segment t&3)
-  const int wlrow = threadIdx.x >> 2;
-  const int wlseg = threadIdx.x & 3;
-  const long wrow = (long)min(n0 + wlrow, N - 1) * K;
   // X tile staging: 16/MT threads per row, MT 16B units each
   constexpr int TPR = 16 / MT;          // threads per X row
   const int xlrow = threadIdx.x / TPR;
   const int xuseg = (threadIdx.x % TPR) * MT;  // first unit of this thread
   const long xrow = (long)min(xlrow, M - 1) * K;
 
-  // staging in batches of 4 float4s keeps live ranges register-sized
-  // (a held gx[MT] array spills to scratch); cross-chunk latency hiding
-  // comes from the second workgroup on the CU, not a register prefetch
-#define SG2M_STAGE(c)                                                       \
-  {                                                                         \
-    const long kcw = k_begin + (long)(c)*SG2_BK + wlseg * 32;               \
-    float4 w0 = *reinterpret_cast<const float4*>(w + wrow + kcw);           \
-    float4 w1 = *reinterpret_cast<const float4*>(w + wrow + kcw + 8);       \
-    float4 w2 = *reinterpret_cast<const float4*>(w + wrow + kcw + 16);      \
-    float4 w3 = *reinterpret_cast<const float4*>(w + wrow + kcw + 24);      \
-    const long kcx = k_begin + (long)(c)*SG2_BK + xuseg * 8;                \
-    _Pragma("unroll") for (int b = 0; b < MT / 4; ++b) {                    \
-      float4 x0 = *reinterpret_cast<const float4*>(x + xrow + kcx + b * 32);\
-      float4 x1 = *reinterpret_cast<const float4*>(x + xrow + kcx + b * 32 + 8);  \
-      float4 x2 = *reinterpret_cast<const float4*>(x + xrow + kcx + b * 32 + 16); \
-      float4 x3 = *reinterpret_cast<const float4*>(x + xrow + kcx + b * 32 + 24); \
-      *reinterpret_cast<float4*>(xt + sg2_addr(xlrow, xuseg + b * 4)) = x0;     \
-      *reinterpret_cast<float4*>(xt + sg2_addr(xlrow, xuseg + b * 4 + 1)) = x1; \
-      *reinterpret_cast<float4*>(xt + sg2_addr(xlrow, xuseg + b * 4 + 2)) = x2; \
-      *reinterpret_cast<float4*>(xt + sg2_addr(xlrow, xuseg + b * 4 + 3)) = x3; \
-    }                                                                       \
-    *reinterpret_cast<float4*>(wt + sg2_addr(wlrow, wlseg * 4 + 0)) = w0;   \
-    *reinterpret_cast<float4*>(wt + sg2_addr(wlrow, wlseg * 4 + 1)) = w1;   \
-    *reinterpret_cast<float4*>(wt + sg2_addr(wlrow, wlseg * 4 + 2)) = w2;   \
-    *reinterpret_cast<float4*>(wt + sg2_addr(wlrow, wlseg * 4 + 3)) = w3;   \
-  }
+  const int bcol = lane & 15;
+  const int kg = lane >> 4;
+  const int n = n0 + wave * 16 + bcol;
+  const long wbase = (long)min(n, N - 1) * K;  // clamped, masked at store
 
   f32x4 acc[MT];
 #pragma unroll
@@ -302,17 +277,28 @@ __global__ __launch_bounds__(256, 2) void skinny_gemm2m_kernel(
 #pragma unroll
     for (int r = 0; r < 4; ++r) acc[t][r] = 0.f;
 
-  const int bcol = lane & 15;
-  const int kg = lane >> 4;
-
   for (int c = 0; c < nchunks; ++c) {
+    const long kc = k_begin + (long)c * SG2_BK;
     __syncthreads();  // all waves done reading the tile of chunk c-1
-    SG2M_STAGE(c);
+    {
+      const long kcx = kc + xuseg * 8;
+#pragma unroll
+      for (int b = 0; b < MT / 4; ++b) {
+        float4 x0 = *reinterpret_cast<const float4*>(x + xrow + kcx + b * 32);
+        float4 x1 = *reinterpret_cast<const float4*>(x + xrow + kcx + b * 32 + 8);
+        float4 x2 = *reinterpret_cast<const float4*>(x + xrow + kcx + b * 32 + 16);
+        float4 x3 = *reinterpret_cast<const float4*>(x + xrow + kcx + b * 32 + 24);
+        *reinterpret_cast<float4*>(xt + sg2_addr(xlrow, xuseg + b * 4)) = x0;
+        *reinterpret_cast<float4*>(xt + sg2_addr(xlrow, xuseg + b * 4 + 1)) = x1;
+        *reinterpret_cast<float4*>(xt + sg2_addr(xlrow, xuseg + b * 4 + 2)) = x2;
+        *reinterpret_cast<float4*>(xt + sg2_addr(xlrow, xuseg + b * 4 + 3)) = x3;
+      }
+    }
     __syncthreads();  // tile of chunk c visible
 #pragma unroll
     for (int s = 0; s < SG2_BK / 32; ++s) {
-      const shortx8 bv = *reinterpret_cast<const shortx8*>(
-          wt + sg2_addr(wave * 16 + bcol, s * 4 + kg));
+      const shortx8 bv =
+          *reinterpret_cast<const shortx8*>(w + wbase + kc + s * 32 + kg * 8);
       sfrag bfr;
       __builtin_memcpy(&bfr, &bv, 16);
 #pragma unroll
@@ -327,7 +313,6 @@ __global__ __launch_bounds__(256, 2) void skinny_gemm2m_kernel(
     }
   }
 
-  const int n = n0 + wave * 16 + bcol;
   if (n >= N) return;
 #pragma unroll
   for (int t = 0; t < MT; ++t) {
@@ -343,7 +328,6 @@ __global__ __launch_bounds__(256, 2) void skinny_gemm2m_kernel(
     }
   }
 }
-#undef SG2M_STAGE
 
 __global__ void skinny_combine_kernel(const float* __restrict__ part,
                                       __hip_bfloat16* __restrict__ out,
