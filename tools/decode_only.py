@@ -13,7 +13,12 @@ for i in range(B):
 for _ in range(40 + CTX // 256):
     eng.step()
 torch.cuda.synchronize()
-for _ in range(100):
+import time
+t0 = time.perf_counter()
+N = 60
+for _ in range(N):
     eng.step()
 torch.cuda.synchronize()
-print("done", eng.counters)
+dt = (time.perf_counter() - t0) / N * 1000
+print(f"done {eng.counters} wall_ms_per_step={dt:.2f}")
+print("timers", {k: round(v, 2) for k, v in eng.timers.items()})
