@@ -161,6 +161,7 @@ class LLMEngine:
             self._host_bt = torch.full((maxb, mp), self._scratch_page,
                                        dtype=torch.int32, pin_memory=True)
             self._bt_staged = [None] * maxb
+        self._dec_staged_ids = None
         # counters (engine observability; surfaced by service.metrics)
         self.counters = {
             "prompt_tokens": 0, "generated_tokens": 0, "prefill_steps": 0,
@@ -329,8 +330,10 @@ class LLMEngine:
             return []
         dev = self.device
         if mode == "mixed":
+            self._dec_staged_ids = None
             return self._step_mixed(*seqs, chunks)
         if mode == "prefill":
+            self._dec_staged_ids = None
             self.counters["prefill_steps"] += 1
             ids: List[int] = []
             positions: List[int] = []
@@ -383,16 +386,17 @@ class LLMEngine:
 
         # decode
         self.counters["decode_steps"] += 1
-        ids = [s.token_ids[-1] for s in seqs]
-        positions = [s.num_tokens - 1 for s in seqs]
-        # note: the last token was appended by sampling but its KV is not yet
-        # written; it is this step's input token.
-        slots = [self._slot(s, s.num_tokens - 1) for s in seqs]
-        ctx_lens = [s.num_tokens for s in seqs]
         _t1 = time.monotonic()
         if self._graphs_enabled() and len(seqs) <= self._BUCKETS[-1]:
-            logits = self._decode_with_graph(seqs, ids, positions, slots, ctx_lens)
+            logits = self._decode_with_graph(seqs)
         else:
+            self._dec_staged_ids = None
+            ids = [s.token_ids[-1] for s in seqs]
+            positions = [s.num_tokens - 1 for s in seqs]
+            # note: the last token was appended by sampling but its KV is
+            # not yet written; it is this step's input token.
+            slots = [self._slot(s, s.num_tokens - 1) for s in seqs]
+            ctx_lens = [s.num_tokens for s in seqs]
             meta = AttnMetadata(
                 mode="decode",
                 positions=torch.tensor(positions, dtype=torch.int32, device=dev),
@@ -496,11 +500,18 @@ class LLMEngine:
         return (self.config.use_hip_graphs and self.device != "cpu"
                 and self.config.tp_size == 1)
 
-    def _decode_with_graph(self, seqs, ids, positions, slots, ctx_lens):
+    def _decode_with_graph(self, seqs):
         """Replay a hipGraph-captured decode forward (per batch-size bucket,
         fixed-shape static buffers; idle rows decode a reserved scratch page
         and are ignored).  Removes the per-layer kernel-launch host time
-        that dominates small-batch decode."""
+        that dominates small-batch decode.
+
+        Staging is incremental: when the batch membership is unchanged
+        since the previous graph decode, position/slot/context rows each
+        advance by exactly one token, so the pinned staging matrix gets a
+        single vectorized += 1 instead of three B-element Python list
+        builds (only page-boundary rows re-derive their slot); any
+        prefill/mixed/eager step invalidates the staged state."""
         B = len(seqs)
         bucket = next(b for b in self._BUCKETS if b >= B)
         g = self._graph_cache.get(bucket)
@@ -509,16 +520,31 @@ class LLMEngine:
             self._graph_cache[bucket] = g
         (graph, in_ids, in_pos, in_slot, in_bt, in_ctx, out_logits) = g
         mp = in_bt.shape[1]
-        # fill the static buffers (idle rows -> scratch page, ctx 1)
+        P = self.kv.page_size
         ids_t = self._host_i64
-        ids_t[:B] = torch.tensor(ids, dtype=torch.int64)
+        ids_t[:B] = torch.tensor([s.token_ids[-1] for s in seqs],
+                                 dtype=torch.int64)
         pos = self._host_i32
-        pos[:B, 0] = torch.tensor(positions, dtype=torch.int32)
-        pos[B:bucket, 0] = 0
-        pos[:B, 1] = torch.tensor(slots, dtype=torch.int32)
-        pos[B:bucket, 1] = self._scratch_slot
-        pos[:B, 2] = torch.tensor(ctx_lens, dtype=torch.int32)
-        pos[B:bucket, 2] = 1
+        batch_ids = [s.req_id for s in seqs]
+        if batch_ids == self._dec_staged_ids:
+            pos[:B] += 1  # position, slot and context each advance one token
+            for i, s in enumerate(seqs):
+                if (s.num_tokens - 1) % P == 0:  # slot jumped to a new page
+                    pos[i, 1] = self._slot(s, s.num_tokens - 1)
+        else:
+            pos[:B, 0] = torch.tensor([s.num_tokens - 1 for s in seqs],
+                                      dtype=torch.int32)
+            pos[B:bucket, 0] = 0
+            # note: the last token was appended by sampling but its KV is
+            # not yet written; it is this step's input token.
+            pos[:B, 1] = torch.tensor(
+                [self._slot(s, s.num_tokens - 1) for s in seqs],
+                dtype=torch.int32)
+            pos[B:bucket, 1] = self._scratch_slot
+            pos[:B, 2] = torch.tensor([s.num_tokens for s in seqs],
+                                      dtype=torch.int32)
+            pos[B:bucket, 2] = 1
+            self._dec_staged_ids = batch_ids
         # block-table rows change only on page append / batch-slot change:
         # rewrite a staged row only when its (seq, npages) key moved
         bt_host = self._host_bt
@@ -554,6 +580,7 @@ class LLMEngine:
             self._host_bt = torch.full((maxb, mp), self._scratch_page,
                                        dtype=torch.int32, pin_memory=True)
             self._bt_staged = [None] * maxb
+        self._dec_staged_ids = None
         in_ids = torch.zeros(bucket, dtype=torch.int64, device=dev)
         in_pos = torch.zeros(bucket, dtype=torch.int32, device=dev)
         in_slot = torch.full((bucket,), self._scratch_slot, dtype=torch.int32,
