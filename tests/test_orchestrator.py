@@ -182,3 +182,24 @@ def test_tool_confirmation_gate(env):
         out = await GLOBAL_REGISTRY.run(ctx3, "terminal_exec", {"command": "uptime"})
         assert "CANCELLED" not in out
     asyncio.run(main())
+
+
+def test_skill_library_inventory():
+    """Integration skills are connection-gated; core/rca always eligible
+    (reference: skills/registry.py:36,430-441 — core/ + integrations/ +
+    rca/ layout)."""
+    from aurora_amd.agent.skills.registry import SkillRegistry
+    reg = SkillRegistry()
+    assert len(reg.skills) >= 20
+    cats = {s.category for s in reg.skills.values()}
+    assert {"core", "integrations", "rca"} <= cats
+    gated = [s for s in reg.skills.values() if s.requires_connection]
+    assert len(gated) >= 8
+    # no connections -> no gated skill bodies in the assembled prompt
+    txt = reg.assemble_rca_prompt(100000, lambda t: len(t) // 4, connected=[])
+    for s in gated:
+        assert f"## Skill: {s.name}" not in txt
+    # rca playbooks are present and ordered by priority before generic core
+    txt2 = reg.assemble_rca_prompt(100000, lambda t: len(t) // 4,
+                                   connected=["datadog"])
+    assert "Skill: datadog" in txt2 and "Skill: rca_deploy_rollback" in txt2
