@@ -400,6 +400,15 @@ def create_app(rt: Runtime) -> FastAPI:
         return store.tool_permissions()
 
     # ---- health ----
+    @app.get("/ui")
+    async def ui():
+        """Built-in single-file operations dashboard (the reference's
+        Next.js client consumes this same REST/SSE surface; SURVEY.md
+        2.10 notes the client can be kept if the API is preserved)."""
+        from fastapi.responses import HTMLResponse
+        from .dashboard import DASHBOARD_HTML
+        return HTMLResponse(DASHBOARD_HTML)
+
     @app.get("/health")
     async def health():
         ok = rt.engine.healthy() if hasattr(rt.engine, "healthy") else True

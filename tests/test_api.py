@@ -287,3 +287,12 @@ def test_client_sdk(env):
     assert sdk.metrics()["incidents_total"] >= 1
     did = sdk.kb_upload("sdk-doc", "restart pgbouncer when pool exhausts")
     assert sdk.kb_search("pgbouncer pool")[0]["doc_id"] == did
+
+
+def test_ui_dashboard(env):
+    _rt, client, _oid, _uid = env
+    r = client.get("/ui")
+    assert r.status_code == 200
+    assert "text/html" in r.headers["content-type"]
+    body = r.text
+    assert "aurora-mi355x" in body and "/incidents/stream" in body
