@@ -100,11 +100,13 @@ async def run_background_chat(rt, args: Dict[str, Any]) -> dict:
             # ORCHESTRATOR_ENABLED=true)
             from ..agent.orchestrator import Orchestrator
             orch = Orchestrator(llm, ctx,
-                                sub_agent_turns=int(args.get("max_turns", 0)) or 6)
+                                sub_agent_turns=int(args.get("max_turns", 0))
+                                or rt.settings.rca_max_turns or 6)
             result = await orch.run(state)
             final = result.final
         else:
-            wf = Workflow(llm, ctx, max_turns=int(args.get("max_turns", 0)) or None)
+            wf = Workflow(llm, ctx, max_turns=int(args.get("max_turns", 0))
+                          or rt.settings.rca_max_turns or None)
             async for ev in wf.stream(state):
                 if ev["type"] == "final":
                     final = ev["content"]

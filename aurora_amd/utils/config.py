@@ -40,6 +40,7 @@ class Settings:
     agent_recursion_limit: int = field(default_factory=lambda: _i("AGENT_RECURSION_LIMIT", 240))
     max_input_tokens: int = field(default_factory=lambda: _i("MAX_INPUT_TOKENS", 20000))
     workflow_timeout_s: int = field(default_factory=lambda: _i("WORKFLOW_TIMEOUT", 1800))
+    rca_max_turns: int = field(default_factory=lambda: _i("RCA_MAX_TURNS", 0))  # 0 = workflow default
     rca_skills_token_budget: int = field(default_factory=lambda: _i("RCA_SKILLS_TOKEN_BUDGET", 4000))
     # guardrails (reference: .env.example:123-128)
     guardrails_enabled: bool = field(default_factory=lambda: _b("GUARDRAILS_ENABLED", True))
