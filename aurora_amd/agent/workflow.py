@@ -62,7 +62,8 @@ class Workflow:
         # call (reference: workflow.py:1076, agent.py:875-910); FAIL CLOSED
         rail_task = asyncio.create_task(check_input(
             state.question, llm=self.llm,
-            enabled=self.settings.guardrails_enabled))
+            enabled=self.settings.guardrails_enabled,
+            embedding_rail=getattr(self.ctx, "extra", {}).get("embedding_rail")))
 
         # 3. prompt assembly concurrent with the rail
         tools = self.registry.available(self.ctx, subset=self.tool_subset)

@@ -2,9 +2,12 @@
 fired before the agent's first LLM call and FAIL-CLOSED (reference:
 guardrails/input_rail.py:244-299).
 
-Two layers:
+Three layers:
 1. host-side heuristic patterns (always on, microseconds);
-2. an LLM self-check on the in-process engine (replaces NeMo's rented
+2. embedding-similarity classifier on the shared on-GPU encoder
+   (guardrails/embedding_rail.py) — catches paraphrases, HBM-resident
+   exemplar bank, skipped when no encoder is wired;
+3. an LLM self-check on the in-process engine (replaces NeMo's rented
    self-check), JSON-grammar constrained, skipped when no engine is given.
 Any rail failure => blocked with a failure classification.
 """
@@ -47,13 +50,14 @@ JSON:"""
 @dataclass
 class InputRailResult:
     allowed: bool
-    classification: str  # "allowed" | "heuristic_block" | "llm_block" | "rail_failure"
+    classification: str  # allowed | heuristic_block | embedding_block | llm_block | rail_failure
     reason: str = ""
     latency_ms: float = 0.0
 
 
 async def check_input(text: str, llm=None, enabled: bool = True,
-                      use_llm: bool = False) -> InputRailResult:
+                      use_llm: bool = False,
+                      embedding_rail=None) -> InputRailResult:
     t0 = time.monotonic()
 
     def _ms() -> float:
@@ -66,6 +70,12 @@ async def check_input(text: str, llm=None, enabled: bool = True,
             if pat.search(text):
                 return InputRailResult(False, "heuristic_block",
                                        f"matched: {pat.pattern[:60]}", _ms())
+        if embedding_rail is not None:
+            hit, score, near = embedding_rail.check(text)
+            if hit:
+                return InputRailResult(
+                    False, "embedding_block",
+                    f"cosine {score:.2f} to exemplar: {near[:60]}", _ms())
         if use_llm and llm is not None:
             res = await llm.complete_text(
                 RAIL_PROMPT.format(text=text[:4000]),

@@ -77,6 +77,7 @@ class Runtime:
             engine_server = AsyncEngineServer(ecfg)
         self.engine = engine_server
         self._rag: Dict[str, RagIndex] = {}
+        self._embedding_rail = None
         self._graph: Dict[str, InfraGraph] = {}
         self._rca_rate: Dict[str, float] = {}  # user -> last background RCA ts
         self.kubectl_agents: Dict[str, Any] = {}  # cluster_id -> AgentConn
@@ -102,6 +103,16 @@ class Runtime:
             self._graph[org_id] = InfraGraph(self.org(org_id))
         return self._graph[org_id]
 
+    def embedding_rail(self):
+        """Shared prompt-injection embedding classifier (lazy singleton;
+        guardrails/embedding_rail.py). Uses the hashed-BoW backend unless
+        a trained encoder is configured — random-init encoder weights
+        would make similarity meaningless."""
+        if self._embedding_rail is None:
+            from ..guardrails.embedding_rail import EmbeddingRail, hashed_bow_embed
+            self._embedding_rail = EmbeddingRail(hashed_bow_embed, threshold=0.55)
+        return self._embedding_rail
+
     def tool_ctx(self, org_id: str, user_id: Optional[str] = None,
                  session_id: Optional[str] = None,
                  incident_id: Optional[str] = None, mode: str = "agent",
@@ -112,7 +123,8 @@ class Runtime:
             incident_id=incident_id, llm=self.llm(org_id), mode=mode,
             rag=self.rag(org_id), graph=self.graph(org_id), queue=self.queue,
             notifier=notifier, settings=self.settings, depth=depth,
-            extra={"kubectl_agents": self.kubectl_agents, "runtime": self})
+            extra={"kubectl_agents": self.kubectl_agents, "runtime": self,
+                   "embedding_rail": self.embedding_rail()})
 
     # ---- RCA trigger rate limit (reference: task.py:336) ----
     def rca_allowed(self, user_key: str) -> bool:
