@@ -231,6 +231,11 @@ async def trigger_rca(ctx: ToolContext, title: str, details: str = "") -> str:
         return "ERROR: nested RCA triggers are blocked"  # trigger_rca_tool.py:38
     if ctx.org_store is None or ctx.queue is None:
         return "ERROR: background pipeline unavailable"
+    rt = (ctx.extra or {}).get("runtime")
+    if rt is not None and not rt.rca_allowed(
+            f"tool:{ctx.user_id or ctx.org_store.org_id}"):
+        return ("ERROR: RCA trigger rate limit — an investigation was "
+                "already dispatched recently")  # trigger_rca_tool.py rate limit
     iid = ctx.org_store.create_incident(
         title=title, source="agent", payload={"details": details},
         user_id=ctx.user_id)

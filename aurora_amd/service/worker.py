@@ -177,10 +177,10 @@ async def trigger_delayed_rca(rt, args: Dict[str, Any]) -> dict:
         return {"error": "incident gone"}
     if inc.get("aurora_chat_session_id"):
         return {"skipped": "rca already dispatched"}
-    user_key = args.get("user_id") or org_id
-    if not rt.rca_allowed(f"bg:{user_key}"):
-        store.lifecycle(incident_id, "rca_rate_limited", {})
-        return {"skipped": "rate limited"}
+    # NOTE: no per-user rate limit here — webhook alert storms are handled
+    # by correlation + the session-link/dedup_key checks above; the per-user
+    # window applies to the agent-initiated trigger_rca TOOL (reference
+    # rate-limits trigger_rca_tool.py, not alert ingestion)
     tid = rt.queue.enqueue(
         "run_background_chat",
         {"org_id": org_id, "incident_id": incident_id,

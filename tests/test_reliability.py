@@ -108,3 +108,56 @@ def test_engine_step_failure_fails_requests_loudly():
         asyncio.run(main())
     finally:
         server.shutdown()
+
+
+@pytest.mark.timeout(240)
+def test_sustained_webhook_load_drains_and_bounds():
+    """Soak: 18 webhooks across 3 waves on a tiny engine; the correlator
+    dedupes the repeat waves into per-service incidents, every incident
+    must reach a terminal aurora_status, the queue must drain, and the
+    engine must stay healthy (session-KV eviction bounds pages)."""
+    import time
+    from fastapi.testclient import TestClient
+    from aurora_amd.engine.engine import EngineConfig
+    from aurora_amd.engine.server import AsyncEngineServer
+    from aurora_amd.service.app import create_app
+    from aurora_amd.service.runtime import Runtime
+    from aurora_amd.utils.config import Settings
+
+    cfg = Settings(db_path=":memory:", secrets_path="/tmp/soak_secrets.json",
+                   rca_max_turns=1)
+    server = AsyncEngineServer(EngineConfig(model="tiny-llama", device="cpu",
+                                            seed=3))
+    rt = Runtime(cfg, engine_server=server)
+    org = rt.admin.create_org("soak")
+    rt.admin.create_user(org, "s@s.io", user_id="soak-user")
+    app = create_app(rt)
+    try:
+        with TestClient(app) as client:
+            h = {"X-User-Id": "soak-user", "X-Org-Id": org}
+            for wave in range(3):
+                for i in range(6):
+                    r = client.post(f"/webhook/pagerduty/{org}", json={
+                        "event": {"eventType": "incident.triggered", "data": {
+                            "id": f"PD-{wave}-{i}",
+                            "title": f"svc-{i} error burst {wave}",
+                            "urgency": "high"}}})
+                    assert r.status_code == 200
+                time.sleep(1.0)
+            deadline = time.time() + 180
+            while time.time() < deadline:
+                xs = client.get("/incidents", headers=h).json()
+                if len(xs) >= 6 and all(
+                        x["aurora_status"] in ("completed", "failed") for x in xs):
+                    break
+                time.sleep(0.5)
+            xs = client.get("/incidents", headers=h).json()
+            # 3 waves x 6 services correlate into 6 incidents (dedup works)
+            assert 6 <= len(xs) < 18, len(xs)
+            states = {x["aurora_status"] for x in xs}
+            assert states <= {"completed", "failed"}, states
+            assert server.healthy()
+            eng = server.engine
+            assert not eng.running and not eng.waiting  # engine drained
+    finally:
+        server.shutdown()
