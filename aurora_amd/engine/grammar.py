@@ -10,8 +10,7 @@ for a byte tokenizer, a high-probability guide for BPE).
 """
 from __future__ import annotations
 
-import copy
-from typing import List, Optional
+from typing import List
 
 # parser states
 _VALUE = 0        # expecting the start of a value

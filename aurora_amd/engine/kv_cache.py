@@ -12,7 +12,7 @@ Replaces nothing in the reference (it rents context windows; SURVEY.md
 """
 from __future__ import annotations
 
-from typing import Dict, List, Tuple
+from typing import List, Tuple
 
 import torch
 

@@ -17,7 +17,6 @@ import json
 import re
 import time
 from dataclasses import dataclass
-from typing import Optional
 
 from ..engine.grammar import JsonGrammar
 from ..engine.sampling_params import SamplingParams

@@ -8,8 +8,7 @@ safetensors checkpoints via models.loader.
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
-from typing import Optional
+from dataclasses import dataclass
 
 
 @dataclass

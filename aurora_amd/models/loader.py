@@ -9,7 +9,6 @@ parallel/tp.py documents)."""
 from __future__ import annotations
 
 import glob
-import json
 import os
 from typing import Dict, Optional
 

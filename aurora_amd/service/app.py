@@ -19,7 +19,7 @@ import time
 from typing import Optional
 
 from fastapi import Depends, FastAPI, Header, HTTPException, Request
-from fastapi.responses import JSONResponse, StreamingResponse
+from fastapi.responses import StreamingResponse
 
 from .runtime import Runtime
 from .store import OrgStore

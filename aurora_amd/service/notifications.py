@@ -5,7 +5,6 @@ POST path is implemented and used when a sink URL is configured."""
 from __future__ import annotations
 
 import json
-from typing import Optional
 
 
 async def dispatch_notification(rt, org_id: str, message: str,

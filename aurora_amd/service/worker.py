@@ -9,7 +9,6 @@ citations, postmortem, notifications, SSE.
 """
 from __future__ import annotations
 
-import asyncio
 import json
 import time
 from typing import Any, Dict, Optional
