@@ -76,11 +76,29 @@ def create_app(rt: Runtime) -> FastAPI:
 
     # ------------------------------------------------------------- webhooks
 
+    def _require_org(org_id: str) -> None:
+        """Webhook URLs embed the org id; an unknown org must 404 instead
+        of silently materialising tenant state (org<->binding middleware,
+        reference: main_compute.py:299)."""
+        if not any(o["id"] == org_id for o in rt.admin.orgs()):
+            raise HTTPException(404, "unknown org")
+
+    async def _json_object(request: Request) -> dict:
+        body = await request.body()
+        try:
+            payload = json.loads(body or b"{}")
+        except (json.JSONDecodeError, UnicodeDecodeError):
+            raise HTTPException(400, "invalid JSON") from None
+        if not isinstance(payload, dict):
+            raise HTTPException(400, "payload must be a JSON object")
+        return payload
+
     @app.post("/webhook/{source}/{org_id}")
     async def webhook(source: str, org_id: str, request: Request,
                       user_id: Optional[str] = None):
         if source not in sources():
             raise HTTPException(404, f"unknown source; known: {sources()}")
+        _require_org(org_id)
         body = await request.body()
         store = rt.org(org_id)
         secret = store.get_token_ref("system", f"webhook_secret:{source}")
@@ -89,9 +107,15 @@ def create_app(rt: Runtime) -> FastAPI:
             raise HTTPException(403, "bad signature")
         try:
             payload = json.loads(body or b"{}")
-        except json.JSONDecodeError:
+        except (json.JSONDecodeError, UnicodeDecodeError):
             raise HTTPException(400, "invalid JSON") from None
-        iid, created = process_webhook(rt, org_id, source, payload, user_id=user_id)
+        if not isinstance(payload, dict):
+            raise HTTPException(400, "payload must be a JSON object")
+        try:
+            iid, created = process_webhook(rt, org_id, source, payload,
+                                           user_id=user_id)
+        except (KeyError, TypeError, AttributeError, ValueError) as e:
+            raise HTTPException(400, f"unprocessable alert payload: {e}") from None
         return {"incident_id": iid, "created": created}
 
     @app.post("/webhook/pr/{provider}/{org_id}")
@@ -99,6 +123,7 @@ def create_app(rt: Runtime) -> FastAPI:
         # change gating (reference: tasks/github_webhook_tasks.py:1090)
         if provider not in ("github", "bitbucket"):
             raise HTTPException(404, "provider must be github|bitbucket")
+        _require_org(org_id)
         body = await request.body()
         store = rt.org(org_id)
         secret = store.get_token_ref("system", f"webhook_secret:{provider}")
@@ -106,8 +131,13 @@ def create_app(rt: Runtime) -> FastAPI:
         if not _vs(provider, body, dict(request.headers), secret):
             raise HTTPException(403, "bad signature")
         from .change_gating import process_pr_webhook
-        tid = process_pr_webhook(rt, org_id, provider,
-                                 json.loads(body or b"{}"))
+        try:
+            payload = json.loads(body or b"{}")
+        except (json.JSONDecodeError, UnicodeDecodeError):
+            raise HTTPException(400, "invalid JSON") from None
+        if not isinstance(payload, dict):
+            raise HTTPException(400, "payload must be a JSON object")
+        tid = process_pr_webhook(rt, org_id, provider, payload)
         return {"task_id": tid}
 
     @app.post("/chat-events/{platform}/{org_id}")
@@ -115,8 +145,9 @@ def create_app(rt: Runtime) -> FastAPI:
         # Slack / Google Chat two-way bridge
         if platform not in ("slack", "google_chat"):
             raise HTTPException(404, "platform must be slack|google_chat")
+        _require_org(org_id)
         from .chat_integrations import normalize_chat_event
-        payload = await request.json()
+        payload = await _json_object(request)
         ev = normalize_chat_event(platform, payload)
         if ev is None:
             return {"ok": True}

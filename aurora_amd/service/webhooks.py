@@ -234,6 +234,11 @@ def process_webhook(rt, org_id: str, source: str, payload: dict,
     if norm_fn is None:
         raise ValueError(f"unknown alert source '{source}'")
     norm = norm_fn(payload or {})
+    # defensive caps: a hostile/buggy alert source must not balloon
+    # incident rows or the correlator's similarity text
+    norm["title"] = str(norm.get("title") or "(untitled alert)")[:500]
+    if norm.get("external_id") is not None:
+        norm["external_id"] = str(norm["external_id"])[:200]
     store = rt.org(org_id)
     corr = AlertCorrelator(store, graph=rt.graph(org_id))
     enriched = dict(payload or {}, **{k: v for k, v in norm.items() if v})
