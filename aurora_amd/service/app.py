@@ -254,6 +254,8 @@ def create_app(rt: Runtime) -> FastAPI:
     # ---- monitor: waterfall + fleet (reference: monitor/waterfall_routes.py) ----
     @app.get("/monitor/waterfall/{iid}")
     async def waterfall(iid: str, store: OrgStore = Depends(auth)):
+        if store.get_incident(iid) is None:  # no cross-org probing oracle
+            raise HTTPException(404, "unknown incident")
         steps = store.steps(iid)
         return {"incident_id": iid, "steps": steps,
                 "total_ms": sum(((s["finished_at"] or s["started_at"])
