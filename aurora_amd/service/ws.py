@@ -87,21 +87,29 @@ def mask_frame(data: bytes, opcode: int = 0x1) -> bytes:
     return head + mask + bytes(b ^ mask[i % 4] for i, b in enumerate(data))
 
 
+MAX_FRAME_BYTES = 16 * 1024 * 1024  # refuse absurd length fields up front
+
+
 async def read_frame(reader: asyncio.StreamReader):
-    """Returns (opcode, payload) or (0x8, b'') on close/EOF."""
+    """Returns (opcode, payload) or (0x8, b'') on close/EOF/oversize.
+    Any mid-frame EOF and any length beyond MAX_FRAME_BYTES map to a
+    close frame — a hostile peer must not be able to raise out of here
+    or make the server buffer a 2**63-byte frame."""
     try:
         h = await reader.readexactly(2)
+        opcode = h[0] & 0x0F
+        masked = h[1] & 0x80
+        n = h[1] & 0x7F
+        if n == 126:
+            n = struct.unpack(">H", await reader.readexactly(2))[0]
+        elif n == 127:
+            n = struct.unpack(">Q", await reader.readexactly(8))[0]
+        if n > MAX_FRAME_BYTES:
+            return 0x8, b""
+        mask = await reader.readexactly(4) if masked else b"\x00" * 4
+        data = await reader.readexactly(n) if n else b""
     except (asyncio.IncompleteReadError, ConnectionResetError):
         return 0x8, b""
-    opcode = h[0] & 0x0F
-    masked = h[1] & 0x80
-    n = h[1] & 0x7F
-    if n == 126:
-        n = struct.unpack(">H", await reader.readexactly(2))[0]
-    elif n == 127:
-        n = struct.unpack(">Q", await reader.readexactly(8))[0]
-    mask = await reader.readexactly(4) if masked else b"\x00" * 4
-    data = await reader.readexactly(n) if n else b""
     if masked:
         data = bytes(b ^ mask[i % 4] for i, b in enumerate(data))
     return opcode, data
