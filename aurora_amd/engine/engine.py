@@ -177,6 +177,13 @@ class LLMEngine:
                     params: SamplingParams, session_id: Optional[str] = None) -> None:
         if request_id in self.seqs:
             raise ValueError(f"duplicate request id {request_id}")
+        if not prompt_token_ids:
+            raise ValueError("empty prompt")
+        v = self.spec.vocab_size
+        if min(prompt_token_ids) < 0 or max(prompt_token_ids) >= v:
+            # fail at the API boundary, not as an embedding IndexError deep
+            # inside the stepping thread
+            raise ValueError(f"prompt token id outside vocab [0, {v})")
         if len(prompt_token_ids) >= self.max_model_len:
             prompt_token_ids = prompt_token_ids[-(self.max_model_len - params.max_tokens - 1):]
         seq = Sequence(request_id, prompt_token_ids, params, session_id)
