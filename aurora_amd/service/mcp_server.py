@@ -99,8 +99,12 @@ class MCPServer:
         return self.rt.admin.get_user(token)
 
     async def handle(self, msg: dict, token: str) -> Optional[dict]:
+        if not isinstance(msg, dict):
+            return self._err(None, -32600, "invalid request")
         mid = msg.get("id")
         method = msg.get("method", "")
+        if not isinstance(method, str):
+            return self._err(mid, -32600, "method must be a string")
         if method == "initialize":
             return self._ok(mid, {
                 "protocolVersion": "2024-11-05",
@@ -121,12 +125,16 @@ class MCPServer:
                 for e in self.entries.values()]})
         if method == "tools/call":
             p = msg.get("params", {})
+            if not isinstance(p, dict):
+                return self._err(mid, -32602, "params must be an object")
             entry = self.entries.get(p.get("name", ""))
             if entry is None:
                 return self._err(mid, -32602, f"unknown tool {p.get('name')}")
+            args = p.get("arguments") or {}
+            if not isinstance(args, dict):
+                return self._err(mid, -32602, "arguments must be an object")
             try:
-                result = await entry.handler(self.rt, store,
-                                             p.get("arguments", {}) or {})
+                result = await entry.handler(self.rt, store, args)
                 return self._ok(mid, {"content": [
                     {"type": "text", "text": json.dumps(result, default=str)[:20000]}]})
             except Exception as e:  # noqa: BLE001
