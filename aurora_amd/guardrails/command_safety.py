@@ -47,6 +47,9 @@ _READONLY_PREFIXES = (
     "gcloud logging read", "az vm list", "az monitor", "journalctl",
     "systemctl status", "git log", "git show", "git diff", "helm list",
     "helm status", "docker ps", "docker logs", "docker inspect",
+    "grep ", "zgrep ", "egrep ", "tail ", "head ", "wc ", "stat ",
+    "dmesg", "lsof", "vmstat", "iostat", "mpstat", "nproc", "file ",
+    "find /var/log", "find /tmp",
 )
 
 
