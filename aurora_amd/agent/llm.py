@@ -34,10 +34,19 @@ class ModelConfig:
 
 class LLMManager:
     def __init__(self, server: AsyncEngineServer, config: Optional[ModelConfig] = None,
-                 store=None):
+                 store=None, engine_resolver=None):
         self.server = server
         self.config = config or ModelConfig()
         self.store = store  # OrgStore for usage tracking (optional)
+        # optional model-name -> engine mapping (per-purpose aux models;
+        # reference llm.py:32-67 ModelConfig); default: everything on the
+        # shared main engine
+        self.engine_resolver = engine_resolver
+
+    def _server_for(self, purpose: str) -> AsyncEngineServer:
+        if self.engine_resolver is None:
+            return self.server
+        return self.engine_resolver(self.config.resolve(purpose))
 
     @property
     def tokenizer(self):
@@ -50,7 +59,8 @@ class LLMManager:
                      purpose: str = "main", session_id: Optional[str] = None) -> GenResult:
         params = params or SamplingParams(temperature=0.2, max_tokens=512)
         t0 = time.monotonic()
-        res = await self.server.complete_text(prompt, params, session_id=session_id)
+        srv = self._server_for(purpose)
+        res = await srv.complete_text(prompt, params, session_id=session_id)
         if self.store is not None:
             self.store.track_usage(
                 model=self.config.resolve(purpose),

@@ -76,6 +76,10 @@ class Runtime:
             )
             engine_server = AsyncEngineServer(ecfg)
         self.engine = engine_server
+        # per-purpose aux engines (reference ModelConfig maps MAIN /
+        # SUMMARIZATION / GUARDRAILS to different models, llm.py:32-67;
+        # 288 GB HBM co-hosts a small aux model beside the main one)
+        self._aux_engines: Dict[str, AsyncEngineServer] = {}
         self._rag: Dict[str, RagIndex] = {}
         self._embedding_rail = None
         self._graph: Dict[str, InfraGraph] = {}
@@ -88,9 +92,27 @@ class Runtime:
     def org(self, org_id: str) -> OrgStore:
         return self.admin.org_store(org_id)
 
+    def engine_for(self, model: str) -> AsyncEngineServer:
+        """Engine serving `model`: the main engine when the name matches
+        (or is empty), else a lazily-started aux engine with a small KV
+        slice of the remaining HBM."""
+        if not model or model == self.settings.main_model:
+            return self.engine
+        srv = self._aux_engines.get(model)
+        if srv is None:
+            srv = AsyncEngineServer(EngineConfig(
+                model=model, device=self.settings.engine_device,
+                kv_memory_fraction=0.08, max_num_seqs=64))
+            self._aux_engines[model] = srv
+        return srv
+
     def llm(self, org_id: str) -> LLMManager:
-        return LLMManager(self.engine, ModelConfig(
-            main_model=self.settings.main_model), store=self.org(org_id))
+        mc = ModelConfig(
+            main_model=self.settings.main_model,
+            guardrails_model=self.settings.guardrails_model,
+            summarization_model=self.settings.summarization_model)
+        return LLMManager(self.engine, mc, store=self.org(org_id),
+                          engine_resolver=self.engine_for)
 
     def rag(self, org_id: str) -> RagIndex:
         if org_id not in self._rag:
@@ -145,3 +167,5 @@ class Runtime:
 
     async def stop(self) -> None:
         await self.queue.stop()
+        for srv in self._aux_engines.values():
+            srv.shutdown()

@@ -303,3 +303,36 @@ def test_trigger_rca_tool_rate_limited():
     r2 = asyncio.run(tool.func(ctx, title="second incident", details="y"))
     assert "rate limit" in r2
     server.shutdown()
+
+
+def test_per_purpose_model_routing():
+    """Purposes with a configured aux model run on a second in-process
+    engine (reference llm.py:32-67 ModelConfig purpose->model mapping);
+    unconfigured purposes share the main engine."""
+    import asyncio
+    from aurora_amd.engine.engine import EngineConfig
+    from aurora_amd.engine.server import AsyncEngineServer
+    from aurora_amd.engine.sampling_params import SamplingParams
+    from aurora_amd.service.runtime import Runtime
+    from aurora_amd.utils.config import Settings
+    server = AsyncEngineServer(EngineConfig(model="tiny-llama", device="cpu"))
+    rt = Runtime(Settings(db_path=":memory:", secrets_path="/tmp/pp_sec.json",
+                          main_model="tiny-llama",
+                          summarization_model="smoke-llama"),
+                 engine_server=server)
+    org = rt.admin.create_org("pp")
+    llm = rt.llm(org)
+    p = SamplingParams(temperature=0.0, max_tokens=2)
+
+    async def go():
+        a = await llm.invoke("hello", p, purpose="main")
+        b = await llm.invoke("hello", p, purpose="summarization")
+        return a, b
+    a, b = asyncio.run(go())
+    assert a.token_ids and b.token_ids
+    aux = rt._aux_engines.get("smoke-llama")
+    assert aux is not None
+    assert aux.engine.counters["generated_tokens"] >= 2
+    assert server.engine.counters["generated_tokens"] >= 2
+    asyncio.run(rt.stop())
+    server.shutdown()
