@@ -320,3 +320,37 @@ def test_engine_gpu_moe():
     eng2 = LLMEngine(EngineConfig(model="smoke-moe", device="cuda", seed=9,
                                   kv_cache_pages=64))
     assert eng2.generate([[5, 6, 7, 8] * 10, [9] * 17], p) == outs
+
+
+def test_engine_gpu_multi_model_cohost():
+    """Two in-process engines co-host on one GPU (per-purpose aux models,
+    service/runtime.py engine_for): both must generate."""
+    import asyncio
+    from aurora_amd.engine.engine import EngineConfig
+    from aurora_amd.engine.server import AsyncEngineServer
+    from aurora_amd.engine.sampling_params import SamplingParams
+    from aurora_amd.service.runtime import Runtime
+    from aurora_amd.utils.config import Settings
+    server = AsyncEngineServer(EngineConfig(
+        model="smoke-llama", device="cuda", seed=1, kv_memory_fraction=0.05,
+        use_hip_graphs=False))
+    rt = Runtime(Settings(db_path=":memory:", secrets_path="/tmp/mm_sec.json",
+                          main_model="smoke-llama",
+                          summarization_model="tiny-llama",
+                          engine_device="cuda"),
+                 engine_server=server)
+    try:
+        org = rt.admin.create_org("mm")
+        llm = rt.llm(org)
+        p = SamplingParams(temperature=0.0, max_tokens=4)
+
+        async def go():
+            a = await llm.invoke("alert: db down", p, purpose="main")
+            b = await llm.invoke("summarize: db down", p, purpose="summarization")
+            return a, b
+        a, b = asyncio.run(go())
+        assert len(a.token_ids) == 4 and len(b.token_ids) == 4
+        assert rt._aux_engines["tiny-llama"].engine.device == "cuda"
+    finally:
+        asyncio.run(rt.stop())
+        server.shutdown()
