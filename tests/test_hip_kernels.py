@@ -336,7 +336,7 @@ def test_engine_gpu_multi_model_cohost():
         use_hip_graphs=False))
     rt = Runtime(Settings(db_path=":memory:", secrets_path="/tmp/mm_sec.json",
                           main_model="smoke-llama",
-                          summarization_model="tiny-llama",
+                          summarization_model="smoke-moe",
                           engine_device="cuda"),
                  engine_server=server)
     try:
@@ -350,7 +350,7 @@ def test_engine_gpu_multi_model_cohost():
             return a, b
         a, b = asyncio.run(go())
         assert len(a.token_ids) == 4 and len(b.token_ids) == 4
-        assert rt._aux_engines["tiny-llama"].engine.device == "cuda"
+        assert rt._aux_engines["smoke-moe"].engine.device == "cuda"
     finally:
         asyncio.run(rt.stop())
         server.shutdown()
