@@ -342,14 +342,14 @@ def test_engine_gpu_multi_model_cohost():
     try:
         org = rt.admin.create_org("mm")
         llm = rt.llm(org)
-        p = SamplingParams(temperature=0.0, max_tokens=4)
+        p = SamplingParams(temperature=0.0, max_tokens=4, ignore_eos=True)
 
         async def go():
             a = await llm.invoke("alert: db down", p, purpose="main")
             b = await llm.invoke("summarize: db down", p, purpose="summarization")
             return a, b
         a, b = asyncio.run(go())
-        assert len(a.token_ids) == 4 and len(b.token_ids) == 4
+        assert len(a.token_ids) == 4 and len(b.token_ids) == 4, (a, b)
         assert rt._aux_engines["smoke-moe"].engine.device == "cuda"
     finally:
         asyncio.run(rt.stop())
