@@ -10,7 +10,11 @@ from typing import List, Tuple
 _PATTERNS: List[Tuple[str, re.Pattern]] = [
     ("aws_access_key", re.compile(r"\b(AKIA|ASIA)[0-9A-Z]{16}\b")),
     ("aws_secret_key", re.compile(r"(?i)aws_secret_access_key[\"'\s:=]+([A-Za-z0-9/+=]{40})")),
-    ("private_key", re.compile(r"-----BEGIN [A-Z ]*PRIVATE KEY-----[\s\S]+?-----END [A-Z ]*PRIVATE KEY-----")),
+    # redact to END marker or end-of-text: tool-output truncation can cut
+    # the END line and must not leak the key body
+    ("private_key", re.compile(
+        r"-----BEGIN [A-Z ]*PRIVATE KEY-----"
+        r"[\s\S]*?(?:-----END [A-Z ]*PRIVATE KEY-----|\Z)")),
     ("github_token", re.compile(r"\b(ghp|gho|ghu|ghs|ghr)_[A-Za-z0-9]{36,}\b")),
     ("slack_token", re.compile(r"\bxox[baprs]-[A-Za-z0-9-]{10,}\b")),
     ("gcp_api_key", re.compile(r"\bAIza[0-9A-Za-z_\-]{35}\b")),
