@@ -157,6 +157,12 @@ class TaskQueue:
 
     def start(self) -> None:
         loop = asyncio.get_running_loop()
+        # acks-late redelivery: rows left 'running' by a previous process
+        # (crash mid-task) are requeued — the handler never acked, so it
+        # must run again (Celery acks_late + visibility-timeout analog,
+        # reference: celery_config.py task_acks_late)
+        self.db.execute(
+            "UPDATE task_queue SET status='pending' WHERE status='running'")
         self._stop.clear()
         self._workers = [loop.create_task(self._worker())
                          for _ in range(self.concurrency)]

@@ -336,3 +336,34 @@ def test_per_purpose_model_routing():
     assert server.engine.counters["generated_tokens"] >= 2
     asyncio.run(rt.stop())
     server.shutdown()
+
+
+def test_queue_crash_redelivery():
+    """Rows stuck in 'running' from a crashed process are requeued when a
+    new queue starts (acks-late redelivery, service/queue.py start)."""
+    import asyncio as aio
+    from aurora_amd.service.queue import TaskQueue
+
+    async def main():
+        db = Database(":memory:")
+        q1 = TaskQueue(db, concurrency=1)
+        ran = []
+
+        async def h(args):
+            ran.append(args["x"])
+            return "ok"
+
+        q1.register("job", h)
+        tid = q1.enqueue("job", {"x": 1})
+        # simulate a crash AFTER claim, BEFORE ack: mark running directly
+        db.execute("UPDATE task_queue SET status='running', attempts=1 "
+                   "WHERE id=?", [tid])
+        # a fresh queue (new process) must redeliver it
+        q2 = TaskQueue(db, concurrency=1)
+        q2.register("job", h)
+        q2.start()
+        assert await q2.drain(timeout_s=20)
+        assert ran == [1]
+        assert q2.get_task(tid)["status"] == "done"
+        await q2.stop()
+    aio.run(main())
