@@ -50,3 +50,35 @@ def test_loaded_model_generates_identically(tmp_path):
     e1 = LLMEngine(EngineConfig(model="tiny-llama", device="cpu"), model=src)
     e2 = LLMEngine(EngineConfig(model="tiny-llama", device="cpu"), model=dst)
     assert e1.generate([[4, 5, 6]], p) == e2.generate([[4, 5, 6]], p)
+
+
+def test_corrupt_checkpoint_fails_cleanly(tmp_path):
+    """Corrupted safetensors and shape mismatches must raise a clear
+    error, not load garbage weights silently (models/loader.py)."""
+    import pytest as _pt
+    import torch
+    from aurora_amd.models.config import get_spec
+    from aurora_amd.models.llama import LlamaForCausalLM
+    from aurora_amd.models.loader import load_llama_safetensors, save_llama_safetensors
+
+    spec = get_spec("tiny-llama")
+    m = LlamaForCausalLM(spec, dtype=torch.float32)
+    m.init_random(3)
+    ck = str(tmp_path / "ck")
+    save_llama_safetensors(m, ck)
+
+    # 1. truncated file
+    import glob, os
+    f = glob.glob(os.path.join(ck, "*.safetensors"))[0]
+    data = open(f, "rb").read()
+    open(f, "wb").write(data[: len(data) // 2])
+    m2 = LlamaForCausalLM(spec, dtype=torch.float32)
+    with _pt.raises(Exception):
+        load_llama_safetensors(m2, ck)
+
+    # 2. wrong-architecture checkpoint against a different spec
+    ck2 = str(tmp_path / "ck2")
+    save_llama_safetensors(m, ck2)
+    other = LlamaForCausalLM(get_spec("smoke-llama"), dtype=torch.float32)
+    with _pt.raises(Exception):
+        load_llama_safetensors(other, ck2)
