@@ -442,6 +442,38 @@ def create_app(rt: Runtime) -> FastAPI:
         from .dashboard import DASHBOARD_HTML
         return HTMLResponse(DASHBOARD_HTML)
 
+    @app.get("/metrics/prometheus")
+    async def metrics_prometheus():
+        """Prometheus exposition of engine + queue + incident gauges
+        (fleet-level, unauthenticated like /health; org-level SRE numbers
+        stay on the authed JSON /metrics)."""
+        from fastapi.responses import PlainTextResponse
+        eng = rt.engine.counters() if hasattr(rt.engine, "counters") else {}
+        lines = [
+            "# TYPE aurora_engine_tokens_total counter",
+            f"aurora_engine_tokens_total{{kind=\"prompt\"}} {eng.get('prompt_tokens', 0)}",
+            f"aurora_engine_tokens_total{{kind=\"generated\"}} {eng.get('generated_tokens', 0)}",
+            "# TYPE aurora_engine_steps_total counter",
+            f"aurora_engine_steps_total{{kind=\"prefill\"}} {eng.get('prefill_steps', 0)}",
+            f"aurora_engine_steps_total{{kind=\"decode\"}} {eng.get('decode_steps', 0)}",
+            "# TYPE aurora_engine_preemptions_total counter",
+            f"aurora_engine_preemptions_total {eng.get('preemptions', 0)}",
+            "# TYPE aurora_queue_depth gauge",
+            f"aurora_queue_depth {rt.queue.pending_count()}",
+            "# TYPE aurora_engine_healthy gauge",
+            f"aurora_engine_healthy {int(rt.engine.healthy()) if hasattr(rt.engine, 'healthy') else 1}",
+        ]
+        for org in rt.admin.orgs():
+            st = rt.org(org["id"])
+            by = {}
+            for r in st.list_incidents(limit=500):
+                by[r["aurora_status"]] = by.get(r["aurora_status"], 0) + 1
+            for status, n in sorted(by.items()):
+                lines.append(
+                    f"aurora_incidents{{org=\"{org['id']}\",status=\"{status}\"}} {n}")
+        return PlainTextResponse("\n".join(lines) + "\n",
+                                 media_type="text/plain; version=0.0.4")
+
     @app.get("/health")
     async def health():
         ok = rt.engine.healthy() if hasattr(rt.engine, "healthy") else True

@@ -339,3 +339,13 @@ def test_cross_org_isolation_every_get_route(env):
     assert checked >= 15, checked
     # sanity: org A itself CAN see its data
     assert "SECRET-ALPHA" in client.get(f"/incidents/{iid}", headers=a).text
+
+
+def test_prometheus_metrics(env):
+    rt, client, oid, uid = env
+    r = client.get("/metrics/prometheus")
+    assert r.status_code == 200
+    body = r.text
+    assert "aurora_engine_tokens_total" in body
+    assert "aurora_queue_depth" in body
+    assert "# TYPE" in body
