@@ -204,3 +204,26 @@ def test_moe_batched_matches_loop():
         if tok.numel():
             ref.index_add_(0, tok, m.experts[e](x[tok]) * w[tok, slot, None])
     assert torch.allclose(batched, ref, atol=1e-5), (batched - ref).abs().max()
+
+
+def test_engine_determinism_across_instances():
+    """Same seed + same requests => identical token streams across two
+    fresh engines (reproducibility; CPU)."""
+    from aurora_amd.engine.engine import EngineConfig, LLMEngine
+    from aurora_amd.engine.sampling_params import SamplingParams
+
+    def run():
+        eng = LLMEngine(EngineConfig(model="tiny-llama", device="cpu", seed=42))
+        p = SamplingParams(temperature=0.8, top_p=0.9, max_tokens=12,
+                           ignore_eos=True)
+        for i in range(3):
+            eng.add_request(f"d{i}", [7 + i, 9, 11], p)
+        toks = {f"d{i}": [] for i in range(3)}
+        for _ in range(60):
+            for o in eng.step():
+                toks[o.request_id].extend(o.token_ids)
+            if not eng.running and not eng.waiting:
+                break
+        return toks
+
+    assert run() == run()
