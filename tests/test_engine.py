@@ -221,7 +221,7 @@ def test_engine_determinism_across_instances():
         toks = {f"d{i}": [] for i in range(3)}
         for _ in range(60):
             for o in eng.step():
-                toks[o.request_id].extend(o.token_ids)
+                toks[o.request_id].extend(o.new_token_ids)
             if not eng.running and not eng.waiting:
                 break
         return toks
