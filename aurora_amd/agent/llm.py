@@ -10,6 +10,7 @@ tracked per call into llm_usage (reference: llm.py:270-350).
 """
 from __future__ import annotations
 
+import asyncio
 import time
 from dataclasses import dataclass
 from typing import Optional
@@ -60,7 +61,22 @@ class LLMManager:
         params = params or SamplingParams(temperature=0.2, max_tokens=512)
         t0 = time.monotonic()
         srv = self._server_for(purpose)
-        res = await srv.complete_text(prompt, params, session_id=session_id)
+        # transient-failure retry (reference: agent.py:1031-1047 retries
+        # network errors 3x; here the analog is an engine step fault that
+        # failed the in-flight request while the stepping thread recovered)
+        last: Optional[BaseException] = None
+        for attempt in range(3):
+            try:
+                res = await srv.complete_text(prompt, params,
+                                              session_id=session_id)
+                break
+            except (RuntimeError, TimeoutError) as e:
+                last = e
+                if attempt == 2:
+                    raise
+                await asyncio.sleep(0.5 * (attempt + 1))
+        else:  # pragma: no cover
+            raise last  # type: ignore[misc]
         if self.store is not None:
             self.store.track_usage(
                 model=self.config.resolve(purpose),

@@ -367,3 +367,45 @@ def test_queue_crash_redelivery():
         assert q2.get_task(tid)["status"] == "done"
         await q2.stop()
     aio.run(main())
+
+
+def test_llm_invoke_retries_transient_engine_failures():
+    """LLMManager retries a transient engine fault before failing the
+    turn (reference agent.py:1031-1047 network retry analog)."""
+    import asyncio
+    from aurora_amd.agent.llm import LLMManager
+    from aurora_amd.engine.sampling_params import SamplingParams
+
+    class FlakyServer:
+        def __init__(self):
+            self.calls = 0
+
+        def tokenizer(self):
+            raise NotImplementedError
+
+        async def complete_text(self, prompt, params, session_id=None):
+            self.calls += 1
+            if self.calls < 3:
+                raise RuntimeError("engine step fault")
+            class R:
+                text = "ok"
+                token_ids = [1]
+                prompt_tokens = 2
+            return R()
+
+    srv = FlakyServer()
+    llm = LLMManager(srv)
+    res = asyncio.run(llm.invoke("x", SamplingParams(max_tokens=1)))
+    assert res.text == "ok" and srv.calls == 3
+
+    class DeadServer(FlakyServer):
+        async def complete_text(self, prompt, params, session_id=None):
+            self.calls += 1
+            raise RuntimeError("still down")
+
+    dead = DeadServer()
+    llm2 = LLMManager(dead)
+    import pytest as _pt
+    with _pt.raises(RuntimeError):
+        asyncio.run(llm2.invoke("x", SamplingParams(max_tokens=1)))
+    assert dead.calls == 3
