@@ -30,6 +30,16 @@ class AuroraClient:
         self.http = httpx.Client(base_url=base_url, headers=headers, timeout=30.0)
         self.user_id = user_id
 
+    def _get(self, path: str):
+        r = self.http.get(path)
+        r.raise_for_status()
+        return r.json()
+
+    def _post(self, path: str, payload: dict):
+        r = self.http.post(path, json=payload)
+        r.raise_for_status()
+        return r.json()
+
     # ---- incidents ----
     def incidents(self, status: Optional[str] = None) -> List[dict]:
         return self.http.get("/incidents", params={"status": status}
@@ -54,6 +64,36 @@ class AuroraClient:
 
     def metrics(self) -> dict:
         return self.http.get("/metrics").json()
+
+    def findings(self, iid: str) -> List[dict]:
+        return self._get(f"/incidents/{iid}/findings")
+
+    def citations(self, iid: str) -> List[dict]:
+        return self._get(f"/incidents/{iid}/citations")
+
+    def suggestions(self, iid: str) -> List[dict]:
+        return self._get(f"/incidents/{iid}/suggestions")
+
+    def timeline(self, iid: str) -> List[dict]:
+        return self._get(f"/incidents/{iid}/timeline")
+
+    def thoughts(self, iid: str) -> List[dict]:
+        return self._get(f"/incidents/{iid}/thoughts")
+
+    def cancel_investigation(self, iid: str) -> dict:
+        return self._post(f"/incidents/{iid}/cancel", {})
+
+    def resolve(self, iid: str) -> dict:
+        return self._post(f"/incidents/{iid}/resolve", {})
+
+    def graph_overview(self) -> dict:
+        return self._get("/graph/overview")
+
+    def impact(self, service: str) -> dict:
+        return self._get(f"/graph/impact/{service}")
+
+    def llm_usage(self) -> List[dict]:
+        return self._get("/llm-usage")
 
     def waterfall(self, iid: str) -> dict:
         return self.http.get(f"/monitor/waterfall/{iid}").json()

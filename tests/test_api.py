@@ -341,6 +341,23 @@ def test_cross_org_isolation_every_get_route(env):
     assert "SECRET-ALPHA" in client.get(f"/incidents/{iid}", headers=a).text
 
 
+def test_sdk_extended_surface(env):
+    """New SDK methods map 1:1 onto routes (client.py)."""
+    rt, client, oid, uid = env
+    from aurora_amd.client import AuroraClient
+    sdk = AuroraClient(user_id=uid, org_id=oid)
+    sdk.http = client
+    client.headers.update({"X-User-Id": uid, "X-Org-Id": oid})
+    iid = rt.org(oid).create_incident(title="sdk probe", source="agent",
+                                      payload={})
+    assert sdk.findings(iid) == []
+    assert sdk.citations(iid) == []
+    assert sdk.timeline(iid)  # incident_created event
+    assert isinstance(sdk.graph_overview(), dict)
+    assert isinstance(sdk.llm_usage(), list)
+    assert sdk.resolve(iid)["status"] in ("resolved", "ok")
+
+
 def test_prometheus_metrics(env):
     rt, client, oid, uid = env
     r = client.get("/metrics/prometheus")
