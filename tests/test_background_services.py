@@ -1,0 +1,81 @@
+"""Tests for the lighter background subsystems: prediscovery context,
+PR change-gating normalization/dispatch, and notification dispatch."""
+import asyncio
+
+import pytest
+
+from aurora_amd.engine.engine import EngineConfig
+from aurora_amd.engine.server import AsyncEngineServer
+from aurora_amd.service.runtime import Runtime
+from aurora_amd.utils.config import Settings
+
+
+@pytest.fixture(scope="module")
+def rt():
+    server = AsyncEngineServer(EngineConfig(model="tiny-llama", device="cpu"))
+    r = Runtime(Settings(db_path=":memory:", secrets_path="/tmp/bg_sec.json",
+                         rca_max_turns=1), engine_server=server)
+    yield r
+    server.shutdown()
+
+
+@pytest.fixture(scope="module")
+def org(rt):
+    o = rt.admin.create_org("bg")
+    rt.admin.create_user(o, "bg@bg.io", user_id="bg-user")
+    return o
+
+
+def test_pr_event_normalization():
+    from aurora_amd.service.change_gating import normalize_pr_event
+    gh = normalize_pr_event("github", {
+        "action": "opened",
+        "pull_request": {"number": 7, "title": "fix: retry logic",
+                         "html_url": "https://x/pr/7",
+                         "head": {"ref": "fix"}, "base": {"ref": "main"},
+                         "user": {"login": "dev"}},
+        "repository": {"full_name": "acme/api"}})
+    assert gh and gh["number"] == 7 and gh["repo"] == "acme/api"
+    assert normalize_pr_event("github", {"action": "labeled"}) is None
+    assert normalize_pr_event("github", {"nonsense": True}) is None
+    bb = normalize_pr_event("bitbucket", {
+        "pullrequest": {"id": 3, "title": "chore", "links": {},
+                        "source": {"branch": {"name": "b"}},
+                        "destination": {"branch": {"name": "main"}}},
+        "repository": {"full_name": "acme/infra"}})
+    assert bb and bb["number"] == 3
+
+
+def test_pr_webhook_dispatches_review_task(rt, org):
+    from aurora_amd.service.change_gating import process_pr_webhook
+    tid = process_pr_webhook(rt, org, "github", {
+        "action": "opened",
+        "pull_request": {"number": 11, "title": "risky migration",
+                         "html_url": "https://x/pr/11",
+                         "head": {"ref": "mig"}, "base": {"ref": "main"},
+                         "user": {"login": "dev"}},
+        "repository": {"full_name": "acme/api"}})
+    assert tid
+    # ignored action -> no task
+    assert process_pr_webhook(rt, org, "github", {"action": "labeled"}) is None
+
+
+def test_prediscovery_context_roundtrip(rt, org):
+    from aurora_amd.service.prediscovery import prediscovery_context
+    store = rt.org(org)
+    ctx = prediscovery_context(store)
+    assert isinstance(ctx, str)
+
+
+def test_notification_dispatch_records(rt, org):
+    from aurora_amd.service.notifications import (
+        notify_investigation_completed, notify_investigation_started)
+    store = rt.org(org)
+    iid = store.create_incident(title="notif probe", source="datadog",
+                                payload={})
+    asyncio.run(notify_investigation_started(rt, org, iid))
+    asyncio.run(notify_investigation_completed(rt, org, iid))
+    rows = store.notifications() if hasattr(store, "notifications") else None
+    # at minimum the calls must not raise; if persisted, they reference iid
+    if rows:
+        assert any(iid in str(r) for r in rows)
