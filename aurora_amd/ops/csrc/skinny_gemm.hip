@@ -104,7 +104,8 @@ __global__ __launch_bounds__(256) void skinny_gemm_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// v2: LDS-staged variant for 17 <= M <= 64.  v1's per-lane 16 B fragment
+// v2: LDS-staged variant for M <= 64 (the dispatched decode path).
+// v1's per-lane 16 B fragment
 // loads are row-gathers (consecutive lanes touch lines K*2 bytes apart);
 // at MT=4 that is 5 gathered loads per 32-k step and caps the kernel at
 // ~1.3 TB/s cold (tools/sweep_skinny2.py).  Here a WG cooperatively
