@@ -176,10 +176,19 @@ async def trigger_delayed_rca(rt, args: Dict[str, Any]) -> dict:
         return {"error": "incident gone"}
     if inc.get("aurora_chat_session_id"):
         return {"skipped": "rca already dispatched"}
-    # NOTE: no per-user rate limit here — webhook alert storms are handled
-    # by correlation + the session-link/dedup_key checks above; the per-user
-    # window applies to the agent-initiated trigger_rca TOOL (reference
-    # rate-limits trigger_rca_tool.py, not alert ingestion)
+    # admission control (reference: is_background_chat_allowed,
+    # task.py:336): when too many investigations are already running for
+    # this org, DEFER (re-enqueue with countdown) rather than drop — a
+    # concurrency cap, not the per-user cooldown (that one gates the
+    # agent-initiated trigger_rca TOOL only)
+    running = sum(1 for r in store.list_incidents(limit=500)
+                  if r.get("aurora_status") == "running")
+    if running >= rt.settings.max_concurrent_rca:
+        store.lifecycle(incident_id, "rca_deferred", {"running": running})
+        rt.queue.enqueue("trigger_delayed_rca", dict(args),
+                         countdown_s=30.0,
+                         dedup_key=f"rca-defer:{incident_id}:{running}")
+        return {"deferred": True}
     tid = rt.queue.enqueue(
         "run_background_chat",
         {"org_id": org_id, "incident_id": incident_id,

@@ -54,6 +54,7 @@ class Settings:
     worker_concurrency: int = field(default_factory=lambda: _i("CELERY_CONCURRENCY", 4))
     ws_rate_limit: int = field(default_factory=lambda: _i("WS_RATE_LIMIT", 5))  # msgs / 60 s
     rca_rate_window_s: int = field(default_factory=lambda: _i("RCA_RATE_WINDOW", 300))
+    max_concurrent_rca: int = field(default_factory=lambda: _i("MAX_CONCURRENT_RCA", 256))
     # feature flags (reference: utils/flags/feature_flags.py:6-36)
     flags: dict = field(default_factory=lambda: {
         name: _b(f"FLAG_{name.upper()}", True)

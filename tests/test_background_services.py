@@ -111,3 +111,24 @@ def test_serve_module_builds_app(rt):
     app = create_app(rt)
     paths = {r.path for r in app.routes}
     assert {"/health", "/ui", "/incidents", "/metrics/prometheus"} <= paths
+
+
+def test_rca_admission_defers_over_cap(rt, org):
+    """Over the MAX_CONCURRENT_RCA cap, new webhook RCAs defer (re-enqueue
+    with countdown) instead of dropping (reference
+    is_background_chat_allowed, task.py:336)."""
+    from aurora_amd.service.worker import trigger_delayed_rca
+    store = rt.org(org)
+    old = rt.settings.max_concurrent_rca
+    rt.settings.max_concurrent_rca = 1
+    try:
+        busy = store.create_incident(title="busy", source="datadog", payload={})
+        store.update_incident(busy, aurora_status="running")
+        iid = store.create_incident(title="queued", source="datadog", payload={})
+        out = asyncio.run(trigger_delayed_rca(
+            rt, {"org_id": org, "incident_id": iid}))
+        assert out.get("deferred") is True
+        evs = [e for e in store.lifecycle_events(iid) if e["event"] == "rca_deferred"]
+        assert evs
+    finally:
+        rt.settings.max_concurrent_rca = old
