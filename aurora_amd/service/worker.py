@@ -184,10 +184,16 @@ async def trigger_delayed_rca(rt, args: Dict[str, Any]) -> dict:
     running = sum(1 for r in store.list_incidents(limit=500)
                   if r.get("aurora_status") == "running")
     if running >= rt.settings.max_concurrent_rca:
+        defers = int(args.get("defer_count", 0))
+        if defers >= 60:  # ~30 min of deferral -> give up loudly
+            store.lifecycle(incident_id, "rca_defer_exhausted",
+                            {"running": running})
+            store.update_incident(incident_id, aurora_status="failed")
+            return {"skipped": "defer exhausted"}
         store.lifecycle(incident_id, "rca_deferred", {"running": running})
-        rt.queue.enqueue("trigger_delayed_rca", dict(args),
-                         countdown_s=30.0,
-                         dedup_key=f"rca-defer:{incident_id}:{running}")
+        rt.queue.enqueue("trigger_delayed_rca",
+                         dict(args, defer_count=defers + 1),
+                         countdown_s=30.0)
         return {"deferred": True}
     tid = rt.queue.enqueue(
         "run_background_chat",

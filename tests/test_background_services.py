@@ -130,5 +130,14 @@ def test_rca_admission_defers_over_cap(rt, org):
         assert out.get("deferred") is True
         evs = [e for e in store.lifecycle_events(iid) if e["event"] == "rca_deferred"]
         assert evs
+        # deferral must re-enqueue even when the running count is unchanged
+        out2 = asyncio.run(trigger_delayed_rca(
+            rt, {"org_id": org, "incident_id": iid, "defer_count": 1}))
+        assert out2.get("deferred") is True
+        # exhaustion fails loudly instead of deferring forever
+        out3 = asyncio.run(trigger_delayed_rca(
+            rt, {"org_id": org, "incident_id": iid, "defer_count": 60}))
+        assert out3.get("skipped")
+        assert store.get_incident(iid)["aurora_status"] == "failed"
     finally:
         rt.settings.max_concurrent_rca = old
