@@ -95,14 +95,24 @@ def build_postmortem(incident: dict, summary: str, citations: List[dict],
                      suggestions: List[dict], timeline: List[dict]) -> str:
     """Deterministic postmortem assembly from the investigation record
     (auto-generated postmortem — reference: postmortem_action.py)."""
+    import datetime as _dt
+
+    def _iso(ts) -> str:
+        try:
+            return _dt.datetime.utcfromtimestamp(float(ts)).strftime(
+                "%Y-%m-%d %H:%M:%SZ")
+        except (TypeError, ValueError, OSError, OverflowError):
+            return str(ts)
+
+    status = incident.get("aurora_status") or incident.get("status") or "open"
     lines = [
         f"# Postmortem: {incident['title']}",
         "",
         f"- **Incident**: {incident['id']}",
         f"- **Source**: {incident['source']}",
         f"- **Severity**: {incident.get('severity') or 'unknown'}",
-        f"- **Opened**: {incident['created_at']}",
-        f"- **Status**: {incident['status']}",
+        f"- **Opened**: {_iso(incident['created_at'])}",
+        f"- **Investigation**: {status}",
         "",
         "## Summary",
         summary,
@@ -110,7 +120,7 @@ def build_postmortem(incident: dict, summary: str, citations: List[dict],
         "## Timeline",
     ]
     for ev in timeline:
-        lines.append(f"- {ev['ts']:.0f}: {ev['event']}")
+        lines.append(f"- {_iso(ev['ts'])}: {ev['event']}")
     if citations:
         lines += ["", "## Evidence"]
         for c in citations[:10]:
