@@ -120,7 +120,8 @@ class AsyncEngineServer:
                     self.engine.abort_request(rid)
 
     async def generate(self, prompt_token_ids: List[int], params: SamplingParams,
-                       session_id: Optional[str] = None) -> GenResult:
+                       session_id: Optional[str] = None,
+                       raise_on_fault: bool = True) -> GenResult:
         t0 = time.monotonic()
         res = GenResult(prompt_tokens=len(prompt_token_ids))
         async for out in self.stream(prompt_token_ids, params, session_id):
@@ -129,6 +130,13 @@ class AsyncEngineServer:
             if out.finished:
                 res.finish_reason = out.finish_reason
         res.elapsed_ms = (time.monotonic() - t0) * 1000.0
+        # engine faults must be EXCEPTIONS to await-ers (so LLMManager's
+        # transient-fault retry can fire), not silent empty generations;
+        # "aborted" stays a normal result (deliberate cancellation)
+        fr = res.finish_reason or ""
+        if raise_on_fault and (fr.startswith("engine step failed")
+                               or fr == "engine shutdown"):
+            raise RuntimeError(fr)
         return res
 
     async def complete_text(self, prompt: str, params: SamplingParams,

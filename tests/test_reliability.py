@@ -102,7 +102,12 @@ def test_engine_step_failure_fails_requests_loudly():
     server.engine.step = boom  # fault injection
 
     async def main():
-        res = await server.generate([1, 2, 3], SamplingParams(max_tokens=4))
+        import pytest as _pt
+        with _pt.raises(RuntimeError, match="engine step failed"):
+            await server.generate([1, 2, 3], SamplingParams(max_tokens=4))
+        # opt-out keeps the old inspectable-result behavior
+        res = await server.generate([1, 2, 3], SamplingParams(max_tokens=4),
+                                    raise_on_fault=False)
         assert "engine step failed" in (res.finish_reason or "")
     try:
         asyncio.run(main())
