@@ -111,6 +111,7 @@ class Workflow:
                       + render_history(state.messages) + "\nThought:")
             # --- thought (streamed) ---
             thought_parts: List[str] = []
+            fault: Optional[str] = None
             async for out in self.llm.server.stream(
                     self.llm.tokenizer.encode(prompt, add_bos=True),
                     SamplingParams(temperature=0.3, max_tokens=160,
@@ -120,6 +121,14 @@ class Workflow:
                 if out.new_text:
                     thought_parts.append(out.new_text)
                     yield {"type": "token", "content": out.new_text}
+                if out.finished and out.finish_reason and (
+                        out.finish_reason.startswith("engine step failed")
+                        or out.finish_reason == "engine shutdown"):
+                    fault = out.finish_reason
+            if fault:
+                # an engine fault mid-thought must fail the investigation
+                # (-> queue retry), not spin out max_turns of empty turns
+                raise RuntimeError(fault)
             thought = "".join(thought_parts).strip()
             state.messages.append({"role": "thought", "content": thought})
             yield {"type": "thought", "content": thought}
