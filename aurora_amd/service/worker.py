@@ -114,6 +114,12 @@ async def run_background_chat(rt, args: Dict[str, Any]) -> dict:
                 elif ev["type"] == "error":
                     error = ev["error"]
     except Exception as e:  # noqa: BLE001
+        msg = str(e)
+        if "engine step failed" in msg or "engine shutdown" in msg:
+            # transient engine fault: re-raise so the queue retries with
+            # backoff (semantic failures below stay terminal)
+            store.lifecycle(incident_id, "rca_engine_fault", {"error": msg})
+            raise
         error = f"workflow failure: {e}"
 
     # ---- post-RCA stage ----
