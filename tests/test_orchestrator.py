@@ -90,6 +90,38 @@ def test_orchestrator_multi_agent_flow(env):
     assert len(rows) == 2 and all(r["status"] == "completed" for r in rows)
 
 
+def test_orchestrator_subagent_failure_isolated(env):
+    """One sub-agent raising must not sink the wave: its finding is
+    marked failed, the other sub-agent completes, synthesis still runs
+    (reference: dispatcher pre-emitted findings + failure capture)."""
+    store, iid = env
+    from aurora_amd.agent.orchestrator import Orchestrator
+    from aurora_amd.agent.state import State
+    script = [
+        json.dumps({"mode": "multi_agent",
+                    "roles": ["runbook_lookup", "ticket_history"], "waves": 1}),
+        json.dumps({"tool": "final_answer", "args": {"answer": "ok finding"}}),
+        json.dumps({"tool": "final_answer", "args": {"answer": "other"}}),
+        json.dumps({"decision": "end", "rca": "Root cause: X."}),
+    ]
+    llm, ctx = _mk(store, iid, script)
+    orch = Orchestrator(llm, ctx, sub_agent_turns=2)
+
+    orig = orch._sub_agent
+
+    async def flaky(state, role, wave, finding_id):
+        if role == "ticket_history":
+            raise RuntimeError("sub-agent crashed")
+        return await orig(state, role, wave, finding_id)
+
+    orch._sub_agent = flaky
+    state = State(question="alert", incident_id=iid)
+    res = asyncio.run(orch.run(state))
+    assert "Root cause" in res.final
+    by_role = {f["role"]: f for f in res.findings}
+    assert "failed" in by_role["ticket_history"]["content"]
+
+
 def test_orchestrator_direct_path(env):
     store, iid = env
     from aurora_amd.agent.orchestrator import Orchestrator
