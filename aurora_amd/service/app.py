@@ -165,9 +165,9 @@ def create_app(rt: Runtime) -> FastAPI:
     # ------------------------------------------------------------ incidents
 
     @app.get("/incidents")
-    async def list_incidents(status: Optional[str] = None,
+    async def list_incidents(status: Optional[str] = None, limit: int = 100,
                              store: OrgStore = Depends(auth)):
-        return store.list_incidents(status)
+        return store.list_incidents(status, limit=max(1, min(limit, 1000)))
 
     @app.get("/incidents/{iid}")
     async def get_incident(iid: str, store: OrgStore = Depends(auth)):
