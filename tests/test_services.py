@@ -409,3 +409,30 @@ def test_llm_invoke_retries_transient_engine_failures():
     with _pt.raises(RuntimeError):
         asyncio.run(llm2.invoke("x", SamplingParams(max_tokens=1)))
     assert dead.calls == 3
+
+
+def test_db_concurrent_writers():
+    """SQLite under the multi-worker runtime: concurrent writers from
+    threads must not raise 'database is locked' (service/store.py
+    Database serialization)."""
+    import threading
+    from aurora_amd.service.store import AdminStore, Database
+    db = Database(":memory:")
+    admin = AdminStore(db)
+    store = admin.org_store(admin.create_org("conc"))
+    errs = []
+
+    def writer(n):
+        try:
+            for i in range(40):
+                store.create_incident(title=f"w{n}-{i}", source="datadog",
+                                      payload={"i": i})
+                store.audit(f"evt-{n}", str(i))
+        except Exception as e:  # noqa: BLE001
+            errs.append(e)
+
+    ts = [threading.Thread(target=writer, args=(n,)) for n in range(6)]
+    [t.start() for t in ts]
+    [t.join(timeout=60) for t in ts]
+    assert not errs, errs
+    assert len(store.list_incidents(limit=1000)) == 240
