@@ -18,6 +18,8 @@ Design (MI355X-first):
 """
 from __future__ import annotations
 
+import dataclasses
+import itertools
 import time
 from collections import OrderedDict, deque
 from dataclasses import dataclass
@@ -63,10 +65,15 @@ class StepOutput:
     finish_reason: Optional[str] = None
 
 
+_SEQ_SERIAL = itertools.count()
+
+
 class Sequence:
     def __init__(self, req_id: str, token_ids: List[int], params: SamplingParams,
                  session_id: Optional[str] = None):
         self.req_id = req_id
+        # unique per Sequence OBJECT (req_id strings may be reused by callers)
+        self.serial = next(_SEQ_SERIAL)
         self.session_id = session_id
         self.token_ids: List[int] = list(token_ids)
         self.num_prompt = len(token_ids)
@@ -79,6 +86,10 @@ class Sequence:
         self.grammar = params.grammar.fork() if params.grammar is not None else None
         self.arrival = time.monotonic()
         self.first_token_time: Optional[float] = None
+        # bumped whenever this seq's pages are freed (preemption/session evict)
+        # so graph block-table staging keyed on (req_id, gen, npages) can never
+        # reuse a stale row after the free list hands back different page ids
+        self.page_gen = 0
 
     @property
     def num_tokens(self) -> int:
@@ -186,6 +197,11 @@ class LLMEngine:
             raise ValueError(f"prompt token id outside vocab [0, {v})")
         if len(prompt_token_ids) >= self.max_model_len:
             prompt_token_ids = prompt_token_ids[-(self.max_model_len - params.max_tokens - 1):]
+        # clamp so prompt + output can never exceed max_model_len (the
+        # per-step max_len check is the backstop; this keeps max_tokens honest)
+        room = self.max_model_len - len(prompt_token_ids)
+        if params.max_tokens > room:
+            params = dataclasses.replace(params, max_tokens=max(room, 1))
         seq = Sequence(request_id, prompt_token_ids, params, session_id)
         if session_id is not None and session_id in self.sessions:
             cached = self.sessions.pop(session_id)
@@ -252,6 +268,7 @@ class LLMEngine:
             self.kv.free(victim.pages)
             victim.pages = []
             victim.num_cached = 0
+            victim.page_gen += 1
             victim.status = WAITING
             self.waiting.appendleft(victim)
             self.counters["preemptions"] += 1
@@ -532,7 +549,7 @@ class LLMEngine:
         ids_t[:B] = torch.tensor([s.token_ids[-1] for s in seqs],
                                  dtype=torch.int64)
         pos = self._host_i32
-        batch_ids = [s.req_id for s in seqs]
+        batch_ids = [(s.serial, s.page_gen) for s in seqs]
         if batch_ids == self._dec_staged_ids:
             pos[:B] += 1  # position, slot and context each advance one token
             for i, s in enumerate(seqs):
@@ -557,7 +574,7 @@ class LLMEngine:
         bt_host = self._host_bt
         staged = self._bt_staged
         for i, s in enumerate(seqs):
-            key = (s.req_id, len(s.pages))
+            key = (s.serial, s.page_gen, len(s.pages))
             if staged[i] != key:
                 n = min(len(s.pages), mp)
                 bt_host[i, :n] = torch.tensor(s.pages[:n], dtype=torch.int32)
@@ -733,14 +750,16 @@ class LLMEngine:
                 finish = "grammar"
             elif seq.num_output >= p.max_tokens:
                 finish = "max_tokens"
+            elif seq.num_tokens >= self.max_model_len:
+                # checked BEFORE stop strings so it can never be shadowed:
+                # growing past max_position would index RoPE tables OOB
+                finish = "max_len"
             elif p.stop:
                 tail = seq.out_bytes[-256:].decode("utf-8", errors="ignore")
                 for s in p.stop:
                     if s in tail:
                         finish = "stop"
                         break
-            elif seq.num_tokens >= self.max_model_len:
-                finish = "max_len"
             if finish:
                 self._finish(seq, finish)
             outputs.append(StepOutput(seq.req_id, [tok], new_text, bool(finish), finish))

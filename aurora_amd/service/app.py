@@ -21,6 +21,7 @@ from typing import Optional
 from fastapi import Depends, FastAPI, Header, HTTPException, Request
 from fastapi.responses import StreamingResponse
 
+from ..utils.config import settings
 from .runtime import Runtime
 from .store import OrgStore
 from .webhooks import process_webhook, sources, verify_signature
@@ -102,7 +103,12 @@ def create_app(rt: Runtime) -> FastAPI:
         body = await request.body()
         store = rt.org(org_id)
         secret = store.get_token_ref("system", f"webhook_secret:{source}")
-        if not verify_signature(source, body, dict(request.headers), secret):
+        if not secret:
+            if settings().require_webhook_secrets:
+                store.audit("webhook_rejected_unsigned", source)
+                raise HTTPException(403, "webhook secret required")
+            store.audit("webhook_accepted_unsigned", source)
+        elif not verify_signature(source, body, dict(request.headers), secret):
             store.audit("webhook_bad_signature", source)
             raise HTTPException(403, "bad signature")
         try:
@@ -128,7 +134,12 @@ def create_app(rt: Runtime) -> FastAPI:
         store = rt.org(org_id)
         secret = store.get_token_ref("system", f"webhook_secret:{provider}")
         from .webhooks import verify_signature as _vs
-        if not _vs(provider, body, dict(request.headers), secret):
+        if not secret:
+            if settings().require_webhook_secrets:
+                store.audit("webhook_rejected_unsigned", provider)
+                raise HTTPException(403, "webhook secret required")
+            store.audit("webhook_accepted_unsigned", provider)
+        elif not _vs(provider, body, dict(request.headers), secret):
             raise HTTPException(403, "bad signature")
         from .change_gating import process_pr_webhook
         try:
@@ -198,7 +209,9 @@ def create_app(rt: Runtime) -> FastAPI:
         return {"task_id": tid}
 
     @app.post("/incidents/{iid}/cancel")
-    async def cancel_rca(iid: str, store: OrgStore = Depends(auth)):
+    async def cancel_rca(iid: str, store: OrgStore = Depends(auth),
+                         x_user_id: Optional[str] = Header(None)):
+        _require("write", x_user_id)
         inc = store.get_incident(iid)
         if inc is None:
             raise HTTPException(404, "no such incident")
@@ -206,7 +219,9 @@ def create_app(rt: Runtime) -> FastAPI:
         return {"cancelled": ok}
 
     @app.post("/incidents/{iid}/resolve")
-    async def resolve(iid: str, store: OrgStore = Depends(auth)):
+    async def resolve(iid: str, store: OrgStore = Depends(auth),
+                      x_user_id: Optional[str] = Header(None)):
+        _require("write", x_user_id)
         store.update_incident(iid, status="resolved", resolved_at=time.time())
         store.lifecycle(iid, "resolved", {})
         return {"ok": True}
@@ -306,7 +321,9 @@ def create_app(rt: Runtime) -> FastAPI:
         return pm
 
     @app.put("/postmortems/{iid}")
-    async def put_pm(iid: str, body: dict, store: OrgStore = Depends(auth)):
+    async def put_pm(iid: str, body: dict, store: OrgStore = Depends(auth),
+                     x_user_id: Optional[str] = Header(None)):
+        _require("write", x_user_id)
         return {"id": store.save_postmortem(iid, body.get("content", ""))}
 
     @app.get("/artifacts")
@@ -345,7 +362,9 @@ def create_app(rt: Runtime) -> FastAPI:
 
     # ---- knowledge base ----
     @app.post("/knowledge-base/documents")
-    async def kb_upload(body: dict, store: OrgStore = Depends(auth)):
+    async def kb_upload(body: dict, store: OrgStore = Depends(auth),
+                        x_user_id: Optional[str] = Header(None)):
+        _require("write", x_user_id)
         name = body.get("name", "untitled")
         content = body.get("content", "")
         did = store.add_kb_document(name, content)
@@ -377,7 +396,9 @@ def create_app(rt: Runtime) -> FastAPI:
         return rt.graph(store.org_id).overview()
 
     @app.post("/graph/services")
-    async def graph_upsert(body: dict, store: OrgStore = Depends(auth)):
+    async def graph_upsert(body: dict, store: OrgStore = Depends(auth),
+                           x_user_id: Optional[str] = Header(None)):
+        _require("write", x_user_id)
         g = rt.graph(store.org_id)
         g.upsert_service(body["name"], kind=body.get("kind", "service"))
         for dep in body.get("depends_on", []):

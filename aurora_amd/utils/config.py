@@ -52,6 +52,9 @@ class Settings:
     ws_port: int = field(default_factory=lambda: _i("AURORA_WS_PORT", 5006))
     internal_api_secret: str = field(default_factory=lambda: os.environ.get("INTERNAL_API_SECRET", ""))
     worker_concurrency: int = field(default_factory=lambda: _i("CELERY_CONCURRENCY", 4))
+    # production deployments set this to reject webhooks with no configured
+    # secret instead of merely auditing the unsigned acceptance
+    require_webhook_secrets: bool = field(default_factory=lambda: _b("AURORA_REQUIRE_WEBHOOK_SECRETS", False))
     ws_rate_limit: int = field(default_factory=lambda: _i("WS_RATE_LIMIT", 5))  # msgs / 60 s
     rca_rate_window_s: int = field(default_factory=lambda: _i("RCA_RATE_WINDOW", 300))
     max_concurrent_rca: int = field(default_factory=lambda: _i("MAX_CONCURRENT_RCA", 256))

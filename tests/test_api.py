@@ -366,3 +366,42 @@ def test_prometheus_metrics(env):
     assert "aurora_engine_tokens_total" in body
     assert "aurora_queue_depth" in body
     assert "# TYPE" in body
+
+
+def test_viewer_cannot_mutate_state(env):
+    """Every state-changing route requires the write/admin verb (ADVICE r1:
+    resolve, postmortem PUT, KB upload, graph upsert, RCA cancel were open
+    to viewers)."""
+    rt, client, oid, uid = env
+    rt.admin.create_user(oid, "viewer2@acme.io", role="viewer", user_id="user-v2")
+    V = {"X-User-Id": "user-v2"}
+    iid = rt.org(oid).create_incident(title="ro probe", source="agent", payload={})
+    assert client.post(f"/incidents/{iid}/resolve", headers=V).status_code == 403
+    assert client.post(f"/incidents/{iid}/cancel", headers=V).status_code == 403
+    assert client.put(f"/postmortems/{iid}", headers=V,
+                      json={"content": "overwrite"}).status_code == 403
+    assert client.post("/knowledge-base/documents", headers=V,
+                       json={"name": "x", "content": "poison"}).status_code == 403
+    assert client.post("/graph/services", headers=V,
+                       json={"name": "evil-svc"}).status_code == 403
+    # viewer can still read all of these surfaces
+    assert client.get("/graph/services", headers=V).status_code == 200
+    assert client.get("/knowledge-base/documents", headers=V).status_code == 200
+
+
+def test_unsigned_webhook_audited_and_gatable(env, monkeypatch):
+    """No-secret webhook acceptance is audited; AURORA_REQUIRE_WEBHOOK_SECRETS
+    rejects it."""
+    rt, client, oid, uid = env
+    payload = {"event": {"data": {"title": "unsigned alert", "id": "u-1"}}}
+    r = client.post(f"/webhook/pagerduty/{oid}", json=payload)
+    assert r.status_code == 200
+    events = [e for e in rt.org(oid).audit_events()
+              if e["kind"] == "webhook_accepted_unsigned"]
+    assert events, "unsigned acceptance must be audited"
+    # flip the production flag: unsigned webhooks now fail closed
+    from aurora_amd.utils import config as cfgmod
+    monkeypatch.setattr(cfgmod.settings(), "require_webhook_secrets", True)
+    r = client.post(f"/webhook/pagerduty/{oid}", json=payload)
+    assert r.status_code == 403
+    monkeypatch.setattr(cfgmod.settings(), "require_webhook_secrets", False)

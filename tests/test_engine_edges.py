@@ -62,3 +62,47 @@ def test_engine_rejects_bad_token_ids(eng):
     with _pt.raises(ValueError):
         eng.add_request("bad2", [],
                         SamplingParams(temperature=0.0, max_tokens=1))
+
+
+def test_max_len_not_shadowed_by_stop_strings():
+    """With stop strings set (the agent workflow always sets them) a sequence
+    must still terminate at max_model_len (the 'elif p.stop' branch
+    shadowed the max_len check, letting seqs grow past max_position)."""
+    e = LLMEngine(EngineConfig(model="tiny-llama", device="cpu", seed=0,
+                               max_model_len=32))
+    # stop string that never occurs in output; generous max_tokens
+    e.add_request("ml-1", list(range(5, 25)),  # 20-token prompt, room = 12
+                  SamplingParams(temperature=0.0, max_tokens=10_000,
+                                 stop=["NEVER-EMITTED-STOP"], ignore_eos=True))
+    outs = _drain(e, max_steps=200)
+    assert outs and outs[-1].finished
+    assert outs[-1].finish_reason in ("max_len", "max_tokens")
+    seq_tokens = 20 + sum(len(o.new_token_ids) for o in outs)
+    assert seq_tokens <= 32
+
+
+def test_preemption_bumps_page_generation():
+    """Preempting a sequence must bump page_gen so graph block-table staging
+    keyed on (serial, gen, npages) can never reuse a stale row (ADVICE r1
+    high: silent KV corruption after preempt+resume)."""
+    e = LLMEngine(EngineConfig(model="tiny-llama", device="cpu", seed=0))
+    e.add_request("pg-1", [5, 6, 7], SamplingParams(max_tokens=4))
+    e.step()  # prefill; pg-1 now running with pages
+    seq = e.seqs["pg-1"]
+    assert seq.pages and seq.page_gen == 0
+    assert e._preempt_lowest()
+    assert seq.page_gen == 1 and not seq.pages
+    # serials are unique even for a reused req_id
+    e2_ids = set()
+    for rid in ("dup", "dup2"):
+        e.add_request(rid, [5], SamplingParams(max_tokens=1))
+        e2_ids.add(e.seqs[rid].serial)
+    assert len(e2_ids) == 2
+
+
+def test_max_tokens_clamped_to_model_len_at_admission():
+    e = LLMEngine(EngineConfig(model="tiny-llama", device="cpu", seed=0,
+                               max_model_len=64))
+    e.add_request("clamp-1", list(range(5, 35)),  # 30-token prompt
+                  SamplingParams(max_tokens=10_000))
+    assert e.seqs["clamp-1"].params.max_tokens <= 34
