@@ -19,7 +19,11 @@ from .service.ws import mask_frame, read_frame
 class AuroraClient:
     def __init__(self, base_url: str = "http://127.0.0.1:5080",
                  user_id: str = "", org_id: str = "",
-                 internal_secret: str = ""):
+                 internal_secret: str = "", auth_token: str = "",
+                 signing_secret: str = ""):
+        """``signing_secret`` (= the server's INTERNAL_API_SECRET) lets a
+        trusted caller mint its own HMAC identity token for ``user_id``;
+        ``auth_token`` supplies a pre-issued one (POST /auth/token)."""
         headers = {}
         if user_id:
             headers["X-User-Id"] = user_id
@@ -27,6 +31,11 @@ class AuroraClient:
             headers["X-Org-Id"] = org_id
         if internal_secret:
             headers["X-Internal-Api-Secret"] = internal_secret
+        if not auth_token and signing_secret and user_id:
+            from .utils.auth_tokens import mint_token
+            auth_token = mint_token(user_id, signing_secret, ttl_s=3600)
+        if auth_token:
+            headers["Authorization"] = f"Bearer {auth_token}"
         self.http = httpx.Client(base_url=base_url, headers=headers, timeout=30.0)
         self.user_id = user_id
 

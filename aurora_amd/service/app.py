@@ -14,6 +14,8 @@ calls (AGENTS.md:96).
 from __future__ import annotations
 
 import asyncio
+import contextvars
+import hmac
 import json
 import time
 from typing import Optional
@@ -21,6 +23,7 @@ from typing import Optional
 from fastapi import Depends, FastAPI, Header, HTTPException, Request
 from fastapi.responses import StreamingResponse
 
+from ..utils.auth_tokens import mint_token, verify_token
 from ..utils.config import settings
 from .runtime import Runtime
 from .store import OrgStore
@@ -48,23 +51,55 @@ def create_app(rt: Runtime) -> FastAPI:
         "admin": {"read", "write", "investigate", "connect", "admin"},
     }
 
-    def _require(verb: str, x_user_id: Optional[str]) -> None:
-        if not x_user_id:
+    # identity resolved by auth() for the current request: "__internal__"
+    # for service-secret calls, else the authenticated user id. _require
+    # checks THIS, never a client-supplied header, so omitting X-User-Id
+    # can't skip RBAC in token mode.
+    _ident: contextvars.ContextVar[Optional[str]] = \
+        contextvars.ContextVar("auth_ident", default=None)
+
+    def _require(verb: str, x_user_id: Optional[str] = None) -> None:
+        ident = _ident.get()
+        if ident == "__internal__":
             return  # internal-secret calls carry full permissions
-        user = rt.admin.get_user(x_user_id)
+        uid = ident or x_user_id
+        if not uid:
+            raise HTTPException(401, "authentication required")
+        user = rt.admin.get_user(uid)
         role = (user or {}).get("role", "viewer")
         if verb not in ROLE_PERMS.get(role, {"read"}):
             raise HTTPException(403, f"role '{role}' lacks '{verb}' permission")
 
-    def auth(x_user_id: Optional[str] = Header(None),
-             x_org_id: Optional[str] = Header(None),
-             x_internal_api_secret: Optional[str] = Header(None)) -> OrgStore:
+    async def auth(x_user_id: Optional[str] = Header(None),
+                   x_org_id: Optional[str] = Header(None),
+                   x_internal_api_secret: Optional[str] = Header(None),
+                   authorization: Optional[str] = Header(None)) -> OrgStore:
+        # async so the ContextVar set here is visible to the route handler
+        # (sync dependencies run in a copied threadpool context)
         s = rt.settings
-        if s.internal_api_secret and x_internal_api_secret == s.internal_api_secret:
+        if s.internal_api_secret and x_internal_api_secret is not None:
+            if not hmac.compare_digest(x_internal_api_secret,
+                                       s.internal_api_secret):
+                raise HTTPException(401, "bad internal API secret")
             if x_org_id:
+                _ident.set("__internal__")
                 return rt.org(x_org_id)
             raise HTTPException(400, "internal call requires X-Org-Id")
-        if not x_user_id:
+        if s.internal_api_secret:
+            # authenticated mode: identity comes ONLY from an HMAC-signed
+            # bearer token (reference signs JWTs, main_chatbot.py:107-158);
+            # a bare X-User-Id is never trusted when a secret is configured
+            # (identity-header anti-spoofing).
+            tok = (authorization or "")
+            if tok.lower().startswith("bearer "):
+                tok = tok[7:].strip()
+            claims = verify_token(tok, s.internal_api_secret)
+            if claims is None:
+                raise HTTPException(401, "valid bearer token required")
+            if x_user_id and x_user_id != claims["user_id"]:
+                raise HTTPException(401, "X-User-Id does not match token")
+            x_user_id = claims["user_id"]
+        elif not x_user_id:
             raise HTTPException(401, "X-User-Id required")
         user = rt.admin.get_user(x_user_id)
         if user is None:
@@ -73,7 +108,27 @@ def create_app(rt: Runtime) -> FastAPI:
         # org id that mismatches the user's org is rejected, not trusted
         if x_org_id and x_org_id != user["org_id"]:
             raise HTTPException(403, "user is not a member of that org")
+        _ident.set(x_user_id)
         return rt.org(user["org_id"])
+
+    @app.post("/auth/token")
+    async def issue_token(body: dict,
+                          x_internal_api_secret: Optional[str] = Header(None)):
+        """Mint a short-lived signed identity token for a user. Only the
+        trusted frontend proxy (which holds INTERNAL_API_SECRET, the
+        reference's service-to-service trust anchor, AGENTS.md:96) may mint."""
+        s = rt.settings
+        if not s.internal_api_secret:
+            raise HTTPException(404, "token auth not enabled")
+        if not (x_internal_api_secret and hmac.compare_digest(
+                x_internal_api_secret, s.internal_api_secret)):
+            raise HTTPException(401, "internal API secret required")
+        uid = body.get("user_id", "")
+        if rt.admin.get_user(uid) is None:
+            raise HTTPException(404, "unknown user")
+        ttl = min(int(body.get("ttl_s", 600)), 86400)
+        return {"token": mint_token(uid, s.internal_api_secret, ttl_s=ttl),
+                "ttl_s": ttl}
 
     # ------------------------------------------------------------- webhooks
 

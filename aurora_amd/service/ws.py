@@ -140,7 +140,9 @@ class ChatServer:
     def __init__(self, rt):
         self.rt = rt
         self.rate = RateLimiter(rt.settings.ws_rate_limit)
-        self._seen_tokens: Dict[str, float] = {}  # jti replay cache (Redis analog)
+        self._seen_tokens: Dict[str, float] = {}  # dev-mode nonce replay cache
+        from ..utils.auth_tokens import ReplayCache
+        self._replay = ReplayCache()  # signed-token jti replay (Redis analog)
         self._server: Optional[asyncio.AbstractServer] = None
         self.port: Optional[int] = None
 
@@ -159,8 +161,20 @@ class ChatServer:
         token = query.get("token", "")
         if not token:
             return None
-        # token format "<user_id>.<nonce>" — the stateless-auth analog; the
-        # nonce is the jti replay guard (reference: main_chatbot.py:107-158)
+        secret = self.rt.settings.internal_api_secret
+        if secret:
+            # authenticated mode: HMAC-signed one-shot token, jti replay
+            # rejected (reference: signed JWT + Redis jti,
+            # main_chatbot.py:107-158). Forged/expired/replayed => refuse.
+            from ..utils.auth_tokens import verify_token
+            claims = verify_token(token, secret)
+            if claims is None:
+                return None
+            if not self._replay.check_and_store(claims["jti"]):
+                return None
+            return self.rt.admin.get_user(claims["user_id"])
+        # dev mode (no secret configured): "<user_id>.<nonce>" identity
+        # assertion with nonce replay guard
         parts = token.split(".", 1)
         user = self.rt.admin.get_user(parts[0])
         if user is None:
