@@ -1,21 +1,73 @@
 """Pluggable secrets backend (reference: utils/secrets/ — Vault KV v2 /
 AWS SM behind SECRETS_BACKEND, ref format 'vault:kv/data/aurora/users/{n}').
 
-Backends here: 'file' (encrypted-at-rest JSON using an XOR keystream keyed
-from AURORA_SECRET_KEY — no cryptography dep guaranteed in the image) and
-'env'.  The store keeps REFERENCES (provider+user), never raw tokens, same
-as the reference ('never log tokens', AGENTS.md:91)."""
+Backends here: 'file' (authenticated-encrypted JSON at rest) and 'env'.
+The store keeps REFERENCES (provider+user), never raw tokens, same as the
+reference ('never log tokens', AGENTS.md:91).
+
+At-rest construction (no third-party crypto in the image, stdlib only):
+encrypt-then-MAC from HMAC-SHA256 primitives —
+
+* per-entry random 16-byte salt;
+* enc_key = HMAC(master, b"enc"|salt), mac_key = HMAC(master, b"mac"|salt)
+  (an HKDF-extract-style domain separation);
+* keystream = HMAC(enc_key, counter_le64) blocks — HMAC as a PRF in counter
+  mode, so the pad is unique per entry and per write;
+* tag = HMAC(mac_key, ref|salt|ciphertext), verified constant-time before
+  decryption, binding each ciphertext to its ref (no entry swapping).
+
+This replaces round 1's deterministic XOR keystream, which reused the pad
+across rewrites of the same ref and had no integrity.
+"""
 from __future__ import annotations
 
 import base64
 import hashlib
+import hmac
 import json
 import os
 import threading
 from typing import Dict, Optional
 
 
-def _keystream(key: bytes, n: int) -> bytes:
+def _hmac(key: bytes, msg: bytes) -> bytes:
+    return hmac.new(key, msg, hashlib.sha256).digest()
+
+
+def _keystream(enc_key: bytes, n: int) -> bytes:
+    out = bytearray()
+    counter = 0
+    while len(out) < n:
+        out += _hmac(enc_key, counter.to_bytes(8, "little"))
+        counter += 1
+    return bytes(out[:n])
+
+
+def _seal(master: bytes, ref: str, raw: bytes) -> str:
+    salt = os.urandom(16)
+    enc_key = _hmac(master, b"enc" + salt)
+    mac_key = _hmac(master, b"mac" + salt)
+    ct = bytes(a ^ b for a, b in zip(raw, _keystream(enc_key, len(raw))))
+    tag = _hmac(mac_key, ref.encode() + salt + ct)
+    return "v2:" + base64.b64encode(salt + ct + tag).decode()
+
+
+def _open(master: bytes, ref: str, blob: str) -> Optional[bytes]:
+    try:
+        data = base64.b64decode(blob[3:])
+    except Exception:  # noqa: BLE001
+        return None
+    if len(data) < 48:  # salt(16) + tag(32)
+        return None
+    salt, ct, tag = data[:16], data[16:-32], data[-32:]
+    mac_key = _hmac(master, b"mac" + salt)
+    if not hmac.compare_digest(_hmac(mac_key, ref.encode() + salt + ct), tag):
+        return None  # tampered / wrong key / swapped entry
+    enc_key = _hmac(master, b"enc" + salt)
+    return bytes(a ^ b for a, b in zip(ct, _keystream(enc_key, len(ct))))
+
+
+def _legacy_keystream(key: bytes, n: int) -> bytes:
     out = b""
     counter = 0
     while len(out) < n:
@@ -46,9 +98,7 @@ class FileSecretsBackend(SecretsBackend):
             return json.load(f)
 
     def put(self, ref: str, value: Dict) -> None:
-        raw = json.dumps(value).encode()
-        ks = _keystream(self.key + ref.encode(), len(raw))
-        enc = base64.b64encode(bytes(a ^ b for a, b in zip(raw, ks))).decode()
+        enc = _seal(self.key, ref, json.dumps(value).encode())
         with self._lock:
             data = self._load()
             data[ref] = enc
@@ -62,11 +112,21 @@ class FileSecretsBackend(SecretsBackend):
         enc = data.get(ref)
         if enc is None:
             return None
+        if enc.startswith("v2:"):
+            raw = _open(self.key, ref, enc)
+            if raw is None:
+                return None
+            try:
+                return json.loads(raw)
+            except Exception:  # noqa: BLE001
+                return None
+        # legacy round-1 entries (unauthenticated XOR): readable for
+        # migration; rewritten in v2 on the next put
         raw = base64.b64decode(enc)
-        ks = _keystream(self.key + ref.encode(), len(raw))
+        ks = _legacy_keystream(self.key + ref.encode(), len(raw))
         try:
             return json.loads(bytes(a ^ b for a, b in zip(raw, ks)))
-        except Exception:
+        except Exception:  # noqa: BLE001
             return None
 
 
