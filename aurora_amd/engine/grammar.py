@@ -238,3 +238,91 @@ class JsonGrammar:
 def validate_json_bytes(data: bytes) -> bool:
     g = JsonGrammar(require_object=False, compact=False)
     return g.advance(data) and g.is_complete()
+
+
+class ToolCallGrammar:
+    """Constrains decoding to exactly ``{"tool":"<name>","args":{...}}``
+    with ``<name>`` drawn from a fixed set — the in-engine analog of the
+    reference's forced tool choice (reference: middleware/force_tool.py,
+    used agent.py:615-622) plus the provider SDK's schema enforcement.
+
+    Byte-level FSM: literal prefix -> trie over tool names -> literal
+    ``","args":`` -> a nested :class:`JsonGrammar` object -> closing ``}``.
+    Same interface as JsonGrammar (fork/advance/is_complete), so the
+    engine's guided sampler uses it unchanged. With this grammar a model
+    can never emit an unknown tool or malformed call — one sampling
+    attempt always parses.
+    """
+
+    _PREFIX = b'{"tool":"'
+    _MID = b'","args":'
+
+    def __init__(self, tools: List[str], max_args_depth: int = 8):
+        if not tools:
+            raise ValueError("ToolCallGrammar needs at least one tool name")
+        self.names = sorted(set(t.encode() for t in tools))
+        self.phase = 0        # 0 prefix, 1 name, 2 mid, 3 args, 4 done
+        self.lit_pos = 0
+        self.name_prefix = b""
+        self.max_args_depth = max_args_depth
+        self.args: JsonGrammar = None  # created lazily at phase 3
+
+    def fork(self) -> "ToolCallGrammar":
+        g = ToolCallGrammar.__new__(ToolCallGrammar)
+        g.names = self.names  # immutable, shared
+        g.phase = self.phase
+        g.lit_pos = self.lit_pos
+        g.name_prefix = self.name_prefix
+        g.max_args_depth = self.max_args_depth
+        g.args = self.args.fork() if self.args is not None else None
+        return g
+
+    def is_complete(self) -> bool:
+        return self.phase == 4
+
+    def tool_name(self) -> str:
+        """The accepted tool name (valid once phase >= 2)."""
+        return self.name_prefix.decode()
+
+    def advance(self, data: bytes) -> bool:
+        for b in data:
+            if not self._advance_byte(b):
+                return False
+        return True
+
+    def _advance_byte(self, b: int) -> bool:
+        if self.phase == 0:
+            if b != self._PREFIX[self.lit_pos]:
+                return False
+            self.lit_pos += 1
+            if self.lit_pos == len(self._PREFIX):
+                self.phase, self.lit_pos = 1, 0
+            return True
+        if self.phase == 1:
+            if b == 0x22:  # '"' closes the name: must match exactly
+                if self.name_prefix in self.names:
+                    self.phase, self.lit_pos = 2, 1  # consumed MID[0]='"'
+                    return True
+                return False
+            cand = self.name_prefix + bytes([b])
+            if any(n.startswith(cand) for n in self.names):
+                self.name_prefix = cand
+                return True
+            return False
+        if self.phase == 2:
+            if b != self._MID[self.lit_pos]:
+                return False
+            self.lit_pos += 1
+            if self.lit_pos == len(self._MID):
+                self.phase = 3
+                self.args = JsonGrammar(max_depth=self.max_args_depth,
+                                        require_object=True, compact=True)
+            return True
+        if self.phase == 3:
+            if self.args.is_complete():
+                if b == 0x7D:  # closing } of the outer object
+                    self.phase = 4
+                    return True
+                return False
+            return self.args._advance_byte(b)
+        return False  # phase 4: nothing after the closing brace

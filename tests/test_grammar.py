@@ -73,3 +73,80 @@ def test_compact_rejects_whitespace():
     g2 = JsonGrammar()
     assert g2.advance(b'{"a b":" c "}')  # ws inside strings still fine
     assert g2.is_complete()
+
+
+# ---------------------------------------------------------- ToolCallGrammar
+
+def test_tool_call_grammar_exact_shape():
+    from aurora_amd.engine.grammar import ToolCallGrammar
+    g = ToolCallGrammar(["kubectl", "final_answer", "knowledge_base_search"])
+    s = b'{"tool":"kubectl","args":{"args":"get pods"}}'
+    assert g.advance(s) and g.is_complete()
+    assert g.tool_name() == "kubectl"
+
+
+def test_tool_call_grammar_rejects_unknown_tool():
+    from aurora_amd.engine.grammar import ToolCallGrammar
+    g = ToolCallGrammar(["kubectl", "final_answer"])
+    assert not g.advance(b'{"tool":"rm_rf"')
+    g = ToolCallGrammar(["kubectl"])
+    # prefix of a valid name is not a valid name
+    assert not g.advance(b'{"tool":"kube"')
+    # name that extends a valid name but matches nothing
+    g = ToolCallGrammar(["kubectl"])
+    assert not g.advance(b'{"tool":"kubectlx')
+
+
+def test_tool_call_grammar_prefix_names():
+    from aurora_amd.engine.grammar import ToolCallGrammar
+    # one tool name is a strict prefix of another: both must be accepted
+    for name in ("get_incident", "get_incident_timeline"):
+        g = ToolCallGrammar(["get_incident", "get_incident_timeline"])
+        s = ('{"tool":"%s","args":{}}' % name).encode()
+        assert g.advance(s) and g.is_complete(), name
+
+
+def test_tool_call_grammar_rejects_malformed():
+    from aurora_amd.engine.grammar import ToolCallGrammar
+    cases = [
+        b'{"tool":"kubectl","args":[]}',      # args must be an object
+        b'{"tool":"kubectl"}',                # missing args
+        b'{"args":{},"tool":"kubectl"}',      # wrong key order (fixed shape)
+        b'{"tool":"kubectl","args":{}}x',     # trailing bytes
+        b'{ "tool":"kubectl","args":{}}',     # whitespace (compact)
+    ]
+    for s in cases:
+        g = ToolCallGrammar(["kubectl"])
+        ok = g.advance(s) and g.is_complete()
+        assert not ok, s
+
+
+def test_tool_call_grammar_fork_isolation():
+    from aurora_amd.engine.grammar import ToolCallGrammar
+    g = ToolCallGrammar(["alpha", "beta"])
+    assert g.advance(b'{"tool":"a')
+    f = g.fork()
+    assert f.advance(b'lpha","args":{}}') and f.is_complete()
+    # original unaffected by the fork's progress
+    assert g.advance(b'lpha","args":{"k":1}}') and g.is_complete()
+
+
+def test_tool_call_grammar_engine_end_to_end():
+    """The CPU engine under ToolCallGrammar always emits a parseable call
+    naming a real tool, even with random weights."""
+    import json as _json
+    from aurora_amd.engine.engine import EngineConfig, LLMEngine
+    from aurora_amd.engine.grammar import ToolCallGrammar
+    from aurora_amd.engine.sampling_params import SamplingParams
+
+    eng = LLMEngine(EngineConfig(model="tiny-llama", device="cpu", seed=7))
+    tools = ["kubectl", "final_answer", "list_incidents"]
+    outs = eng.generate(
+        [[5, 6, 7], [8, 9, 10]],
+        SamplingParams(temperature=0.8, max_tokens=96,
+                       grammar=ToolCallGrammar(tools)))
+    for toks in outs:
+        text = eng.tokenizer.decode(toks)
+        doc = _json.loads(text)
+        assert doc["tool"] in tools
+        assert isinstance(doc["args"], dict)
