@@ -4,16 +4,22 @@ concurrent awaitable generations.
 This is the in-process replacement for the reference's provider HTTP
 boundary *and* its Celery worker concurrency: every concurrent
 investigation/chat turn is just another sequence in the same continuously
-batched engine (SURVEY.md §2.12 row 'Concurrent investigations')."""
+batched engine (SURVEY.md §2.12 row 'Concurrent investigations').
+
+Delivery path: the stepping thread groups each step's outputs by consumer
+event loop and posts ONE ``call_soon_threadsafe`` batch per loop per step
+onto per-request ``asyncio.Queue``s. No executor threads are parked in
+blocking reads, so hundreds of concurrent streams (the flagship B=256
+investigation batch) cost one self-pipe write per step instead of
+saturating the default 32-thread pool."""
 from __future__ import annotations
 
 import asyncio
-import queue as _queue
 import threading
 import time
 import uuid
 from dataclasses import dataclass, field
-from typing import AsyncIterator, Dict, List, Optional
+from typing import AsyncIterator, Dict, List, Optional, Tuple
 
 from .engine import EngineConfig, LLMEngine, StepOutput
 from .sampling_params import SamplingParams
@@ -31,7 +37,8 @@ class GenResult:
 class AsyncEngineServer:
     def __init__(self, config: EngineConfig, engine: Optional[LLMEngine] = None):
         self.engine = engine or LLMEngine(config)
-        self._subs: Dict[str, _queue.SimpleQueue] = {}
+        # rid -> (consumer loop, asyncio queue); mutated only under _lock
+        self._subs: Dict[str, Tuple[asyncio.AbstractEventLoop, asyncio.Queue]] = {}
         self._lock = threading.Lock()
         self._stop = threading.Event()
         self._wake = threading.Event()
@@ -61,18 +68,37 @@ class AsyncEngineServer:
                 err = f"engine step failed: {e}"
                 traceback.print_exc()
                 with self._lock:
-                    for rid in list(self._subs):
+                    rids = list(self._subs)
+                    for rid in rids:
                         try:
                             self.engine.abort_request(rid)
                         except Exception:  # noqa: BLE001
                             pass
-                for rid, q in list(self._subs.items()):
-                    q.put(StepOutput(rid, [], "", True, err))
+                self._deliver([StepOutput(rid, [], "", True, err) for rid in rids])
                 continue
-            for out in outs:
-                q = self._subs.get(out.request_id)
-                if q is not None:
-                    q.put(out)
+            self._deliver(outs)
+        # shutdown: fail any remaining consumers so awaiters wake up
+        with self._lock:
+            rids = list(self._subs)
+        self._deliver([StepOutput(rid, [], "", True, "engine shutdown")
+                       for rid in rids])
+
+    def _deliver(self, outs: List[StepOutput]) -> None:
+        """Post step outputs to their consumers, one loop wakeup per batch."""
+        if not outs:
+            return
+        by_loop: Dict[asyncio.AbstractEventLoop, List[Tuple[asyncio.Queue, StepOutput]]] = {}
+        for out in outs:
+            sub = self._subs.get(out.request_id)
+            if sub is not None:
+                by_loop.setdefault(sub[0], []).append((sub[1], out))
+        for loop, items in by_loop.items():
+            try:
+                loop.call_soon_threadsafe(_post_batch, items)
+            except RuntimeError:
+                # consumer loop already closed; its finally-block cleanup
+                # (or abort) will free the sequences
+                pass
 
     def shutdown(self) -> None:
         self._stop.set()
@@ -88,35 +114,28 @@ class AsyncEngineServer:
                      session_id: Optional[str] = None,
                      request_id: Optional[str] = None) -> AsyncIterator[StepOutput]:
         rid = request_id or f"req-{uuid.uuid4().hex[:12]}"
-        q: _queue.SimpleQueue = _queue.SimpleQueue()
-        self._subs[rid] = q
-        with self._lock:
-            self.engine.add_request(rid, prompt_token_ids, params, session_id=session_id)
-        self._wake.set()
+        if self._stop.is_set():
+            yield StepOutput(rid, [], "", True, "engine shutdown")
+            return
         loop = asyncio.get_running_loop()
+        q: asyncio.Queue = asyncio.Queue()
+        with self._lock:
+            self._subs[rid] = (loop, q)
+            self.engine.add_request(rid, prompt_token_ids, params,
+                                    session_id=session_id)
+        self._wake.set()
         finished = False
-
-        def _get():  # bounded wait so executor threads never leak
-            while True:
-                try:
-                    return q.get(timeout=0.5)
-                except _queue.Empty:
-                    if self._stop.is_set():
-                        return StepOutput(rid, [], "", True, "engine shutdown")
-
         try:
             while True:
-                out = await loop.run_in_executor(None, _get)
+                out = await q.get()
                 yield out
                 if out.finished:
                     finished = True
                     return
         finally:
-            self._subs.pop(rid, None)
-            # release any executor thread still blocked in _get
-            q.put(StepOutput(rid, [], "", True, "stream closed"))
-            if not finished:  # consumer bailed (cancel / WS drop): free the seq
-                with self._lock:
+            with self._lock:
+                self._subs.pop(rid, None)
+                if not finished:  # consumer bailed (cancel / WS drop)
                     self.engine.abort_request(rid)
 
     async def generate(self, prompt_token_ids: List[int], params: SamplingParams,
@@ -148,9 +167,14 @@ class AsyncEngineServer:
     def abort(self, request_id: str) -> None:
         with self._lock:
             self.engine.abort_request(request_id)
-        q = self._subs.pop(request_id, None)
-        if q is not None:
-            q.put(StepOutput(request_id, [], "", True, "aborted"))
+            sub = self._subs.pop(request_id, None)
+        if sub is not None:
+            loop, q = sub
+            try:
+                loop.call_soon_threadsafe(
+                    q.put_nowait, StepOutput(request_id, [], "", True, "aborted"))
+            except RuntimeError:
+                pass
 
     def counters(self) -> Dict[str, int]:
         return dict(self.engine.counters)
@@ -162,3 +186,8 @@ class AsyncEngineServer:
             busy = self.engine.has_work()
         return (not busy) or (time.monotonic() - self._last_progress
                               < self.watchdog_timeout_s)
+
+
+def _post_batch(items: List[Tuple[asyncio.Queue, StepOutput]]) -> None:
+    for q, out in items:
+        q.put_nowait(out)
