@@ -177,6 +177,7 @@ class LLMEngine:
         self.counters = {
             "prompt_tokens": 0, "generated_tokens": 0, "prefill_steps": 0,
             "decode_steps": 0, "preemptions": 0, "grammar_fallbacks": 0,
+            "grammar_bucket_picks": 0,
         }
         # phase timers (seconds, cumulative) for perf triage
         self.timers = {"schedule": 0.0, "build": 0.0, "forward": 0.0,
@@ -647,6 +648,48 @@ class LLMEngine:
             if g.advance(b):
                 seq.grammar = g
                 return tok
+        # byte-constrained pick: when the FSM pins the legal next bytes
+        # (literal phases of ToolCallGrammar, JSON punctuation states),
+        # gather only tokens starting with those bytes — O(bucket) instead
+        # of a full-vocab argsort. With the byte tokenizer a bucket is ONE
+        # token, so this is O(|allowed|).
+        allowed = None
+        fn = getattr(seq.grammar, "allowed_next_bytes", None)
+        if fn is not None:
+            allowed = fn()
+        if fn is not None:
+            if allowed == b"":
+                return self.tokenizer.eos_id
+            self.counters["grammar_bucket_picks"] += 1
+            buckets = self.tokenizer.first_byte_buckets()
+            ids: List[int] = []
+            if allowed is None:
+                # unconstrained state (inside a string): every token with a
+                # byte expansion is a candidate — for the byte tokenizer
+                # that is 256 ids vs a full-vocab argsort
+                for lst in buckets.values():
+                    ids.extend(lst)
+            else:
+                for by in allowed:
+                    ids.extend(buckets.get(by, ()))
+            if ids:
+                if len(ids) > 1:
+                    t = torch.tensor(ids, device=row.device)
+                    vals = row[t]
+                    temp = seq.params.temperature
+                    if temp > 0:  # gumbel-max keeps sampling stochastic
+                        u = torch.rand(vals.shape, device=vals.device,
+                                       generator=self._dev_gen(vals.device))
+                        vals = (vals / max(temp, 1e-6)
+                                - torch.log(-torch.log(u.clamp(min=1e-20))))
+                    order = t[vals.argsort(descending=True)].tolist()
+                else:
+                    order = ids
+                for tok in order:
+                    g = seq.grammar.fork()
+                    if g.advance(self.tokenizer.token_bytes(tok)):
+                        seq.grammar = g
+                        return tok
         # fallback: full-vocab scan in logit order (rare)
         self.counters["grammar_fallbacks"] += 1
         order = row.argsort(descending=True).tolist()

@@ -73,6 +73,53 @@ class JsonGrammar:
                 return False
         return True
 
+    def allowed_next_bytes(self):
+        """The set of legal next bytes (as ``bytes``), or None when the
+        state is effectively unconstrained (inside a string). Lets the
+        sampler pick from tiny per-byte candidate buckets instead of
+        scanning the vocabulary when the FSM pins the continuation.
+        Computed for compact generation (no inter-token whitespace)."""
+        s = self.state
+        if s == _DONE:
+            return b""
+        if s == _VALUE:
+            if self.require_object and not self.started:
+                return b"{"
+            out = b'"{[-0123456789tfn'
+            if self.arr_fresh and self.stack and self.stack[-1] == 0x5B:
+                out += b"]"
+            return out
+        if s == _STRING:
+            return None  # any byte >= 0x20: unconstrained enough
+        if s == _STR_ESC:
+            return b'"\\/bfnrtu'
+        if s == _STR_U:
+            return b"0123456789abcdefABCDEF"
+        if s in (_NUM_INT, _NUM_FRAC, _NUM_EXP):
+            out = bytes(_DIGITS)
+            if s == _NUM_INT and self.num_had_digit:
+                out += b".eE"
+            elif s == _NUM_FRAC and self.num_had_digit:
+                out += b"eE"
+            elif s == _NUM_EXP and not self.num_had_digit:
+                out += b"+-"
+            if self.num_had_digit and self.stack:
+                out += b",}" if self.stack[-1] == 0x7B else b",]"
+            return out
+        if s == _LITERAL:
+            return bytes([self.lit[self.lit_pos]])
+        if s == _OBJ_KEY_OR_END:
+            return b'"}'
+        if s == _OBJ_COLON:
+            return b":"
+        if s == _OBJ_KEY_REQUIRED:
+            return b'"'
+        if s == _OBJ_COMMA_OR_END:
+            return b",}"
+        if s == _ARR_COMMA_OR_END:
+            return b",]"
+        return None
+
     # -- single byte -------------------------------------------------------
 
     def _pop_value_done(self) -> bool:
@@ -257,7 +304,8 @@ class ToolCallGrammar:
     _PREFIX = b'{"tool":"'
     _MID = b'","args":'
 
-    def __init__(self, tools: List[str], max_args_depth: int = 8):
+    def __init__(self, tools: List[str], max_args_depth: int = 8,
+                 soft_close_bytes: int = 160):
         if not tools:
             raise ValueError("ToolCallGrammar needs at least one tool name")
         self.names = sorted(set(t.encode() for t in tools))
@@ -266,6 +314,11 @@ class ToolCallGrammar:
         self.name_prefix = b""
         self.max_args_depth = max_args_depth
         self.args: JsonGrammar = None  # created lazily at phase 3
+        # past this many bytes the sampler is steered toward closing the
+        # call (so it completes before the token budget instead of being
+        # truncated mid-string into unparseable JSON)
+        self.soft_close_bytes = soft_close_bytes
+        self.nbytes = 0
 
     def fork(self) -> "ToolCallGrammar":
         g = ToolCallGrammar.__new__(ToolCallGrammar)
@@ -275,6 +328,8 @@ class ToolCallGrammar:
         g.name_prefix = self.name_prefix
         g.max_args_depth = self.max_args_depth
         g.args = self.args.fork() if self.args is not None else None
+        g.soft_close_bytes = self.soft_close_bytes
+        g.nbytes = self.nbytes
         return g
 
     def is_complete(self) -> bool:
@@ -288,7 +343,39 @@ class ToolCallGrammar:
         for b in data:
             if not self._advance_byte(b):
                 return False
+            self.nbytes += 1
         return True
+
+    def allowed_next_bytes(self):
+        """Legal next bytes for the sampler's bucket pick (see JsonGrammar)."""
+        if self.phase == 0:
+            return bytes([self._PREFIX[self.lit_pos]])
+        if self.phase == 1:
+            n = len(self.name_prefix)
+            out = {m[n] for m in self.names
+                   if len(m) > n and m.startswith(self.name_prefix)}
+            if self.name_prefix in self.names:
+                out.add(0x22)  # closing quote
+            return bytes(sorted(out))
+        if self.phase == 2:
+            return bytes([self._MID[self.lit_pos]])
+        if self.phase == 3:
+            if self.args.is_complete():
+                return b"}"
+            inner = self.args.allowed_next_bytes()
+            if self.nbytes >= self.soft_close_bytes:
+                # over budget: steer toward closing (legality unchanged —
+                # this only narrows what the sampler is offered)
+                if inner is None:  # inside a string: close it
+                    return b'"'
+                closers = bytes(c for c in inner if c in b'"}]')
+                if closers:
+                    return closers
+            return inner
+        return b""  # phase 4: only EOS
+
+    def advance_bytes_hint(self) -> int:
+        return self.nbytes
 
     def _advance_byte(self, b: int) -> bool:
         if self.phase == 0:

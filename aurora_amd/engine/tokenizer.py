@@ -38,6 +38,20 @@ class Tokenizer:
     def count(self, text: str) -> int:
         return len(self.encode(text))
 
+    def first_byte_buckets(self) -> dict:
+        """byte -> list of token ids whose byte expansion starts with it.
+        Lets grammar-constrained sampling gather candidates for a known
+        legal next byte instead of scanning the whole vocabulary."""
+        cache = getattr(self, "_fb_buckets", None)
+        if cache is None:
+            cache = {}
+            for t in range(self.vocab_size):
+                b = self.token_bytes(t)
+                if b:
+                    cache.setdefault(b[0], []).append(t)
+            self._fb_buckets = cache
+        return cache
+
 
 class ByteTokenizer(Tokenizer):
     BOS = 256

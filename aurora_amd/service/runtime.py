@@ -81,6 +81,7 @@ class Runtime:
         # 288 GB HBM co-hosts a small aux model beside the main one)
         self._aux_engines: Dict[str, AsyncEngineServer] = {}
         self._rag: Dict[str, RagIndex] = {}
+        self._encoder = None  # None = unprobed, False = unavailable (CPU)
         self._embedding_rail = None
         self._graph: Dict[str, InfraGraph] = {}
         self._rca_rate: Dict[str, float] = {}  # user -> last background RCA ts
@@ -114,10 +115,35 @@ class Runtime:
         return LLMManager(self.engine, mc, store=self.org(org_id),
                           engine_resolver=self.engine_for)
 
+    def encoder(self):
+        """Shared on-GPU sentence encoder for RAG + the embedding rail
+        (the Weaviate text2vec sidecar replacement, SURVEY preamble table).
+        Built lazily on first use and only when the engine runs on a GPU —
+        CPU deployments fall back to BM25-only search and the hashed-BoW
+        rail (a random-init fp32 encoder on CPU would cost seconds per
+        query for meaningless vectors)."""
+        if self._encoder is False:
+            return None
+        if self._encoder is None:
+            import torch as _t
+            dev = str(getattr(self.engine.engine, "device", "cpu"))
+            if not dev.startswith("cuda"):
+                self._encoder = False
+                return None
+            from ..models.encoder import SentenceEncoder
+            enc = SentenceEncoder("bge-base", dtype=_t.bfloat16, device=dev)
+            enc.init_random(3)  # no network: random-init, BASELINE.md
+            self._encoder = enc
+        return self._encoder
+
     def rag(self, org_id: str) -> RagIndex:
         if org_id not in self._rag:
+            enc = self.encoder()
             self._rag[org_id] = RagIndex.from_store(
-                self.org(org_id), tokenizer=self.engine.tokenizer())
+                self.org(org_id), encoder=enc,
+                tokenizer=self.engine.tokenizer(),
+                device=str(getattr(self.engine.engine, "device", "cpu"))
+                if enc is not None else "cpu")
         return self._rag[org_id]
 
     def graph(self, org_id: str) -> InfraGraph:
@@ -127,12 +153,26 @@ class Runtime:
 
     def embedding_rail(self):
         """Shared prompt-injection embedding classifier (lazy singleton;
-        guardrails/embedding_rail.py). Uses the hashed-BoW backend unless
-        a trained encoder is configured — random-init encoder weights
-        would make similarity meaningless."""
+        guardrails/embedding_rail.py). AURORA_RAIL_BACKEND=encoder runs it
+        on the shared GPU encoder (BASELINE config 3: classifier on every
+        agent turn); default is the hashed-BoW backend — a random-init
+        encoder makes similarity meaningless, so the encoder backend is
+        opt-in for trained weights / perf measurement."""
         if self._embedding_rail is None:
+            import os as _os
             from ..guardrails.embedding_rail import EmbeddingRail, hashed_bow_embed
-            self._embedding_rail = EmbeddingRail(hashed_bow_embed, threshold=0.55)
+            embed_fn = hashed_bow_embed
+            threshold = 0.55
+            if _os.environ.get("AURORA_RAIL_BACKEND", "") == "encoder":
+                enc = self.encoder()
+                if enc is not None:
+                    tok = self.engine.tokenizer()
+
+                    def embed_fn(texts):  # noqa: F811 - encoder backend
+                        return enc.encode_ids(
+                            [tok.encode(t)[:256] for t in texts])
+                    threshold = 0.86
+            self._embedding_rail = EmbeddingRail(embed_fn, threshold=threshold)
         return self._embedding_rail
 
     def tool_ctx(self, org_id: str, user_id: Optional[str] = None,

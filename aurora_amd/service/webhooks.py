@@ -227,9 +227,11 @@ def verify_signature(source: str, body: bytes, headers: Dict[str, str],
 
 def process_webhook(rt, org_id: str, source: str, payload: dict,
                     user_id: Optional[str] = None,
-                    rca_countdown_s: float = 1.0) -> Tuple[str, bool]:
+                    rca_countdown_s: float = 1.0,
+                    rca_args: Optional[dict] = None) -> Tuple[str, bool]:
     """Normalise -> correlate -> upsert incident -> dispatch delayed RCA.
-    Returns (incident_id, created_new)."""
+    Returns (incident_id, created_new). ``rca_args`` (e.g. max_turns,
+    tool_subset) are forwarded to the background investigation."""
     norm_fn = NORMALIZERS.get(source)
     if norm_fn is None:
         raise ValueError(f"unknown alert source '{source}'")
@@ -249,7 +251,8 @@ def process_webhook(rt, org_id: str, source: str, payload: dict,
         # delayed RCA: the countdown lets provider runbook fields settle
         # (reference: trigger_delayed_rca countdown, tasks.py:235)
         rt.queue.enqueue("trigger_delayed_rca",
-                         {"org_id": org_id, "incident_id": iid, "user_id": user_id},
+                         dict(rca_args or {}, org_id=org_id, incident_id=iid,
+                              user_id=user_id),
                          countdown_s=rca_countdown_s,
                          dedup_key=f"delayed-rca:{iid}")
         rt.events.publish(org_id, {"type": "incident_created", "incident_id": iid,

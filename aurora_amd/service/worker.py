@@ -105,7 +105,8 @@ async def run_background_chat(rt, args: Dict[str, Any]) -> dict:
             final = result.final
         else:
             wf = Workflow(llm, ctx, max_turns=int(args.get("max_turns", 0))
-                          or rt.settings.rca_max_turns or None)
+                          or rt.settings.rca_max_turns or None,
+                          tool_subset=args.get("tool_subset"))
             async for ev in wf.stream(state):
                 if ev["type"] == "final":
                     final = ev["content"]
@@ -205,7 +206,8 @@ async def trigger_delayed_rca(rt, args: Dict[str, Any]) -> dict:
         "run_background_chat",
         {"org_id": org_id, "incident_id": incident_id,
          "user_id": args.get("user_id"),
-         "max_turns": args.get("max_turns", 0)},
+         "max_turns": args.get("max_turns", 0),
+         "tool_subset": args.get("tool_subset")},
         dedup_key=f"rca:{incident_id}")
     if tid:
         store.update_incident(incident_id, rca_task_id=tid)

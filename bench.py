@@ -7,19 +7,32 @@ For N>1 the driver launches this under torch.distributed.run with one rank
 per GPU (DP: each rank serves its own shard of concurrent investigations —
 weak scaling, the reference's Celery-worker model mapped onto GPUs).
 
-One STEP = one batch of --concurrency synthetic incident investigations
-processed end-to-end on this rank's engine: per investigation, --turns agent
-turns, each turn appending a synthetic alert/tool-output prompt chunk to the
-session (KV reused across turns) and decoding a fixed thought/tool-call
-budget.  Fixed token counts per step (no early exit) so the timed work is
-identical across steps and ranks.  Data: synthetic token streams,
-random-init weights (no network in this environment — BASELINE.md).
+Two modes:
+
+* ``--mode e2e`` (default, the headline): one STEP = --concurrency synthetic
+  PagerDuty webhooks driven through the REAL service pipeline on this rank —
+  webhook normalise/correlate -> incident row -> queue -> admission ->
+  run_background_chat -> the guardrailed agent workflow (input rail, per-turn
+  ToolCallGrammar-constrained action decoding, real tool execution against
+  the org store / GPU RAG index, per-turn embedding-rail classification of
+  tool outputs) -> forced final RCA -> summarization LLM pass ->
+  recommendations -> citations -> postmortem row. The timed region contains
+  ALL of it (reference pipeline: chat/background/task.py:447-697; metric
+  definition BASELINE.json). p50 time-to-RCA = webhook send -> completion
+  event per incident.
+
+* ``--mode engine``: the round-1 engine-level path (per-turn RAG + rail
+  around raw eng.step() decode turns) — quoted alongside for comparability.
+
+Data: synthetic alert payloads / token streams, random-init weights
+(no network in this environment — BASELINE.md).
 
 Rank 0 prints ONE JSON line; value = whole-job incidents/hr over all ranks.
 """
 from __future__ import annotations
 
 import argparse
+import asyncio
 import json
 import os
 import statistics
@@ -33,6 +46,8 @@ def parse_args():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
+    ap.add_argument("--mode", choices=("e2e", "engine"),
+                    default=os.environ.get("BENCH_MODE", "e2e"))
     ap.add_argument("--model", default=os.environ.get("BENCH_MODEL", "llama3-8b"))
     ap.add_argument("--concurrency", type=int,
                     default=int(os.environ.get("BENCH_CONCURRENCY", "256")))
@@ -43,7 +58,8 @@ def parse_args():
     ap.add_argument("--device", default="auto")
     ap.add_argument("--rag", type=int, default=int(os.environ.get("BENCH_RAG", "1")),
                     help="per-turn on-GPU RAG retrieval + input-rail check (1=on)")
-    ap.add_argument("--rag-docs", type=int, default=10000)
+    ap.add_argument("--rag-docs", type=int,
+                    default=int(os.environ.get("BENCH_RAG_DOCS", "10000")))
     return ap.parse_args()
 
 
@@ -51,8 +67,12 @@ def synth_tokens(rng: torch.Generator, n: int, vocab: int) -> list:
     return torch.randint(0, min(vocab - 4, 250), (n,), generator=rng).tolist()
 
 
+# --------------------------------------------------------------------------
+# engine mode (round-1 comparability path)
+# --------------------------------------------------------------------------
+
 class RagPath:
-    """BASELINE config 2/3: per-turn retrieval over an HBM-resident corpus
+    """Engine mode: per-turn retrieval over an HBM-resident corpus
     (on-GPU sentence encoder + kNN) + the input-rail heuristic check."""
 
     def __init__(self, args, device: str):
@@ -66,7 +86,6 @@ class RagPath:
             device=device)
         self.encoder.init_random(3)
         rng = _t.Generator().manual_seed(17)
-        # synthetic corpus: embed args.rag_docs chunks once at setup
         chunks = [_t.randint(1, 250, (96,), generator=rng).tolist()
                   for _ in range(args.rag_docs)]
         embs = []
@@ -83,15 +102,14 @@ class RagPath:
     def retrieve(self, query_tokens) -> list:
         qv = self.encoder.encode_ids([query_tokens[:96]])
         vals, idx = self.ops.knn_topk(qv, self.docs, 4)
-        # deterministic "retrieved chunk" tokens derived from the hit ids
         out = []
         for i in idx[0].tolist():
             out.extend(((i * 131 + j * 7) % 250 for j in range(32)))
         return out
 
 
-def run_step(eng, args, rng, vocab, step_idx: int, rag=None):
-    """One full investigation batch; returns per-incident latencies."""
+def run_step_engine(eng, args, rng, vocab, step_idx: int, rag=None):
+    """One engine-level investigation batch; returns per-incident latencies."""
     from aurora_amd.engine.sampling_params import SamplingParams
 
     B, turns = args.concurrency, args.turns
@@ -105,8 +123,6 @@ def run_step(eng, args, rng, vocab, step_idx: int, rag=None):
     def submit(i):
         rid = f"s{step_idx}-inc{i}-t{turn_no[i]}"
         if rag is not None:
-            # input rail on the webhook/tool-authored turn text + RAG
-            # retrieval over the HBM corpus, inside the timed path
             rag.rail_check(ctx_tokens[i][-args.turn_prompt:])
             retrieved = rag.retrieve(ctx_tokens[i][-96:])
             ctx_tokens[i][-len(retrieved):] = retrieved  # fixed token count
@@ -135,18 +151,10 @@ def run_step(eng, args, rng, vocab, step_idx: int, rag=None):
     return [t_done[i] - t_start[i] for i in range(B)]
 
 
-def main():
-    args = parse_args()
-    rank = int(os.environ.get("RANK", "0"))
-    world = int(os.environ.get("WORLD_SIZE", "1"))
-    use_cuda = torch.cuda.is_available() and args.device != "cpu"
-
+def main_engine(args, rank, world, use_cuda):
     from aurora_amd.engine.engine import EngineConfig, LLMEngine
     from aurora_amd.models.config import get_spec
     from aurora_amd.parallel import dist as pdist
-
-    if world > 1:
-        pdist.init_distributed(tp_size=1)
 
     spec = get_spec(args.model)
     cfg = EngineConfig(
@@ -170,25 +178,227 @@ def main():
                 torch.cuda.synchronize()
 
     for w in range(args.warmup):
-        run_step(eng, args, rng, vocab, -1 - w, rag)
+        run_step_engine(eng, args, rng, vocab, -1 - w, rag)
     sync()
     t0 = time.monotonic()
     lat = []
     for s in range(args.steps):
-        lat.extend(run_step(eng, args, rng, vocab, s, rag))
+        lat.extend(run_step_engine(eng, args, rng, vocab, s, rag))
+    sync()
+    elapsed = time.monotonic() - t0
+    extra = {
+        "pipeline": "engine",
+        "rag": bool(rag), "rag_docs": args.rag_docs if rag else 0,
+        "guardrail_per_turn": bool(rag),
+        "decode_tokens_per_turn": args.decode_per_turn,
+        "prompt_tokens_per_incident": args.first_prompt
+        + (args.turns - 1) * args.turn_prompt,
+        "engine_timers_s": {k: round(v, 2) for k, v in eng.timers.items()},
+        "engine_counters": eng.counters,
+    }
+    return elapsed, lat, extra
+
+
+# --------------------------------------------------------------------------
+# e2e mode (the headline): real webhook -> worker -> workflow -> summary
+# --------------------------------------------------------------------------
+
+#: fast read-only tools the random-weight agent may call during the bench —
+#: real registry tools hitting the org store / infra graph / GPU RAG
+#: ("stub-fast" tool execution per the metric definition; exec-style tools
+#: are excluded so a random-weight model can't run shell commands)
+BENCH_TOOLS = [
+    "knowledge_base_search", "list_incidents", "get_incident",
+    "get_incident_timeline", "get_alert_field", "list_services",
+    "get_service_impact", "get_infrastructure_context", "final_answer",
+]
+
+_LOREM = ("deploy rollout canary latency p99 spike postgres connection pool "
+          "exhausted kafka lag consumer rebalance oom killed node pressure "
+          "dns resolution timeout ingress 502 upstream retry storm cache "
+          "eviction hot shard gc pause tls handshake certificate expired").split()
+
+
+def _synth_doc(rng: torch.Generator, n_words: int = 160) -> str:
+    idx = torch.randint(0, len(_LOREM), (n_words,), generator=rng).tolist()
+    return " ".join(_LOREM[i] for i in idx)
+
+
+def _build_runtime(args, rank, use_cuda):
+    os.environ.setdefault("AURORA_RAIL_BACKEND", "encoder")
+    from aurora_amd.engine.engine import EngineConfig
+    from aurora_amd.engine.server import AsyncEngineServer
+    from aurora_amd.service.runtime import Runtime
+    from aurora_amd.utils.config import Settings
+
+    scfg = Settings(db_path=":memory:",
+                    secrets_path=f"/tmp/bench_secrets_{rank}.json")
+    scfg.worker_concurrency = args.concurrency + 8
+    scfg.max_concurrent_rca = args.concurrency * 2
+    scfg.rca_max_turns = args.turns
+    scfg.guardrails_enabled = True
+    ecfg = EngineConfig(
+        model=args.model,
+        device="cpu" if not use_cuda else "auto",
+        max_num_seqs=max(args.concurrency + 2, 8),
+        max_batched_tokens=16384,
+        seed=1234 + rank,
+    )
+    server = AsyncEngineServer(ecfg)
+    rt = Runtime(scfg, engine_server=server)
+    org_id = rt.admin.create_org(f"bench-{rank}")
+    rt.admin.create_user(org_id, f"bench{rank}@aurora.local",
+                         user_id=f"bench-user-{rank}")
+    # infra graph for the graph tools
+    g = rt.graph(org_id)
+    for svc in ("api", "db", "cache", "queue", "ingress"):
+        g.upsert_service(svc)
+    for a, b in (("api", "db"), ("api", "cache"), ("ingress", "api"),
+                 ("queue", "db")):
+        g.add_dependency(a, b)
+    # RAG corpus: args.rag_docs synthetic chunks embedded on the GPU encoder
+    # (HBM-resident kNN bank) without round-tripping SQLite at bench setup
+    rag = rt.rag(org_id)
+    rng = torch.Generator().manual_seed(17)
+    docs = [_synth_doc(rng) for _ in range(args.rag_docs)]
+    BATCH = 512
+    for i in range(0, len(docs), BATCH):
+        batch = docs[i:i + BATCH]
+        vecs = rag._embed(batch)
+        for j, d in enumerate(batch):
+            rag.chunks.append({"text": d, "doc_id": f"kb-{i + j}",
+                               "doc_name": f"runbook-{(i + j) % 97}"})
+            rag.bm25.add(d)
+        if vecs is not None:
+            vecs = vecs.to(rag.device)
+            rag.emb = vecs if rag.emb is None else torch.cat([rag.emb, vecs])
+    return rt, org_id
+
+
+def _payload(rng: torch.Generator, step: int, i: int) -> dict:
+    """Unique synthetic PagerDuty v3 event. Every alert must open a NEW
+    incident, so its text is made lexically unique (the correlator's
+    similarity strategy would otherwise merge same-vocabulary alerts —
+    which is correct behavior for real alert storms)."""
+    tag = f"s{step}i{i}"
+    doc = " ".join(f"{w}{tag}" for w in _synth_doc(rng, 60).split())
+    return {"event": {"data": {
+        "id": f"PD-{step}-{i}-{torch.randint(0, 1 << 30, (1,), generator=rng).item()}",
+        "title": f"[bench s{step}] {doc[:80]} #{i}",
+        "service": {"summary": f"svc-{step}-{i}"},
+        "priority": {"summary": "P2"},
+        "urgency": "high",
+        "body": {"details": doc},
+    }}}
+
+
+async def run_step_e2e(rt, org_id, args, rng, step: int):
+    from aurora_amd.service.webhooks import process_webhook
+    B = args.concurrency
+    ev_q = rt.events.subscribe(org_id)
+    t_start: dict = {}
+    pending = set()
+    statuses: dict = {}
+    for i in range(B):
+        t_start_i = time.monotonic()
+        iid, created = process_webhook(
+            rt, org_id, "pagerduty", _payload(rng, step, i),
+            rca_countdown_s=0.0,
+            rca_args={"max_turns": args.turns, "tool_subset": BENCH_TOOLS})
+        assert created, f"correlator merged bench alert {i} into {iid}"
+        t_start[iid] = t_start_i
+        pending.add(iid)
+    lat = []
+    while pending:
+        ev = await asyncio.wait_for(ev_q.get(), timeout=600.0)
+        if (ev.get("type") == "incident_update"
+                and ev.get("incident_id") in pending
+                and ev.get("aurora_status") in ("completed", "failed")):
+            iid = ev["incident_id"]
+            pending.discard(iid)
+            statuses[ev["aurora_status"]] = statuses.get(ev["aurora_status"], 0) + 1
+            lat.append(time.monotonic() - t_start[iid])
+    rt.events.unsubscribe(org_id, ev_q)
+    return lat, statuses
+
+
+async def main_e2e_async(args, rank, world, use_cuda):
+    from aurora_amd.parallel import dist as pdist
+
+    rt, org_id = _build_runtime(args, rank, use_cuda)
+    rt.start()
+    rng = torch.Generator().manual_seed(999 + rank)
+
+    def sync():
+        if use_cuda:
+            torch.cuda.synchronize()
+        if world > 1:
+            pdist.barrier()
+            if use_cuda:
+                torch.cuda.synchronize()
+
+    statuses_all: dict = {}
+    for w in range(args.warmup):
+        await run_step_e2e(rt, org_id, args, rng, -1 - w)
+    sync()
+    t0 = time.monotonic()
+    lat = []
+    for s in range(args.steps):
+        l, st = await run_step_e2e(rt, org_id, args, rng, s)
+        lat.extend(l)
+        for k, v in st.items():
+            statuses_all[k] = statuses_all.get(k, 0) + v
     sync()
     elapsed = time.monotonic() - t0
 
+    eng = rt.engine.engine
+    extra = {
+        "pipeline": "webhook->queue->workflow->summary (end-to-end)",
+        "in_timed_region": ["webhook normalise+correlate", "queue dispatch",
+                           "input rail", "per-turn grammar tool decode",
+                           "tool execution", "per-turn tool-output rail",
+                           "GPU RAG (kb tool)", "forced final RCA",
+                           "summarization pass", "recommendations",
+                           "citations", "postmortem"],
+        "rag_docs": args.rag_docs,
+        "guardrail_per_turn": True,
+        "tool_subset": BENCH_TOOLS,
+        "completed": statuses_all.get("completed", 0),
+        "failed": statuses_all.get("failed", 0),
+        "engine_timers_s": {k: round(v, 2) for k, v in eng.timers.items()},
+        "engine_counters": eng.counters,
+    }
+    await rt.stop()
+    rt.engine.shutdown()
+    return elapsed, lat, extra
+
+
+def main():
+    args = parse_args()
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    use_cuda = torch.cuda.is_available() and args.device != "cpu"
+
+    from aurora_amd.parallel import dist as pdist
+    if world > 1:
+        pdist.init_distributed(tp_size=1)
+
+    if args.mode == "engine":
+        elapsed, lat, extra = main_engine(args, rank, world, use_cuda)
+    else:
+        elapsed, lat, extra = asyncio.run(
+            main_e2e_async(args, rank, world, use_cuda))
+
     # max over ranks (slowest rank defines the job)
     if world > 1:
-        t = torch.tensor([elapsed])
         import torch.distributed as dist
+        t = torch.tensor([elapsed])
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t[0])
 
     incidents = args.steps * args.concurrency * world
     value = incidents / (elapsed / 3600.0)
-    p50 = statistics.median(lat)
+    p50 = statistics.median(lat) if lat else 0.0
     tokens_per_incident = (args.first_prompt + (args.turns - 1) * args.turn_prompt
                            + args.turns * args.decode_per_turn)
     if rank == 0:
@@ -207,19 +417,14 @@ def main():
             "data": "synthetic",
             "config": {
                 "model": args.model,
+                "mode": args.mode,
                 "global_batch": args.concurrency * world,
                 "seq_len": tokens_per_incident,
                 "parallelism": f"dp{world}",
                 "concurrency_per_gpu": args.concurrency,
                 "turns": args.turns,
-                "decode_tokens_per_turn": args.decode_per_turn,
-                "rag": bool(rag), "rag_docs": args.rag_docs if rag else 0,
-                "guardrail_per_turn": bool(rag),
                 "p50_time_to_rca_s": round(p50, 3),
-                "prompt_tokens_per_incident": args.first_prompt
-                + (args.turns - 1) * args.turn_prompt,
-                "engine_timers_s": {k: round(v, 2) for k, v in eng.timers.items()},
-                "engine_counters": eng.counters,
+                **extra,
             },
         }))
 

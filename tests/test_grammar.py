@@ -144,7 +144,7 @@ def test_tool_call_grammar_engine_end_to_end():
     outs = eng.generate(
         [[5, 6, 7], [8, 9, 10]],
         SamplingParams(temperature=0.8, max_tokens=96,
-                       grammar=ToolCallGrammar(tools)))
+                       grammar=ToolCallGrammar(tools, soft_close_bytes=40)))
     for toks in outs:
         text = eng.tokenizer.decode(toks)
         doc = _json.loads(text)
