@@ -192,6 +192,16 @@ def linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
         if w_bytes <= max_mb * 1024 * 1024:
             s_target = 256 if w_bytes <= 40 * 1024 * 1024 else 512
             return _hip.skinny_gemm2_ex(x.contiguous(), weight, s_target)
+    elif (x.is_cuda and x.dim() == 2 and 64 < x.shape[0] <= 256
+            and x.shape[1] % 64 == 0 and weight.shape[0] % 16 == 0
+            and x.dtype == torch.bfloat16 and _hip is not None
+            and weight.shape[0] * weight.shape[1] * 2 < 512 * 1024 * 1024):
+        # M=65..256 (the B=256 decode regime): hipBLASLt runs these cold at
+        # 1.1-3.7 TB/s (docs/ROUND2.md §1); the glds double-buffered v3
+        # streams W once through LDS. Gated off with AURORA_SKINNY3=0.
+        import os
+        if os.environ.get("AURORA_SKINNY3", "1") != "0":
+            return _hip.skinny_gemm3_ex(x.contiguous(), weight, 0)
     return F.linear(x, weight)
 
 

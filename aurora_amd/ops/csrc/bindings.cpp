@@ -27,6 +27,7 @@ torch::Tensor fused_rope_kv_write(torch::Tensor qkv, torch::Tensor positions,
                                   long Hq, long Hkv);
 torch::Tensor skinny_gemm_ex(torch::Tensor x, torch::Tensor w, long s_target, long unroll);
 torch::Tensor skinny_gemm2_ex(torch::Tensor x, torch::Tensor w, long s_target);
+torch::Tensor skinny_gemm3_ex(torch::Tensor x, torch::Tensor w, long s_target);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "fused RMSNorm (bf16, gfx950)");
@@ -45,4 +46,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused qkv split + RoPE + paged KV scatter");
   m.def("skinny_gemm_ex", &skinny_gemm_ex, "skinny GEMM with tuning knobs");
   m.def("skinny_gemm2_ex", &skinny_gemm2_ex, "LDS-staged skinny GEMM (M<=64)");
+  m.def("skinny_gemm3_ex", &skinny_gemm3_ex, "glds double-buffered projection GEMM (M<=256)");
 }
