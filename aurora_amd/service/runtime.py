@@ -204,8 +204,20 @@ class Runtime:
         self.queue.add_beat("action_scheduler", 60.0, {})
         self.queue.add_beat("run_prediscovery", 3600.0, {})  # hourly (reference: celery_config.py:126-129)
         self.queue.start()
+        # external MCP servers (reference: mcp_preloader started at boot,
+        # main_chatbot.py:1684): spawn + register their tools when
+        # AURORA_MCP_SERVERS is configured
+        from ..agent.tools.mcp_client import MCPManager
+        from ..utils.secrets import secrets_backend
+        self.mcp = MCPManager(secrets=secrets_backend())
+        if self.mcp.specs:
+            asyncio.get_event_loop().create_task(
+                self.mcp.register_tools(GLOBAL_REGISTRY))
 
     async def stop(self) -> None:
         await self.queue.stop()
+        mcp = getattr(self, "mcp", None)
+        if mcp is not None:
+            await mcp.stop_all()
         for srv in self._aux_engines.values():
             srv.shutdown()
