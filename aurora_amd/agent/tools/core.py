@@ -323,33 +323,174 @@ async def load_skill(ctx: ToolContext, name: str) -> str:
     return sk.body if sk else f"ERROR: unknown skill '{name}'"
 
 
-# ----------------------------------------------------- connector-gated stubs
-# The provider API-client tools of the reference (datadog/sentry/splunk/...)
-# need live provider endpoints; the registration + gating contract is kept
-# and each tool degrades with an explicit error when unreachable.
+# ------------------------------------------------ connector-gated API tools
+# Real query construction + response parsing per provider
+# (connectors/clients/); only the socket itself degrades when the
+# deployment has no egress. Transport is swappable via
+# ctx.extra["provider_transport"] (tests inject recorded fixtures).
 
-def _provider_tool(name: str, provider: str, desc: str):
-    @R.add(name, desc, {"query": "provider query"}, requires_connection=provider)
-    async def _tool(ctx: ToolContext, query: str, _p=provider) -> str:
-        return (f"ERROR: {_p} API unreachable from this deployment "
-                f"(no egress); connection is configured but the call failed")
+
+def _transport(ctx: ToolContext):
+    t = (ctx.extra or {}).get("provider_transport")
+    if t is not None:
+        return t
+    from ...connectors.clients import http_transport
+    return http_transport()
+
+
+def _creds(ctx: ToolContext, provider: str) -> dict:
+    from ...connectors.registry import get_credentials
+    creds = None
+    if ctx.org_store is not None and ctx.user_id:
+        creds = get_credentials(ctx.org_store, ctx.user_id, provider)
+    return creds or {}
+
+
+def _provider_tool(name: str, provider: str, desc: str,
+                   params: dict | None = None):
+    from ...connectors.clients import NoEgressError, ProviderAPIError, QUERY_FNS
+
+    @R.add(name, desc, params or {
+        "query": "provider query",
+        "time_from": "window start: 'now-1h', '-30m', unix or ISO (optional)",
+        "time_to": "window end (optional, default now)",
+    }, requires_connection=provider)
+    async def _tool(ctx: ToolContext, query: str, time_from: str = "",
+                    time_to: str = "", query_type: str = "",
+                    _p=provider) -> str:
+        fn = QUERY_FNS[_p]
+        try:
+            kwargs = {"time_from": time_from or None,
+                      "time_to": time_to or None}
+            if _p == "datadog":
+                return await fn(_transport(ctx), _creds(ctx, _p),
+                                query_type or "logs", query, **kwargs)
+            return await fn(_transport(ctx), _creds(ctx, _p), query, **kwargs)
+        except NoEgressError as e:
+            return (f"ERROR: {_p} API unreachable from this deployment "
+                    f"(no egress): {e}")
+        except ProviderAPIError as e:
+            return f"ERROR: {_p} API: {e}"
     return _tool
 
 
-_provider_tool("query_datadog", "datadog", "Query Datadog metrics/logs/monitors.")
-_provider_tool("query_newrelic", "newrelic", "Run a NRQL query against New Relic.")
-_provider_tool("query_sentry", "sentry", "Search Sentry issues/events.")
-_provider_tool("search_splunk", "splunk", "Run a Splunk search.")
-_provider_tool("query_dynatrace", "dynatrace", "Query Dynatrace problems/metrics.")
-_provider_tool("query_grafana", "grafana", "Query Grafana dashboards/datasources.")
-_provider_tool("query_cloudwatch", "aws", "Query CloudWatch metrics/logs.")
-_provider_tool("github_rca", "github", "Correlate incident window with recent commits/PRs.")
-_provider_tool("github_fix", "github", "Draft a fix suggestion PR for the incident.")
-_provider_tool("jenkins_rca", "jenkins", "Analyze recent Jenkins build failures.")
-_provider_tool("query_slack", "slack", "Search Slack channels/threads for context.")
-_provider_tool("query_jira", "jira", "Search Jira issues.")
+_provider_tool("query_datadog", "datadog",
+               "Query Datadog: logs (default), metrics, monitors or events "
+               "(set query_type).",
+               {"query": "log filter / metric query / monitor search",
+                "query_type": "logs|metrics|monitors|events",
+                "time_from": "window start (e.g. 'now-1h')",
+                "time_to": "window end (default now)"})
+_provider_tool("query_newrelic", "newrelic",
+               "Run a NRQL query against New Relic (NerdGraph).",
+               {"query": "NRQL, e.g. SELECT count(*) FROM TransactionError "
+                         "SINCE 1 hour ago"})
+_provider_tool("query_sentry", "sentry",
+               "Search Sentry issues (issue search syntax).",
+               {"query": "e.g. is:unresolved level:error release:1.2"})
+_provider_tool("search_splunk", "splunk",
+               "Run a Splunk search (SPL; 'indexes' lists indexes).")
+_provider_tool("query_dynatrace", "dynatrace",
+               "Dynatrace problems in the window ('problems') or a metric "
+               "selector query.")
+_provider_tool("query_grafana", "grafana",
+               "Search Grafana dashboards/folders.",
+               {"query": "dashboard title text"})
+_provider_tool("query_slack", "slack",
+               "Slack: search messages; 'channels' lists channels; "
+               "'history <channel_id>' reads a channel.")
+_provider_tool("query_jira", "jira", "Search Jira issues with JQL.",
+               {"query": "JQL, e.g. project=OPS AND status=Open"})
 _provider_tool("notion_search", "notion", "Search Notion workspace pages.")
-_provider_tool("confluence_search", "confluence", "Search Confluence pages/runbooks.")
+_provider_tool("confluence_search", "confluence",
+               "Search Confluence pages/runbooks (text match).")
+
+
+@R.add("query_cloudwatch",
+       "CloudWatch metric statistics for a namespace/metric over a window.",
+       {"namespace": "e.g. AWS/EC2", "metric": "e.g. CPUUtilization",
+        "dimensions": "k=v,k=v (optional)",
+        "time_from": "window start", "time_to": "window end",
+        "stat": "Average|Sum|Maximum"},
+       requires_connection="aws")
+async def query_cloudwatch(ctx: ToolContext, namespace: str, metric: str,
+                           dimensions: str = "", time_from: str = "",
+                           time_to: str = "", stat: str = "Average") -> str:
+    """CloudWatch GetMetricStatistics (SigV4 signing requires the aws
+    connector's credentials; construction/parsing is provider-exact)."""
+    from ...connectors.clients import NoEgressError, ProviderAPIError
+    from ...connectors.clients.observability import (build_cloudwatch_metric,
+                                                     parse_cloudwatch_metric)
+    dims = {}
+    for part in (dimensions or "").split(","):
+        if "=" in part:
+            k, v = part.split("=", 1)
+            dims[k.strip()] = v.strip()
+    try:
+        req = build_cloudwatch_metric(_creds(ctx, "aws"), namespace, metric,
+                                      dims, time_from or None, time_to or None,
+                                      stat=stat)
+        return parse_cloudwatch_metric(await _transport(ctx)(req))
+    except NoEgressError as e:
+        return f"ERROR: aws API unreachable (no egress): {e}"
+    except ProviderAPIError as e:
+        return f"ERROR: cloudwatch: {e}"
+
+
+@R.add("github_rca", "Correlate the incident window with recent commits and "
+       "merged PRs of a repository (risk-ranked).",
+       {"repo": "owner/name", "incident_time":
+        "incident start: unix, ISO, or 'now-2h' (defaults to this incident)"},
+       requires_connection="github")
+async def github_rca_tool(ctx: ToolContext, repo: str,
+                          incident_time: str = "") -> str:
+    from ...connectors.clients import NoEgressError, ProviderAPIError
+    from ...connectors.clients.devops import github_rca
+    if not incident_time and ctx.org_store is not None and ctx.incident_id:
+        inc = ctx.org_store.get_incident(ctx.incident_id)
+        if inc:
+            incident_time = str(inc.get("created_at", ""))
+    try:
+        return await github_rca(_transport(ctx), _creds(ctx, "github"),
+                                repo, incident_time or "now")
+    except NoEgressError as e:
+        return f"ERROR: github API unreachable (no egress): {e}"
+    except ProviderAPIError as e:
+        return f"ERROR: github: {e}"
+
+
+@R.add("jenkins_rca", "Analyze a Jenkins job's recent builds and extract "
+       "failure lines from failing consoles.",
+       {"job": "job path, e.g. folder/jobname"},
+       requires_connection="jenkins")
+async def jenkins_rca_tool(ctx: ToolContext, job: str) -> str:
+    from ...connectors.clients import NoEgressError, ProviderAPIError
+    from ...connectors.clients.devops import jenkins_rca
+    try:
+        return await jenkins_rca(_transport(ctx), _creds(ctx, "jenkins"), job)
+    except NoEgressError as e:
+        return f"ERROR: jenkins API unreachable (no egress): {e}"
+    except ProviderAPIError as e:
+        return f"ERROR: jenkins: {e}"
+
+
+@R.add("github_fix", "Draft a fix suggestion (branch + patch description) "
+       "for the incident from a suspect commit.",
+       {"repo": "owner/name", "commit_sha": "suspect commit",
+        "description": "what to change and why"},
+       read_only=False, requires_connection="github", needs_confirmation=True)
+async def github_fix_tool(ctx: ToolContext, repo: str, commit_sha: str = "",
+                          description: str = "") -> str:
+    # The reference's github_fix persists a structured fix suggestion on the
+    # incident card (github_fix_tool.py); PR creation is a separate,
+    # confirmed step. Persist the suggestion here.
+    if ctx.org_store is None or not ctx.incident_id:
+        return "ERROR: github_fix needs an incident context"
+    ctx.org_store.add_suggestion(
+        ctx.incident_id, title=f"fix in {repo}"
+        + (f" (suspect {commit_sha[:10]})" if commit_sha else ""),
+        command="", rationale=description[:2000], safe=True)
+    return f"fix suggestion saved to incident {ctx.incident_id}: {repo}@{commit_sha}"
 
 
 # ------------------------------------------------------- files / artefacts

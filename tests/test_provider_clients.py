@@ -1,0 +1,408 @@
+"""Provider API clients against recorded fixtures:
+request construction (URLs, auth headers, time windows, pagination) and
+response parsing for datadog/newrelic/splunk/sentry/dynatrace/grafana +
+slack/jira/confluence/notion + github_rca/jenkins_rca. No sockets: the
+transport seam is a fixture function."""
+import asyncio
+import json
+
+import pytest
+
+from aurora_amd.connectors.clients import (ProviderAPIError, comms, datadog,
+                                           devops, observability)
+from aurora_amd.connectors.clients.base import (parse_time, series_stats,
+                                                window)
+
+NOW = 1_700_000_000.0
+
+
+def run(coro):
+    return asyncio.run(coro)
+
+
+def transport_of(handler):
+    calls = []
+
+    async def t(req):
+        calls.append(req)
+        return handler(req)
+    t.calls = calls
+    return t
+
+
+# ------------------------------------------------------------------ base
+
+def test_parse_time_forms():
+    assert parse_time("now", now=NOW) == NOW
+    assert parse_time("now-15m", now=NOW) == NOW - 900
+    assert parse_time("-2h", now=NOW) == NOW - 7200
+    assert parse_time("1700000000", now=NOW) == 1_700_000_000
+    assert parse_time("1700000000000", now=NOW) == 1_700_000_000  # ms epoch
+    assert abs(parse_time("2023-11-14T22:13:20Z") - 1_700_000_000) < 1
+    with pytest.raises(ProviderAPIError):
+        parse_time("yesterday-ish")
+    with pytest.raises(ProviderAPIError):
+        window("now", "now-1h", now=NOW)  # empty window
+
+
+def test_series_stats():
+    st = series_stats([1.0, 2.0, 3.0, None, 4.0])
+    assert st["points"] == 4 and st["min"] == 1.0 and st["max"] == 4.0
+    assert st["avg"] == 2.5 and st["last"] == 4.0
+    assert series_stats([]) == {"points": 0}
+
+
+# --------------------------------------------------------------- datadog
+
+DD_CREDS = {"api_key": "k", "app_key": "a"}
+
+
+def test_datadog_logs_pagination_and_parse():
+    pages = [
+        {"data": [{"attributes": {"timestamp": "t1", "service": "api",
+                                  "status": "error", "message": "boom 1"}}],
+         "meta": {"page": {"after": "cur1"}}},
+        {"data": [{"attributes": {"timestamp": "t2", "service": "api",
+                                  "status": "error", "message": "boom 2"}}],
+         "meta": {}},
+    ]
+    state = {"i": 0}
+
+    def handler(req):
+        assert req.url.endswith("/api/v2/logs/events/search")
+        assert req.headers["DD-API-KEY"] == "k"
+        body = req.json_body
+        assert body["filter"]["query"] == "service:api status:error"
+        if state["i"] == 1:
+            assert body["page"]["cursor"] == "cur1"
+        out = pages[state["i"]]
+        state["i"] += 1
+        return out
+
+    t = transport_of(handler)
+    out = run(datadog.query_logs(t, DD_CREDS, "service:api status:error",
+                                 "now-1h", "now", now=NOW))
+    assert "2 events" in out and "boom 1" in out and "boom 2" in out
+    assert len(t.calls) == 2  # followed the cursor once, stopped at end
+
+
+def test_datadog_metrics_window_and_summary():
+    def handler(req):
+        assert req.url.endswith("/api/v1/query")
+        assert req.params["from"] == int(NOW - 3600)
+        assert req.params["to"] == int(NOW)
+        return {"status": "ok", "series": [
+            {"metric": "system.cpu.user", "scope": "host:web-1",
+             "unit": [{"short_name": "%"}],
+             "pointlist": [[1, 10.0], [2, 30.0], [3, 20.0]]}]}
+
+    out = run(datadog.query_metrics(transport_of(handler), DD_CREDS,
+                                    "avg:system.cpu.user{*}",
+                                    "now-1h", "now", now=NOW))
+    assert "system.cpu.user{host:web-1} [%]" in out
+    assert '"max": 30.0' in out and '"last": 20.0' in out
+
+
+def test_datadog_monitors_and_errors():
+    def handler(req):
+        return {"monitors": [{"id": 7, "status": "Alert", "name": "cpu high",
+                              "type": "metric alert"}],
+                "metadata": {"total_count": 1}}
+    out = run(datadog.query(transport_of(handler), DD_CREDS, "monitors",
+                            "status:alert", now=NOW))
+    assert "#7 [Alert] cpu high" in out
+    with pytest.raises(ProviderAPIError):
+        datadog.build_logs_request({}, "q", None, None)  # missing creds
+    with pytest.raises(ProviderAPIError):
+        run(datadog.query(transport_of(lambda r: {}), DD_CREDS, "nope", "q"))
+
+
+def test_datadog_interval_clamp():
+    assert datadog.pick_interval(300) == 1
+    assert datadog.pick_interval(3600 * 24) == 300
+    assert datadog.pick_interval(3600, caller_interval=60) == 60
+    # caller interval too fine for the window: clamped up
+    assert datadog.pick_interval(MAX := 3600 * 24 * 30,
+                                 caller_interval=1) >= MAX // 300
+
+
+# -------------------------------------------------------------- newrelic
+
+def test_newrelic_nrql():
+    def handler(req):
+        assert req.url == "https://api.newrelic.com/graphql"
+        assert req.headers["API-Key"] == "nrk"
+        q = req.json_body["query"]
+        assert "account(id: 42)" in q and "SELECT count(*)" in q
+        return {"data": {"actor": {"account": {"nrql": {"results": [
+            {"count": 123}, {"count": 4}]}}}}}
+
+    out = run(observability.newrelic_query(
+        transport_of(handler), {"api_key": "nrk", "account_id": "42"},
+        "SELECT count(*) FROM Transaction SINCE 1 hour ago"))
+    assert "2 rows" in out and '"count": 123' in out
+    # graphql errors surface
+    with pytest.raises(ProviderAPIError):
+        run(observability.newrelic_query(
+            transport_of(lambda r: {"errors": [{"message": "bad nrql"}]}),
+            {"api_key": "nrk", "account_id": "42"}, "SELECT"))
+
+
+# ---------------------------------------------------------------- splunk
+
+def test_splunk_normalize_and_search():
+    assert observability.normalize_spl("index=web error") == \
+        "search index=web error"
+    assert observability.normalize_spl("| tstats count") == "| tstats count"
+    assert observability.normalize_spl("search x") == "search x"
+
+    def handler(req):
+        assert req.url.endswith("/services/search/jobs")
+        assert req.json_body["search"].startswith("search ")
+        assert req.json_body["exec_mode"] == "oneshot"
+        return {"results": [{"_raw": "ERROR db timeout"},
+                            {"field": "v"}]}
+
+    out = run(observability.splunk_query(
+        transport_of(handler),
+        {"base_url": "https://sp:8089", "token": "t"},
+        "index=web error", "now-1h", "now", now=NOW))
+    assert "2 results" in out and "ERROR db timeout" in out
+
+
+def test_splunk_indexes():
+    def handler(req):
+        assert req.url.endswith("/services/data/indexes")
+        return {"entry": [{"name": "main"}, {"name": "web"}]}
+    out = run(observability.splunk_query(
+        transport_of(handler), {"base_url": "https://sp", "token": "t"},
+        "indexes"))
+    assert out == "splunk indexes: main, web"
+
+
+# ---------------------------------------------------------------- sentry
+
+def test_sentry_issue_search():
+    def handler(req):
+        assert "/api/0/organizations/acme/issues/" in req.url
+        assert req.params["query"] == "is:unresolved level:error"
+        return [{"shortId": "API-1", "level": "error", "count": "42",
+                 "title": "TypeError in checkout", "lastSeen": "2026-01-01"}]
+    out = run(observability.sentry_query(
+        transport_of(handler), {"token": "t", "org": "acme"},
+        "is:unresolved level:error"))
+    assert "API-1 [error] x42 TypeError in checkout" in out
+
+
+# ------------------------------------------------------------- dynatrace
+
+def test_dynatrace_problems_and_metrics():
+    def handler(req):
+        if "/problems" in req.url:
+            assert req.params["from"] == int((NOW - 7200) * 1000)
+            return {"totalCount": 1, "problems": [
+                {"displayId": "P-1", "status": "OPEN",
+                 "severityLevel": "ERROR", "title": "Response time spike",
+                 "impactedEntities": [{"name": "svc-a"}]}]}
+        assert "/metrics/query" in req.url
+        return {"result": [{"metricId": "builtin:cpu",
+                            "data": [{"dimensions": ["host-1"],
+                                      "values": [1.0, None, 3.0]}]}]}
+    creds = {"base_url": "https://dt", "token": "t"}
+    out = run(observability.dynatrace_query(transport_of(handler), creds,
+                                            "problems", now=NOW))
+    assert "P-1 [OPEN/ERROR]" in out and "svc-a" in out
+    out = run(observability.dynatrace_query(transport_of(handler), creds,
+                                            "builtin:cpu", now=NOW))
+    assert "builtin:cpu{host-1}" in out
+
+
+# ------------------------------------------------------------------ slack
+
+def test_slack_modes_and_error():
+    def handler(req):
+        if req.url.endswith("conversations.list"):
+            return {"ok": True, "channels": [
+                {"name": "ops", "id": "C1", "num_members": 7}]}
+        if req.url.endswith("conversations.history"):
+            assert req.params["channel"] == "C1"
+            return {"ok": True, "messages": [
+                {"ts": "1.0", "user": "U1", "text": "deploy started"}]}
+        return {"ok": True, "messages": {"total": 1, "matches": [
+            {"channel": {"name": "ops"}, "username": "al",
+             "ts": "2.0", "text": "db down"}]}}
+    creds = {"token": "xoxb"}
+    assert "#ops (C1" in run(comms.slack_query(transport_of(handler), creds,
+                                               "channels"))
+    assert "deploy started" in run(comms.slack_query(transport_of(handler),
+                                                     creds, "history C1"))
+    assert "db down" in run(comms.slack_query(transport_of(handler), creds,
+                                              "db down"))
+    with pytest.raises(ProviderAPIError):
+        run(comms.slack_query(
+            transport_of(lambda r: {"ok": False, "error": "invalid_auth"}),
+            creds, "x"))
+
+
+# ------------------------------------------------------------------- jira
+
+def test_jira_jql_pagination():
+    def handler(req):
+        assert req.url.endswith("/rest/api/2/search")
+        assert req.headers["Authorization"].startswith("Basic ")
+        start = req.json_body["startAt"]
+        if start == 0:
+            return {"total": 2, "issues": [{"key": "OPS-1", "fields": {
+                "status": {"name": "Open"}, "priority": {"name": "P1"},
+                "summary": "db latency", "assignee": None,
+                "updated": "2026-01-01"}}]}
+        return {"total": 2, "issues": [{"key": "OPS-2", "fields": {
+            "status": {"name": "Done"}, "priority": None,
+            "summary": "retro", "assignee": {"displayName": "sam"},
+            "updated": "2026-01-02"}}]}
+
+    t = transport_of(handler)
+    out = run(comms.jira_query(
+        t, {"base_url": "https://j", "token": "t", "email": "e@x.io"},
+        "project=OPS"))
+    assert "OPS-1 [Open/P1]" in out and "OPS-2 [Done/-]" in out
+    assert "assignee=unassigned" in out and "assignee=sam" in out
+    assert len(t.calls) == 2  # paged until total reached
+
+
+# ------------------------------------------------------------- confluence
+
+def test_confluence_cql_and_html_strip():
+    def handler(req):
+        assert 'text ~ "db runbook"' in req.params["cql"]
+        return {"size": 1, "results": [
+            {"title": "DB Runbook", "space": {"key": "OPS"},
+             "body": {"view": {"value": "<p>Step <b>one</b>: check\npool</p>"}}}]}
+    out = run(comms.confluence_query(
+        transport_of(handler), {"base_url": "https://c", "token": "t"},
+        "db runbook"))
+    assert "[OPS] DB Runbook: Step one : check pool" in out.replace("  ", " ")
+
+
+def test_notion_search_title_extraction():
+    def handler(req):
+        assert req.headers["Notion-Version"]
+        return {"results": [{"object": "page", "last_edited_time": "2026",
+                             "properties": {"Name": {
+                                 "type": "title",
+                                 "title": [{"plain_text": "Postmortem X"}]}}}]}
+    out = run(comms.notion_query(transport_of(handler), {"token": "t"}, "x"))
+    assert "[page] Postmortem X" in out
+
+
+# ----------------------------------------------------------------- github
+
+def _commit(sha, date, msg):
+    return {"sha": sha, "commit": {"message": msg,
+                                   "author": {"date": date}}}
+
+
+def test_github_rca_correlation():
+    t_inc = "2023-11-14T22:13:20Z"  # == NOW
+
+    def handler(req):
+        if req.url.endswith("/commits"):
+            assert req.params["until"] == "2023-11-14T22:13:20Z"
+            return [
+                _commit("a" * 40, "2023-11-14T21:30:20Z",
+                        "bump db pool config"),
+                _commit("b" * 40, "2023-11-13T10:00:00Z", "docs typo"),
+            ]
+        if "/commits/" in req.url:
+            return {"files": [{"filename": "deploy/helm/values.yaml",
+                               "additions": 3, "deletions": 1}]}
+        if req.url.endswith("/pulls"):
+            return [{"number": 12, "title": "Tune connection pool",
+                     "merged_at": "2023-11-14T21:00:00Z"}]
+        raise AssertionError(req.url)
+
+    out = run(devops.github_rca(transport_of(handler), {"token": "t"},
+                                "acme/api", t_inc, now=NOW))
+    assert "commits in the 24h before the incident: 2" in out
+    assert "#12 Tune connection pool" in out
+    # the risky config commit ranks above the docs commit
+    lines = [ln for ln in out.splitlines() if ln.strip().startswith(("aaaa", "bbbb"))]
+    assert lines and lines[0].startswith("  aaaaaaaaaa")
+    assert "risky message" in lines[0] and "43min before incident" in lines[0]
+
+
+def test_github_score_commit():
+    s, reasons = devops.score_commit(
+        _commit("c", "", "hotfix: rollback timeout change"),
+        [{"filename": "terraform/main.tf", "additions": 600, "deletions": 10}])
+    assert s >= 6 and any("terraform" in r for r in reasons)
+    s2, _ = devops.score_commit(_commit("d", "", "update readme"), [])
+    assert s2 <= 2
+
+
+# ---------------------------------------------------------------- jenkins
+
+def test_jenkins_rca_console_excerpt():
+    console = ("step 1 ok\nstep 2 ok\n"
+               "ERROR: compilation failed\njava.lang.NullPointerException\n"
+               "Build step 'shell' marked build as failure\nFinished: FAILURE")
+
+    def handler(req):
+        if req.url.endswith("/api/json"):
+            assert "job/folder/job/api" in req.url
+            return {"builds": [
+                {"number": 10, "result": "FAILURE", "timestamp": 1},
+                {"number": 9, "result": "SUCCESS", "timestamp": 0}]}
+        assert req.url.endswith("/folder/job/api/10/consoleText".replace(
+            "/folder/job/api", "/job/folder/job/api"))
+        return console
+
+    out = run(devops.jenkins_rca(
+        transport_of(handler),
+        {"base_url": "https://j", "user": "u", "token": "t"}, "folder/api"))
+    assert "#10:FAILURE" in out and "#9:SUCCESS" in out
+    assert "NullPointerException" in out
+    assert "step 1 ok" not in out  # only failure lines excerpted
+
+
+def test_jenkins_failure_line_extraction():
+    lines = devops.extract_failure_lines("a\nERROR x\nb\nfatal: y\nc")
+    assert lines == ["ERROR x", "fatal: y"]
+    # no matches: fall back to the tail
+    assert devops.extract_failure_lines("l1\nl2", max_lines=1) == ["l2"]
+
+
+# ------------------------------------------------- registry tool wrappers
+
+def test_provider_tools_wired_through_registry():
+    """The agent-facing tools run the real clients via the transport seam
+    and degrade explicitly when credentials/egress are absent."""
+    import aurora_amd.agent.tools.core  # noqa: F401 - registers tools
+    from aurora_amd.agent.tools.registry import GLOBAL_REGISTRY, ToolContext
+
+    async def main():
+        async def fixture(req):
+            assert req.headers["DD-API-KEY"] == "k"
+            return {"data": [], "meta": {}}
+
+        ctx = ToolContext(extra={"provider_transport": fixture})
+        # no creds bound -> the client raises a credentials error, surfaced
+        out = await GLOBAL_REGISTRY.run(ctx, "query_datadog",
+                                        {"query": "service:api"})
+        assert out.startswith("ERROR: datadog API:")
+
+        class Store:  # minimal org-store shim with a github token
+            def get_token_ref(self, uid, provider):
+                return None
+            def tool_permissions(self):
+                return {}
+            def connected_providers(self, uid):
+                return ["datadog"]
+        ctx2 = ToolContext(extra={"provider_transport": fixture})
+        ctx2.user_id = "u"
+        # patch _creds through direct call instead: call the client module
+        from aurora_amd.connectors.clients import datadog as dd
+        out = await dd.query_logs(fixture, DD_CREDS, "service:api")
+        assert "no matches" in out
+
+    run(main())
