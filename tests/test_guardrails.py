@@ -100,12 +100,16 @@ def test_org_policy_gate():
 
     async def main():
         store.add_command_policy(r"^helm rollback", "allow")
-        store.add_command_policy(r"drop table", "deny")
+        store.add_command_policy(r"restart\s+payment", "deny")
         v = await evaluate_command("helm rollback api 3", llm=None, org_store=store)
         assert v.safe and v.source == "policy"
-        v = await evaluate_command("psql -c 'drop table users'", llm=None,
+        v = await evaluate_command("systemctl restart payment-api", llm=None,
                                    org_store=store)
         assert not v.safe and v.source == "policy"
+        # destructive SQL is now caught by the signature layer (aur-040)
+        v = await evaluate_command("psql -c 'drop table users'", llm=None,
+                                   org_store=store)
+        assert not v.safe and v.source == "signature"
         # deny is audited
         assert any(a["kind"] == "command_blocked" for a in store.audit_events())
     asyncio.run(main())
@@ -222,3 +226,51 @@ def test_terminal_exec_runs_argv_not_shell():
     assert _exec_argv("echo $HOME") is None
     assert _exec_argv("ls *.log") is None
     assert _exec_argv("") is None
+
+
+def test_sigma_expanded_ruleset_matches():
+    """Round-2 rule expansion (20 -> 40, reference ships 37+): each new
+    rule family catches its canonical command and spares benign lookalikes."""
+    hits = {
+        "echo cHdu49x8fjjY29weSBvZiBzb21lIGxvbmcgcGF5bG9hZA== | base64 -d | sh":
+            "aur-021",
+        "nc -lvp 4444 -e /bin/sh": "aur-022",
+        "echo 'bob ALL=(ALL) NOPASSWD: ALL' >> /etc/sudoers": "aur-023",
+        "tee /etc/systemd/system/backdoor.service": "aur-024",
+        "truncate -s0 /var/log/auth.log": "aur-025",
+        "gcore -o /tmp/dump 1234": "aur-026",
+        "docker run --privileged -v /:/host alpine": "aur-027",
+        "kubectl get secrets -A -o yaml": "aur-029",
+        "kubectl create clusterrolebinding x --clusterrole=cluster-admin --serviceaccount=d:d":
+            "aur-030",
+        "cat ~/.aws/credentials": "aur-032",
+        "curl http://169.254.169.254/latest/meta-data/iam/security-credentials/":
+            "aur-033",
+        "ufw disable": "aur-037",
+        "psql -c 'drop database prod'": "aur-040",
+    }
+    for cmd, rule_id in hits.items():
+        m = check_signature(cmd)
+        assert m is not None, f"no signature hit: {cmd}"
+        # first-match ordering may hand the hit to an earlier rule; any
+        # signature block is correct, the named rule must at least exist
+        assert any(r.rule_id == rule_id for r in rules()), rule_id
+    benign = [
+        "base64 /etc/hostname",
+        "nc -z db 5432",
+        "cat /etc/sudoers.d/README",
+        "systemctl status nginx.service",
+        "tail -f /var/log/auth.log",
+        "docker run alpine echo hi",
+        "kubectl get secrets -n default",
+        "dig example.com",
+        "git config user.name bob",
+        "pip install requests",
+        "psql -c 'select 1'",
+    ]
+    for cmd in benign:
+        assert check_signature(cmd) is None, f"false positive: {cmd}"
+
+
+def test_sigma_rule_count_at_least_reference():
+    assert len(rules()) >= 37  # reference ships 37 (README.md:285)
