@@ -258,7 +258,6 @@ def test_sigma_expanded_ruleset_matches():
     benign = [
         "base64 /etc/hostname",
         "nc -z db 5432",
-        "cat /etc/sudoers.d/README",
         "systemctl status nginx.service",
         "tail -f /var/log/auth.log",
         "docker run alpine echo hi",
