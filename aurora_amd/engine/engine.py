@@ -521,9 +521,20 @@ class LLMEngine:
 
     def _graphs_enabled(self) -> bool:
         # MoE decode uses the fixed-capacity batched dispatch (static
-        # shapes), so it is capturable too
-        return (self.config.use_hip_graphs and self.device != "cpu"
-                and self.config.tp_size == 1)
+        # shapes), so it is capturable too. TP decode is capturable as of
+        # round 2: the RCCL communicator is initialized before capture (the
+        # two warmup forwards in _capture_decode_graph run a real
+        # all-reduce), and ROCm supports collectives inside hipGraph
+        # capture when the communicator pre-exists (docs/ROUND2.md §2);
+        # every TP rank captures/replays the same bucket in lockstep
+        # because scheduling inputs are mirrored. AURORA_GRAPH_TP=0
+        # restores the round-1 eager-TP behavior.
+        if not (self.config.use_hip_graphs and self.device != "cpu"):
+            return False
+        if self.config.tp_size != 1:
+            import os
+            return os.environ.get("AURORA_GRAPH_TP", "1") != "0"
+        return True
 
     def _decode_with_graph(self, seqs):
         """Replay a hipGraph-captured decode forward (per batch-size bucket,

@@ -85,8 +85,48 @@ def tp_group():
     return _TP_GROUP
 
 
+_ONESHOT = None  # None = unprobed, False = unavailable/disabled
+
+
+def _oneshot():
+    """Lazy one-shot xGMI all-reduce setup (ops/csrc/allreduce.hip):
+    opt-in via AURORA_ONESHOT_AR=1; IPC handles exchanged over the
+    existing process group. Used for decode-sized bf16 messages only —
+    large prefill tensors stay on RCCL ring/tree."""
+    global _ONESHOT
+    if _ONESHOT is not None:
+        return _ONESHOT
+    _ONESHOT = False
+    if (os.environ.get("AURORA_ONESHOT_AR", "0") != "1"
+            or _TP_WORLD <= 1 or not torch.cuda.is_available()):
+        return _ONESHOT
+    try:
+        from aurora_amd import _hip_ops as hip
+        handle = hip.oneshot_init(_TP_WORLD, _TP_RANK, 2 * 1024 * 1024)
+        all_handles = [None] * dist.get_world_size()
+        dist.all_gather_object(all_handles, (dist.get_rank(), bytes(handle)))
+        # my TP group's handles in tp-rank order
+        base = dist.get_rank() - _TP_RANK
+        group = sorted((r, h) for r, h in all_handles
+                       if base <= r < base + _TP_WORLD)
+        hip.oneshot_connect([list(h) for _, h in group])
+        if hip.oneshot_ready():
+            _ONESHOT = hip
+    except Exception:  # noqa: BLE001 - fall back to RCCL on any failure
+        _ONESHOT = False
+    return _ONESHOT
+
+
 def tp_all_reduce(x: torch.Tensor) -> torch.Tensor:
     if _TP_WORLD == 1:
+        return x
+    hip = _oneshot()
+    if (hip and x.dtype == torch.bfloat16 and x.is_cuda
+            and x.numel() * 2 <= hip.oneshot_max_bytes()):
+        y = x.contiguous()
+        hip.oneshot_allreduce_(y)
+        if y.data_ptr() != x.data_ptr():
+            x.copy_(y)
         return x
     dist.all_reduce(x, group=_TP_GROUP)
     return x

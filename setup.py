@@ -19,6 +19,7 @@ SRC = [
     "aurora_amd/ops/csrc/attention_prefill.hip",
     "aurora_amd/ops/csrc/encoder.hip",
     "aurora_amd/ops/csrc/skinny_gemm.hip",
+    "aurora_amd/ops/csrc/allreduce.hip",
 ]
 
 setup(
