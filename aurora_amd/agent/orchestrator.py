@@ -164,7 +164,7 @@ class Orchestrator:
         try:
             res = await self.llm.invoke(
                 prompt, SamplingParams(temperature=0.1, max_tokens=220,
-                                       grammar=JsonGrammar()),
+                                       grammar=JsonGrammar(soft_close_bytes=220)),
                 purpose="orchestrator")
             doc = json.loads(res.text)
             return doc if isinstance(doc, dict) else {}

@@ -17,7 +17,7 @@ import json
 import time
 from typing import Any, AsyncIterator, Dict, List, Optional
 
-from ..engine.grammar import JsonGrammar, ToolCallGrammar
+from ..engine.grammar import JsonGrammar, ReActTurnGrammar
 from ..engine.sampling_params import SamplingParams
 from ..guardrails.input_rail import check_input
 from ..utils.config import settings as get_settings
@@ -109,47 +109,45 @@ class Workflow:
             self._inject_context_updates(state)
             prompt = (system + "\n\n# Investigation so far\n"
                       + render_history(state.messages) + "\nThought:")
-            # --- thought (streamed) ---
-            thought_parts: List[str] = []
-            fault: Optional[str] = None
-            async for out in self.llm.server.stream(
-                    self.llm.tokenizer.encode(prompt, add_bos=True),
-                    SamplingParams(temperature=0.3, max_tokens=160,
-                                   stop=["\nAction", "\nObservation", "\nThought"],
-                                   ignore_eos=False),
-                    session_id=state.session_id):
-                if out.new_text:
-                    thought_parts.append(out.new_text)
-                    yield {"type": "token", "content": out.new_text}
-                if out.finished and out.finish_reason and (
-                        out.finish_reason.startswith("engine step failed")
-                        or out.finish_reason == "engine shutdown"):
-                    fault = out.finish_reason
-            if fault:
-                # an engine fault mid-thought must fail the investigation
-                # (-> queue retry), not spin out max_turns of empty turns
-                raise RuntimeError(fault)
-            thought = "".join(thought_parts).strip()
-            state.messages.append({"role": "thought", "content": thought})
-            yield {"type": "thought", "content": thought}
-            if store is not None and state.incident_id:
-                store.add_thought(state.incident_id, thought)
-
-            # --- action: grammar-constrained tool JSON. ToolCallGrammar
-            # makes an unknown tool or malformed call unrepresentable
-            # (forced tool choice, reference: middleware/force_tool.py),
-            # so one attempt normally parses; retries guard pathological
-            # sampling only.
-            act_prompt = prompt + " " + thought + "\nAction:"
+            # --- fused thought + action: ONE engine request per turn.
+            # ReActTurnGrammar streams free thought text, then pins the
+            # '\nAction: {"tool":...}' transition with the tool name drawn
+            # from the registered set (forced tool choice, reference:
+            # middleware/force_tool.py) — an unknown tool or malformed
+            # call is unrepresentable, and the action needs no second
+            # prefill of prompt+thought.
+            turn_text = ""
             action_raw = ""
             action: Optional[dict] = None
-            for attempt in range(3):  # retry malformed/unknown (agent.py:1031)
-                res = await self.llm.invoke(
-                    act_prompt,
-                    SamplingParams(temperature=0.1 + 0.3 * attempt, max_tokens=256,
-                                   grammar=ToolCallGrammar(sorted(tool_names))),
-                    purpose="main", session_id=state.session_id)
-                action_raw = res.text.strip()
+            thought = ""
+            for attempt in range(3):  # retries guard pathological sampling
+                parts: List[str] = []
+                fault: Optional[str] = None
+                async for out in self.llm.server.stream(
+                        self.llm.tokenizer.encode(prompt, add_bos=True),
+                        SamplingParams(
+                            temperature=0.3 + 0.2 * attempt, max_tokens=300,
+                            grammar=ReActTurnGrammar(sorted(tool_names),
+                                                     thought_bytes=160),
+                            ignore_eos=False),
+                        session_id=state.session_id):
+                    if out.new_text:
+                        parts.append(out.new_text)
+                        yield {"type": "token", "content": out.new_text}
+                    if out.finished and out.finish_reason and (
+                            out.finish_reason.startswith("engine step failed")
+                            or out.finish_reason == "engine shutdown"):
+                        fault = out.finish_reason
+                if fault:
+                    # an engine fault mid-thought must fail the investigation
+                    # (-> queue retry), not spin out max_turns of empty turns
+                    raise RuntimeError(fault)
+                turn_text = "".join(parts)
+                head, sep, tail = turn_text.rpartition("\nAction:")
+                if not sep:
+                    continue
+                thought = head.strip()
+                action_raw = tail.strip()
                 try:
                     doc = json.loads(action_raw)
                 except json.JSONDecodeError:
@@ -157,8 +155,10 @@ class Workflow:
                 if isinstance(doc, dict) and doc.get("tool") in tool_names:
                     action = doc
                     break
-                act_prompt += (f" {action_raw}\nObservation: ERROR unknown tool; "
-                               f"valid tools: {sorted(tool_names)}\nAction:")
+            state.messages.append({"role": "thought", "content": thought})
+            yield {"type": "thought", "content": thought}
+            if store is not None and state.incident_id:
+                store.add_thought(state.incident_id, thought)
             if action is None:
                 state.messages.append({
                     "role": "observation",

@@ -37,13 +37,19 @@ class JsonGrammar:
     """Incremental JSON validator; max_depth bounds container nesting."""
 
     def __init__(self, max_depth: int = 16, require_object: bool = True,
-                 compact: bool = True):
+                 compact: bool = True, soft_close_bytes: int = 0):
         # compact: reject whitespace outside strings — JSON allows unbounded
         # inter-token whitespace, which lets a badly-steered model spin on
         # '\r' forever without completing; generation uses compact=True,
         # validation compact=False.
+        # soft_close_bytes > 0: past that many emitted bytes the sampler is
+        # steered toward closing the document (allowed_next_bytes narrows to
+        # closers) so generation completes before max_tokens instead of
+        # truncating mid-string. Legality is unchanged.
         self.compact = compact
         self.max_depth = max_depth
+        self.soft_close_bytes = soft_close_bytes
+        self.nbytes = 0
         self.stack: List[int] = []  # b'{' or b'[' markers (as ints)
         self.state = _VALUE
         self.lit: bytes = b""
@@ -71,6 +77,7 @@ class JsonGrammar:
         for b in data:
             if not self._advance_byte(b):
                 return False
+            self.nbytes += 1
         return True
 
     def allowed_next_bytes(self):
@@ -79,6 +86,16 @@ class JsonGrammar:
         sampler pick from tiny per-byte candidate buckets instead of
         scanning the vocabulary when the FSM pins the continuation.
         Computed for compact generation (no inter-token whitespace)."""
+        out = self._allowed_raw()
+        if self.soft_close_bytes and self.nbytes >= self.soft_close_bytes:
+            if out is None:  # inside a string: steer to the closing quote
+                return b'"' if self.state == _STRING else out
+            closers = bytes(c for c in out if c in b'"}]')
+            if closers:
+                return closers
+        return out
+
+    def _allowed_raw(self):
         s = self.state
         if s == _DONE:
             return b""
@@ -285,6 +302,87 @@ class JsonGrammar:
 def validate_json_bytes(data: bytes) -> bool:
     g = JsonGrammar(require_object=False, compact=False)
     return g.advance(data) and g.is_complete()
+
+
+class ReActTurnGrammar:
+    """One grammar for a whole ReAct turn: free-text *thought* bytes, then
+    the literal ``\\nAction: `` transition, then a :class:`ToolCallGrammar`
+    tool call. Lets the agent decode thought+action in a SINGLE engine
+    request (one prefill, one stream) instead of two — the round-1
+    workflow re-prefilled the prompt for the action call and paid an extra
+    host round-trip per turn.
+
+    The thought phase is unconstrained; past ``thought_bytes`` the sampler
+    is steered onto the transition literal (legality unchanged). The
+    emitted text is ``<thought>\\nAction: {"tool":...,"args":{...}}``.
+    """
+
+    LIT = b"\nAction:"
+
+    def __init__(self, tools: List[str], thought_bytes: int = 400,
+                 tool_soft_close: int = 160, max_args_depth: int = 8):
+        self.tool = ToolCallGrammar(tools, max_args_depth=max_args_depth,
+                                    soft_close_bytes=tool_soft_close)
+        self.thought_bytes = thought_bytes
+        self.phase = 0      # 0 = thought, 1 = tool call
+        self.lit_pos = 0    # progress through LIT (match or forced)
+        self.sp_ok = True   # one optional space after the colon
+        self.nbytes = 0
+
+    def fork(self) -> "ReActTurnGrammar":
+        g = ReActTurnGrammar.__new__(ReActTurnGrammar)
+        g.tool = self.tool.fork()
+        g.thought_bytes = self.thought_bytes
+        g.phase = self.phase
+        g.lit_pos = self.lit_pos
+        g.sp_ok = self.sp_ok
+        g.nbytes = self.nbytes
+        return g
+
+    def is_complete(self) -> bool:
+        return self.phase == 1 and self.tool.is_complete()
+
+    def tool_name(self) -> str:
+        return self.tool.tool_name()
+
+    def advance(self, data: bytes) -> bool:
+        for b in data:
+            if not self._advance_byte(b):
+                return False
+            self.nbytes += 1
+        return True
+
+    def _advance_byte(self, b: int) -> bool:
+        if self.phase == 1:
+            if self.sp_ok and b == 0x20:
+                self.sp_ok = False
+                return True
+            self.sp_ok = False
+            ok = self.tool._advance_byte(b)
+            if ok:
+                self.tool.nbytes += 1  # keep its soft-close budget live
+            return ok
+        # thought phase: every byte is legal; track the transition literal
+        if b == self.LIT[self.lit_pos]:
+            self.lit_pos += 1
+            if self.lit_pos == len(self.LIT):
+                self.phase = 1
+        elif b == self.LIT[0]:
+            self.lit_pos = 1
+        else:
+            self.lit_pos = 0
+        return True
+
+    def allowed_next_bytes(self):
+        if self.phase == 1:
+            inner = self.tool.allowed_next_bytes()
+            if self.sp_ok and inner is not None:
+                return bytes({0x20} | set(inner))
+            return inner
+        if self.nbytes >= self.thought_bytes:
+            # over the thought budget: steer onto the transition literal
+            return bytes([self.LIT[self.lit_pos]])
+        return None
 
 
 class ToolCallGrammar:

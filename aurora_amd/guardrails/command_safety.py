@@ -111,7 +111,8 @@ async def evaluate_command(command: str, llm=None, org_store=None,
     try:
         res = await llm.complete_text(
             JUDGE_PROMPT.format(command=command[:2000]),
-            SamplingParams(temperature=0.0, max_tokens=120, grammar=JsonGrammar()),
+            SamplingParams(temperature=0.0, max_tokens=120,
+                           grammar=JsonGrammar(soft_close_bytes=80)),
         )
         doc = json.loads(res.text)
         safe = bool(doc.get("safe", False))

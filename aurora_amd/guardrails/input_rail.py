@@ -78,7 +78,8 @@ async def check_input(text: str, llm=None, enabled: bool = True,
         if use_llm and llm is not None:
             res = await llm.complete_text(
                 RAIL_PROMPT.format(text=text[:4000]),
-                SamplingParams(temperature=0.0, max_tokens=100, grammar=JsonGrammar()),
+                SamplingParams(temperature=0.0, max_tokens=100,
+                               grammar=JsonGrammar(soft_close_bytes=70)),
             )
             doc = json.loads(res.text)
             if not bool(doc.get("allowed", False)):

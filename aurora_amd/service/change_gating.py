@@ -74,7 +74,8 @@ async def investigate_pr(rt, args: Dict[str, Any]) -> dict:
         res = await rt.llm(org_id).invoke(
             REVIEW_PROMPT.format(pr=json.dumps(pr)[:2000],
                                  hot_services=hot[:10], services=services),
-            SamplingParams(temperature=0.1, max_tokens=250, grammar=JsonGrammar()),
+            SamplingParams(temperature=0.1, max_tokens=250,
+                           grammar=JsonGrammar(soft_close_bytes=190)),
             purpose="rca")
         verdict = json.loads(res.text)
     except Exception:  # noqa: BLE001

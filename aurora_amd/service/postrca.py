@@ -71,7 +71,8 @@ async def generate_recommendations(llm, store, incident_id: str,
     try:
         res = await llm.invoke(
             RECO_PROMPT.format(transcript=render_transcript(messages, 8000)),
-            SamplingParams(temperature=0.2, max_tokens=400, grammar=JsonGrammar()),
+            SamplingParams(temperature=0.2, max_tokens=400,
+                           grammar=JsonGrammar(soft_close_bytes=320)),
             purpose="rca")
         doc = json.loads(res.text)
         raw = doc.get("suggestions", [])

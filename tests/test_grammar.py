@@ -150,3 +150,55 @@ def test_tool_call_grammar_engine_end_to_end():
         doc = _json.loads(text)
         assert doc["tool"] in tools
         assert isinstance(doc["args"], dict)
+
+
+def test_react_turn_grammar_shapes():
+    from aurora_amd.engine.grammar import ReActTurnGrammar
+    g = ReActTurnGrammar(["kubectl", "final_answer"], thought_bytes=300)
+    s = (b"The pod is crashlooping; check recent events."
+         b'\nAction: {"tool":"kubectl","args":{"args":"get events"}}')
+    assert g.advance(s) and g.is_complete()
+    assert g.tool_name() == "kubectl"
+    # no space after the colon also parses
+    g = ReActTurnGrammar(["kubectl"])
+    assert g.advance(b'x\nAction:{"tool":"kubectl","args":{}}')
+    assert g.is_complete()
+    # unknown tool is unrepresentable even after free text
+    g = ReActTurnGrammar(["kubectl"])
+    assert not g.advance(b'hm\nAction: {"tool":"rm_rf"')
+    # a partial "\nAction" inside the thought does not switch phases
+    g = ReActTurnGrammar(["kubectl"])
+    assert g.advance(b"\nAct later. More thought. ")
+    assert g.phase == 0
+
+
+def test_react_turn_grammar_soft_transition():
+    from aurora_amd.engine.grammar import ReActTurnGrammar
+    g = ReActTurnGrammar(["kubectl"], thought_bytes=10)
+    assert g.advance(b"0123456789extra")  # over budget, still legal
+    allowed = g.allowed_next_bytes()
+    assert allowed == b"\n"  # steered onto the literal
+    assert g.advance(b"\n")
+    assert g.allowed_next_bytes() == b"A"
+
+
+def test_react_turn_grammar_engine_end_to_end():
+    """A random-weight CPU engine under ReActTurnGrammar emits a parseable
+    fused turn every time."""
+    import json as _json
+    from aurora_amd.engine.engine import EngineConfig, LLMEngine
+    from aurora_amd.engine.grammar import ReActTurnGrammar
+    from aurora_amd.engine.sampling_params import SamplingParams
+
+    eng = LLMEngine(EngineConfig(model="tiny-llama", device="cpu", seed=11))
+    tools = ["kubectl", "final_answer", "list_incidents"]
+    outs = eng.generate(
+        [[5, 6, 7]],
+        SamplingParams(temperature=0.7, max_tokens=220,
+                       grammar=ReActTurnGrammar(tools, thought_bytes=60,
+                                                tool_soft_close=60)))
+    text = eng.tokenizer.decode(outs[0])
+    head, sep, tail = text.rpartition("\nAction:")
+    assert sep, text
+    doc = _json.loads(tail.strip())
+    assert doc["tool"] in tools and isinstance(doc["args"], dict)

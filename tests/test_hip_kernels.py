@@ -250,7 +250,8 @@ def test_engine_gpu_grammar_json():
 
     eng = LLMEngine(EngineConfig(model="smoke-llama", device="cuda", seed=5,
                                  kv_cache_pages=64))
-    p = SamplingParams(temperature=0.8, max_tokens=200, grammar=JsonGrammar())
+    p = SamplingParams(temperature=0.8, max_tokens=200,
+                       grammar=JsonGrammar(soft_close_bytes=140))
     out = eng.generate([[1, 2, 3]], p)[0]
     data = bytes(t for t in out if t < 256)
     assert validate_json_bytes(data), data

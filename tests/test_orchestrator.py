@@ -31,6 +31,13 @@ class ScriptedServer:
 
     async def stream(self, ids, params, session_id=None, request_id=None):
         from aurora_amd.engine.engine import StepOutput
+        g = params.grammar
+        if g is not None and hasattr(g, "thought_bytes") and self.script:
+            # fused ReAct turn: thought + scripted action JSON
+            yield StepOutput("x", [1],
+                             "thinking...\nAction: " + self.script.pop(0),
+                             True, "grammar")
+            return
         yield StepOutput("x", [1], "thinking...", True, "stop")
 
     async def complete_text(self, prompt, params, session_id=None):
