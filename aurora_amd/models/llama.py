@@ -115,6 +115,23 @@ class LlamaMLP(nn.Module):
         return self.down(ops.silu_mul(self.gate_up(x)))
 
 
+class _FullWidthMLP(nn.Module):
+    """Unsharded SiLU MLP for expert-parallel mode: the whole expert lives
+    on its owner rank (no intra-expert TP, no per-expert all-reduce)."""
+
+    def __init__(self, spec: ModelSpec, dtype):
+        super().__init__()
+        self.gate_up = nn.Parameter(torch.empty(
+            2 * spec.intermediate_size, spec.hidden_size, dtype=dtype),
+            requires_grad=False)
+        self.down = nn.Parameter(torch.empty(
+            spec.hidden_size, spec.intermediate_size, dtype=dtype),
+            requires_grad=False)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return ops.linear(ops.silu_mul(ops.linear(x, self.gate_up)), self.down)
+
+
 class MoEMLP(nn.Module):
     """Mixtral-style sparse MoE block (top-k routing over dense experts).
 
@@ -126,21 +143,45 @@ class MoEMLP(nn.Module):
 
     def __init__(self, spec: ModelSpec, dtype):
         super().__init__()
+        from ..parallel import ep as pep
         self.n_experts = spec.n_experts
         self.top_k = spec.experts_per_token
         self.router = nn.Linear(spec.hidden_size, spec.n_experts, bias=False, dtype=dtype)
         self.router.weight.requires_grad_(False)
-        self.experts = nn.ModuleList(
-            [LlamaMLP(spec, dtype) for _ in range(spec.n_experts)]
-        )
+        self.ep_size = pep.ep_size()
+        if self.ep_size > 1:
+            # expert parallelism (AURORA_EP=1): this rank owns
+            # n_experts/ep FULL-width experts; tokens travel to their
+            # experts over all-to-all instead of sharding each expert
+            assert spec.n_experts % self.ep_size == 0
+            self.e_local = spec.n_experts // self.ep_size
+            self.experts = nn.ModuleList(
+                [_FullWidthMLP(spec, dtype) for _ in range(self.e_local)])
+        else:
+            self.experts = nn.ModuleList(
+                [LlamaMLP(spec, dtype) for _ in range(spec.n_experts)]
+            )
         self._w_gu = None  # lazy stacked-weight cache for the batched path
         self._w_dn = None
+
+    def _local_expert_fn(self, xp: torch.Tensor,
+                         local_sel: torch.Tensor) -> torch.Tensor:
+        y = torch.empty_like(xp)
+        for le in range(self.e_local):
+            m = (local_sel == le)
+            if m.any():
+                y[m] = self.experts[le](xp[m])
+        return y
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         T, H = x.shape
         logits = self.router(x).float()  # [T, E]
         weights, sel = logits.topk(self.top_k, dim=-1)
         weights = weights.softmax(-1).to(x.dtype)  # [T, k]
+        if self.ep_size > 1:
+            from ..parallel import ep as pep
+            return pep.ep_moe_forward(x, weights, sel, self.n_experts,
+                                      self._local_expert_fn)
         if T <= 256:
             return self._forward_batched(x, weights, sel)
         # prefill: per-expert token batches are large, so plain per-expert
@@ -155,39 +196,48 @@ class MoEMLP(nn.Module):
 
     def _forward_batched(self, x: torch.Tensor, weights: torch.Tensor,
                          sel: torch.Tensor) -> torch.Tensor:
-        """Decode path: fixed-capacity batched dispatch — every op has a
-        static shape (bmm over [E, C, *]), so the step is hipGraph-capturable
-        and the per-expert kernel-launch storm (8 experts x 3 GEMMs x layers)
-        collapses into 3 bmms.  C = T  =>  no token is ever dropped (a token
-        appears at most once per expert)."""
+        """Decode path: tokens sorted by expert into a PACKED buffer
+        (no capacity padding), then per-expert grouped GEMMs via the
+        streaming moe_grouped_gemm kernel — unrouted experts skip their
+        whole weight stream, activations/silu run over A = T*top_k rows
+        instead of E*C (docs/ROUND2.md §4). Static shapes + device-read
+        counts keep the step hipGraph-capturable. CPU / no-extension
+        fallback: the round-1 fixed-capacity bmm dispatch."""
         T, H = x.shape
         E, k = self.n_experts, self.top_k
-        C = T
         flat = sel.reshape(-1)  # [T*k]
         order = flat.argsort(stable=True)  # tokens grouped by expert
         tok_of = order // k  # token index of each sorted assignment
-        # rank within the expert group = position - group start
-        # (bincount host-syncs to size its output -> not stream-capturable;
-        # a scatter_add histogram is)
+        # histogram via scatter_add (bincount host-syncs -> not capturable)
         counts = torch.zeros(E, dtype=torch.long, device=x.device).scatter_add_(
             0, flat, torch.ones_like(flat))
         starts = torch.cumsum(counts, 0) - counts
-        rank = torch.arange(T * k, device=x.device) - starts[flat[order]]
-        slot_idx = flat[order] * C + rank  # flat [E*C] position (rank < C always)
-        # gather tokens into the capacity buffer
-        xe = x.new_zeros(E * C, H)
-        xe[slot_idx] = x[tok_of]
-        xe = xe.view(E, C, H)
         if self._w_gu is None:  # stacked expert weights, cached (constant)
             self._w_gu = torch.stack([ex.gate_up.weight for ex in self.experts])
             self._w_dn = torch.stack([ex.down.weight for ex in self.experts])
         w_gu, w_dn = self._w_gu, self._w_dn
-        gu = torch.bmm(xe, w_gu.transpose(1, 2))  # [E, C, 2I']
-        act = ops.silu_mul(gu.reshape(E * C, -1)).view(E, C, -1)
-        ye = torch.bmm(act, w_dn.transpose(1, 2)).view(E * C, H)  # [E*C, H]
-        out = torch.zeros_like(x)
         wflat = weights.reshape(-1)[order].unsqueeze(1).to(x.dtype)
-        out.index_add_(0, tok_of, ye[slot_idx] * wflat)
+        if ops.has_moe_grouped(x):
+            xp = x[tok_of]  # [A, H] packed, expert-sorted
+            st32 = starts.to(torch.int32)
+            ct32 = counts.to(torch.int32)
+            gu = ops.moe_grouped_gemm(xp, w_gu, st32, ct32)     # [A, 2I']
+            act = ops.silu_mul(gu)                              # [A, I']
+            ye = ops.moe_grouped_gemm(act, w_dn, st32, ct32)    # [A, H]
+            out = torch.zeros_like(x)
+            out.index_add_(0, tok_of, ye * wflat)
+        else:
+            C = T
+            rank = torch.arange(T * k, device=x.device) - starts[flat[order]]
+            slot_idx = flat[order] * C + rank  # rank < C always
+            xe = x.new_zeros(E * C, H)
+            xe[slot_idx] = x[tok_of]
+            xe = xe.view(E, C, H)
+            gu = torch.bmm(xe, w_gu.transpose(1, 2))  # [E, C, 2I']
+            act = ops.silu_mul(gu.reshape(E * C, -1)).view(E, C, -1)
+            ye = torch.bmm(act, w_dn.transpose(1, 2)).view(E * C, H)
+            out = torch.zeros_like(x)
+            out.index_add_(0, tok_of, ye[slot_idx] * wflat)
         from ..parallel import dist as pdist
         return pdist.tp_all_reduce(out) if pdist.tp_world_size() > 1 else out
 

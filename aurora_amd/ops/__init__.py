@@ -205,6 +205,23 @@ def linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
     return F.linear(x, weight)
 
 
+def has_moe_grouped(x: torch.Tensor) -> bool:
+    """True when the packed MoE grouped-GEMM kernel serves this tensor."""
+    import os
+    return (x.is_cuda and x.dtype == torch.bfloat16 and _hip is not None
+            and x.shape[1] % 64 == 0
+            and os.environ.get("AURORA_MOE_GROUPED", "1") != "0")
+
+
+def moe_grouped_gemm(xp: torch.Tensor, w: torch.Tensor, starts: torch.Tensor,
+                     counts: torch.Tensor) -> torch.Tensor:
+    """Packed per-expert GEMM: yp[A, N] with tokens sorted by expert.
+    Requires per-expert counts <= 256 (the decode regime; prefill uses
+    per-expert dense GEMMs)."""
+    return _require_hip("moe_grouped_gemm").moe_grouped_gemm(
+        xp.contiguous(), w, starts, counts)
+
+
 def softmax_sample(logits, temperature, top_p, top_k, generator=None):
     # Sampling is [B, V] with B small; keep it in torch on both devices for now
     # (the GPU cost is one softmax over V — negligible next to the forward).

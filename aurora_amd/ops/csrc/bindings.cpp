@@ -28,6 +28,8 @@ torch::Tensor fused_rope_kv_write(torch::Tensor qkv, torch::Tensor positions,
 torch::Tensor skinny_gemm_ex(torch::Tensor x, torch::Tensor w, long s_target, long unroll);
 torch::Tensor skinny_gemm2_ex(torch::Tensor x, torch::Tensor w, long s_target);
 torch::Tensor skinny_gemm3_ex(torch::Tensor x, torch::Tensor w, long s_target);
+torch::Tensor moe_grouped_gemm(torch::Tensor xp, torch::Tensor w,
+                               torch::Tensor starts, torch::Tensor counts);
 std::vector<uint8_t> oneshot_init(long world, long rank, long max_bytes);
 void oneshot_connect(std::vector<std::vector<uint8_t>> handles);
 bool oneshot_ready();
@@ -53,6 +55,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("skinny_gemm_ex", &skinny_gemm_ex, "skinny GEMM with tuning knobs");
   m.def("skinny_gemm2_ex", &skinny_gemm2_ex, "LDS-staged skinny GEMM (M<=64)");
   m.def("skinny_gemm3_ex", &skinny_gemm3_ex, "glds double-buffered projection GEMM (M<=256)");
+  m.def("moe_grouped_gemm", &moe_grouped_gemm,
+        "packed per-expert grouped GEMM (skips unrouted experts)");
   m.def("oneshot_init", &oneshot_init, "one-shot allreduce: alloc + IPC handle");
   m.def("oneshot_connect", &oneshot_connect, "one-shot allreduce: map peer buffers");
   m.def("oneshot_ready", &oneshot_ready);

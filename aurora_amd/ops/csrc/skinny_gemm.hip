@@ -471,6 +471,141 @@ __global__ __launch_bounds__(512, 4) void skinny_gemm3_kernel(
 }
 #undef SG3_ISSUE
 
+// ---------------------------------------------------------------------------
+// MoE grouped GEMM (packed): yp[A, N] = per-expert xp-slice @ w[e]^T.
+// Tokens arrive SORTED by expert (xp[A, K], device starts[E]/counts[E]);
+// block t covers expert e = t / (N/64), N-tile nt = t % (N/64). Empty
+// experts cost one scalar load (the block exits before touching W), so
+// the dominant expert-weight stream is skipped for unrouted experts —
+// vs torch.bmm over the fixed-capacity [E, C, *] buffer, which streams
+// every expert's weights and pays E*C rows of activation traffic every
+// decode step (docs/ROUND2.md §4). Same glds double-buffered body as
+// skinny_gemm3 (v3); all shapes static + counts read on device => the
+// step stays hipGraph-capturable.
+// ---------------------------------------------------------------------------
+__global__ __launch_bounds__(512, 4) void moe_gemm_kernel(
+    const __hip_bfloat16* __restrict__ xp,  // [A, K] expert-sorted tokens
+    const __hip_bfloat16* __restrict__ w,   // [E, N, K]
+    __hip_bfloat16* __restrict__ yp,        // [A, N]
+    const int* __restrict__ starts,         // [E]
+    const int* __restrict__ counts,         // [E]
+    int N, int K) {
+  const int NT = N >> 6;
+  const int e = blockIdx.x / NT;
+  const int n0 = (blockIdx.x % NT) * 64;
+  const int cnt = counts[e];
+  if (cnt == 0) return;  // unrouted expert: no weight stream at all
+  const int row0 = starts[e];
+  const __hip_bfloat16* x = xp + (long)row0 * K;
+  const __hip_bfloat16* we = w + (long)e * N * K;
+  __hip_bfloat16* out = yp + (long)row0 * N;
+
+  __shared__ char lds[2][SG3_WTILE_B + SG3_XTILE_B];
+  const int lane = threadIdx.x % WAVE;
+  const int wave = threadIdx.x / WAVE;
+  const int nchunks = (K + SG3_BK - 1) / SG3_BK;
+
+  const int wfr = 8 * wave + (lane >> 3);
+  const int wfu = lane & 7;
+  const long wfg = (long)min(n0 + wfr, N - 1) * K + ((wfu ^ (wfr & 7)) << 3);
+  const int xfr0 = 32 * wave + (lane >> 3);
+  const int xfu = lane & 7;
+
+  f32x4 acc[2][4];
+#pragma unroll
+  for (int t = 0; t < 2; ++t)
+#pragma unroll
+    for (int j = 0; j < 4; ++j)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acc[t][j][r] = 0.f;
+
+  const int col = lane & 15;
+  const int kg = lane >> 4;
+
+#define MG_ISSUE(buf, c)                                                     \
+  {                                                                          \
+    const long kc = (long)(c)*SG3_BK;                                        \
+    char* wt = lds[buf];                                                     \
+    char* xt = lds[buf] + SG3_WTILE_B;                                       \
+    SG3_GLDS(we + wfg + kc, wt + 8 * wave * SG3_ROWB + 16 * lane, 2 /*nt*/); \
+    _Pragma("unroll") for (int j = 0; j < 4; ++j) {                          \
+      const int r = xfr0 + 8 * j;                                            \
+      const long g = (long)min(r, cnt - 1) * K + ((xfu ^ (r & 7)) << 3);     \
+      SG3_GLDS(x + g + kc, xt + (32 * wave + 8 * j) * SG3_ROWB + 16 * lane,  \
+               0);                                                           \
+    }                                                                        \
+  }
+
+  MG_ISSUE(0, 0);
+  for (int c = 0; c < nchunks; ++c) {
+    const int buf = c & 1;
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __syncthreads();
+    if (c + 1 < nchunks) MG_ISSUE(!buf, c + 1);
+    const char* wt = lds[buf];
+    const char* xt = lds[buf] + SG3_WTILE_B;
+    const int klim = K - c * SG3_BK;
+#pragma unroll
+    for (int s = 0; s < SG3_BK / 32; ++s) {
+      if (s * 32 >= klim) break;
+      sfrag bf[4], af[2];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const shortx8 bv = *reinterpret_cast<const shortx8*>(
+            wt + sg3_addr(j * 16 + col, s * 4 + kg));
+        __builtin_memcpy(&bf[j], &bv, 16);
+      }
+#pragma unroll
+      for (int t = 0; t < 2; ++t) {
+        const shortx8 av = *reinterpret_cast<const shortx8*>(
+            xt + sg3_addr(32 * wave + t * 16 + col, s * 4 + kg));
+        __builtin_memcpy(&af[t], &av, 16);
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[t][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[t].v, bf[j].v, acc[t][j], 0, 0, 0);
+      }
+    }
+  }
+#undef MG_ISSUE
+
+#pragma unroll
+  for (int t = 0; t < 2; ++t) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int n = n0 + j * 16 + col;
+      if (n >= N) continue;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = 32 * wave + t * 16 + kg * 4 + r;
+        if (m >= cnt) continue;  // rows past this expert's tokens
+        *(unsigned short*)(out + (long)m * N + n) = f32_to_bf16(acc[t][j][r]);
+      }
+    }
+  }
+}
+
+torch::Tensor moe_grouped_gemm(torch::Tensor xp, torch::Tensor w,
+                               torch::Tensor starts, torch::Tensor counts) {
+  TORCH_CHECK(xp.is_cuda() && xp.dtype() == torch::kBFloat16);
+  TORCH_CHECK(w.dim() == 3 && xp.dim() == 2 && xp.size(1) == w.size(2));
+  TORCH_CHECK(starts.dtype() == torch::kInt32 && counts.dtype() == torch::kInt32);
+  const int E = w.size(0), N = w.size(1), K = w.size(2);
+  const long A = xp.size(0);
+  TORCH_CHECK(K % SG3_BK == 0, "K must be a multiple of ", SG3_BK);
+  TORCH_CHECK(N % 64 == 0, "N must be a multiple of 64");
+  auto out = torch::empty({A, (long)N}, xp.options());
+  if (A == 0) return out;
+  dim3 grid(E * (N / 64)), block(512);
+  hipLaunchKernelGGL(moe_gemm_kernel, grid, block, 0, CUR_STREAM,
+                     (const __hip_bfloat16*)xp.data_ptr(),
+                     (const __hip_bfloat16*)w.data_ptr(),
+                     (__hip_bfloat16*)out.data_ptr(),
+                     starts.data_ptr<int>(), counts.data_ptr<int>(), N, K);
+  HIP_CHECK_KERNEL();
+  return out;
+}
+
 __global__ void skinny_combine_kernel(const float* __restrict__ part,
                                       __hip_bfloat16* __restrict__ out,
                                       int S, long MN) {
