@@ -69,11 +69,16 @@ async def run_prediscovery(rt, args: Dict[str, Any]) -> dict:
         store = rt.org(org_id)
         found = await discover_local(graph)
         found += await discover_kubernetes(graph)
+        # dependency inference over the discovered assets (reference:
+        # services/discovery/inference/ modules)
+        from .discovery import run_inference
+        inferred = await run_inference(graph)
         ctx = {
             "discovered_at": time.time(),
             "services": [s["name"] for s in graph.list_services()][:100],
             "spof": graph.single_points_of_failure(),
             "stale": graph.stale_services()[:20],
+            "inferred_edges": inferred,
         }
         store.save_artifact("prediscovery_context", json.dumps(ctx, indent=1))
         total += len(found)

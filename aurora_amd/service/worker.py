@@ -140,7 +140,7 @@ async def run_background_chat(rt, args: Dict[str, Any]) -> dict:
     severity = determine_severity(summary, inc.get("severity"))
     n_cits = save_citations(store, incident_id, state.messages)
     suggestions = await generate_recommendations(llm, store, incident_id,
-                                                 state.messages)
+                                                 state.messages, ctx=ctx)
     store.update_incident(incident_id, aurora_status="completed",
                           summary=summary, severity=severity,
                           root_cause=_extract_root_cause(summary))
