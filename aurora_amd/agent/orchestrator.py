@@ -22,40 +22,61 @@ from .state import State
 from .workflow import Workflow
 
 # role briefs (reference: orchestrator/roles/*.md)
-ROLES: Dict[str, Dict[str, Any]] = {
-    "general_investigator": {
-        "brief": "You broadly investigate the incident: read the alert, check "
-                 "service health, and identify the most suspicious subsystem.",
-        "tools": ["get_alert_field", "get_infrastructure_context",
-                  "list_incidents", "terminal_exec", "kubectl"],
-    },
-    "error_signal_investigator": {
-        "brief": "You focus on error signals: logs, error rates, stack traces "
-                 "around the incident start time.",
-        "tools": ["terminal_exec", "kubectl", "get_alert_field"],
-    },
-    "recent_change_investigator": {
-        "brief": "You focus on recent changes: deploys, config changes, "
-                 "migrations in the 60 minutes before the incident.",
-        "tools": ["get_incident_timeline", "list_incidents", "github_rca",
-                  "get_alert_field"],
-    },
-    "runtime_state_investigator": {
-        "brief": "You inspect live runtime state: resource saturation, "
-                 "process health, connectivity.",
-        "tools": ["terminal_exec", "kubectl", "get_service_impact"],
-    },
-    "runbook_lookup": {
-        "brief": "You find and summarise the most relevant runbooks and past "
-                 "incidents for this alert.",
-        "tools": ["knowledge_base_search", "list_incidents", "get_postmortem"],
-    },
-    "ticket_history": {
-        "brief": "You mine past incident history for recurrences of this "
-                 "failure signature.",
-        "tools": ["list_incidents", "get_incident", "get_postmortem"],
-    },
-}
+def _load_roles() -> Dict[str, Dict[str, Any]]:
+    """Role briefs live as markdown with YAML frontmatter (reference:
+    orchestrator/roles/*.md + role_registry.py:45) — name, tool subset and
+    skill keywords in the frontmatter, the investigation method as the
+    brief body."""
+    import os
+
+    import yaml
+    roles: Dict[str, Dict[str, Any]] = {}
+    d = os.path.join(os.path.dirname(__file__), "roles")
+    if not os.path.isdir(d):
+        return roles
+    for fn in sorted(os.listdir(d)):
+        if not fn.endswith(".md"):
+            continue
+        with open(os.path.join(d, fn)) as f:
+            raw = f.read()
+        body = raw
+        meta: Dict[str, Any] = {}
+        if raw.startswith("---"):
+            _, fm, body = raw.split("---", 2)
+            meta = yaml.safe_load(fm) or {}
+        name = meta.get("name") or fn[:-3]
+        roles[name] = {
+            "brief": body.strip(),
+            "tools": list(meta.get("tools") or []),
+            "skills": list(meta.get("skills") or []),
+        }
+    return roles
+
+
+ROLES: Dict[str, Dict[str, Any]] = _load_roles()
+
+
+def select_skills_for_role(role: str, budget_tokens: int = 1200,
+                           count_tokens=None) -> str:
+    """Per-role skill selection (reference: orchestrator/select_skills.py):
+    skills whose name/tags overlap the role's keywords, under a budget."""
+    from .skills.registry import get_skill_registry
+    keywords = set(ROLES.get(role, {}).get("skills", []))
+    keywords |= set(role.split("_"))
+    reg = get_skill_registry()
+    count = count_tokens or (lambda t: len(t) // 4)
+    picked = []
+    used = 0
+    for sk in reg.all():
+        hay = (sk.name + " " + " ".join(getattr(sk, "tags", []) or [])).lower()
+        if not any(k in hay for k in keywords):
+            continue
+        cost = count(sk.body)
+        if used + cost > budget_tokens:
+            continue
+        picked.append(f"## skill: {sk.name}\n{sk.body}")
+        used += cost
+    return "\n\n".join(picked)
 
 TRIAGE_PROMPT = """You are the RCA orchestrator. Given the alert, pick the
 investigation plan as a JSON object:

@@ -117,6 +117,7 @@ class RagIndex:
         return len(self.chunks)
 
     def search(self, query: str, k: int = 5) -> List[dict]:
+        from ..utils.streams import on_aux_stream
         with self._lock:
             n = len(self.chunks)
             if n == 0:
@@ -125,12 +126,16 @@ class RagIndex:
             lex_max = max(lex.values()) if lex else 1.0
             vec_scores: Dict[int, float] = {}
             if self.emb is not None:
-                qv = self._embed([query])
-                if qv is not None:
-                    vals, idx = ops.knn_topk(qv.to(self.emb.device).to(self.emb.dtype),
-                                             self.emb, min(4 * k, n))
-                    for v, i in zip(vals[0].tolist(), idx[0].tolist()):
-                        vec_scores[i] = v
+                # aux stream: the encoder + kNN + host read must not queue
+                # behind the engine's decode steps (utils/streams.py)
+                with on_aux_stream():
+                    qv = self._embed([query])
+                    if qv is not None:
+                        vals, idx = ops.knn_topk(
+                            qv.to(self.emb.device).to(self.emb.dtype),
+                            self.emb, min(4 * k, n))
+                        for v, i in zip(vals[0].tolist(), idx[0].tolist()):
+                            vec_scores[i] = v
             cands = set(vec_scores) | set(lex)
             scored = []
             for i in cands:

@@ -50,7 +50,7 @@ def parse_args():
                     default=os.environ.get("BENCH_MODE", "e2e"))
     ap.add_argument("--model", default=os.environ.get("BENCH_MODEL", "llama3-8b"))
     ap.add_argument("--concurrency", type=int,
-                    default=int(os.environ.get("BENCH_CONCURRENCY", "256")))
+                    default=int(os.environ.get("BENCH_CONCURRENCY", "128")))
     ap.add_argument("--turns", type=int, default=int(os.environ.get("BENCH_TURNS", "4")))
     ap.add_argument("--first-prompt", type=int, default=768)
     ap.add_argument("--turn-prompt", type=int, default=512)
@@ -242,6 +242,7 @@ def _build_runtime(args, rank, use_cuda):
         device="cpu" if not use_cuda else "auto",
         max_num_seqs=max(args.concurrency + 2, 8),
         max_batched_tokens=16384,
+        kv_memory_fraction=float(os.environ.get("KV_MEMORY_FRACTION", "0.8")),
         seed=1234 + rank,
     )
     server = AsyncEngineServer(ecfg)
@@ -292,13 +293,16 @@ def _payload(rng: torch.Generator, step: int, i: int) -> dict:
     }}}
 
 
-async def run_step_e2e(rt, org_id, args, rng, step: int):
+async def run_step_e2e(rt, org_id, args, rng, step: int,
+                       deadline_s: float = 1500.0):
+    import sys
     from aurora_amd.service.webhooks import process_webhook
     B = args.concurrency
     ev_q = rt.events.subscribe(org_id)
     t_start: dict = {}
     pending = set()
     statuses: dict = {}
+    t0 = time.monotonic()
     for i in range(B):
         t_start_i = time.monotonic()
         iid, created = process_webhook(
@@ -309,9 +313,27 @@ async def run_step_e2e(rt, org_id, args, rng, step: int):
         t_start[iid] = t_start_i
         pending.add(iid)
     lat = []
+    last_beat = t0
     while pending:
-        ev = await asyncio.wait_for(ev_q.get(), timeout=600.0)
-        if (ev.get("type") == "incident_update"
+        left = deadline_s - (time.monotonic() - t0)
+        if left <= 0:
+            # stragglers are failures, not silently-dropped work: they
+            # count in the denominator via `failed`
+            print(f"[bench] step {step}: deadline, {len(pending)} pending",
+                  file=sys.stderr, flush=True)
+            statuses["failed"] = statuses.get("failed", 0) + len(pending)
+            break
+        try:
+            ev = await asyncio.wait_for(ev_q.get(), timeout=min(left, 30.0))
+        except asyncio.TimeoutError:
+            ev = None
+        now = time.monotonic()
+        if now - last_beat > 30.0:
+            last_beat = now
+            c = rt.engine.engine.counters
+            print(f"[bench] step {step} t={now - t0:.0f}s done="
+                  f"{len(lat)}/{B} counters={c}", file=sys.stderr, flush=True)
+        if (ev and ev.get("type") == "incident_update"
                 and ev.get("incident_id") in pending
                 and ev.get("aurora_status") in ("completed", "failed")):
             iid = ev["incident_id"]
