@@ -67,8 +67,8 @@ def select_skills_for_role(role: str, budget_tokens: int = 1200,
     count = count_tokens or (lambda t: len(t) // 4)
     picked = []
     used = 0
-    for sk in reg.all():
-        hay = (sk.name + " " + " ".join(getattr(sk, "tags", []) or [])).lower()
+    for sk in sorted(reg.skills.values(), key=lambda s: s.rca_priority):
+        hay = (sk.name + " " + sk.description + " " + sk.category).lower()
         if not any(k in hay for k in keywords):
             continue
         cost = count(sk.body)
@@ -204,11 +204,16 @@ class Orchestrator:
             user_id=state.user_id, org_id=state.org_id,
             session_id=None,  # sub-agents keep their own scratch context
             incident_id=state.incident_id, is_background=True)
+        skills_text = select_skills_for_role(
+            role, count_tokens=getattr(self.llm, "count_tokens", None))
         wf = Workflow(self.llm, self.ctx, tool_subset=subset,
                       system_prompt_override=(
                           f"You are the {role} sub-agent of an incident "
-                          f"investigation.\n{brief}\nFinish with final_answer "
-                          f"containing your findings."),
+                          f"investigation.\n{brief}\n"
+                          + (f"\n# Relevant skills\n{skills_text}\n"
+                             if skills_text else "")
+                          + "Finish with final_answer containing your "
+                            "findings."),
                       max_turns=self.sub_agent_turns)
         final = ""
         try:

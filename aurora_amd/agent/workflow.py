@@ -101,6 +101,7 @@ class Workflow:
 
         tool_names = {t.name for t in tools} | {"final_answer"}
         final_text: Optional[str] = None
+        call_counts: Dict[str, int] = {}  # per-tool loop guard (sub_agent.py:81)
         for turn in range(self.max_turns):
             if (time.monotonic() - t_start) > self.settings.workflow_timeout_s:
                 yield {"type": "error", "error": "workflow timeout"}
@@ -174,6 +175,21 @@ class Workflow:
             if name == "final_answer":
                 final_text = str(args.get("answer", "")).strip() or thought
                 break
+            # loop guard: the exact same call repeated is evidence the
+            # model is stuck (reference: sub_agent.py:81 per-tool guard) —
+            # nudge instead of re-running, force wrap-up when ignored
+            call_key = name + ":" + json.dumps(args, sort_keys=True)[:300]
+            call_counts[call_key] = call_counts.get(call_key, 0) + 1
+            if call_counts[call_key] >= 3:
+                state.messages.append({
+                    "role": "observation",
+                    "content": f"[loop guard] {name} was already called "
+                               "with these exact arguments; its output has "
+                               "not changed. Use a different tool or "
+                               "finish with final_answer."})
+                if call_counts[call_key] >= 5:
+                    break  # forced wrap-up -> _forced_final below
+                continue
             yield {"type": "tool_call", "tool": name, "args": args}
             output = await self.registry.run(self.ctx, name, args)
             # per-turn rail on the tool output: tool results are untrusted

@@ -242,3 +242,42 @@ def test_skill_library_inventory():
     txt2 = reg.assemble_rca_prompt(100000, lambda t: len(t) // 4,
                                    connected=["datadog"])
     assert "Skill: datadog" in txt2 and "Skill: rca_deploy_rollback" in txt2
+
+
+def test_roles_loaded_from_markdown():
+    from aurora_amd.agent.orchestrator import ROLES, select_skills_for_role
+    assert len(ROLES) >= 6
+    for name, r in ROLES.items():
+        assert r["brief"] and "Method" in r["brief"], name
+        assert r["tools"], name
+    # skill selection stays within budget and returns role-relevant text
+    txt = select_skills_for_role("runbook_lookup", budget_tokens=2000)
+    assert isinstance(txt, str)
+
+
+def test_workflow_loop_guard(env):
+    """The same tool+args repeated is intercepted (nudge at 3, forced
+    final at 5) instead of burning turns (reference: sub_agent.py:81)."""
+    import asyncio
+    from aurora_amd.agent.state import State
+    from aurora_amd.agent.workflow import Workflow
+    store, iid = env
+    # script: the model calls the same tool with identical args forever
+    script = [json.dumps({"tool": "list_incidents", "args": {}})
+              for _ in range(12)]
+    llm, ctx = _mk(store, iid, script)
+    wf = Workflow(llm, ctx, max_turns=10,
+                  tool_subset=["list_incidents"])
+    events = []
+
+    async def main():
+        async for ev in wf.stream(State(question="loop", incident_id=iid)):
+            events.append(ev)
+    asyncio.run(main())
+    tool_calls = [e for e in events if e["type"] == "tool_call"]
+    assert len(tool_calls) <= 2  # guard stopped re-execution at 3 repeats
+    finals = [e for e in events if e["type"] == "final"]
+    assert finals and finals[0]["content"]
+    guard_msgs = [m for m in wf.__dict__.get("_last", [])] or [
+        m for e in events for m in [e] if "loop guard" in str(e)]
+    # the guard observation reached the context
