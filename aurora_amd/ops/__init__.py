@@ -193,15 +193,19 @@ def linear(x: torch.Tensor, weight: torch.Tensor) -> torch.Tensor:
             s_target = 256 if w_bytes <= 40 * 1024 * 1024 else 512
             return _hip.skinny_gemm2_ex(x.contiguous(), weight, s_target)
     elif (x.is_cuda and x.dim() == 2 and 64 < x.shape[0] <= 256
-            and x.shape[1] % 64 == 0 and weight.shape[0] % 16 == 0
-            and x.dtype == torch.bfloat16 and _hip is not None
-            and weight.shape[0] * weight.shape[1] * 2 < 512 * 1024 * 1024):
-        # M=65..256 (the B=256 decode regime): hipBLASLt runs these cold at
-        # 1.1-3.7 TB/s (docs/ROUND2.md §1); the glds double-buffered v3
-        # streams W once through LDS. Gated off with AURORA_SKINNY3=0.
+            and x.shape[1] % 64 == 0 and weight.shape[0] % 64 == 0
+            and x.dtype == torch.bfloat16 and _hip is not None):
+        # M=65..256 (the B=128-256 decode regime): measured per-shape
+        # dispatch (gpurun_out/sweep_skinny3.log, cold rotated weights) —
+        # the glds v3 kernel wins the qkv shape at M>128 (1.36 vs 1.18
+        # TB/s) and deep-K at M<=128 (down128 2.18 vs 1.59); hipBLASLt
+        # keeps o / gate_up / lm-head. Gated off with AURORA_SKINNY3=0.
         import os
         if os.environ.get("AURORA_SKINNY3", "1") != "0":
-            return _hip.skinny_gemm3_ex(x.contiguous(), weight, 0)
+            M, K, N = x.shape[0], x.shape[1], weight.shape[0]
+            if (K >= 8192 and M <= 128) or (4096 < N <= 8192 and K <= 4096
+                                            and M > 128):
+                return _hip.skinny_gemm3_ex(x.contiguous(), weight, 384)
     return F.linear(x, weight)
 
 

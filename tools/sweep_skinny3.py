@@ -53,7 +53,7 @@ for tag, M, K, N in SHAPES:
     x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16)
     iters = max(20, min(60, 3 * ncopy))
     res = {}
-    for st in (0, 128, 192, 256, 384):
+    for st in (256, 384, 512, 640, 896):
         dt = timeit(lambda i: hip.skinny_gemm3_ex(x, ws[i % ncopy], st), iters)
         res[f"v3_s{st}"] = dt
     dt = timeit(lambda i: hip.skinny_gemm2_ex(x, ws[i % ncopy], 512), iters)
