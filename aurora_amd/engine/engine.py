@@ -134,6 +134,16 @@ class LLMEngine:
         page_size = config.page_size or (256 if self.device != "cpu" else 16)
         tp = pdist.tp_world_size()
         kv_heads_local = max(self.spec.n_kv_heads // tp, 1)
+        # fp8 (e4m3) KV cache, env-gated (AURORA_KV_FP8=1): attention is at
+        # the HBM stream limit in bf16 (6.15 TB/s measured), so fewer bytes
+        # is the remaining lever — and the page budget doubles
+        # (docs/ROUND2.md §3). Compute stays bf16; KV is converted at the
+        # cache boundary by the HIP kernels.
+        import os as _os
+        kv_fp8 = (_os.environ.get("AURORA_KV_FP8", "0") == "1"
+                  and self.device != "cpu")
+        kv_dtype = torch.float8_e4m3fn if kv_fp8 else self.dtype
+        kv_bytes = 1 if kv_fp8 else (2 if self.dtype != torch.float32 else 4)
         if config.kv_cache_pages:
             pages = config.kv_cache_pages
         elif self.device != "cpu":
@@ -141,14 +151,14 @@ class LLMEngine:
             budget = int(free * config.kv_memory_fraction)
             pages = PagedKVCache.pages_for_memory(
                 budget, self.spec.n_layers, kv_heads_local, self.spec.head_dim,
-                page_size, 2 if self.dtype != torch.float32 else 4,
+                page_size, kv_bytes,
             )
         else:
             pages = 512
         self.kv = PagedKVCache(
             self.spec.n_layers, kv_heads_local, self.spec.head_dim,
             page_size=page_size, num_pages=pages, device=self.device,
-            dtype=self.dtype,
+            dtype=kv_dtype,
         )
         self.max_model_len = min(config.max_model_len, self.spec.max_position)
 

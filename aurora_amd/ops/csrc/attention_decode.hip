@@ -25,11 +25,11 @@
 #define TUNR 4  // tokens prefetched per group iteration (8 measured neutral)
 #endif
 
-template <int D, int G>
+template <int D, int G, bool FP8>
 __global__ __launch_bounds__(256) void paged_decode_kernel(
     const __hip_bfloat16* __restrict__ q,   // [B, Hq, D]
-    const __hip_bfloat16* __restrict__ kc,  // [NP, Hkv, P, D]
-    const __hip_bfloat16* __restrict__ vc,
+    const void* __restrict__ kc,  // [NP, Hkv, P, D] bf16 | fp8 e4m3
+    const void* __restrict__ vc,
     __hip_bfloat16* __restrict__ out,       // [B, Hq, D] (CS==1)
     float* __restrict__ part,               // [CS, B, Hq, D+2] (CS>1)
     const int* __restrict__ block_tables,   // [B, MP]
@@ -76,13 +76,34 @@ __global__ __launch_bounds__(256) void paged_decode_kernel(
        t0 += stride * TUNR) {
     shortx8 kf[TUNR], vf[TUNR];
     int nt = 0;
+    if constexpr (FP8) {
+      // fp8 cache: half the KV stream bytes (the kernel is KV-bound at
+      // ~6.15 TB/s in bf16 — docs/ROUND2.md §3); 8-byte loads, convert
+      // to bf16 fragments in-register
+      u64b k8[TUNR], v8[TUNR];
 #pragma unroll
-    for (int u = 0; u < TUNR; ++u) {
-      const int t = t0 + u;
-      const int tc = min(t, t_hi - 1);  // clamped prefetch (masked below)
-      const long row = (((long)bt[tc / P] * Hkv + kvh) * P + (tc % P)) * D + sub * 8;
-      kf[u] = *reinterpret_cast<const shortx8*>(kc + row);
-      vf[u] = *reinterpret_cast<const shortx8*>(vc + row);
+      for (int u = 0; u < TUNR; ++u) {
+        const int t = t0 + u;
+        const int tc = min(t, t_hi - 1);
+        const long row =
+            (((long)bt[tc / P] * Hkv + kvh) * P + (tc % P)) * D + sub * 8;
+        k8[u] = *reinterpret_cast<const u64b*>((const unsigned char*)kc + row);
+        v8[u] = *reinterpret_cast<const u64b*>((const unsigned char*)vc + row);
+      }
+#pragma unroll
+      for (int u = 0; u < TUNR; ++u) {
+        kf[u] = fp8x8_to_bf16x8(k8[u]);
+        vf[u] = fp8x8_to_bf16x8(v8[u]);
+      }
+    } else {
+#pragma unroll
+      for (int u = 0; u < TUNR; ++u) {
+        const int t = t0 + u;
+        const int tc = min(t, t_hi - 1);  // clamped prefetch (masked below)
+        const long row = (((long)bt[tc / P] * Hkv + kvh) * P + (tc % P)) * D + sub * 8;
+        kf[u] = *reinterpret_cast<const shortx8*>((const __hip_bfloat16*)kc + row);
+        vf[u] = *reinterpret_cast<const shortx8*>((const __hip_bfloat16*)vc + row);
+      }
     }
 #pragma unroll
     for (int u = 0; u < TUNR; ++u) {
@@ -185,6 +206,7 @@ torch::Tensor paged_attention_decode(torch::Tensor q, torch::Tensor kc,
   const int Hkv = kc.size(1), P = kc.size(2), MP = bt.size(1);
   const int G = Hq / Hkv;
   TORCH_CHECK(Hq % Hkv == 0, "GQA requires Hq % Hkv == 0");
+  const bool fp8 = kc.dtype() == torch::kFloat8_e4m3fn;
   auto out = torch::empty_like(q);
   if (B == 0) return out;
   // context splits: fill the chip at small batch (>=512 WGs target)
@@ -197,15 +219,16 @@ torch::Tensor paged_attention_decode(torch::Tensor q, torch::Tensor kc,
                         q.options().dtype(torch::kFloat32));
   dim3 grid(B, Hkv, CS), block(256);
   auto stream = CUR_STREAM;
-#define LAUNCH(DD, GG)                                                        \
-  hipLaunchKernelGGL((paged_decode_kernel<DD, GG>), grid, block, 0, stream,   \
-                     (const __hip_bfloat16*)q.data_ptr(),                     \
-                     (const __hip_bfloat16*)kc.data_ptr(),                    \
-                     (const __hip_bfloat16*)vc.data_ptr(),                    \
+#define LAUNCH1(DD, GG, F8)                                                   \
+  hipLaunchKernelGGL((paged_decode_kernel<DD, GG, F8>), grid, block, 0,       \
+                     stream, (const __hip_bfloat16*)q.data_ptr(),             \
+                     kc.data_ptr(), vc.data_ptr(),                            \
                      (__hip_bfloat16*)out.data_ptr(),                         \
                      CS > 1 ? part.data_ptr<float>() : nullptr,               \
                      bt.data_ptr<int>(), ctx.data_ptr<int>(), Hkv, P, MP, CS, \
                      (float)scale)
+#define LAUNCH(DD, GG)                                                        \
+  if (fp8) { LAUNCH1(DD, GG, true); } else { LAUNCH1(DD, GG, false); }
 #define LAUNCH_D(DD)                                                          \
   switch (G) {                                                                \
     case 1: LAUNCH(DD, 1); break;                                             \
@@ -235,6 +258,7 @@ torch::Tensor paged_attention_decode(torch::Tensor q, torch::Tensor kc,
   }
 #undef LAUNCH_D
 #undef LAUNCH
+#undef LAUNCH1
   HIP_CHECK_KERNEL();
   return out;
 }

@@ -38,11 +38,11 @@ union frag_u {
   bf16x8 v;
 };
 
-template <int D, int HG>  // HG = query heads per workgroup
+template <int D, int HG, bool FP8>  // HG = query heads per workgroup
 __global__ __launch_bounds__(512, 1) void prefill_kernel(
     const __hip_bfloat16* __restrict__ q,   // [T, Hq, D]
-    const __hip_bfloat16* __restrict__ kc,  // [NP, Hkv, P, D]
-    const __hip_bfloat16* __restrict__ vc,
+    const void* __restrict__ kc,  // [NP, Hkv, P, D] bf16 | fp8 e4m3
+    const void* __restrict__ vc,
     __hip_bfloat16* __restrict__ out,       // [T, Hq, D]
     const int* __restrict__ block_seq,      // [NB] seq index of this q-block
     const int* __restrict__ block_qoff,     // [NB] q-row offset within the seq
@@ -133,8 +133,17 @@ __global__ __launch_bounds__(512, 1) void prefill_kernel(
       if (t < ctx) {
         const long base =
             (((long)bt[t / P] * Hkv + kvh) * P + (t % P)) * D + d0;
-        kv8 = *reinterpret_cast<const shortx8*>(kc + base);
-        vv8 = *reinterpret_cast<const shortx8*>(vc + base);
+        if constexpr (FP8) {
+          kv8 = fp8x8_to_bf16x8(
+              *reinterpret_cast<const u64b*>((const unsigned char*)kc + base));
+          vv8 = fp8x8_to_bf16x8(
+              *reinterpret_cast<const u64b*>((const unsigned char*)vc + base));
+        } else {
+          kv8 = *reinterpret_cast<const shortx8*>(
+              (const __hip_bfloat16*)kc + base);
+          vv8 = *reinterpret_cast<const shortx8*>(
+              (const __hip_bfloat16*)vc + base);
+        }
       }
       const int kb = (row * D * 2 + ((d0 * 2) ^ ((row & SWZ) << 4))) / 2;
       *reinterpret_cast<shortx8*>(kbuf + kb) = kv8;
@@ -327,17 +336,19 @@ torch::Tensor attention_prefill_paged(torch::Tensor q, torch::Tensor kc,
 
   const int VROW = 64 + 8;
   const size_t lds = 64 * D * 2 + D * VROW * 2 + 8 * 32 * 4;
+  const bool fp8 = kc.dtype() == torch::kFloat8_e4m3fn;
   dim3 grid(NB, Hq / HG), block(512);
   auto stream = CUR_STREAM;
-#define PLAUNCH(DD, HGV)                                                      \
-  hipLaunchKernelGGL((prefill_kernel<DD, HGV>), grid, block, lds, stream,     \
+#define PLAUNCH1(DD, HGV, F8)                                                 \
+  hipLaunchKernelGGL((prefill_kernel<DD, HGV, F8>), grid, block, lds, stream, \
                      (const __hip_bfloat16*)q.data_ptr(),                     \
-                     (const __hip_bfloat16*)kc.data_ptr(),                    \
-                     (const __hip_bfloat16*)vc.data_ptr(),                    \
+                     kc.data_ptr(), vc.data_ptr(),                            \
                      (__hip_bfloat16*)out.data_ptr(),                         \
                      bseq_t.data_ptr<int>(), boff_t.data_ptr<int>(),          \
                      cu_q.data_ptr<int>(), bt.data_ptr<int>(),                \
                      ctx.data_ptr<int>(), Hq, Hkv, P, MP, (float)scale)
+#define PLAUNCH(DD, HGV)                                                      \
+  if (fp8) { PLAUNCH1(DD, HGV, true); } else { PLAUNCH1(DD, HGV, false); }
 #define PLAUNCH_D(DD)                                                         \
   switch (HG) {                                                               \
     case 1: PLAUNCH(DD, 1); break;                                            \
@@ -351,6 +362,7 @@ torch::Tensor attention_prefill_paged(torch::Tensor q, torch::Tensor kc,
   }
 #undef PLAUNCH_D
 #undef PLAUNCH
+#undef PLAUNCH1
   HIP_CHECK_KERNEL();
   return out;
 }

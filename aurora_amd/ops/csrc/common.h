@@ -82,6 +82,46 @@ DEVINL float group_max(float x) {
   return x;
 }
 
+// ---- fp8 (OCP e4m3) KV-cache conversion helpers (gfx950 native cvt;
+// the chip's fp8 is e4m3fn, NOT MI300X fnuz — platform guide §4) -------
+typedef unsigned long long u64b;
+
+// 8 fp8 bytes -> 8 bf16 (used where attention kernels read the cache)
+DEVINL shortx8 fp8x8_to_bf16x8(u64b v) {
+  shortx8 out;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const floatx2 f = __builtin_amdgcn_cvt_pk_f32_fp8(
+        (unsigned int)(v >> (16 * i)) & 0xffffu, false);
+    out[2 * i] = (short)f32_to_bf16(f[0]);
+    out[2 * i + 1] = (short)f32_to_bf16(f[1]);
+  }
+  return out;
+}
+
+// 8 bf16 -> 8 fp8 bytes (cache write path; saturates at +-448)
+DEVINL u64b bf16x8_to_fp8x8(const shortx8 v) {
+  u64b out = 0;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const unsigned int p = __builtin_amdgcn_cvt_pk_fp8_f32(
+        bf16_to_f32((unsigned short)v[2 * i]),
+        bf16_to_f32((unsigned short)v[2 * i + 1]), 0u, false);
+    out |= (u64b)(p & 0xffffu) << (16 * i);
+  }
+  return out;
+}
+
+DEVINL unsigned char f32_to_fp8(float a) {
+  return (unsigned char)(__builtin_amdgcn_cvt_pk_fp8_f32(a, 0.f, 0u, false)
+                         & 0xffu);
+}
+
+DEVINL float fp8_to_f32(unsigned char b) {
+  const floatx2 f = __builtin_amdgcn_cvt_pk_f32_fp8((unsigned int)b, false);
+  return f[0];
+}
+
 #define HIP_CHECK_KERNEL()                                                     \
   do {                                                                         \
     hipError_t e = hipGetLastError();                                          \

@@ -208,9 +208,10 @@ void rope_apply_(torch::Tensor q, torch::Tensor k, torch::Tensor positions,
 
 // Scatter new K/V rows into the paged cache.
 // k/v: [T, Hkv, D]; caches: [NP, Hkv, P, D]; slot = page*P + off.
+template <bool FP8>
 __global__ void kv_cache_write_kernel(
     const __hip_bfloat16* __restrict__ k, const __hip_bfloat16* __restrict__ v,
-    __hip_bfloat16* __restrict__ kc, __hip_bfloat16* __restrict__ vc,
+    void* __restrict__ kc, void* __restrict__ vc,
     const int* __restrict__ slots, int T, int Hkv, int P, int D) {
   // one wave per (token, head); lane copies 8 elements at a time
   const int waves_per_block = blockDim.x / WAVE;
@@ -225,10 +226,17 @@ __global__ void kv_cache_write_kernel(
   const long src = ((long)t * Hkv + h) * D;
   const long dst = (((long)page * Hkv + h) * P + off) * D;
   for (int i = lane * 8; i < D; i += WAVE * 8) {
-    *reinterpret_cast<shortx8*>(kc + dst + i) =
-        *reinterpret_cast<const shortx8*>(k + src + i);
-    *reinterpret_cast<shortx8*>(vc + dst + i) =
-        *reinterpret_cast<const shortx8*>(v + src + i);
+    if constexpr (FP8) {
+      *reinterpret_cast<u64b*>((unsigned char*)kc + dst + i) =
+          bf16x8_to_fp8x8(*reinterpret_cast<const shortx8*>(k + src + i));
+      *reinterpret_cast<u64b*>((unsigned char*)vc + dst + i) =
+          bf16x8_to_fp8x8(*reinterpret_cast<const shortx8*>(v + src + i));
+    } else {
+      *reinterpret_cast<shortx8*>((__hip_bfloat16*)kc + dst + i) =
+          *reinterpret_cast<const shortx8*>(k + src + i);
+      *reinterpret_cast<shortx8*>((__hip_bfloat16*)vc + dst + i) =
+          *reinterpret_cast<const shortx8*>(v + src + i);
+    }
   }
 }
 
@@ -240,12 +248,19 @@ void kv_cache_write(torch::Tensor k, torch::Tensor v, torch::Tensor kc,
   if (T == 0) return;
   long waves = (long)T * Hkv;
   int wpb = 4;
-  hipLaunchKernelGGL(kv_cache_write_kernel, dim3((waves + wpb - 1) / wpb),
-                     dim3(wpb * WAVE), 0, CUR_STREAM,
-                     (const __hip_bfloat16*)k.data_ptr(),
-                     (const __hip_bfloat16*)v.data_ptr(),
-                     (__hip_bfloat16*)kc.data_ptr(), (__hip_bfloat16*)vc.data_ptr(),
-                     slots.data_ptr<int>(), T, Hkv, P, D);
+  const bool fp8 = kc.dtype() == torch::kFloat8_e4m3fn;
+  if (fp8)
+    hipLaunchKernelGGL(kv_cache_write_kernel<true>,
+                       dim3((waves + wpb - 1) / wpb), dim3(wpb * WAVE), 0,
+                       CUR_STREAM, (const __hip_bfloat16*)k.data_ptr(),
+                       (const __hip_bfloat16*)v.data_ptr(), kc.data_ptr(),
+                       vc.data_ptr(), slots.data_ptr<int>(), T, Hkv, P, D);
+  else
+    hipLaunchKernelGGL(kv_cache_write_kernel<false>,
+                       dim3((waves + wpb - 1) / wpb), dim3(wpb * WAVE), 0,
+                       CUR_STREAM, (const __hip_bfloat16*)k.data_ptr(),
+                       (const __hip_bfloat16*)v.data_ptr(), kc.data_ptr(),
+                       vc.data_ptr(), slots.data_ptr<int>(), T, Hkv, P, D);
   HIP_CHECK_KERNEL();
 }
 
@@ -255,10 +270,11 @@ void kv_cache_write(torch::Tensor k, torch::Tensor v, torch::Tensor kc,
 // rocprof showed those as ~5 launches + 3 copies per layer of pure launch
 // floor in the decode step (profiles/decode_kernel_stats_r01.csv).
 // qkv: [T, (Hq+2*Hkv)*D] -> writes q_out [T, Hq, D] (roped), kc/vc slots.
+template <bool FP8>
 __global__ void fused_rope_kv_write_kernel(
     const __hip_bfloat16* __restrict__ qkv,
     __hip_bfloat16* __restrict__ q_out,
-    __hip_bfloat16* __restrict__ kc, __hip_bfloat16* __restrict__ vc,
+    void* __restrict__ kc, void* __restrict__ vc,
     const int* __restrict__ positions, const int* __restrict__ slots,
     const float* __restrict__ cos_sin,
     int T, int Hq, int Hkv, int D, int P) {
@@ -274,30 +290,50 @@ __global__ void fused_rope_kv_write_kernel(
   if (h < Hq + Hkv) {
     // roped write (q -> q_out, k -> cache)
     const int pos = positions[t];
-    __hip_bfloat16* dst;
-    if (h < Hq) {
-      dst = q_out + ((long)t * Hq + h) * D;
+    const bool is_q = h < Hq;
+    long dst_row;
+    if (is_q) {
+      dst_row = ((long)t * Hq + h) * D;
     } else {
       const int slot = slots[t];
       if (slot < 0) return;
-      dst = kc + ((((long)(slot / P) * Hkv + (h - Hq)) * P + slot % P)) * D;
+      dst_row = ((((long)(slot / P) * Hkv + (h - Hq)) * P + slot % P)) * D;
     }
     for (int i = lane; i < half; i += WAVE) {
       const float c = cos_sin[(long)pos * D + i];
       const float s = cos_sin[(long)pos * D + half + i];
       const float x1 = bf16_to_f32(*(const unsigned short*)(src + i));
       const float x2 = bf16_to_f32(*(const unsigned short*)(src + half + i));
-      *(unsigned short*)(dst + i) = f32_to_bf16(x1 * c - x2 * s);
-      *(unsigned short*)(dst + half + i) = f32_to_bf16(x2 * c + x1 * s);
+      const float r1 = x1 * c - x2 * s;
+      const float r2 = x2 * c + x1 * s;
+      if (is_q) {
+        __hip_bfloat16* dq = q_out + dst_row;
+        *(unsigned short*)(dq + i) = f32_to_bf16(r1);
+        *(unsigned short*)(dq + half + i) = f32_to_bf16(r2);
+      } else if constexpr (FP8) {
+        unsigned char* dk = (unsigned char*)kc + dst_row;
+        dk[i] = f32_to_fp8(r1);
+        dk[half + i] = f32_to_fp8(r2);
+      } else {
+        __hip_bfloat16* dk = (__hip_bfloat16*)kc + dst_row;
+        *(unsigned short*)(dk + i) = f32_to_bf16(r1);
+        *(unsigned short*)(dk + half + i) = f32_to_bf16(r2);
+      }
     }
   } else {
     const int slot = slots[t];
     if (slot < 0) return;
-    __hip_bfloat16* dst =
-        vc + ((((long)(slot / P) * Hkv + (h - Hq - Hkv)) * P + slot % P)) * D;
-    for (int i = lane * 8; i < D; i += WAVE * 8)
-      *reinterpret_cast<shortx8*>(dst + i) =
-          *reinterpret_cast<const shortx8*>(src + i);
+    const long dst_row =
+        ((((long)(slot / P) * Hkv + (h - Hq - Hkv)) * P + slot % P)) * D;
+    for (int i = lane * 8; i < D; i += WAVE * 8) {
+      if constexpr (FP8) {
+        *reinterpret_cast<u64b*>((unsigned char*)vc + dst_row + i) =
+            bf16x8_to_fp8x8(*reinterpret_cast<const shortx8*>(src + i));
+      } else {
+        *reinterpret_cast<shortx8*>((__hip_bfloat16*)vc + dst_row + i) =
+            *reinterpret_cast<const shortx8*>(src + i);
+      }
+    }
   }
 }
 
@@ -313,13 +349,23 @@ torch::Tensor fused_rope_kv_write(torch::Tensor qkv, torch::Tensor positions,
   if (T == 0) return q_out;
   const long waves = (long)T * (Hq + 2 * Hkv);
   const int wpb = 4;
-  hipLaunchKernelGGL(fused_rope_kv_write_kernel,
-                     dim3((waves + wpb - 1) / wpb), dim3(wpb * WAVE), 0,
-                     CUR_STREAM, (const __hip_bfloat16*)qkv.data_ptr(),
-                     (__hip_bfloat16*)q_out.data_ptr(),
-                     (__hip_bfloat16*)kc.data_ptr(), (__hip_bfloat16*)vc.data_ptr(),
-                     positions.data_ptr<int>(), slots.data_ptr<int>(),
-                     cos_sin.data_ptr<float>(), T, (int)Hq, (int)Hkv, D, P);
+  const bool fp8 = kc.dtype() == torch::kFloat8_e4m3fn;
+  if (fp8)
+    hipLaunchKernelGGL(fused_rope_kv_write_kernel<true>,
+                       dim3((waves + wpb - 1) / wpb), dim3(wpb * WAVE), 0,
+                       CUR_STREAM, (const __hip_bfloat16*)qkv.data_ptr(),
+                       (__hip_bfloat16*)q_out.data_ptr(), kc.data_ptr(),
+                       vc.data_ptr(), positions.data_ptr<int>(),
+                       slots.data_ptr<int>(), cos_sin.data_ptr<float>(),
+                       T, (int)Hq, (int)Hkv, D, P);
+  else
+    hipLaunchKernelGGL(fused_rope_kv_write_kernel<false>,
+                       dim3((waves + wpb - 1) / wpb), dim3(wpb * WAVE), 0,
+                       CUR_STREAM, (const __hip_bfloat16*)qkv.data_ptr(),
+                       (__hip_bfloat16*)q_out.data_ptr(), kc.data_ptr(),
+                       vc.data_ptr(), positions.data_ptr<int>(),
+                       slots.data_ptr<int>(), cos_sin.data_ptr<float>(),
+                       T, (int)Hq, (int)Hkv, D, P);
   HIP_CHECK_KERNEL();
   return q_out;
 }

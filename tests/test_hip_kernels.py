@@ -487,3 +487,66 @@ def test_engine_long_context_prefill_decode_evict():
             break
     assert "sess-long" not in eng.sessions  # evicted under pressure
     assert eng.kv.num_free() > 0
+
+
+# ----------------------------------------------------------- fp8 KV cache
+
+
+def test_fp8_kv_roundtrip_and_decode():
+    """e4m3 KV cache: write kernel converts bf16->fp8, decode reads fp8 —
+    numerics vs the fp32 reference within the documented fp8 tolerance
+    (docs/ROUND2.md §3)."""
+    from aurora_amd import _hip_ops
+    torch.manual_seed(4)
+    D, Hkv, G, P, NP, B = 128, 2, 4, 64, 12, 3
+    Hq = G * Hkv
+    lens = [17, 130, 301]
+    # build both caches through the same write kernel
+    kc8 = torch.zeros(NP, Hkv, P, D, device="cuda",
+                      dtype=torch.float8_e4m3fn)
+    vc8 = torch.zeros_like(kc8)
+    kc16 = torch.zeros(NP, Hkv, P, D, device="cuda", dtype=torch.bfloat16)
+    vc16 = torch.zeros_like(kc16)
+    bt = torch.zeros(B, (max(lens) + P - 1) // P, dtype=torch.int32)
+    used = torch.randperm(NP)
+    u = 0
+    slot_rows = []
+    kv_rows = []
+    for b, L in enumerate(lens):
+        need = (L + P - 1) // P
+        pages = used[u:u + need]
+        bt[b, :need] = pages.to(torch.int32)
+        u += need
+        for t in range(L):
+            slot_rows.append(int(pages[t // P]) * P + t % P)
+    T = len(slot_rows)
+    k = torch.randn(T, Hkv, D, device="cuda", dtype=torch.bfloat16) / 4
+    v = torch.randn(T, Hkv, D, device="cuda", dtype=torch.bfloat16) / 4
+    slots = torch.tensor(slot_rows, dtype=torch.int32, device="cuda")
+    _hip_ops.kv_cache_write(k, v, kc8, vc8, slots)
+    _hip_ops.kv_cache_write(k, v, kc16, vc16, slots)
+    # round-trip error of the fp8 cache itself
+    rt = kc8.float() - kc16.float()
+    assert rt.abs().max().item() < 0.05
+    q = torch.randn(B, Hq, D, device="cuda", dtype=torch.bfloat16)
+    ctx = torch.tensor(lens, dtype=torch.int32)
+    scale = 1.0 / math.sqrt(D)
+    out8 = ops.paged_attention_decode(q, kc8, vc8, bt.cuda(), ctx.cuda(), scale)
+    ref = R.paged_attention_decode(q.cpu(), kc16.cpu(), vc16.cpu(), bt, ctx,
+                                   scale)
+    # fp8 e4m3: ~2^-3 relative element error, averaged down by the dot
+    assert _err(out8, ref) < 8e-2
+
+
+def test_engine_fp8_kv_end_to_end(monkeypatch):
+    """Engine with AURORA_KV_FP8=1: page budget uses 1 byte/elem, chunked
+    prefill reads the fp8 cache, decode completes."""
+    monkeypatch.setenv("AURORA_KV_FP8", "1")
+    from aurora_amd.engine.engine import EngineConfig, LLMEngine
+    from aurora_amd.engine.sampling_params import SamplingParams
+    eng = LLMEngine(EngineConfig(model="smoke-llama", device="cuda", seed=6,
+                                 kv_cache_pages=64, max_batched_tokens=512))
+    assert eng.kv.k_caches[0].dtype == torch.float8_e4m3fn
+    prompts = [torch.randint(1, 900, (700,)).tolist() for _ in range(3)]
+    outs = eng.generate(prompts, SamplingParams(max_tokens=6, ignore_eos=True))
+    assert all(len(o) == 6 for o in outs)
