@@ -40,38 +40,25 @@ async def terminal_exec(ctx: ToolContext, command: str, timeout: int = 60) -> st
     if not verdict.safe:
         return (f"BLOCKED by guardrail ({verdict.source}"
                 f"{':' + verdict.rule_id if verdict.rule_id else ''}): {verdict.reason}")
-    # sandbox: clean env, temp cwd, hard timeout (the reference runs these in
-    # per-user K8s pods — terminal_pod_manager.py:68; locally we subprocess
-    # with a stripped environment)
-    env = {"PATH": "/usr/bin:/bin:/usr/sbin:/sbin", "HOME": tempfile.gettempdir(),
-           "LANG": "C.UTF-8"}
-    try:
-        argv = _exec_argv(command)
-        if argv is not None:
-            # no shell metacharacters: run the tokenized argv directly so the
-            # judged string IS the executed command (no shell re-interpretation)
-            proc = await asyncio.create_subprocess_exec(
-                *argv, stdout=asyncio.subprocess.PIPE,
-                stderr=asyncio.subprocess.PIPE, env=env,
-                cwd=tempfile.gettempdir())
-        else:
-            # pipelines/redirections reach here only after an explicit judge or
-            # org-policy allow (the read-only fast path rejects shell metachars)
-            proc = await asyncio.create_subprocess_shell(
-                command, stdout=asyncio.subprocess.PIPE,
-                stderr=asyncio.subprocess.PIPE, env=env,
-                cwd=tempfile.gettempdir())
-        out, err = await asyncio.wait_for(proc.communicate(), timeout=timeout)
-        text = out.decode("utf-8", "replace")
-        if err:
-            text += "\n[stderr]\n" + err.decode("utf-8", "replace")
-        return text or f"(exit {proc.returncode}, no output)"
-    except asyncio.TimeoutError:
-        try:
-            proc.kill()
-        except ProcessLookupError:
-            pass
+    # per-session sandbox: scratch dir, rlimits, network isolation when
+    # the host supports unprivileged namespaces, idle cleanup (the
+    # reference runs these in per-user K8s pods with a NetworkPolicy —
+    # terminal_pod_manager.py:68; sandbox.py is the host-primitive rebuild)
+    from .sandbox import sandbox_manager
+    argv = _exec_argv(command)
+    # argv path: the judged string IS the executed command (no shell
+    # re-interpretation); pipelines reach the shell path only after an
+    # explicit judge or org-policy allow
+    res = await sandbox_manager().run(
+        ctx.session_id or ctx.user_id or "anon",
+        argv, shell_command=None if argv is not None else command,
+        timeout_s=timeout)
+    if res["timed_out"]:
         return f"ERROR: command timed out after {timeout}s"
+    text = res["stdout"]
+    if res["stderr"]:
+        text += "\n[stderr]\n" + res["stderr"]
+    return text or f"(exit {res['returncode']}, no output)"
 
 
 @R.add("kubectl",

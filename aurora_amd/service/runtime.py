@@ -202,7 +202,8 @@ class Runtime:
         # beat schedules (reference: celery_config.py:113-146)
         self.queue.add_beat("cleanup_stale_investigations", 300.0, {})
         self.queue.add_beat("action_scheduler", 60.0, {})
-        self.queue.add_beat("run_prediscovery", 3600.0, {})  # hourly (reference: celery_config.py:126-129)
+        self.queue.add_beat("run_prediscovery", 3600.0, {})
+        self.queue.add_beat("cleanup_sandboxes", 600.0, {})  # reference: 10 min  # hourly (reference: celery_config.py:126-129)
         self.queue.start()
         # external MCP servers (reference: mcp_preloader started at boot,
         # main_chatbot.py:1684): spawn + register their tools when

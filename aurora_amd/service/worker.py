@@ -235,6 +235,13 @@ async def dispatch_on_incident_actions(rt, args: Dict[str, Any]) -> dict:
     return await run_incident_actions(rt, args["org_id"], args["incident_id"])
 
 
+async def cleanup_sandboxes(rt, args: Dict[str, Any]) -> dict:
+    """Beat: remove idle terminal sandboxes (reference:
+    terminal_pod_cleanup.py:85, every 10 min)."""
+    from ..agent.tools.sandbox import sandbox_manager
+    return {"removed": sandbox_manager().cleanup_idle()}
+
+
 async def action_scheduler(rt, args: Dict[str, Any]) -> dict:
     """Per-minute scheduled-action beat (reference: scheduler.py:38)."""
     from .actions import run_scheduled_actions
@@ -250,6 +257,7 @@ def register_tasks(rt) -> None:
     q.register("dispatch_on_incident_actions",
                lambda a: dispatch_on_incident_actions(rt, a))
     q.register("action_scheduler", lambda a: action_scheduler(rt, a))
+    q.register("cleanup_sandboxes", lambda a: cleanup_sandboxes(rt, a))
     from .change_gating import investigate_pr
     from .prediscovery import run_prediscovery
     from .chat_integrations import handle_chat_command
