@@ -470,6 +470,59 @@ def create_app(rt: Runtime) -> FastAPI:
         return [dict(c, connected=c["name"] in connected)
                 for c in connector_catalog()]
 
+    @app.get("/connectors/{provider}/oauth/start")
+    async def oauth_start(provider: str, redirect_uri: str = "",
+                          store: OrgStore = Depends(auth),
+                          x_user_id: Optional[str] = Header(None)):
+        """Begin the OAuth authorization-code flow (reference: per-connector
+        OAuth routes, connectors/<name>_connector; state is one-shot CSRF)."""
+        _require("connect", x_user_id)
+        from ..connectors import oauth as _oauth
+        from ..connectors.registry import get_connector
+        spec = get_connector(provider)
+        if spec is None or spec.auth != "oauth":
+            raise HTTPException(404, f"{provider} is not an oauth connector")
+        app_creds = store.get_token_ref("system", f"oauth_app:{provider}")
+        from ..utils.secrets import secrets_backend
+        cfg = secrets_backend().get(app_creds) if app_creds else None
+        if not cfg or not cfg.get("client_id"):
+            raise HTTPException(
+                400, f"no OAuth app registered for {provider}: store "
+                     f"client_id/client_secret under oauth_app:{provider}")
+        state = rt.oauth_states.issue(x_user_id or "", provider)
+        url = _oauth.authorize_url(provider, cfg["client_id"],
+                                   redirect_uri or cfg.get("redirect_uri", ""),
+                                   state)
+        return {"authorize_url": url, "state": state}
+
+    @app.get("/connectors/oauth/callback")
+    async def oauth_callback(code: str, state: str,
+                             store: OrgStore = Depends(auth)):
+        """Code -> token exchange; credentials land in the secrets backend,
+        the DB keeps only the reference (AGENTS.md:33)."""
+        from ..connectors import oauth as _oauth
+        from ..connectors.clients import http_transport
+        from ..utils.secrets import secrets_backend
+        st = rt.oauth_states.consume(state)
+        if st is None:
+            raise HTTPException(403, "unknown or expired OAuth state")
+        provider = st["provider"]
+        app_ref = store.get_token_ref("system", f"oauth_app:{provider}")
+        cfg = secrets_backend().get(app_ref) if app_ref else None
+        if not cfg:
+            raise HTTPException(400, "OAuth app config disappeared")
+        transport = getattr(rt, "provider_transport", None) or http_transport()
+        try:
+            creds = await _oauth.exchange_code(
+                transport, provider, cfg["client_id"],
+                cfg.get("client_secret", ""), code,
+                cfg.get("redirect_uri", ""))
+        except Exception as e:  # noqa: BLE001 - no egress / provider error
+            raise HTTPException(502, f"token exchange failed: {e}") from None
+        from ..connectors.registry import connect_provider
+        return connect_provider(rt, store, st["user_id"], provider,
+                                {"credentials": creds})
+
     @app.post("/connectors/{provider}/connect")
     async def connect(provider: str, body: dict, store: OrgStore = Depends(auth),
                       x_user_id: Optional[str] = Header(None)):

@@ -86,6 +86,8 @@ class Runtime:
         self._graph: Dict[str, InfraGraph] = {}
         self._rca_rate: Dict[str, float] = {}  # user -> last background RCA ts
         self.kubectl_agents: Dict[str, Any] = {}  # cluster_id -> AgentConn
+        from ..connectors.oauth import StateCache
+        self.oauth_states = StateCache()  # one-shot OAuth CSRF states
         from . import worker as _worker  # registers task handlers
         _worker.register_tasks(self)
 
