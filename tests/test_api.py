@@ -159,7 +159,7 @@ def test_websocket_chat():
             {"type": "query", "content": "hello"}).encode()))
         await writer.drain()
         types = []
-        for _ in range(400):
+        for _ in range(8000):
             opcode, data = await asyncio.wait_for(read_frame(reader), timeout=120)
             ev = json.loads(data)
             types.append(ev["type"])
