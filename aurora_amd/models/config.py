@@ -107,6 +107,11 @@ register(ModelSpec(  # GPU smoke/test model: real head_dim for the HIP kernels
     head_dim=128, intermediate_size=1024, vocab_size=1024, max_position=8192,
     tie_embeddings=True,
 ))
+register(ModelSpec(  # GPU long-context smoke: 64k positions, tiny layers
+    name="smoke-llama-64k", hidden_size=512, n_layers=4, n_heads=4,
+    n_kv_heads=2, head_dim=128, intermediate_size=1024, vocab_size=1024,
+    max_position=65536, tie_embeddings=True,
+))
 register(ModelSpec(  # GPU MoE smoke: real head_dim for the HIP kernels
     name="smoke-moe", hidden_size=512, n_layers=4, n_heads=4, n_kv_heads=2,
     head_dim=128, intermediate_size=768, vocab_size=1024, max_position=8192,

@@ -460,7 +460,7 @@ def test_engine_long_context_prefill_decode_evict():
     from aurora_amd.engine.engine import EngineConfig, LLMEngine
     from aurora_amd.engine.sampling_params import SamplingParams
 
-    eng = LLMEngine(EngineConfig(model="smoke-llama", device="cuda", seed=9,
+    eng = LLMEngine(EngineConfig(model="smoke-llama-64k", device="cuda", seed=9,
                                  max_model_len=40960, kv_cache_pages=200,
                                  max_batched_tokens=8192))
     P = eng.kv.page_size
