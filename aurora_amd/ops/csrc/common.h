@@ -86,15 +86,21 @@ DEVINL float group_max(float x) {
 // the chip's fp8 is e4m3fn, NOT MI300X fnuz — platform guide §4) -------
 typedef unsigned long long u64b;
 
-// 8 fp8 bytes -> 8 bf16 (used where attention kernels read the cache)
+// 8 fp8 bytes -> 8 bf16 (used where attention kernels read the cache).
+// Every e4m3 value (4-bit exp, 3-bit mantissa) is EXACTLY representable
+// in bf16 (8-bit exp, 7-bit mantissa), so the f32->bf16 step is a pure
+// high-half truncation — no rounding VALU; ~3 ops per converted pair.
 DEVINL shortx8 fp8x8_to_bf16x8(u64b v) {
   shortx8 out;
+  unsigned int* o32 = (unsigned int*)&out;
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
     const floatx2 f = __builtin_amdgcn_cvt_pk_f32_fp8(
         (unsigned int)(v >> (16 * i)) & 0xffffu, false);
-    out[2 * i] = (short)f32_to_bf16(f[0]);
-    out[2 * i + 1] = (short)f32_to_bf16(f[1]);
+    union { float f; unsigned int u; } a, b;
+    a.f = f[0];
+    b.f = f[1];
+    o32[i] = (a.u >> 16) | (b.u & 0xffff0000u);
   }
   return out;
 }
