@@ -222,9 +222,20 @@ async def knowledge_base_search(ctx: ToolContext, query: str, k: int = 5) -> str
 @R.add("web_search", "Search the web for relevant context.",
        {"query": "search query"})
 async def web_search(ctx: ToolContext, query: str) -> str:
-    # The reference proxies a self-hosted SearXNG (web_search_service.py:162);
-    # this environment has no egress, so fall back to the knowledge base and
+    # Self-hosted SearXNG when configured (reference:
+    # web_search_service.py:162); else fall back to the knowledge base and
     # say so explicitly rather than fabricate results.
+    import os
+    base = os.environ.get("AURORA_SEARXNG_URL", "")
+    if base:
+        from ...connectors.clients import NoEgressError, ProviderAPIError
+        from ...connectors.clients.websearch import compose_query, web_search as _ws
+        try:
+            return await _ws(_transport(ctx), base, compose_query(query))
+        except NoEgressError as e:
+            return f"ERROR: SearXNG unreachable (no egress): {e}"
+        except ProviderAPIError as e:
+            return f"ERROR: web search: {e}"
     if ctx.rag is not None:
         hits = ctx.rag.search(query, k=3)
         if hits:

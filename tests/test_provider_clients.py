@@ -406,3 +406,50 @@ def test_provider_tools_wired_through_registry():
         assert "no matches" in out
 
     run(main())
+
+
+# -------------------------------------------------------------- websearch
+
+def test_websearch_compose_build_parse():
+    from aurora_amd.connectors.clients import websearch as ws
+    q = ws.compose_query(
+        "CRITICAL alert: checkout-svc p99 latency is firing on prod-cluster-7")
+    assert "checkout-svc" in q and "prod-cluster-7" in q
+    assert "alert" not in q.lower().split()
+    req = ws.build_search_request("http://searx:8080", q, time_range="week")
+    assert req.url == "http://searx:8080/search"
+    assert req.params["format"] == "json" and req.params["time_range"] == "week"
+    results = ws.parse_search_response({"results": [
+        {"title": "Fixing p99 latency", "url": "https://a.io/x",
+         "content": "tune the pool", "score": 2.0},
+        {"title": "Fixing p99 latency", "url": "https://a.io/y",
+         "content": "dup host+title", "score": 1.9},
+        {"title": "Other", "url": "https://b.io/z", "content": "c",
+         "score": 0.5},
+    ]})
+    assert len(results) == 2  # deduped by host+title
+    assert results[0]["url"] == "https://a.io/x"  # score-ranked
+    out = ws.render_results(results, q)
+    assert "Fixing p99 latency" in out and "https://b.io/z" in out
+
+
+def test_websearch_tool_via_fixture_transport():
+    import aurora_amd.agent.tools.core  # noqa: F401
+    from aurora_amd.agent.tools.registry import GLOBAL_REGISTRY, ToolContext
+
+    async def main():
+        async def fixture(req):
+            assert req.params["q"]
+            return {"results": [{"title": "T", "url": "https://x.io/1",
+                                 "content": "hit", "score": 1}]}
+        import os
+        os.environ["AURORA_SEARXNG_URL"] = "http://searx:8080"
+        try:
+            out = await GLOBAL_REGISTRY.run(
+                ToolContext(extra={"provider_transport": fixture}),
+                "web_search", {"query": "redis timeout spike"})
+            assert "https://x.io/1" in out
+        finally:
+            del os.environ["AURORA_SEARXNG_URL"]
+
+    run(main())
