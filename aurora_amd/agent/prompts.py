@@ -45,10 +45,13 @@ def assemble_system_prompt(segs: List[str]) -> str:
 
 
 def render_history(messages: List[dict], max_messages: int = 24,
-                   tool_output_chars: int = 4000) -> str:
+                   tool_output_chars: int = 1600) -> str:
     """Flatten the conversation for the next completion; last-N message
     window + per-observation truncation (reference: agent.py:86,662-664,
-    691-692 history cleanup)."""
+    691-692 history cleanup). The reference's 4 KB ToolMessage cap is a
+    ~1k-BPE-token budget; with the byte tokenizer (1 token/byte) the
+    equivalent information budget is ~1.6 KB — 4000 here quadrupled every
+    prompt, prefill and KV footprint for the same content."""
     parts: List[str] = []
     for m in messages[-max_messages:]:
         role = m.get("role")
