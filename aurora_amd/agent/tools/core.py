@@ -214,8 +214,10 @@ async def knowledge_base_search(ctx: ToolContext, query: str, k: int = 5) -> str
     hits = ctx.rag.search(query, k=int(k))
     if not hits:
         return "No matching knowledge-base content."
+    # excerpt per hit (full chunks quadruple the observation for no
+    # extra signal; the agent can re-query with a narrower question)
     return "\n\n".join(
-        f"[{i+1}] (score {h['score']:.3f}) {h['doc_name']}:\n{h['text']}"
+        f"[{i+1}] (score {h['score']:.3f}) {h['doc_name']}:\n{h['text'][:400]}"
         for i, h in enumerate(hits))
 
 
