@@ -525,9 +525,10 @@ def test_fp8_kv_roundtrip_and_decode():
     slots = torch.tensor(slot_rows, dtype=torch.int32, device="cuda")
     _hip_ops.kv_cache_write(k, v, kc8, vc8, slots)
     _hip_ops.kv_cache_write(k, v, kc16, vc16, slots)
-    # round-trip error of the fp8 cache itself
+    # round-trip error of the fp8 cache itself: e4m3 has a 2^-3 relative
+    # step, so abs error near |x|~1.5 (randn/4 tail) is up to 0.0625
     rt = kc8.float() - kc16.float()
-    assert rt.abs().max().item() < 0.05
+    assert rt.abs().max().item() < 0.07
     q = torch.randn(B, Hq, D, device="cuda", dtype=torch.bfloat16)
     ctx = torch.tensor(lens, dtype=torch.int32)
     scale = 1.0 / math.sqrt(D)
