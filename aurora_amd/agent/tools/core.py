@@ -402,6 +402,24 @@ _provider_tool("query_slack", "slack",
 _provider_tool("query_jira", "jira", "Search Jira issues with JQL.",
                {"query": "JQL, e.g. project=OPS AND status=Open"})
 _provider_tool("notion_search", "notion", "Search Notion workspace pages.")
+_provider_tool("coroot", "coroot", "Coroot application health overview.")
+_provider_tool("thousandeyes", "thousandeyes",
+               "ThousandEyes: list tests or fetch a test's network results "
+               "(query = test id).")
+_provider_tool("netdata", "netdata",
+               "Netdata chart data with stats (query = chart id, e.g. "
+               "system.cpu).")
+_provider_tool("cloudflare", "cloudflare",
+               "Cloudflare zone HTTP analytics (requests/cache/5xx/threats; "
+               "query = zone tag).")
+_provider_tool("query_opsgenie", "opsgenie",
+               "Opsgenie alert search (query syntax, default status:open).")
+_provider_tool("list_incidentio_incidents", "incidentio",
+               "incident.io incidents ('list') or an incident's timeline "
+               "(query = incident id).")
+_provider_tool("query_flyio_metrics", "flyio",
+               "Fly.io Prometheus metrics (query = PromQL; default HTTP "
+               "status rates).")
 _provider_tool("confluence_search", "confluence",
                "Search Confluence pages/runbooks (text match).")
 

@@ -7,7 +7,7 @@ Reference parity: tools/datadog_tool.py, query_newrelic.py,
 splunk_tool.py, query_sentry.py, dynatrace, slack/jira/confluence/notion
 clusters, github_rca_tool.py, jenkins_rca (SURVEY §2.3 rows
 'Observability query tools' / 'Comms/docs tools' / 'RCA tools')."""
-from . import comms, datadog, devops, observability  # noqa: F401
+from . import comms, datadog, devops, observability, observability2  # noqa: F401
 from .base import (NoEgressError, ProviderAPIError, Request,  # noqa: F401
                    http_transport, parse_time, window)
 
@@ -23,4 +23,11 @@ QUERY_FNS = {
     "jira": comms.jira_query,
     "confluence": comms.confluence_query,
     "notion": comms.notion_query,
+    "coroot": observability2.coroot_query,
+    "thousandeyes": observability2.thousandeyes_query,
+    "netdata": observability2.netdata_query,
+    "cloudflare": observability2.cloudflare_query,
+    "opsgenie": observability2.opsgenie_query,
+    "incidentio": observability2.incidentio_query,
+    "flyio": observability2.flyio_query,
 }

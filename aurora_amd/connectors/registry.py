@@ -34,7 +34,7 @@ def _reg(spec: ConnectorSpec) -> None:
 
 for name, cat, auth, alert, tools in [
     ("pagerduty", "alerting", "token", True, []),
-    ("opsgenie", "alerting", "token", True, []),
+    ("opsgenie", "alerting", "token", True, ["query_opsgenie"]),
     ("datadog", "observability", "token", True, ["query_datadog"]),
     ("grafana", "observability", "token", True, ["query_grafana"]),
     ("newrelic", "observability", "token", True, ["query_newrelic"]),
@@ -42,14 +42,17 @@ for name, cat, auth, alert, tools in [
     ("splunk", "observability", "token", True, ["search_splunk"]),
     ("dynatrace", "observability", "token", True, ["query_dynatrace"]),
     ("bigpanda", "alerting", "token", True, []),
-    ("netdata", "observability", "webhook_only", True, []),
-    ("incidentio", "alerting", "token", True, []),
+    ("netdata", "observability", "token", True, ["netdata"]),
+    ("coroot", "observability", "token", True, ["coroot"]),
+    ("thousandeyes", "observability", "token", False, ["thousandeyes"]),
+    ("cloudflare", "infra", "token", False, ["cloudflare"]),
+    ("incidentio", "alerting", "token", True, ["list_incidentio_incidents"]),
     ("aws", "cloud", "token", True, ["query_cloudwatch"]),
     ("gcp", "cloud", "oauth", False, []),
     ("azure", "cloud", "oauth", False, []),
     ("ovh", "cloud", "token", False, []),
     ("scaleway", "cloud", "token", False, []),
-    ("flyio", "cloud", "token", False, []),
+    ("flyio", "cloud", "token", False, ["query_flyio_metrics"]),
     ("tailscale", "infra", "token", False, []),
     ("kubernetes", "infra", "token", False, []),
     ("github", "code", "oauth", False, ["github_rca", "github_fix"]),

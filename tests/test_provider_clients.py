@@ -453,3 +453,94 @@ def test_websearch_tool_via_fixture_transport():
             del os.environ["AURORA_SEARXNG_URL"]
 
     run(main())
+
+
+# ------------------------------------------------- second observability set
+
+def test_coroot_overview():
+    from aurora_amd.connectors.clients import observability2 as o2
+    out = run(o2.coroot_query(transport_of(lambda r: [
+        {"id": {"name": "checkout"}, "status": "warning",
+         "indicators": [{"message": "p99 above SLO"}]},
+        {"id": {"name": "db"}, "status": "ok", "indicators": []}]),
+        {"base_url": "https://c", "project": "p1"}, ""))
+    assert "2 applications, 1 unhealthy" in out
+    assert "[warning] checkout: p99 above SLO" in out
+
+
+def test_thousandeyes_modes():
+    from aurora_amd.connectors.clients import observability2 as o2
+    creds = {"token": "t"}
+
+    def handler(req):
+        if req.url.endswith("/tests"):
+            return {"tests": [{"testId": 9, "testName": "dc-ping",
+                               "type": "agent-to-server", "interval": 60}]}
+        assert "/test-results/9/network" in req.url
+        return {"results": [{"agent": {"agentName": "fra"}, "loss": 12.5,
+                             "avgLatency": 80, "jitter": 3}]}
+    assert "9: dc-ping" in run(o2.thousandeyes_query(transport_of(handler),
+                                                     creds, "tests"))
+    out = run(o2.thousandeyes_query(transport_of(handler), creds, "9"))
+    assert "max loss 12.5%" in out and "agent fra" in out
+
+
+def test_netdata_series_stats():
+    from aurora_amd.connectors.clients import observability2 as o2
+    out = run(o2.netdata_query(transport_of(lambda r: {
+        "id": "system.cpu", "labels": ["time", "user", "system"],
+        "data": [[1, 10.0, 5.0], [2, 30.0, None], [3, 20.0, 7.0]]}),
+        {"base_url": "http://n:19999"}, "system.cpu", now=NOW))
+    assert "netdata system.cpu" in out
+    assert '"max": 30.0' in out
+
+
+def test_cloudflare_analytics_rollup():
+    from aurora_amd.connectors.clients import observability2 as o2
+
+    def handler(req):
+        assert "graphql" in req.url
+        assert req.json_body["variables"]["zone"] == "ztag"
+        return {"data": {"viewer": {"zones": [{"httpRequests1mGroups": [
+            {"sum": {"requests": 100, "cachedRequests": 60, "threats": 2,
+                     "responseStatusMap": [
+                         {"edgeResponseStatus": 200, "requests": 90},
+                         {"edgeResponseStatus": 502, "requests": 10}]}},
+            {"sum": {"requests": 50, "cachedRequests": 10, "threats": 0,
+                     "responseStatusMap": [
+                         {"edgeResponseStatus": 200, "requests": 50}]}},
+        ]}]}}}
+    out = run(o2.cloudflare_query(transport_of(handler), {"token": "t"},
+                                  "ztag", now=NOW))
+    assert "150 requests" in out and "10 5xx" in out and "200=140" in out
+
+
+def test_opsgenie_and_incidentio_and_flyio():
+    from aurora_amd.connectors.clients import observability2 as o2
+    out = run(o2.opsgenie_query(transport_of(lambda r: {"data": [
+        {"priority": "P1", "status": "open", "message": "db down",
+         "acknowledged": False, "createdAt": "2026"}]}),
+        {"api_key": "k"}, ""))
+    assert "[P1/open] db down" in out
+
+    def io_handler(req):
+        if "incident_updates" in req.url:
+            return {"incident_updates": [
+                {"created_at": "t1", "new_incident_status": {"name": "fixed"},
+                 "message": "rolled back"}]}
+        return {"incidents": [{"reference": "INC-7", "name": "API errors",
+                               "severity": {"name": "major"},
+                               "incident_status": {"name": "open"}}]}
+    assert "INC-7 [major/open]" in run(o2.incidentio_query(
+        transport_of(io_handler), {"api_key": "k"}, "list"))
+    assert "rolled back" in run(o2.incidentio_query(
+        transport_of(io_handler), {"api_key": "k"}, "abc123"))
+
+    def fly_handler(req):
+        assert "query_range" in req.url
+        return {"status": "success", "data": {"result": [
+            {"metric": {"status": "500"},
+             "values": [[1, "0.5"], [2, "1.5"]]}]}}
+    out = run(o2.flyio_query(transport_of(fly_handler),
+                             {"token": "t", "app": "web"}, "", now=NOW))
+    assert '"status": "500"' in out and '"max": 1.5' in out
