@@ -402,6 +402,8 @@ _provider_tool("query_slack", "slack",
 _provider_tool("query_jira", "jira", "Search Jira issues with JQL.",
                {"query": "JQL, e.g. project=OPS AND status=Open"})
 _provider_tool("notion_search", "notion", "Search Notion workspace pages.")
+_provider_tool("sharepoint_search", "sharepoint",
+               "Search SharePoint/OneDrive content (Microsoft Graph).")
 _provider_tool("coroot", "coroot", "Coroot application health overview.")
 _provider_tool("thousandeyes", "thousandeyes",
                "ThousandEyes: list tests or fetch a test's network results "
@@ -512,6 +514,30 @@ async def github_fix_tool(ctx: ToolContext, repo: str, commit_sha: str = "",
 
 
 # ------------------------------------------------------- files / artefacts
+
+
+@R.add("publish_postmortem", "Publish this incident's postmortem to a "
+       "Notion page.", {"parent_page_id": "Notion parent page id"},
+       read_only=False, requires_connection="notion",
+       needs_confirmation=True)
+async def publish_postmortem(ctx: ToolContext, parent_page_id: str) -> str:
+    from ...connectors.clients import NoEgressError, ProviderAPIError
+    from ...connectors.clients.comms import notion_publish_postmortem
+    if ctx.org_store is None or not ctx.incident_id:
+        return "ERROR: no incident bound"
+    pm = ctx.org_store.get_postmortem(ctx.incident_id)
+    if not pm:
+        return "ERROR: this incident has no postmortem yet"
+    inc = ctx.org_store.get_incident(ctx.incident_id) or {}
+    try:
+        return await notion_publish_postmortem(
+            _transport(ctx), _creds(ctx, "notion"), parent_page_id,
+            f"Postmortem: {inc.get('title', ctx.incident_id)}",
+            pm["content"] if isinstance(pm, dict) else str(pm))
+    except NoEgressError as e:
+        return f"ERROR: notion unreachable (no egress): {e}"
+    except ProviderAPIError as e:
+        return f"ERROR: notion: {e}"
 
 
 @R.add("analyze_zip_file",

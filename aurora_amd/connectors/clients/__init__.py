@@ -30,4 +30,5 @@ QUERY_FNS = {
     "opsgenie": observability2.opsgenie_query,
     "incidentio": observability2.incidentio_query,
     "flyio": observability2.flyio_query,
+    "sharepoint": comms.sharepoint_query,
 }

@@ -200,6 +200,88 @@ def _notion_title(page: dict) -> str:
     return page.get("id", "?")
 
 
+def build_notion_postmortem(creds: dict, parent_page_id: str, title: str,
+                            markdown: str) -> Request:
+    """Publish a postmortem as a Notion page (reference: notion postmortem
+    suite ~2.6k LoC): markdown -> Notion block objects (headings, bullets,
+    paragraphs — the shapes build_postmortem emits)."""
+    if not creds.get("token"):
+        raise ProviderAPIError("notion credentials need token")
+    blocks: List[dict] = []
+    for raw in markdown.splitlines():
+        line = raw.rstrip()
+        if not line.strip():
+            continue
+        if line.startswith("### "):
+            t, key = line[4:], "heading_3"
+        elif line.startswith("## "):
+            t, key = line[3:], "heading_2"
+        elif line.startswith("# "):
+            t, key = line[2:], "heading_1"
+        elif line.lstrip().startswith("- "):
+            t, key = line.lstrip()[2:], "bulleted_list_item"
+        else:
+            t, key = line, "paragraph"
+        blocks.append({"object": "block", "type": key,
+                       key: {"rich_text": [{"type": "text",
+                                            "text": {"content": t[:1990]}}]}})
+        if len(blocks) >= 95:  # Notion caps children per request at 100
+            break
+    return Request("POST", "https://api.notion.com/v1/pages",
+                   headers={"Authorization": f"Bearer {creds['token']}",
+                            "Notion-Version": "2022-06-28",
+                            "Content-Type": "application/json"},
+                   json_body={"parent": {"page_id": parent_page_id},
+                              "properties": {"title": {"title": [
+                                  {"type": "text",
+                                   "text": {"content": title[:200]}}]}},
+                              "children": blocks})
+
+
+async def notion_publish_postmortem(transport: Transport, creds: dict,
+                                    parent_page_id: str, title: str,
+                                    markdown: str) -> str:
+    doc = await transport(build_notion_postmortem(creds, parent_page_id,
+                                                  title, markdown))
+    if doc.get("object") == "error":
+        raise ProviderAPIError(str(doc.get("message", "notion error"))[:200])
+    return f"postmortem published to Notion: {doc.get('url', doc.get('id', '?'))}"
+
+
+# -------------------------------------------------------------- SharePoint
+
+
+def build_sharepoint_search(creds: dict, text: str) -> Request:
+    """Microsoft Graph search over SharePoint/OneDrive content
+    (reference: sharepoint search tool)."""
+    if not creds.get("token"):
+        raise ProviderAPIError("sharepoint credentials need token")
+    return Request("POST", "https://graph.microsoft.com/v1.0/search/query",
+                   headers={"Authorization": f"Bearer {creds['token']}",
+                            "Content-Type": "application/json"},
+                   json_body={"requests": [{
+                       "entityTypes": ["driveItem", "listItem"],
+                       "query": {"queryString": text}, "size": 10}]})
+
+
+def parse_sharepoint_search(doc: dict) -> str:
+    rows = []
+    for v in doc.get("value", []):
+        for c in v.get("hitsContainers", []):
+            for h in c.get("hits", [])[:10]:
+                res = h.get("resource", {})
+                rows.append(f"{res.get('name', res.get('id', '?'))}: "
+                            f"{(h.get('summary') or '')[:200]}")
+    return ("sharepoint search:" + ("\n" + "\n".join(rows)
+                                    if rows else " no results"))
+
+
+async def sharepoint_query(transport: Transport, creds: dict, text: str,
+                           **_kw) -> str:
+    return parse_sharepoint_search(await transport(
+        build_sharepoint_search(creds, text)))
+
+
 def parse_notion_search(doc: dict) -> str:
     rows = [f"[{r.get('object', '?')}] {_notion_title(r)} "
             f"(edited {r.get('last_edited_time', '')})"

@@ -544,3 +544,45 @@ def test_opsgenie_and_incidentio_and_flyio():
     out = run(o2.flyio_query(transport_of(fly_handler),
                              {"token": "t", "app": "web"}, "", now=NOW))
     assert '"status": "500"' in out and '"max": 1.5' in out
+
+
+def test_notion_postmortem_markdown_to_blocks():
+    from aurora_amd.connectors.clients import comms
+    md = ("# Postmortem: db outage\n\n## Summary\nPool exhausted.\n"
+          "## Timeline\n- 10:00 alert\n- 10:05 picked up\n")
+    req = comms.build_notion_postmortem({"token": "t"}, "page-1",
+                                        "Postmortem: db outage", md)
+    body = req.json_body
+    assert body["parent"]["page_id"] == "page-1"
+    kinds = [b["type"] for b in body["children"]]
+    assert kinds == ["heading_1", "heading_2", "paragraph", "heading_2",
+                     "bulleted_list_item", "bulleted_list_item"]
+    assert body["children"][0]["heading_1"]["rich_text"][0]["text"][
+        "content"] == "Postmortem: db outage"
+
+    async def main():
+        out = await comms.notion_publish_postmortem(
+            transport_of(lambda r: {"object": "page",
+                                    "url": "https://notion.so/x"}),
+            {"token": "t"}, "page-1", "T", md)
+        assert "https://notion.so/x" in out
+        with pytest.raises(ProviderAPIError):
+            await comms.notion_publish_postmortem(
+                transport_of(lambda r: {"object": "error",
+                                        "message": "denied"}),
+                {"token": "t"}, "page-1", "T", md)
+    run(main())
+
+
+def test_sharepoint_search_graph():
+    from aurora_amd.connectors.clients import comms
+
+    def handler(req):
+        assert req.url.endswith("/search/query")
+        assert req.json_body["requests"][0]["query"]["queryString"] == "runbook db"
+        return {"value": [{"hitsContainers": [{"hits": [
+            {"resource": {"name": "db-runbook.docx"},
+             "summary": "failover steps"}]}]}]}
+    out = run(comms.sharepoint_query(transport_of(handler), {"token": "t"},
+                                     "runbook db"))
+    assert "db-runbook.docx: failover steps" in out
