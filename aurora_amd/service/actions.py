@@ -56,6 +56,55 @@ async def _execute(rt, org_id: str, action: dict, incident_id: str = "") -> None
             pm = build_postmortem(inc, inc["summary"], store.citations(incident_id),
                                   [], store.lifecycle_events(incident_id))
             store.save_postmortem(incident_id, pm)
+    elif kind == "alert_gap":
+        # alert-coverage gap detection (reference: system alert-gap action):
+        # services in the dependency graph that have NEVER produced an
+        # alert-sourced incident have no monitoring wired to Aurora
+        graph = rt.graph(org_id)
+        covered = set()
+        for inc in store.list_incidents(limit=500):
+            svc = (json.loads(inc.get("payload") or "{}")
+                   .get("service") or "")
+            if isinstance(svc, dict):
+                svc = svc.get("summary", "")
+            if svc:
+                covered.add(str(svc).split(".")[0].lower())
+        gaps = [s["name"] for s in graph.list_services()
+                if s["name"].split(":")[0].lower() not in covered]
+        store.save_artifact("alert_gap_report", json.dumps({
+            "generated_at": time.time(), "covered": sorted(covered),
+            "uncovered_services": gaps[:100],
+        }, indent=1))
+        if gaps:
+            from .notifications import dispatch_notification
+            await dispatch_notification(
+                rt, org_id,
+                f"alert-coverage gap: {len(gaps)} graph services have no "
+                f"alert source wired (see alert_gap_report artifact)",
+                incident_id=incident_id)
+    elif kind == "rightsizing":
+        # scheduled right-sizing sweep: run the deterministic analysis over
+        # kubectl top output when a cluster answers, persist the report
+        # (reference: hpa_vpa scheduled action)
+        from ..agent.tools.core import rightsizing_analysis
+        from ..agent.tools.registry import ToolContext
+        usage = ""
+        try:
+            import asyncio as _aio
+            proc = await _aio.create_subprocess_shell(
+                "kubectl top pods -A --no-headers 2>/dev/null",
+                stdout=_aio.subprocess.PIPE)
+            out, _ = await _aio.wait_for(proc.communicate(), timeout=15)
+            usage = out.decode("utf-8", "replace")
+        except Exception:  # noqa: BLE001
+            usage = ""
+        if usage.strip():
+            report = await rightsizing_analysis(ToolContext(), usage)
+            store.save_artifact("rightsizing_report", report)
+        else:
+            store.save_artifact("rightsizing_report",
+                                "no cluster metrics available "
+                                "(kubectl top unreachable)")
     elif kind == "agent_prompt":
         # headless one-shot agent run with a fixed prompt (right-sizing /
         # alert-gap style automations)

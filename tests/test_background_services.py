@@ -141,3 +141,36 @@ def test_rca_admission_defers_over_cap(rt, org):
         assert store.get_incident(iid)["aurora_status"] == "failed"
     finally:
         rt.settings.max_concurrent_rca = old
+
+
+def test_alert_gap_system_action():
+    """alert_gap finds graph services with no alert-sourced incidents
+    (reference: system alert-gap action)."""
+    import asyncio
+    import json as _json
+    from aurora_amd.engine.engine import EngineConfig
+    from aurora_amd.engine.server import AsyncEngineServer
+    from aurora_amd.service.actions import _execute
+    from aurora_amd.service.runtime import Runtime
+    from aurora_amd.utils.config import Settings
+
+    srv = AsyncEngineServer(EngineConfig(model="tiny-llama", device="cpu"))
+    rt = Runtime(Settings(db_path=":memory:",
+                          secrets_path="/tmp/ag_sec.json"),
+                 engine_server=srv)
+    oid = rt.admin.create_org("gap")
+    store = rt.org(oid)
+    g = rt.graph(oid)
+    for svc in ("api", "db", "shadow-svc"):
+        g.upsert_service(svc)
+    store.create_incident("api 5xx", "pagerduty", {"service": "api"})
+
+    async def main():
+        await _execute(rt, oid, {"id": "a1", "name": "gap-check",
+                                 "config": _json.dumps({"kind": "alert_gap"})})
+    asyncio.run(main())
+    rep = _json.loads(store.get_artifact("alert_gap_report")["content"])
+    assert "api" in rep["covered"]
+    assert "shadow-svc" in rep["uncovered_services"]
+    assert "db" in rep["uncovered_services"]
+    srv.shutdown()
