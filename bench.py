@@ -414,7 +414,8 @@ def main():
     # max over ranks (slowest rank defines the job)
     if world > 1:
         import torch.distributed as dist
-        t = torch.tensor([elapsed])
+        dev = "cuda" if (use_cuda and dist.get_backend() == "nccl") else "cpu"
+        t = torch.tensor([elapsed], device=dev)  # RCCL needs device tensors
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t[0])
 
