@@ -145,6 +145,27 @@ def series_stats(points: List[float]) -> dict:
             "last": round(vals[-1], 4)}
 
 
+def fail_closed(fn):
+    """Parser guard: a hostile or half-changed provider response must
+    degrade into an explicit ProviderAPIError (which tools render as an
+    ERROR observation), never leak a TypeError/KeyError into the agent
+    turn."""
+    import functools
+
+    @functools.wraps(fn)
+    def wrapped(*a, **k):
+        try:
+            return fn(*a, **k)
+        except ProviderAPIError:
+            raise
+        except (AttributeError, TypeError, KeyError, ValueError,
+                IndexError) as e:
+            raise ProviderAPIError(
+                f"malformed provider response ({type(e).__name__}: "
+                f"{str(e)[:120]})") from e
+    return wrapped
+
+
 def truncate_items(items: List[str], max_items: int = 25,
                    max_chars: int = 8000) -> tuple:
     """(kept_items, truncated?) under both count and char budgets

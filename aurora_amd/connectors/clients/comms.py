@@ -8,7 +8,8 @@ import json
 import re
 from typing import Dict, List, Optional
 
-from .base import (ProviderAPIError, Request, Transport, truncate_items,
+from .base import (fail_closed,
+                   ProviderAPIError, Request, Transport, truncate_items,
                    window)
 
 # ------------------------------------------------------------------ Slack
@@ -46,6 +47,7 @@ def _slack_ok(doc: dict) -> dict:
     return doc
 
 
+@fail_closed
 def parse_slack_search(doc: dict) -> str:
     _slack_ok(doc)
     matches = doc.get("messages", {}).get("matches", [])
@@ -56,6 +58,7 @@ def parse_slack_search(doc: dict) -> str:
             + ("\n" + "\n".join(rows) if rows else " — none"))
 
 
+@fail_closed
 def parse_slack_history(doc: dict) -> str:
     _slack_ok(doc)
     rows = [f"[{m.get('ts', '')}] {m.get('user', m.get('bot_id', '?'))}: "
@@ -64,6 +67,7 @@ def parse_slack_history(doc: dict) -> str:
     return ("slack history:" + ("\n" + "\n".join(rows) if rows else " empty"))
 
 
+@fail_closed
 def parse_slack_channels(doc: dict) -> str:
     _slack_ok(doc)
     rows = [f"#{c.get('name')} ({c.get('id')}, members={c.get('num_members', '?')})"
@@ -107,6 +111,7 @@ def build_jira_search(creds: dict, jql: str, start_at: int = 0,
                               "fields": JIRA_FIELDS})
 
 
+@fail_closed
 def parse_jira_search(doc: dict) -> tuple:
     """-> (rows, total)."""
     rows: List[str] = []
@@ -159,6 +164,7 @@ def build_confluence_search(creds: dict, text: str,
                            "expand": "body.view,space,version"})
 
 
+@fail_closed
 def parse_confluence_search(doc: dict) -> str:
     rows = []
     for p in doc.get("results", [])[:10]:
@@ -264,6 +270,7 @@ def build_sharepoint_search(creds: dict, text: str) -> Request:
                        "query": {"queryString": text}, "size": 10}]})
 
 
+@fail_closed
 def parse_sharepoint_search(doc: dict) -> str:
     rows = []
     for v in doc.get("value", []):
@@ -282,6 +289,7 @@ async def sharepoint_query(transport: Transport, creds: dict, text: str,
         build_sharepoint_search(creds, text)))
 
 
+@fail_closed
 def parse_notion_search(doc: dict) -> str:
     rows = [f"[{r.get('object', '?')}] {_notion_title(r)} "
             f"(edited {r.get('last_edited_time', '')})"

@@ -10,7 +10,8 @@ from __future__ import annotations
 import json
 from typing import Dict, List, Optional
 
-from .base import (ProviderAPIError, Request, Transport, iso, series_stats,
+from .base import (fail_closed,
+                   ProviderAPIError, Request, Transport, iso, series_stats,
                    truncate_items, window)
 
 MAX_LOG_PAGES = 3
@@ -57,6 +58,7 @@ def build_logs_request(creds: dict, query: str, time_from: Optional[str],
                    headers=_auth(creds), json_body=body)
 
 
+@fail_closed
 def parse_logs_response(doc: dict) -> tuple:
     """-> (lines, next_cursor)."""
     lines: List[str] = []
@@ -101,6 +103,7 @@ def build_metrics_request(creds: dict, query: str, time_from: Optional[str],
                    params={"query": query, "from": int(f), "to": int(t)})
 
 
+@fail_closed
 def parse_metrics_response(doc: dict) -> str:
     if doc.get("status") == "error":
         raise ProviderAPIError(doc.get("error", "datadog query error"))
@@ -137,6 +140,7 @@ def build_monitors_request(creds: dict, query: str) -> Request:
                    params={"query": query or "status:alert", "per_page": 30})
 
 
+@fail_closed
 def parse_monitors_response(doc: dict) -> str:
     rows = []
     for m in doc.get("monitors", [])[:30]:
@@ -160,6 +164,7 @@ def build_events_request(creds: dict, time_from: Optional[str],
                    headers=_auth(creds), params=params)
 
 
+@fail_closed
 def parse_events_response(doc: dict) -> str:
     rows = []
     for e in doc.get("events", [])[:40]:

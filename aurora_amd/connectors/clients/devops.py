@@ -12,7 +12,8 @@ import json
 import re
 from typing import Dict, List, Optional
 
-from .base import (ProviderAPIError, Request, Transport, iso, parse_time,
+from .base import (fail_closed,
+                   ProviderAPIError, Request, Transport, iso, parse_time,
                    truncate_items)
 
 # ------------------------------------------------------------------ GitHub
@@ -167,6 +168,7 @@ def build_console(creds: dict, job: str, number: int) -> Request:
                    headers=_jenkins_headers(creds))
 
 
+@fail_closed
 def parse_builds(doc: dict) -> List[dict]:
     return doc.get("builds", [])
 

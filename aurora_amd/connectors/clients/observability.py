@@ -11,7 +11,8 @@ from __future__ import annotations
 import json
 from typing import Dict, List, Optional
 
-from .base import (ProviderAPIError, Request, Transport, iso, series_stats,
+from .base import (fail_closed,
+                   ProviderAPIError, Request, Transport, iso, series_stats,
                    truncate_items, window)
 
 # ------------------------------------------------------------- New Relic
@@ -32,6 +33,7 @@ def build_nrql_request(creds: dict, nrql: str) -> Request:
                    json_body={"query": gql})
 
 
+@fail_closed
 def parse_nrql_response(doc: dict) -> str:
     if doc.get("errors"):
         raise ProviderAPIError(
@@ -78,6 +80,7 @@ def build_splunk_search(creds: dict, query: str, time_from: Optional[str],
                    "count": limit})
 
 
+@fail_closed
 def parse_splunk_results(doc: dict) -> str:
     rows = []
     for r in doc.get("results", [])[:60]:
@@ -99,6 +102,7 @@ def build_splunk_indexes(creds: dict) -> Request:
                    params={"output_mode": "json", "count": 50})
 
 
+@fail_closed
 def parse_splunk_indexes(doc: dict) -> str:
     names = [e.get("name", "?") for e in doc.get("entry", [])]
     return "splunk indexes: " + (", ".join(names) if names else "(none)")
@@ -129,6 +133,7 @@ def build_sentry_issues(creds: dict, query: str,
                 "limit": 25, "sort": "freq"})
 
 
+@fail_closed
 def parse_sentry_issues(doc) -> str:
     if isinstance(doc, dict):  # error envelope
         raise ProviderAPIError(str(doc.get("detail", doc))[:300])
@@ -162,6 +167,7 @@ def build_dynatrace_problems(creds: dict, time_from: Optional[str],
                            "pageSize": 25})
 
 
+@fail_closed
 def parse_dynatrace_problems(doc: dict) -> str:
     rows = []
     for p in doc.get("problems", [])[:25]:
@@ -186,6 +192,7 @@ def build_dynatrace_metrics(creds: dict, selector: str,
                            "resolution": "120"})
 
 
+@fail_closed
 def parse_dynatrace_metrics(doc: dict) -> str:
     rows = []
     for res in doc.get("result", [])[:10]:
@@ -222,6 +229,7 @@ def build_grafana_search(creds: dict, query: str) -> Request:
                    params={"query": query or "", "limit": 30})
 
 
+@fail_closed
 def parse_grafana_search(doc) -> str:
     rows = [f"[{d.get('type', '?')}] {d.get('title', '?')} "
             f"(uid={d.get('uid', '?')}, folder={d.get('folderTitle', '')})"
@@ -262,6 +270,7 @@ def build_cloudwatch_metric(creds: dict, namespace: str, metric: str,
                    json_body=body)
 
 
+@fail_closed
 def parse_cloudwatch_metric(doc: dict) -> str:
     pts = sorted(doc.get("Datapoints", []),
                  key=lambda p: p.get("Timestamp", ""))

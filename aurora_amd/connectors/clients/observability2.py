@@ -12,7 +12,8 @@ from __future__ import annotations
 import json
 from typing import List, Optional
 
-from .base import (ProviderAPIError, Request, Transport, iso, series_stats,
+from .base import (fail_closed,
+                   ProviderAPIError, Request, Transport, iso, series_stats,
                    truncate_items, window)
 
 # ------------------------------------------------------------------ coroot
@@ -29,6 +30,7 @@ def build_coroot_overview(creds: dict) -> Request:
                    + f"/api/project/{pid}/overview/applications", headers=h)
 
 
+@fail_closed
 def parse_coroot_overview(doc) -> str:
     apps = doc if isinstance(doc, list) else doc.get("applications", [])
     rows = []
@@ -66,6 +68,7 @@ def build_te_results(creds: dict, test_id: str) -> Request:
                    headers={"Authorization": f"Bearer {creds['token']}"})
 
 
+@fail_closed
 def parse_te_results(doc: dict) -> str:
     results = doc.get("results", [])
     rows = []
@@ -108,6 +111,7 @@ def build_netdata_data(creds: dict, chart: str, time_from: Optional[str],
                            "points": 120, "format": "json"})
 
 
+@fail_closed
 def parse_netdata_data(doc: dict) -> str:
     labels = doc.get("labels", [])
     data = doc.get("data", [])
@@ -151,6 +155,7 @@ def build_cf_analytics(creds: dict, zone: str, time_from: Optional[str],
                                             "until": iso(t)}})
 
 
+@fail_closed
 def parse_cf_analytics(doc: dict) -> str:
     if doc.get("errors"):
         raise ProviderAPIError(
@@ -197,6 +202,7 @@ def build_opsgenie_alerts(creds: dict, query: str) -> Request:
                            "sort": "createdAt", "order": "desc"})
 
 
+@fail_closed
 def parse_opsgenie_alerts(doc: dict) -> str:
     rows = [f"[{a.get('priority', '?')}/{a.get('status', '?')}] "
             f"{a.get('message', '')[:140]} "
@@ -229,6 +235,7 @@ def build_incidentio_timeline(creds: dict, incident_id: str) -> Request:
                    params={"incident_id": incident_id, "page_size": 50})
 
 
+@fail_closed
 def parse_incidentio_list(doc: dict) -> str:
     rows = []
     for i in doc.get("incidents", [])[:25]:
@@ -239,6 +246,7 @@ def parse_incidentio_list(doc: dict) -> str:
     return ("incident.io:" + ("\n" + "\n".join(rows) if rows else " none"))
 
 
+@fail_closed
 def parse_incidentio_timeline(doc: dict) -> str:
     rows = [f"[{u.get('created_at', '')}] "
             f"{(u.get('new_incident_status') or {}).get('name', '')}: "
@@ -275,6 +283,7 @@ def build_flyio_metrics(creds: dict, app: str, promql: str,
                            "start": int(f), "end": int(t), "step": 60})
 
 
+@fail_closed
 def parse_flyio_metrics(doc: dict) -> str:
     if doc.get("status") != "success":
         raise ProviderAPIError(str(doc.get("error", "prometheus error"))[:200])

@@ -12,7 +12,7 @@ from __future__ import annotations
 import re
 from typing import List, Optional
 
-from .base import ProviderAPIError, Request, Transport, truncate_items
+from .base import fail_closed, ProviderAPIError, Request, Transport, truncate_items
 
 #: noise words stripped when composing a search query from alert text
 _STOP = {"the", "a", "an", "is", "are", "was", "on", "in", "of", "for",
@@ -48,6 +48,7 @@ def build_search_request(base_url: str, query: str,
     return Request("GET", base_url.rstrip("/") + "/search", params=params)
 
 
+@fail_closed
 def parse_search_response(doc: dict, max_results: int = 6) -> List[dict]:
     """Dedup by host+title, rank by SearXNG score, strip tracking junk."""
     out: List[dict] = []
