@@ -461,6 +461,150 @@ def create_app(rt: Runtime) -> FastAPI:
             g.add_dependency(body["name"], dep)
         return {"ok": True}
 
+    # ---- actions (reference: services/actions routes) ----
+    @app.get("/actions")
+    async def list_actions(store: OrgStore = Depends(auth)):
+        return store._select("actions")
+
+    @app.post("/actions")
+    async def create_action(body: dict, store: OrgStore = Depends(auth),
+                            x_user_id: Optional[str] = Header(None)):
+        _require("write", x_user_id)
+        if body.get("trigger") not in ("on_incident", "schedule"):
+            raise HTTPException(400, "trigger must be on_incident|schedule")
+        from .store import new_id
+        aid = new_id("act")
+        store._insert("actions", dict(
+            id=aid, name=str(body.get("name", "unnamed"))[:200],
+            trigger=body["trigger"], enabled=1, created_at=time.time(),
+            config=json.dumps(body.get("config") or {}), last_run=0))
+        return {"id": aid}
+
+    @app.post("/actions/{aid}/toggle")
+    async def toggle_action(aid: str, store: OrgStore = Depends(auth),
+                            x_user_id: Optional[str] = Header(None)):
+        _require("write", x_user_id)
+        row = store._select("actions", "id=?", [aid])
+        if not row:
+            raise HTTPException(404, "no such action")
+        new = 0 if row[0]["enabled"] else 1
+        store._update("actions", {"enabled": new}, "id=?", [aid])
+        return {"id": aid, "enabled": bool(new)}
+
+    @app.delete("/actions/{aid}")
+    async def delete_action(aid: str, store: OrgStore = Depends(auth),
+                            x_user_id: Optional[str] = Header(None)):
+        _require("write", x_user_id)
+        return {"deleted": store._delete("actions", "id=?", [aid]) > 0}
+
+    # ---- skills catalog (reference: skills registry surface) ----
+    @app.get("/skills")
+    async def skills_catalog(store: OrgStore = Depends(auth)):
+        from ..agent.skills.registry import get_skill_registry
+        return get_skill_registry().catalog()
+
+    @app.get("/skills/{name}")
+    async def skill_body(name: str, store: OrgStore = Depends(auth)):
+        from ..agent.skills.registry import get_skill_registry
+        sk = get_skill_registry().get(name)
+        if sk is None:
+            raise HTTPException(404, "unknown skill")
+        return {"name": sk.name, "description": sk.description,
+                "category": sk.category, "body": sk.body}
+
+    # ---- artifacts by name (reference: artifact_routes.py) ----
+    @app.get("/artifacts/{name}")
+    async def get_artifact(name: str, store: OrgStore = Depends(auth)):
+        row = store.get_artifact(name)
+        if row is None:
+            raise HTTPException(404, "no such artifact")
+        return row
+
+    @app.put("/artifacts/{name}")
+    async def put_artifact(name: str, body: dict,
+                           store: OrgStore = Depends(auth),
+                           x_user_id: Optional[str] = Header(None)):
+        _require("write", x_user_id)
+        store.save_artifact(name, str(body.get("content", ""))[:200000],
+                            session_id=body.get("session_id"))
+        return {"ok": True}
+
+    # ---- incident mutation/extras ----
+    @app.patch("/incidents/{iid}")
+    async def patch_incident(iid: str, body: dict,
+                             store: OrgStore = Depends(auth),
+                             x_user_id: Optional[str] = Header(None)):
+        _require("write", x_user_id)
+        allowed = {k: v for k, v in body.items()
+                   if k in ("title", "severity", "status", "user_id")}
+        if not allowed:
+            raise HTTPException(400, "no mutable fields given "
+                                     "(title/severity/status/user_id)")
+        if store.update_incident(iid, **allowed) == 0:
+            raise HTTPException(404, "no such incident")
+        store.lifecycle(iid, "updated", allowed)
+        return {"ok": True, "updated": sorted(allowed)}
+
+    @app.get("/incidents/{iid}/hypotheses")
+    async def hypotheses(iid: str, store: OrgStore = Depends(auth)):
+        row = store.get_artifact(f"hypothesis_ledger:{iid}")
+        if row is None:
+            return {"hypotheses": [], "ruled_out": [], "unexplored": []}
+        return json.loads(row["content"])
+
+    @app.get("/sessions")
+    async def list_sessions(limit: int = 50,
+                            store: OrgStore = Depends(auth)):
+        rows = store._select("chat_sessions", order="created_at DESC",
+                             limit=max(1, min(int(limit), 200)))
+        return [{k: r[k] for k in ("id", "user_id", "is_background",
+                                   "incident_id", "created_at")}
+                for r in rows]
+
+    # ---- knowledge base delete ----
+    @app.delete("/knowledge-base/documents/{did}")
+    async def kb_delete(did: str, store: OrgStore = Depends(auth),
+                        x_user_id: Optional[str] = Header(None)):
+        _require("write", x_user_id)
+        store.delete_kb_document(did)
+        rt._rag.pop(store.org_id, None)  # rebuild the index lazily
+        return {"deleted": did}
+
+    # ---- webhook secret management (reference: per-connector secret) ----
+    @app.post("/org/webhook-secrets")
+    async def set_webhook_secret(body: dict, store: OrgStore = Depends(auth),
+                                 x_user_id: Optional[str] = Header(None)):
+        _require("admin", x_user_id)
+        source = body.get("source", "")
+        if source not in sources() and source not in ("github", "bitbucket"):
+            raise HTTPException(400, f"unknown source {source!r}")
+        store.store_tokens("system", f"webhook_secret:{source}",
+                           str(body.get("secret", "")))
+        store.audit("webhook_secret_set", source)
+        return {"ok": True, "source": source}
+
+    # ---- engine status (reference: engine counters surface) ----
+    @app.get("/engine/status")
+    async def engine_status(store: OrgStore = Depends(auth)):
+        eng = rt.engine.engine
+        return {"healthy": rt.engine.healthy(),
+                "model": rt.settings.main_model,
+                "device": str(getattr(eng, "device", "cpu")),
+                "counters": rt.engine.counters(),
+                "kv_pages_free": eng.kv.num_free(),
+                "kv_pages_total": eng.kv.num_pages,
+                "running": eng.num_running(),
+                "aux_models": sorted(rt._aux_engines)}
+
+    @app.post("/connectors/{provider}/disconnect")
+    async def disconnect(provider: str, store: OrgStore = Depends(auth),
+                         x_user_id: Optional[str] = Header(None)):
+        _require("connect", x_user_id)
+        n = store._delete("connector_tokens", "user_id=? AND provider=?",
+                          [x_user_id or "", provider])
+        store.audit("connector_disconnected", f"{provider} by {x_user_id}")
+        return {"disconnected": n > 0}
+
     # ---- connectors ----
     @app.get("/connectors")
     async def connectors(store: OrgStore = Depends(auth),

@@ -405,3 +405,70 @@ def test_unsigned_webhook_audited_and_gatable(env, monkeypatch):
     r = client.post(f"/webhook/pagerduty/{oid}", json=payload)
     assert r.status_code == 403
     monkeypatch.setattr(cfgmod.settings(), "require_webhook_secrets", False)
+
+
+def test_round2_route_batch(env):
+    """Actions CRUD, skills catalog, artifacts by name, incident patch,
+    sessions, kb delete, webhook secrets, engine status, disconnect."""
+    rt, client, oid, uid = env
+    rt.db.execute("UPDATE users SET role='admin' WHERE id=?", [uid])
+    H2 = {"X-User-Id": uid}
+    # actions
+    r = client.post("/actions", headers=H2, json={
+        "name": "notify-oncall", "trigger": "on_incident",
+        "config": {"kind": "notify", "message": "incident!"}})
+    assert r.status_code == 200
+    aid = r.json()["id"]
+    assert any(a["id"] == aid for a in client.get("/actions",
+                                                  headers=H2).json())
+    assert client.post(f"/actions/{aid}/toggle",
+                       headers=H2).json()["enabled"] is False
+    assert client.delete(f"/actions/{aid}", headers=H2).json()["deleted"]
+    assert client.post("/actions", headers=H2,
+                       json={"trigger": "bogus"}).status_code == 400
+    # skills
+    cat = client.get("/skills", headers=H2).json()
+    assert len(cat) >= 10
+    name = cat[0]["name"]
+    assert client.get(f"/skills/{name}", headers=H2).json()["body"]
+    assert client.get("/skills/notaskill", headers=H2).status_code == 404
+    # artifacts by name
+    assert client.put("/artifacts/runbook-x", headers=H2,
+                      json={"content": "steps"}).status_code == 200
+    assert client.get("/artifacts/runbook-x",
+                      headers=H2).json()["content"] == "steps"
+    assert client.get("/artifacts/ghost", headers=H2).status_code == 404
+    # incident patch
+    iid = rt.org(oid).create_incident(title="patchme", source="agent",
+                                      payload={})
+    r = client.patch(f"/incidents/{iid}", headers=H2,
+                     json={"severity": "critical", "bogus": "x"})
+    assert r.json()["updated"] == ["severity"]
+    assert rt.org(oid).get_incident(iid)["severity"] == "critical"
+    assert client.patch(f"/incidents/{iid}", headers=H2,
+                        json={"bogus": 1}).status_code == 400
+    # hypotheses (empty default)
+    assert client.get(f"/incidents/{iid}/hypotheses",
+                      headers=H2).json()["hypotheses"] == []
+    # sessions
+    rt.org(oid).create_session(user_id=uid)
+    assert len(client.get("/sessions", headers=H2).json()) >= 1
+    # kb delete
+    did = client.post("/knowledge-base/documents", headers=H2,
+                      json={"name": "d", "content": "c"}).json()["document_id"]
+    assert client.delete(f"/knowledge-base/documents/{did}",
+                         headers=H2).json()["deleted"] == did
+    # webhook secrets (admin only — user-a was promoted earlier)
+    r = client.post("/org/webhook-secrets", headers=H2,
+                    json={"source": "pagerduty", "secret": "s3"})
+    assert r.status_code == 200
+    assert client.post("/org/webhook-secrets", headers=H2,
+                       json={"source": "nope"}).status_code == 400
+    # engine status
+    st = client.get("/engine/status", headers=H2).json()
+    assert st["kv_pages_total"] > 0 and "counters" in st
+    # connector disconnect
+    client.post("/connectors/pagerduty/connect", headers=H2,
+                json={"credentials": {"token": "t"}})
+    assert client.post("/connectors/pagerduty/disconnect",
+                       headers=H2).json()["disconnected"]
