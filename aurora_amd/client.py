@@ -115,6 +115,58 @@ class AuroraClient:
         return self.http.get("/knowledge-base/search",
                              params={"q": query, "k": k}).json()
 
+    # ---- round-2 surface ----
+    def actions(self) -> List[dict]:
+        return self._get("/actions")
+
+    def create_action(self, name: str, trigger: str, config: dict) -> str:
+        return self._post("/actions", {"name": name, "trigger": trigger,
+                                       "config": config})["id"]
+
+    def toggle_action(self, aid: str) -> dict:
+        return self._post(f"/actions/{aid}/toggle", {})
+
+    def delete_action(self, aid: str) -> dict:
+        r = self.http.delete(f"/actions/{aid}")
+        r.raise_for_status()
+        return r.json()
+
+    def skills(self) -> List[dict]:
+        return self._get("/skills")
+
+    def skill(self, name: str) -> dict:
+        return self._get(f"/skills/{name}")
+
+    def artifact(self, name: str) -> dict:
+        return self._get(f"/artifacts/{name}")
+
+    def save_artifact(self, name: str, content: str) -> dict:
+        r = self.http.put(f"/artifacts/{name}", json={"content": content})
+        r.raise_for_status()
+        return r.json()
+
+    def update_incident(self, iid: str, **fields) -> dict:
+        r = self.http.patch(f"/incidents/{iid}", json=fields)
+        r.raise_for_status()
+        return r.json()
+
+    def hypotheses(self, iid: str) -> dict:
+        return self._get(f"/incidents/{iid}/hypotheses")
+
+    def sessions(self, limit: int = 50) -> List[dict]:
+        return self._get(f"/sessions?limit={limit}")
+
+    def engine_status(self) -> dict:
+        return self._get("/engine/status")
+
+    def oauth_start(self, provider: str, redirect_uri: str = "") -> dict:
+        return self._get(f"/connectors/{provider}/oauth/start"
+                         + (f"?redirect_uri={redirect_uri}"
+                            if redirect_uri else ""))
+
+    def disconnect_provider(self, provider: str) -> dict:
+        return self._post(f"/connectors/{provider}/disconnect", {})
+
     def connect_provider(self, provider: str, credentials: dict) -> dict:
         return self.http.post(f"/connectors/{provider}/connect",
                               json={"credentials": credentials}).json()

@@ -472,3 +472,23 @@ def test_round2_route_batch(env):
                 json={"credentials": {"token": "t"}})
     assert client.post("/connectors/pagerduty/disconnect",
                        headers=H2).json()["disconnected"]
+
+
+def test_sdk_round2_surface(env):
+    rt, client, oid, uid = env
+    from aurora_amd.client import AuroraClient
+    sdk = AuroraClient(user_id=uid, org_id=oid)
+    sdk.http = client
+    client.headers.update({"X-User-Id": uid, "X-Org-Id": oid})
+    aid = sdk.create_action("pm", "on_incident", {"kind": "postmortem"})
+    assert any(a["id"] == aid for a in sdk.actions())
+    sdk.toggle_action(aid)
+    sdk.delete_action(aid)
+    assert sdk.skills() and sdk.skill(sdk.skills()[0]["name"])["body"]
+    sdk.save_artifact("sdk-doc", "hello")
+    assert sdk.artifact("sdk-doc")["content"] == "hello"
+    iid = rt.org(oid).create_incident(title="sdk2", source="agent", payload={})
+    assert sdk.update_incident(iid, severity="low")["ok"]
+    assert sdk.hypotheses(iid)["hypotheses"] == []
+    assert isinstance(sdk.sessions(), list)
+    assert sdk.engine_status()["kv_pages_total"] > 0
