@@ -80,3 +80,20 @@ def test_ep_moe_matches_reference():
     for rank, err in outs:
         assert not isinstance(err, str), err
         assert err < 1e-4, (rank, err)
+
+
+def test_ep_moe_four_ranks():
+    """EP=4 on gloo: uneven routing across 4 owners still matches the
+    reference computation (the Mixtral TP=4 shape, BASELINE config 4)."""
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_ep_worker, args=(r, 4, 29571, q))
+             for r in range(4)]
+    for p in procs:
+        p.start()
+    outs = [q.get(timeout=300) for _ in range(4)]
+    for p in procs:
+        p.join(timeout=60)
+    for rank, err in outs:
+        assert not isinstance(err, str), err
+        assert err < 1e-4, (rank, err)
