@@ -2,7 +2,6 @@
 execution, introspection, knowledge and artifact surface rebuilt)."""
 from __future__ import annotations
 
-import asyncio
 import json
 import shlex
 import tempfile

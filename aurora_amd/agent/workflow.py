@@ -17,7 +17,7 @@ import json
 import time
 from typing import Any, AsyncIterator, Dict, List, Optional
 
-from ..engine.grammar import JsonGrammar, ReActTurnGrammar
+from ..engine.grammar import ReActTurnGrammar
 from ..engine.sampling_params import SamplingParams
 from ..guardrails.input_rail import check_input
 from ..utils.config import settings as get_settings

@@ -12,7 +12,7 @@ sends — same math, test-only transport.
 from __future__ import annotations
 
 import os
-from typing import List, Tuple
+from typing import List
 
 import torch
 import torch.distributed as dist

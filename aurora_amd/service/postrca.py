@@ -7,7 +7,7 @@ from __future__ import annotations
 
 import json
 import re
-from typing import List, Optional
+from typing import List
 
 from ..engine.grammar import JsonGrammar
 from ..engine.sampling_params import SamplingParams
