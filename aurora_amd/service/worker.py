@@ -118,7 +118,11 @@ async def run_background_chat(rt, args: Dict[str, Any]) -> dict:
         msg = str(e)
         if "engine step failed" in msg or "engine shutdown" in msg:
             # transient engine fault: re-raise so the queue retries with
-            # backoff (semantic failures below stay terminal)
+            # backoff (semantic failures below stay terminal). Status must
+            # leave 'running' FIRST or the retry dedups against our own
+            # crashed attempt and the incident strands (found by the
+            # fault-injection test; Celery acks-late redelivers fresh).
+            store.update_incident(incident_id, aurora_status="pending")
             store.lifecycle(incident_id, "rca_engine_fault", {"error": msg})
             raise
         error = f"workflow failure: {e}"

@@ -166,3 +166,69 @@ def test_sustained_webhook_load_drains_and_bounds():
             assert not eng.running and not eng.waiting  # engine drained
     finally:
         server.shutdown()
+
+
+def test_chaos_engine_faults_requeue_until_complete():
+    """Fault injection: an engine that fails transiently N
+    times mid-investigation must not strand the incident — the queue's
+    acks-late retry relaunches run_background_chat until it completes."""
+    import asyncio
+    import json as _json
+
+    from aurora_amd.engine.engine import EngineConfig, StepOutput
+    from aurora_amd.engine.server import AsyncEngineServer
+    from aurora_amd.service.runtime import Runtime
+    from aurora_amd.service.webhooks import process_webhook
+    from aurora_amd.utils.config import Settings
+
+    class FaultyServer(AsyncEngineServer):
+        """Real CPU engine server whose stream injects transient engine
+        faults for the first few requests."""
+
+        def __init__(self, cfg, faults=3):
+            super().__init__(cfg)
+            self.faults_left = faults
+
+        async def stream(self, ids, params, session_id=None, request_id=None):
+            if self.faults_left > 0:
+                self.faults_left -= 1
+                yield StepOutput(request_id or "x", [], "", True,
+                                 "engine step failed: injected fault")
+                return
+            async for out in super().stream(ids, params,
+                                            session_id=session_id,
+                                            request_id=request_id):
+                yield out
+
+    async def main():
+        srv = FaultyServer(EngineConfig(model="tiny-llama", device="cpu",
+                                        seed=4), faults=2)
+        cfg = Settings(db_path=":memory:", secrets_path="/tmp/chaos_sec.json")
+        cfg.worker_concurrency = 2
+        rt = Runtime(cfg, engine_server=srv)
+        oid = rt.admin.create_org("chaos")
+        rt.start()
+        iid, created = process_webhook(
+            rt, oid, "pagerduty",
+            {"event": {"data": {"id": "c1", "title": "chaos alert",
+                                "service": {"summary": "svc"}}}},
+            rca_countdown_s=0.0,
+            rca_args={"max_turns": 2,
+                      "tool_subset": ["list_incidents", "final_answer"]})
+        assert created
+        store = rt.org(oid)
+        for _ in range(600):
+            inc = store.get_incident(iid)
+            if inc["aurora_status"] in ("completed", "failed"):
+                break
+            await asyncio.sleep(0.2)
+        inc = store.get_incident(iid)
+        # the transient faults were retried through the queue, not terminal
+        assert inc["aurora_status"] == "completed", inc["aurora_status"]
+        events = [e["event"] for e in store.lifecycle_events(iid)]
+        assert "rca_engine_fault" in events  # the injected faults were seen
+        assert "rca_completed" in events
+        await rt.stop()
+        srv.shutdown()
+
+    asyncio.run(main())
