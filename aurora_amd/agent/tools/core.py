@@ -33,16 +33,23 @@ def _exec_argv(command: str) -> list[str] | None:
        "(read-only infra commands; every command is safety-gated).",
        {"command": "the shell command to run"}, read_only=False,
        needs_confirmation=True)
-async def terminal_exec(ctx: ToolContext, command: str, timeout: int = 60) -> str:
+async def terminal_exec(ctx: ToolContext, command: str, timeout: int = 60,
+                        extra_env: dict = None) -> str:
     verdict = await evaluate_command(command, llm=ctx.llm, org_store=ctx.org_store,
                                      user_id=ctx.user_id)
     if not verdict.safe:
         return (f"BLOCKED by guardrail ({verdict.source}"
                 f"{':' + verdict.rule_id if verdict.rule_id else ''}): {verdict.reason}")
-    # per-session sandbox: scratch dir, rlimits, network isolation when
-    # the host supports unprivileged namespaces, idle cleanup (the
-    # reference runs these in per-user K8s pods with a NetworkPolicy —
-    # terminal_pod_manager.py:68; sandbox.py is the host-primitive rebuild)
+    return await _run_sandboxed(ctx, command, extra_env or {}, timeout)
+
+
+async def _run_sandboxed(ctx: ToolContext, command: str, extra_env: dict,
+                         timeout: int = 60) -> str:
+    """Post-judge execution: per-session sandbox (scratch dir, rlimits,
+    unshare network isolation, idle cleanup — the reference's per-user K8s
+    pods, terminal_pod_manager.py:68; sandbox.py is the host-primitive
+    rebuild). extra_env injects per-invocation provider credentials into
+    the child only."""
     from .sandbox import sandbox_manager
     argv = _exec_argv(command)
     # argv path: the judged string IS the executed command (no shell
@@ -51,7 +58,7 @@ async def terminal_exec(ctx: ToolContext, command: str, timeout: int = 60) -> st
     res = await sandbox_manager().run(
         ctx.session_id or ctx.user_id or "anon",
         argv, shell_command=None if argv is not None else command,
-        timeout_s=timeout)
+        timeout_s=timeout, extra_env=extra_env)
     if res["timed_out"]:
         return f"ERROR: command timed out after {timeout}s"
     text = res["stdout"]
@@ -67,16 +74,98 @@ async def kubectl(ctx: ToolContext, args: str) -> str:
     return await terminal_exec(ctx, f"kubectl {args}")
 
 
+#: CLI binary -> (connector name, credential -> env-var map). Mirrors the
+#: reference's per-provider isolated env setup (reference:
+#: cloud_exec_tool.py:125-823 setup_*_environment_isolated): credentials
+#: are injected per invocation into the sandboxed child only, never into
+#: the service process environment.
+_CLOUD_ENV = {
+    "aws": ("aws", {"access_key_id": "AWS_ACCESS_KEY_ID",
+                    "secret_access_key": "AWS_SECRET_ACCESS_KEY",
+                    "session_token": "AWS_SESSION_TOKEN",
+                    "region": "AWS_DEFAULT_REGION"}),
+    "az": ("azure", {"access_token": "AZURE_ACCESS_TOKEN",
+                     "subscription": "AZURE_SUBSCRIPTION_ID"}),
+    "gcloud": ("gcp", {"access_token": "CLOUDSDK_AUTH_ACCESS_TOKEN",
+                       "project": "CLOUDSDK_CORE_PROJECT"}),
+    "scw": ("scaleway", {"access_key": "SCW_ACCESS_KEY",
+                         "secret_key": "SCW_SECRET_KEY",
+                         "project_id": "SCW_DEFAULT_PROJECT_ID"}),
+    "ovhcli": ("ovh", {"app_key": "OVH_APPLICATION_KEY",
+                       "app_secret": "OVH_APPLICATION_SECRET",
+                       "consumer_key": "OVH_CONSUMER_KEY"}),
+    "tailscale": ("tailscale", {"api_key": "TS_API_KEY"}),
+    "flyctl": ("flyio", {"token": "FLY_API_TOKEN"}),
+}
+
+
+def cloud_env_for(ctx: ToolContext, binname: str,
+                  account: str = "") -> dict:
+    """Per-invocation credential env for the sandboxed child. ``account``
+    selects an alternate credential set stored under
+    '<provider>:<account>' (the reference's AWS multi-account fan-out,
+    cloud_exec_tool.py:1199)."""
+    spec = _CLOUD_ENV.get(binname)
+    if spec is None:
+        return {}
+    provider, mapping = spec
+    from ...connectors.registry import get_credentials
+    creds = None
+    if ctx.org_store is not None and ctx.user_id:
+        name = f"{provider}:{account}" if account else provider
+        creds = get_credentials(ctx.org_store, ctx.user_id, name)
+        if creds is None and account:
+            return {}
+    if not creds:
+        return {}
+    return {env: str(creds[k]) for k, env in mapping.items() if creds.get(k)}
+
+
 @R.add("cloud_exec",
        "Run a cloud-provider CLI command (aws/az/gcloud...) with isolated "
-       "credentials (safety-gated).",
-       {"command": "full CLI command starting with the provider binary"},
+       "per-invocation credentials (safety-gated).",
+       {"command": "full CLI command starting with the provider binary",
+        "account": "optional named account/profile (multi-account orgs)"},
        read_only=False)
-async def cloud_exec(ctx: ToolContext, command: str) -> str:
+async def cloud_exec(ctx: ToolContext, command: str, account: str = "") -> str:
     binname = shlex.split(command)[0] if command.strip() else ""
-    if binname not in ("aws", "az", "gcloud", "scw", "ovhcli", "tailscale", "flyctl"):
-        return "ERROR: cloud_exec only accepts aws/az/gcloud/scw/ovhcli/tailscale/flyctl"
-    return await terminal_exec(ctx, command)
+    if binname not in _CLOUD_ENV:
+        return ("ERROR: cloud_exec only accepts "
+                + "/".join(sorted(_CLOUD_ENV)))
+    env = cloud_env_for(ctx, binname, account)
+    if account and not env:
+        return f"ERROR: no credentials stored for {binname} account {account!r}"
+    return await terminal_exec(ctx, command, extra_env=env)
+
+
+@R.add("aws_exec_all_accounts",
+       "Run a read-only aws CLI command across every connected AWS "
+       "account and aggregate the outputs (safety-gated once).",
+       {"command": "aws CLI command"}, read_only=False)
+async def aws_exec_all_accounts(ctx: ToolContext, command: str) -> str:
+    """Reference: AWS multi-account fan-out (cloud_exec_tool.py:1199)."""
+    binname = shlex.split(command)[0] if command.strip() else ""
+    if binname != "aws":
+        return "ERROR: this tool only accepts aws commands"
+    if ctx.org_store is None or not ctx.user_id:
+        return "ERROR: no credential store bound"
+    accounts = [p.split(":", 1)[1] for p
+                in ctx.org_store.connected_providers(ctx.user_id)
+                if p.startswith("aws:")]
+    targets = [""] + accounts  # default account + named ones
+    verdict = await evaluate_command(command, llm=ctx.llm,
+                                     org_store=ctx.org_store,
+                                     user_id=ctx.user_id)
+    if not verdict.safe:
+        return f"BLOCKED by guardrail ({verdict.source}): {verdict.reason}"
+    outs = []
+    for acct in targets:
+        env = cloud_env_for(ctx, "aws", acct)
+        if acct and not env:
+            continue
+        out = await _run_sandboxed(ctx, command, env)
+        outs.append(f"## account: {acct or 'default'}\n{out[:4000]}")
+    return "\n\n".join(outs) if outs else "ERROR: no AWS accounts connected"
 
 
 @R.add("on_prem_kubectl",

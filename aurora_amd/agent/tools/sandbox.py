@@ -103,7 +103,8 @@ class SandboxManager:
 
     async def run(self, session_id: str, argv: Optional[List[str]],
                   shell_command: Optional[str] = None,
-                  timeout_s: float = 60.0) -> dict:
+                  timeout_s: float = 60.0,
+                  extra_env: Optional[Dict[str, str]] = None) -> dict:
         """Run argv (exec, preferred) or a shell command inside the
         session's sandbox. Returns {stdout, stderr, returncode,
         timed_out, isolated}."""
@@ -112,6 +113,8 @@ class SandboxManager:
         box.commands_run += 1
         env = {"PATH": "/usr/bin:/bin:/usr/sbin:/sbin", "HOME": box.root,
                "TMPDIR": box.root, "LANG": "C.UTF-8"}
+        if extra_env:
+            env.update({str(k): str(v) for k, v in extra_env.items()})
         if argv is not None:
             cmd = list(argv)
         else:

@@ -81,7 +81,10 @@ def get_connector(name: str) -> Optional[ConnectorSpec]:
 def connect_provider(rt, store, user_id: str, provider: str, body: dict) -> dict:
     """Store credentials via the secrets backend; DB keeps only the ref
     (reference: store_tokens_in_db / vault ref format)."""
-    spec = _CATALOG.get(provider)
+    # "<provider>:<account>" stores an alternate credential set for the
+    # same connector (multi-account orgs — reference: AWS fan-out)
+    base = provider.split(":", 1)[0]
+    spec = _CATALOG.get(base)
     if spec is None:
         return {"error": f"unknown provider '{provider}'"}
     creds = body.get("credentials") or {}
