@@ -582,6 +582,39 @@ async def jenkins_rca_tool(ctx: ToolContext, job: str) -> str:
         return f"ERROR: jenkins: {e}"
 
 
+@R.add("spinnaker_rca", "Recent Spinnaker pipeline executions of an "
+       "application with failed stages.",
+       {"application": "Spinnaker application name"},
+       requires_connection="spinnaker")
+async def spinnaker_rca_tool(ctx: ToolContext, application: str) -> str:
+    from ...connectors.clients import NoEgressError, ProviderAPIError
+    from ...connectors.clients.devops import spinnaker_rca
+    try:
+        return await spinnaker_rca(_transport(ctx), _creds(ctx, "spinnaker"),
+                                   application)
+    except NoEgressError as e:
+        return f"ERROR: spinnaker API unreachable (no egress): {e}"
+    except ProviderAPIError as e:
+        return f"ERROR: spinnaker: {e}"
+
+
+@R.add("cloudbees_rca", "Analyze a CloudBees (Jenkins-based) job's recent "
+       "builds and extract failure lines.",
+       {"job": "job path"}, requires_connection="cloudbees")
+async def cloudbees_rca_tool(ctx: ToolContext, job: str) -> str:
+    # CloudBees CI speaks the Jenkins remote API (reference: cloudbees_rca
+    # reuses the Jenkins analysis against the CloudBees endpoint)
+    from ...connectors.clients import NoEgressError, ProviderAPIError
+    from ...connectors.clients.devops import jenkins_rca
+    try:
+        return await jenkins_rca(_transport(ctx), _creds(ctx, "cloudbees"),
+                                 job)
+    except NoEgressError as e:
+        return f"ERROR: cloudbees API unreachable (no egress): {e}"
+    except ProviderAPIError as e:
+        return f"ERROR: cloudbees: {e}"
+
+
 @R.add("github_fix", "Draft a fix suggestion (branch + patch description) "
        "for the incident from a suspect commit.",
        {"repo": "owner/name", "commit_sha": "suspect commit",

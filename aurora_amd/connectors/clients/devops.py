@@ -137,6 +137,46 @@ async def github_rca(transport: Transport, creds: dict, repo: str,
 
 # ----------------------------------------------------------------- Jenkins
 
+# ---------------------------------------------------------------- spinnaker
+
+
+def build_spinnaker_executions(creds: dict, application: str) -> Request:
+    """Recent pipeline executions via Gate (reference: spinnaker_rca)."""
+    if not creds.get("base_url"):
+        raise ProviderAPIError("spinnaker credentials need base_url (gate)")
+    h = {}
+    if creds.get("token"):
+        h["Authorization"] = f"Bearer {creds['token']}"
+    return Request("GET", creds["base_url"].rstrip("/")
+                   + f"/applications/{application}/pipelines",
+                   headers=h, params={"limit": 15, "expand": "false"})
+
+
+@fail_closed
+def parse_spinnaker_executions(doc) -> str:
+    if isinstance(doc, dict):
+        doc = doc.get("executions", [])
+    rows = []
+    for e in doc[:15]:
+        stages = e.get("stages", [])
+        failed = [s.get("name", "?") for s in stages
+                  if s.get("status") in ("TERMINAL", "FAILED_CONTINUE")]
+        rows.append(f"{e.get('name', '?')} [{e.get('status', '?')}] "
+                    f"build {e.get('buildTime', e.get('startTime', ''))}"
+                    + (f" failed stages: {', '.join(failed)}" if failed
+                       else ""))
+    bad = sum(1 for e in doc if e.get("status") not in
+              ("SUCCEEDED", "RUNNING", "NOT_STARTED"))
+    return (f"spinnaker: {len(doc)} recent executions, {bad} failed"
+            + ("\n" + "\n".join(rows) if rows else ""))
+
+
+async def spinnaker_rca(transport: Transport, creds: dict,
+                        application: str, **_kw) -> str:
+    return parse_spinnaker_executions(await transport(
+        build_spinnaker_executions(creds, application)))
+
+
 _FAIL_LINE = re.compile(
     r"(error|exception|failed|fatal|traceback|build step.*failed)",
     re.IGNORECASE)

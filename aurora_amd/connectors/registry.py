@@ -58,7 +58,8 @@ for name, cat, auth, alert, tools in [
     ("github", "code", "oauth", False, ["github_rca", "github_fix"]),
     ("bitbucket", "code", "oauth", False, []),
     ("jenkins", "code", "token", True, ["jenkins_rca"]),
-    ("spinnaker", "code", "webhook_only", True, []),
+    ("spinnaker", "code", "token", True, ["spinnaker_rca"]),
+    ("cloudbees", "code", "token", True, ["cloudbees_rca"]),
     ("slack", "comms", "oauth", False, ["query_slack"]),
     ("jira", "docs", "oauth", True, ["query_jira"]),
     ("notion", "docs", "oauth", False, ["notion_search"]),

@@ -586,3 +586,20 @@ def test_sharepoint_search_graph():
     out = run(comms.sharepoint_query(transport_of(handler), {"token": "t"},
                                      "runbook db"))
     assert "db-runbook.docx: failover steps" in out
+
+
+def test_spinnaker_executions():
+    from aurora_amd.connectors.clients import devops
+
+    def handler(req):
+        assert "/applications/checkout/pipelines" in req.url
+        return [{"name": "deploy-prod", "status": "TERMINAL",
+                 "startTime": 1700000000000,
+                 "stages": [{"name": "bake", "status": "SUCCEEDED"},
+                            {"name": "deploy", "status": "TERMINAL"}]},
+                {"name": "deploy-prod", "status": "SUCCEEDED",
+                 "startTime": 1699990000000, "stages": []}]
+    out = run(devops.spinnaker_rca(transport_of(handler),
+                                   {"base_url": "https://gate"}, "checkout"))
+    assert "2 recent executions, 1 failed" in out
+    assert "failed stages: deploy" in out
