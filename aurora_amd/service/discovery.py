@@ -138,6 +138,118 @@ def infer_connection_dependencies(service_ports: Dict[str, int],
     return sorted(edges)
 
 
+# ----------------------------------------------------- provider assets
+# (reference: services/discovery/providers/ — asset discovery per cloud;
+# parsers are pure over the CLI/API JSON so they fixture-test without
+# credentials; collection rides cloud_exec's per-invocation creds)
+
+
+def parse_aws_instances(describe_json: str) -> List[dict]:
+    """`aws ec2 describe-instances` -> asset nodes."""
+    try:
+        doc = json.loads(describe_json)
+    except json.JSONDecodeError:
+        return []
+    out = []
+    for res in doc.get("Reservations", []):
+        for inst in res.get("Instances", []):
+            name = next((t.get("Value") for t in inst.get("Tags", [])
+                         if t.get("Key") == "Name"), None)
+            out.append({
+                "name": name or inst.get("InstanceId", "?"),
+                "kind": "aws_ec2",
+                "id": inst.get("InstanceId", ""),
+                "state": (inst.get("State") or {}).get("Name", ""),
+                "type": inst.get("InstanceType", ""),
+                "private_ip": inst.get("PrivateIpAddress", ""),
+            })
+    return out
+
+
+def parse_aws_rds(describe_json: str) -> List[dict]:
+    """`aws rds describe-db-instances` -> asset nodes."""
+    try:
+        doc = json.loads(describe_json)
+    except json.JSONDecodeError:
+        return []
+    return [{
+        "name": db.get("DBInstanceIdentifier", "?"), "kind": "aws_rds",
+        "id": db.get("DbiResourceId", ""),
+        "state": db.get("DBInstanceStatus", ""),
+        "engine": db.get("Engine", ""),
+        "endpoint": (db.get("Endpoint") or {}).get("Address", ""),
+    } for db in doc.get("DBInstances", [])]
+
+
+def parse_aws_lambda(list_json: str) -> List[dict]:
+    try:
+        doc = json.loads(list_json)
+    except json.JSONDecodeError:
+        return []
+    return [{"name": f.get("FunctionName", "?"), "kind": "aws_lambda",
+             "id": f.get("FunctionArn", ""), "runtime": f.get("Runtime", "")}
+            for f in doc.get("Functions", [])]
+
+
+def parse_gcp_instances(list_json: str) -> List[dict]:
+    """`gcloud compute instances list --format json`."""
+    try:
+        doc = json.loads(list_json)
+    except json.JSONDecodeError:
+        return []
+    if not isinstance(doc, list):
+        return []
+    return [{"name": i.get("name", "?"), "kind": "gcp_instance",
+             "id": str(i.get("id", "")), "state": i.get("status", ""),
+             "zone": (i.get("zone", "").rsplit("/", 1) or [""])[-1]}
+            for i in doc if isinstance(i, dict)]
+
+
+def write_assets(graph, assets: List[dict],
+                 endpoints_to_services: bool = True) -> int:
+    """Asset nodes into the graph; RDS/LB endpoints become dependency
+    targets discoverable by the env/dns inference (resource_mapper.py
+    analog)."""
+    n = 0
+    for a in assets:
+        props = {k: v for k, v in a.items() if k not in ("name", "kind") and v}
+        graph.upsert_service(a["name"], kind=a["kind"], **props)
+        n += 1
+        ep = a.get("endpoint", "")
+        if endpoints_to_services and ep:
+            base = ep.split(".")[0]
+            if base and base != a["name"]:
+                graph.upsert_service(base, kind="endpoint")
+                graph.add_dependency(base, a["name"], via="endpoint")
+    return n
+
+
+async def discover_cloud_assets(ctx, graph) -> int:
+    """Collect via the guarded cloud_exec path when provider credentials
+    are connected; every CLI absence degrades to zero assets."""
+    from ..agent.tools.registry import GLOBAL_REGISTRY
+    total = 0
+    cmds = [
+        ("aws", "aws ec2 describe-instances --output json",
+         parse_aws_instances),
+        ("aws", "aws rds describe-db-instances --output json", parse_aws_rds),
+        ("aws", "aws lambda list-functions --output json", parse_aws_lambda),
+        ("gcp", "gcloud compute instances list --format json",
+         parse_gcp_instances),
+    ]
+    connected = []
+    if ctx.org_store is not None and ctx.user_id:
+        connected = ctx.org_store.connected_providers(ctx.user_id)
+    for provider, cmd, parser in cmds:
+        if provider not in connected:
+            continue
+        out = await GLOBAL_REGISTRY.run(ctx, "cloud_exec", {"command": cmd})
+        if out.startswith(("ERROR", "BLOCKED", "CANCELLED")):
+            continue
+        total += write_assets(graph, parser(out))
+    return total
+
+
 # ------------------------------------------------------------- writer
 
 def write_edges(graph, edges: List[Edge]) -> int:

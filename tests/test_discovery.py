@@ -133,3 +133,53 @@ def test_self_diagnostics_only_runs_readonly():
         assert "load" in ran[0]["rationale"]
 
     asyncio.run(main())
+
+
+def test_provider_asset_parsers():
+    """Cloud asset discovery parsers over recorded CLI JSON (reference:
+    services/discovery/providers/)."""
+    from aurora_amd.service import discovery as d
+    ec2 = json.dumps({"Reservations": [{"Instances": [
+        {"InstanceId": "i-1", "InstanceType": "m5.large",
+         "State": {"Name": "running"}, "PrivateIpAddress": "10.0.0.5",
+         "Tags": [{"Key": "Name", "Value": "api-1"}]},
+        {"InstanceId": "i-2", "State": {"Name": "stopped"}, "Tags": []},
+    ]}]})
+    assets = d.parse_aws_instances(ec2)
+    assert assets[0]["name"] == "api-1" and assets[0]["kind"] == "aws_ec2"
+    assert assets[1]["name"] == "i-2"
+    rds = json.dumps({"DBInstances": [
+        {"DBInstanceIdentifier": "db-primary", "Engine": "postgres",
+         "DBInstanceStatus": "available",
+         "Endpoint": {"Address": "db-primary.abc.rds.amazonaws.com"}}]})
+    r = d.parse_aws_rds(rds)
+    assert r[0]["endpoint"].startswith("db-primary.")
+    lam = json.dumps({"Functions": [{"FunctionName": "resize",
+                                     "Runtime": "python3.12"}]})
+    assert d.parse_aws_lambda(lam)[0]["kind"] == "aws_lambda"
+    gcp = json.dumps([{"name": "web-1", "id": 123, "status": "RUNNING",
+                       "zone": "https://.../zones/europe-west1-b"}])
+    g = d.parse_gcp_instances(gcp)
+    assert g[0]["zone"] == "europe-west1-b"
+    # malformed inputs degrade to empty
+    for fn in (d.parse_aws_instances, d.parse_aws_rds, d.parse_aws_lambda,
+               d.parse_gcp_instances):
+        assert fn("not json") == []
+        assert fn("{}") == [] or fn("[]") == []
+
+
+def test_write_assets_endpoints_become_dependencies():
+    from aurora_amd.graph.infra_graph import InfraGraph
+    from aurora_amd.service.discovery import write_assets
+    from aurora_amd.service.store import AdminStore, Database
+    db = Database(":memory:")
+    store = AdminStore(db).org_store(AdminStore(db).create_org("a"))
+    g = InfraGraph(store)
+    n = write_assets(g, [
+        {"name": "db-primary", "kind": "aws_rds",
+         "endpoint": "db-primary.abc.rds.amazonaws.com",
+         "engine": "postgres"},
+        {"name": "api-1", "kind": "aws_ec2", "state": "running"}])
+    assert n == 2
+    names = {s["name"] for s in g.list_services()}
+    assert {"db-primary", "api-1"} <= names
