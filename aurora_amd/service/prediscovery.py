@@ -71,14 +71,24 @@ async def run_prediscovery(rt, args: Dict[str, Any]) -> dict:
         found += await discover_kubernetes(graph)
         # dependency inference over the discovered assets (reference:
         # services/discovery/inference/ modules)
-        from .discovery import run_inference
+        from .discovery import discover_cloud_assets, run_inference
         inferred = await run_inference(graph)
+        # provider asset discovery through the first org member with
+        # connected cloud credentials (reference: providers/ discovery)
+        assets = 0
+        for u in rt.db.query("SELECT id FROM users WHERE org_id=?", [org_id]):
+            ctx = rt.tool_ctx(org_id, user_id=u["id"])
+            if any(p.split(":")[0] in ("aws", "gcp", "azure")
+                   for p in store.connected_providers(u["id"])):
+                assets = await discover_cloud_assets(ctx, graph)
+                break
         ctx = {
             "discovered_at": time.time(),
             "services": [s["name"] for s in graph.list_services()][:100],
             "spof": graph.single_points_of_failure(),
             "stale": graph.stale_services()[:20],
             "inferred_edges": inferred,
+            "provider_assets": assets,
         }
         store.save_artifact("prediscovery_context", json.dumps(ctx, indent=1))
         total += len(found)
