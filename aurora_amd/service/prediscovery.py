@@ -77,10 +77,10 @@ async def run_prediscovery(rt, args: Dict[str, Any]) -> dict:
         # connected cloud credentials (reference: providers/ discovery)
         assets = 0
         for u in rt.db.query("SELECT id FROM users WHERE org_id=?", [org_id]):
-            ctx = rt.tool_ctx(org_id, user_id=u["id"])
             if any(p.split(":")[0] in ("aws", "gcp", "azure")
                    for p in store.connected_providers(u["id"])):
-                assets = await discover_cloud_assets(ctx, graph)
+                tctx = rt.tool_ctx(org_id, user_id=u["id"])
+                assets = await discover_cloud_assets(tctx, graph)
                 break
         ctx = {
             "discovered_at": time.time(),
