@@ -30,7 +30,8 @@ input{background:#0d1117;color:#c9d1d9;border:1px solid #30363d;border-radius:4p
 </style></head><body>
 <header><h1>aurora-mi355x</h1>
 <span class="kpi">user <input id="u" size="8" value="admin"> org <input id="o" size="8" value="default"></span>
-<span class="kpi" id="kpis">loading…</span></header>
+<span class="kpi" id="kpis">loading…</span>
+<span class="kpi" id="eng"></span></header>
 <main><section id="list"><table id="tbl"><thead><tr>
 <th>incident</th><th>sev</th><th>status</th><th>source</th><th>title</th></tr></thead>
 <tbody></tbody></table></section>
@@ -59,11 +60,18 @@ await get('/incidents/'+id+'/thoughts','Thoughts');
 await get('/incidents/'+id+'/citations','Citations');
 await get('/incidents/'+id+'/suggestions','Suggestions');
 await get('/incidents/'+id+'/timeline','Timeline');
+await get('/incidents/'+id+'/hypotheses','Hypothesis ledger');
 await get('/monitor/waterfall/'+id,'Tool waterfall');
+try{const pm=await api('/postmortems/'+id);
+if(pm&&pm.content)out.push(`<h3>Postmortem</h3><pre>${esc(pm.content.slice(0,6000))}</pre>`);}catch(e){}
 d.innerHTML=out.join('')||'<p class="kpi">no data</p>';}
+async function engine(){try{const st=await api('/engine/status');
+$('#eng').innerHTML=`engine <b>${st.healthy?'healthy':'STUCK'}</b> · ${esc(st.model)} on ${esc(st.device)}`+
+` · kv ${st.kv_pages_total-st.kv_pages_free}/${st.kv_pages_total} pages · ${st.running} running`+
+` · ${(st.counters.generated_tokens||0).toLocaleString()} tokens`;}catch(e){$('#eng').textContent='';}}
 async function sse(){try{const r=await fetch('/incidents/stream',{headers:H()});
 const rd=r.body.getReader();const dec=new TextDecoder();
 for(;;){const {done,value}=await rd.read();if(done)break;
 if(dec.decode(value).includes('data:')){list();kpis();}}}catch(e){setTimeout(sse,3000);}}
-list();kpis();sse();setInterval(()=>{list();kpis();},10000);
+list();kpis();engine();sse();setInterval(()=>{list();kpis();engine();},10000);
 </script></body></html>"""
