@@ -551,3 +551,29 @@ def test_engine_fp8_kv_end_to_end(monkeypatch):
     prompts = [torch.randint(1, 900, (700,)).tolist() for _ in range(3)]
     outs = eng.generate(prompts, SamplingParams(max_tokens=6, ignore_eos=True))
     assert all(len(o) == 6 for o in outs)
+
+
+def test_engine_gpu_fused_react_turn():
+    """The fused thought+action grammar on the GPU engine: every output
+    parses into thought + a registered tool call (the e2e bench's hot
+    request shape)."""
+    import json as _json
+    from aurora_amd.engine.engine import EngineConfig, LLMEngine
+    from aurora_amd.engine.grammar import ReActTurnGrammar
+    from aurora_amd.engine.sampling_params import SamplingParams
+
+    eng = LLMEngine(EngineConfig(model="smoke-llama", device="cuda", seed=8,
+                                 kv_cache_pages=64))
+    tools = ["kubectl", "final_answer", "knowledge_base_search"]
+    outs = eng.generate(
+        [[1, 2, 3], [4, 5, 6], [7, 8, 9]],
+        SamplingParams(temperature=0.6, max_tokens=260,
+                       grammar=ReActTurnGrammar(tools, thought_bytes=120,
+                                                tool_soft_close=80)))
+    for toks in outs:
+        text = eng.tokenizer.decode(toks)
+        head, sep, tail = text.rpartition("\nAction:")
+        assert sep, text[-200:]
+        doc = _json.loads(tail.strip())
+        assert doc["tool"] in tools and isinstance(doc["args"], dict)
+    assert eng.counters["grammar_fallbacks"] == 0  # byte-bucket pick covers
