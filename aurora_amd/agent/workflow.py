@@ -124,6 +124,7 @@ class Workflow:
             for attempt in range(3):  # retries guard pathological sampling
                 parts: List[str] = []
                 fault: Optional[str] = None
+                streamed = 0  # chars already sent as UI tokens
                 async for out in self.llm.server.stream(
                         self.llm.tokenizer.encode(prompt, add_bos=True),
                         SamplingParams(
@@ -134,7 +135,17 @@ class Workflow:
                         session_id=state.session_id):
                     if out.new_text:
                         parts.append(out.new_text)
-                        yield {"type": "token", "content": out.new_text}
+                        # stream THOUGHT text only: once the turn crosses
+                        # into '\nAction:', the JSON bytes are machine
+                        # payload, not UI tokens (the action is surfaced
+                        # as a tool_call event below)
+                        text_so_far = "".join(parts)
+                        cut = text_so_far.find("\nAction:")
+                        visible = text_so_far if cut < 0 else text_so_far[:cut]
+                        if len(visible) > streamed:
+                            yield {"type": "token",
+                                   "content": visible[streamed:]}
+                            streamed = len(visible)
                     if out.finished and out.finish_reason and (
                             out.finish_reason.startswith("engine step failed")
                             or out.finish_reason == "engine shutdown"):
