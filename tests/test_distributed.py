@@ -193,3 +193,45 @@ def test_bench_dp2_cpu():
     assert doc["config"]["global_batch"] == 4  # 2 ranks x 2 incidents
     assert doc["config"]["parallelism"] == "dp2"
     assert doc["value"] > 0
+
+
+@pytest.mark.timeout(300)
+def test_bench_e2e_deadline_straggler_accounting():
+    """run_step_e2e's deadline marks stragglers failed (they stay in the
+    denominator) instead of hanging the bench forever."""
+    import asyncio
+    import importlib.util
+    spec = importlib.util.spec_from_file_location(
+        "bench_mod", os.path.join(ROOT, "bench.py"))
+    bench = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(bench)
+
+    import torch as _t
+    from aurora_amd.engine.engine import EngineConfig
+    from aurora_amd.engine.server import AsyncEngineServer
+    from aurora_amd.service.runtime import Runtime
+    from aurora_amd.utils.config import Settings
+
+    class Args:
+        concurrency = 2
+        turns = 2
+        rag_docs = 5
+        model = "tiny-llama"
+
+    async def main():
+        cfg = Settings(db_path=":memory:",
+                       secrets_path="/tmp/bench_dl_sec.json")
+        cfg.worker_concurrency = 0  # workers never start -> nothing completes
+        srv = AsyncEngineServer(EngineConfig(model="tiny-llama",
+                                             device="cpu", seed=5))
+        rt = Runtime(cfg, engine_server=srv)
+        org = rt.admin.create_org("dl")
+        rng = _t.Generator().manual_seed(1)
+        lat, statuses = await bench.run_step_e2e(rt, org, Args(), rng, 0,
+                                                 deadline_s=3.0)
+        assert statuses.get("failed", 0) == 2  # both stragglers accounted
+        assert lat == []
+        await rt.stop()
+        srv.shutdown()
+
+    asyncio.run(main())
