@@ -736,6 +736,17 @@ def create_app(rt: Runtime) -> FastAPI:
             "# TYPE aurora_engine_healthy gauge",
             f"aurora_engine_healthy {int(rt.engine.healthy()) if hasattr(rt.engine, 'healthy') else 1}",
         ]
+        e = getattr(rt.engine, "engine", None)
+        if e is not None:
+            lines += [
+                "# TYPE aurora_kv_pages gauge",
+                f"aurora_kv_pages{{state=\"free\"}} {e.kv.num_free()}",
+                f"aurora_kv_pages{{state=\"total\"}} {e.kv.num_pages}",
+                "# TYPE aurora_sequences_running gauge",
+                f"aurora_sequences_running {e.num_running()}",
+                "# TYPE aurora_sessions_cached gauge",
+                f"aurora_sessions_cached {len(e.sessions)}",
+            ]
         for org in rt.admin.orgs():
             st = rt.org(org["id"])
             by = {}
