@@ -62,3 +62,42 @@ def test_random_interleavings_preserve_invariants(seed):
     held = sum(len(s.pages) for s in eng.sessions.values())
     assert eng.kv.num_free() + held == total_pages, (
         eng.kv.num_free(), held, total_pages)
+
+
+def test_staging_key_uniquely_identifies_page_assignment():
+    """The hipGraph block-table staging key (serial, page_gen, len(pages))
+    must change whenever a sequence's page list changes identity — under
+    preemption/eviction churn on a tiny KV (stale rows
+    attended over freed pages)."""
+    import random
+    from aurora_amd.engine.engine import EngineConfig, LLMEngine
+    from aurora_amd.engine.sampling_params import SamplingParams
+
+    rnd = random.Random(7)
+    eng = LLMEngine(EngineConfig(model="tiny-llama", device="cpu", seed=3,
+                                 kv_cache_pages=12, max_num_seqs=6))
+    key_to_pages = {}
+    live = 0
+    for step in range(400):
+        if live < 5 and rnd.random() < 0.5:
+            rid = f"pk-{step}"
+            eng.add_request(rid, [5] * rnd.randint(3, 40),
+                            SamplingParams(max_tokens=rnd.randint(1, 30),
+                                           ignore_eos=True),
+                            session_id=f"s-{step % 3}")
+            live += 1
+        try:
+            outs = eng.step()
+        except MemoryError:
+            break
+        live -= sum(1 for o in outs if o.finished)
+        for seq in list(eng.running) + list(eng.waiting):
+            key = (seq.serial, seq.page_gen, len(seq.pages))
+            pages = tuple(seq.pages)
+            if key in key_to_pages:
+                assert key_to_pages[key] == pages, (
+                    f"staging key {key} reused for different pages "
+                    f"{key_to_pages[key]} -> {pages}")
+            else:
+                key_to_pages[key] = pages
+    assert eng.counters["preemptions"] > 0  # churn actually happened
