@@ -225,12 +225,16 @@ async def generate_recommendations(llm, store, incident_id: str,
     """Hypothesis ledger -> optional safe self-run diagnostics -> LLM
     suggestions -> parse -> safety + inventory filter -> persist
     (reference: recommender.py:243-1064; suggestion_extractor.py:53-77)."""
-    ledger = await extract_hypotheses(llm, messages)
+    # hypothesis extraction and diagnostics planning both depend only on
+    # the transcript — overlap their LLM passes to keep the decode batch full
+    import asyncio as _aio
+    import os
+    ledger_task = _aio.create_task(extract_hypotheses(llm, messages))
     diags: List[dict] = []
-    if ctx is not None:
-        import os
-        if os.environ.get("AURORA_SELF_DIAGNOSTICS", "1") != "0":
-            diags = await self_execute_diagnostics(llm, ctx, messages)
+    if (ctx is not None
+            and os.environ.get("AURORA_SELF_DIAGNOSTICS", "1") != "0"):
+        diags = await self_execute_diagnostics(llm, ctx, messages)
+    ledger = await ledger_task
     diag_text = ""
     if diags:
         diag_text = "ADDITIONAL DIAGNOSTICS RUN POST-INVESTIGATION:\n" + \

@@ -137,14 +137,22 @@ async def run_background_chat(rt, args: Dict[str, Any]) -> dict:
                                    "aurora_status": "failed"})
         return {"error": error}
 
-    try:
-        summary = await generate_incident_summary(llm, state.messages, session_id)
-    except Exception:
-        summary = alert_only_summary(inc["title"], payload)
-    severity = determine_severity(summary, inc.get("severity"))
+    # summary and recommendations are independent LLM passes: run them
+    # concurrently so the post-RCA tail keeps the engine's decode batch
+    # occupied instead of serializing 4 calls per incident
+    import asyncio as _aio
+    summary_task = _aio.create_task(
+        generate_incident_summary(llm, state.messages, session_id))
+    summary_task.add_done_callback(
+        lambda t: None if t.cancelled() else t.exception())
     n_cits = save_citations(store, incident_id, state.messages)
     suggestions = await generate_recommendations(llm, store, incident_id,
                                                  state.messages, ctx=ctx)
+    try:
+        summary = await summary_task
+    except Exception:
+        summary = alert_only_summary(inc["title"], payload)
+    severity = determine_severity(summary, inc.get("severity"))
     store.update_incident(incident_id, aurora_status="completed",
                           summary=summary, severity=severity,
                           root_cause=_extract_root_cause(summary))
