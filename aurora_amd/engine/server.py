@@ -121,8 +121,14 @@ class AsyncEngineServer:
         q: asyncio.Queue = asyncio.Queue()
         with self._lock:
             self._subs[rid] = (loop, q)
-            self.engine.add_request(rid, prompt_token_ids, params,
-                                    session_id=session_id)
+            try:
+                self.engine.add_request(rid, prompt_token_ids, params,
+                                        session_id=session_id)
+            except BaseException:  # rejection at the API boundary (empty /
+                # out-of-vocab prompt): drop the subscription we just made
+                # or it leaks until shutdown
+                self._subs.pop(rid, None)
+                raise
         self._wake.set()
         finished = False
         try:

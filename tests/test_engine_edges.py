@@ -106,3 +106,34 @@ def test_max_tokens_clamped_to_model_len_at_admission():
     e.add_request("clamp-1", list(range(5, 35)),  # 30-token prompt
                   SamplingParams(max_tokens=10_000))
     assert e.seqs["clamp-1"].params.max_tokens <= 34
+
+
+def test_rejected_request_leaves_no_subscription():
+    """A request that add_request rejects at the API boundary (empty or
+    out-of-vocab prompt) must not leak its server subscription — the
+    stepping thread would otherwise hold a dead queue until shutdown."""
+    import asyncio
+
+    import pytest as _pytest
+
+    from aurora_amd.engine.engine import EngineConfig
+    from aurora_amd.engine.sampling_params import SamplingParams
+    from aurora_amd.engine.server import AsyncEngineServer
+
+    srv = AsyncEngineServer(EngineConfig(model="tiny-llama", device="cpu",
+                                         seed=0))
+
+    async def go():
+        with _pytest.raises(ValueError):
+            await srv.generate([], SamplingParams(max_tokens=2))
+        with _pytest.raises(ValueError):
+            await srv.generate([10 ** 9], SamplingParams(max_tokens=2))
+        assert srv._subs == {}
+        r = await srv.generate([1, 2, 3],
+                               SamplingParams(max_tokens=2, ignore_eos=True))
+        assert len(r.token_ids) == 2 and srv._subs == {}
+
+    try:
+        asyncio.run(go())
+    finally:
+        srv.shutdown()
