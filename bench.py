@@ -44,7 +44,7 @@ import torch
 def parse_args():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--steps", type=int, default=2)
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--mode", choices=("e2e", "engine"),
                     default=os.environ.get("BENCH_MODE", "e2e"))
@@ -317,11 +317,11 @@ async def run_step_e2e(rt, org_id, args, rng, step: int,
     while pending:
         left = deadline_s - (time.monotonic() - t0)
         if left <= 0:
-            # stragglers are failures, not silently-dropped work: they
-            # count in the denominator via `failed`
+            # stragglers the deadline cut off: reported, and EXCLUDED from
+            # the throughput numerator (work the timed region didn't finish)
             print(f"[bench] step {step}: deadline, {len(pending)} pending",
                   file=sys.stderr, flush=True)
-            statuses["failed"] = statuses.get("failed", 0) + len(pending)
+            statuses["straggler"] = statuses.get("straggler", 0) + len(pending)
             break
         try:
             ev = await asyncio.wait_for(ev_q.get(), timeout=min(left, 30.0))
@@ -386,7 +386,8 @@ async def main_e2e_async(args, rank, world, use_cuda):
         "guardrail_per_turn": True,
         "tool_subset": BENCH_TOOLS,
         "completed": statuses_all.get("completed", 0),
-        "failed": statuses_all.get("failed", 0),
+        "failed_terminal": statuses_all.get("failed", 0),
+        "stragglers": statuses_all.get("straggler", 0),
         "engine_timers_s": {k: round(v, 2) for k, v in eng.timers.items()},
         "engine_counters": eng.counters,
     }
@@ -411,15 +412,26 @@ def main():
         elapsed, lat, extra = asyncio.run(
             main_e2e_async(args, rank, world, use_cuda))
 
-    # max over ranks (slowest rank defines the job)
+    # incidents that reached a terminal status inside the deadline; in
+    # engine mode every submitted incident completes by construction.
+    # Stragglers cut off by the e2e deadline must NOT count in the
+    # numerator — that would credit work the timed region didn't finish.
+    done_local = float(extra.get("completed", 0) + extra.get("failed_terminal", 0)
+                       if args.mode == "e2e"
+                       else args.steps * args.concurrency)
+
+    # max over ranks (slowest rank defines the job); sum of terminal counts
     if world > 1:
         import torch.distributed as dist
         dev = "cuda" if (use_cuda and dist.get_backend() == "nccl") else "cpu"
         t = torch.tensor([elapsed], device=dev)  # RCCL needs device tensors
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         elapsed = float(t[0])
-
-    incidents = args.steps * args.concurrency * world
+        d = torch.tensor([done_local], device=dev)
+        dist.all_reduce(d, op=dist.ReduceOp.SUM)
+        incidents = int(d[0])
+    else:
+        incidents = int(done_local)
     value = incidents / (elapsed / 3600.0)
     p50 = statistics.median(lat) if lat else 0.0
     tokens_per_incident = (args.first_prompt + (args.turns - 1) * args.turn_prompt
