@@ -28,6 +28,8 @@ class ModelSpec:
     # MoE (Mixtral-style); n_experts == 0 => dense MLP
     n_experts: int = 0
     experts_per_token: int = 2
+    # Qwen2-family attention projections carry a bias (q/k/v only)
+    qkv_bias: bool = False
     # encoder-only (BGE-style) models use bidirectional attention + pooling
     is_encoder: bool = False
 
@@ -81,6 +83,16 @@ register(ModelSpec(  # BASELINE config 4 (MoE tool-calling agent)
     head_dim=128, intermediate_size=14336, vocab_size=32000, rope_theta=1e6,
     n_experts=8, experts_per_token=2,
 ))
+register(ModelSpec(  # Qwen2/2.5-7B: llama arch + QKV bias, 1e6 rope, 152k vocab
+    name="qwen2-7b", hidden_size=3584, n_layers=28, n_heads=28, n_kv_heads=4,
+    head_dim=128, intermediate_size=18944, vocab_size=152064, rope_theta=1e6,
+    max_position=32768, rms_eps=1e-6, qkv_bias=True,
+))
+register(ModelSpec(
+    name="qwen2-72b", hidden_size=8192, n_layers=80, n_heads=64, n_kv_heads=8,
+    head_dim=128, intermediate_size=29568, vocab_size=152064, rope_theta=1e6,
+    max_position=32768, rms_eps=1e-6, qkv_bias=True,
+))
 register(ModelSpec(  # BGE-base-style embedding encoder (RAG path)
     name="bge-base", hidden_size=768, n_layers=12, n_heads=12, n_kv_heads=12,
     head_dim=64, intermediate_size=3072, vocab_size=30522, is_encoder=True,
@@ -91,6 +103,11 @@ register(ModelSpec(
     name="tiny-llama", hidden_size=64, n_layers=2, n_heads=4, n_kv_heads=2,
     head_dim=16, intermediate_size=128, vocab_size=512, max_position=2048,
     tie_embeddings=True,
+))
+register(ModelSpec(  # Qwen2-shaped tiny config (qkv_bias path on CPU)
+    name="tiny-qwen", hidden_size=64, n_layers=2, n_heads=4, n_kv_heads=2,
+    head_dim=16, intermediate_size=128, vocab_size=512, max_position=2048,
+    rope_theta=1e6, qkv_bias=True, tie_embeddings=True,
 ))
 register(ModelSpec(
     name="tiny-moe", hidden_size=64, n_layers=2, n_heads=4, n_kv_heads=2,

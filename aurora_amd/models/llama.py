@@ -1,4 +1,5 @@
-"""Llama-3-family decoder (GQA, RoPE, SiLU-MLP) built on the aurora_amd op set.
+"""Llama-3 / Qwen2-family decoder (GQA, RoPE, SiLU-MLP, optional QKV bias)
+built on the aurora_amd op set.
 
 Every hot op routes through aurora_amd.ops (HIP/CDNA4 kernels on GPU, fp32
 torch reference on CPU); plain projections stay on hipBLASLt via F.linear.
@@ -62,7 +63,8 @@ class LlamaAttention(nn.Module):
         self.head_dim = spec.head_dim
         self.scale = spec.head_dim ** -0.5
         self.qkv_proj = QKVParallelLinear(
-            spec.hidden_size, spec.n_heads, spec.n_kv_heads, spec.head_dim, dtype=dtype
+            spec.hidden_size, spec.n_heads, spec.n_kv_heads, spec.head_dim,
+            dtype=dtype, bias=spec.qkv_bias
         )
         self.o_proj = RowParallelLinear(spec.q_size, spec.hidden_size, dtype=dtype)
 

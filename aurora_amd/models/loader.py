@@ -2,8 +2,8 @@
 
 This environment has no network, so every run is random-init — but the
 production path a reference user needs is real weights.  Maps HuggingFace
-Llama/Mixtral tensor names onto this model's fused layouts (qkv_proj,
-gate_up) and carves each rank's TP shard while reading (head-sharded qkv,
+Llama/Qwen2/Mixtral tensor names onto this model's fused layouts (qkv_proj
+weight+bias, gate_up) and carves each rank's TP shard while reading (head-sharded qkv,
 row-sharded o/down, per-half-sharded gate_up — the layout
 parallel/tp.py documents)."""
 from __future__ import annotations
@@ -96,6 +96,17 @@ def load_llama_safetensors(model, path: str, strict: bool = True) -> int:
                 _shard(v, 0, rank, tp)])
             put(layer.attn.qkv_proj.weight, fused)
             put(layer.attn.o_proj.weight, _shard(o, 1, rank, tp))
+        if layer.attn.qkv_proj.bias is not None:
+            # Qwen2-family: per-projection biases fuse/shard like the rows
+            qb = get(pre + "self_attn.q_proj.bias")
+            kb = get(pre + "self_attn.k_proj.bias")
+            vb = get(pre + "self_attn.v_proj.bias")
+            if qb is None and strict:
+                raise KeyError(pre + "self_attn.q_proj.bias")
+            if qb is not None:
+                put(layer.attn.qkv_proj.bias, torch.cat([
+                    _shard(qb, 0, rank, tp), _shard(kb, 0, rank, tp),
+                    _shard(vb, 0, rank, tp)]))
         if spec.n_experts == 0:
             g = get(pre + "mlp.gate_proj.weight")
             u = get(pre + "mlp.up_proj.weight")
@@ -149,6 +160,11 @@ def save_llama_safetensors(model, path: str) -> str:
         out[pre + "self_attn.k_proj.weight"] = qkv[nq:nq + nkv]
         out[pre + "self_attn.v_proj.weight"] = qkv[nq + nkv:]
         out[pre + "self_attn.o_proj.weight"] = layer.attn.o_proj.weight.float().cpu()
+        if layer.attn.qkv_proj.bias is not None:  # Qwen2-family
+            qkvb = layer.attn.qkv_proj.bias.float().cpu()
+            out[pre + "self_attn.q_proj.bias"] = qkvb[:nq]
+            out[pre + "self_attn.k_proj.bias"] = qkvb[nq:nq + nkv]
+            out[pre + "self_attn.v_proj.bias"] = qkvb[nq + nkv:]
         if spec.n_experts == 0:
             gu = layer.mlp.gate_up.weight.float().cpu()
             I = spec.intermediate_size

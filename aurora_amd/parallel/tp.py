@@ -19,7 +19,8 @@ from . import dist as pdist
 class ColumnParallelLinear(nn.Module):
     """Y_local = X @ W_local^T, W_local = W[rank-th output shard]."""
 
-    def __init__(self, in_features: int, out_features: int, dtype=None, gather_output: bool = False):
+    def __init__(self, in_features: int, out_features: int, dtype=None,
+                 gather_output: bool = False, bias: bool = False):
         super().__init__()
         tp = pdist.tp_world_size()
         assert out_features % tp == 0, (out_features, tp)
@@ -30,9 +31,16 @@ class ColumnParallelLinear(nn.Module):
         self.weight = nn.Parameter(
             torch.empty(self.out_local, in_features, dtype=dtype), requires_grad=False
         )
+        # output-feature bias shards with the weight rows (Qwen2-family
+        # attention projections carry a bias; Llama/Mixtral do not)
+        self.bias = nn.Parameter(
+            torch.empty(self.out_local, dtype=dtype), requires_grad=False
+        ) if bias else None
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         y = ops.linear(x, self.weight)
+        if self.bias is not None:
+            y = y + self.bias
         if self.gather_output:
             y = pdist.tp_all_gather(y, dim=-1)
         return y
@@ -61,14 +69,15 @@ class QKVParallelLinear(ColumnParallelLinear):
     """Fused QKV projection, head-sharded: local output is
     [q_heads_local*D | kv_heads_local*D | kv_heads_local*D]."""
 
-    def __init__(self, hidden: int, n_heads: int, n_kv_heads: int, head_dim: int, dtype=None):
+    def __init__(self, hidden: int, n_heads: int, n_kv_heads: int, head_dim: int,
+                 dtype=None, bias: bool = False):
         tp = pdist.tp_world_size()
         assert n_heads % tp == 0 and n_kv_heads % tp == 0, (n_heads, n_kv_heads, tp)
         self.n_heads_local = n_heads // tp
         self.n_kv_heads_local = n_kv_heads // tp
         self.head_dim = head_dim
         out = (n_heads + 2 * n_kv_heads) * head_dim
-        super().__init__(hidden, out, dtype=dtype)
+        super().__init__(hidden, out, dtype=dtype, bias=bias)
 
     def split_qkv(self, qkv: torch.Tensor):
         d = self.head_dim

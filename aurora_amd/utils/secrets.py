@@ -122,11 +122,11 @@ class FileSecretsBackend(SecretsBackend):
                 return None
         # legacy round-1 entries (unauthenticated XOR): readable for
         # migration; rewritten in v2 on the next put
-        raw = base64.b64decode(enc)
-        ks = _legacy_keystream(self.key + ref.encode(), len(raw))
         try:
+            raw = base64.b64decode(enc)
+            ks = _legacy_keystream(self.key + ref.encode(), len(raw))
             return json.loads(bytes(a ^ b for a, b in zip(raw, ks)))
-        except Exception:  # noqa: BLE001
+        except Exception:  # noqa: BLE001 - malformed legacy blob fails closed
             return None
 
 

@@ -82,3 +82,36 @@ def test_corrupt_checkpoint_fails_cleanly(tmp_path):
     other = LlamaForCausalLM(get_spec("smoke-llama"), dtype=torch.float32)
     with _pt.raises(Exception):
         load_llama_safetensors(other, ck2)
+
+
+def test_roundtrip_qwen_bias(tmp_path):
+    """Qwen2-family: q/k/v biases export to HF layout and fuse back."""
+    from aurora_amd.engine.engine import EngineConfig, LLMEngine
+    from aurora_amd.engine.sampling_params import SamplingParams
+    spec = get_spec("tiny-qwen")
+    m1 = LlamaForCausalLM(spec, dtype=torch.float32)
+    m1.init_random(14)
+    gen = torch.Generator().manual_seed(5)
+    for layer in m1.layers:  # non-constant bias: catch fuse-order mistakes
+        layer.attn.qkv_proj.bias.copy_(
+            torch.randn(layer.attn.qkv_proj.bias.shape, generator=gen))
+    save_llama_safetensors(m1, str(tmp_path))
+    m2 = LlamaForCausalLM(spec, dtype=torch.float32)
+    load_llama_safetensors(m2, str(tmp_path))
+    for (na, a), (nb, b) in zip(m1.named_parameters(), m2.named_parameters()):
+        assert torch.equal(a, b), na
+    p = SamplingParams(temperature=0.0, max_tokens=6, ignore_eos=True)
+    e1 = LLMEngine(EngineConfig(model="tiny-qwen", device="cpu"), model=m1)
+    e2 = LLMEngine(EngineConfig(model="tiny-qwen", device="cpu"), model=m2)
+    assert e1.generate([[4, 5, 6]], p) == e2.generate([[4, 5, 6]], p)
+
+
+def test_qwen_bias_checkpoint_missing_bias_fails(tmp_path):
+    """A biasless (Llama-layout) checkpoint must not silently load into a
+    qkv_bias spec (strict mode)."""
+    llama = LlamaForCausalLM(get_spec("tiny-llama"), dtype=torch.float32)
+    llama.init_random(1)
+    save_llama_safetensors(llama, str(tmp_path))
+    qwen = LlamaForCausalLM(get_spec("tiny-qwen"), dtype=torch.float32)
+    with pytest.raises(KeyError):
+        load_llama_safetensors(qwen, str(tmp_path))
