@@ -129,6 +129,11 @@ register(ModelSpec(  # GPU long-context smoke: 64k positions, tiny layers
     n_kv_heads=2, head_dim=128, intermediate_size=1024, vocab_size=1024,
     max_position=65536, tie_embeddings=True,
 ))
+register(ModelSpec(  # GPU Qwen2-shaped smoke: qkv_bias on the HIP path
+    name="smoke-qwen", hidden_size=512, n_layers=4, n_heads=4, n_kv_heads=2,
+    head_dim=128, intermediate_size=1024, vocab_size=1024, max_position=8192,
+    rope_theta=1e6, qkv_bias=True, tie_embeddings=True,
+))
 register(ModelSpec(  # GPU MoE smoke: real head_dim for the HIP kernels
     name="smoke-moe", hidden_size=512, n_layers=4, n_heads=4, n_kv_heads=2,
     head_dim=128, intermediate_size=768, vocab_size=1024, max_position=8192,
