@@ -101,3 +101,60 @@ def test_staging_key_uniquely_identifies_page_assignment():
             else:
                 key_to_pages[key] = pages
     assert eng.counters["preemptions"] > 0  # churn actually happened
+
+
+def test_kv_page_accounting_invariant_fuzz():
+    """Randomized add/step/abort/session churn on a tight page budget:
+    after every operation, every page is owned by exactly one of
+    {free list, a live sequence, a cached session, the scratch page} and
+    the total equals num_pages (no leaks, no double ownership)."""
+    import random
+
+    from aurora_amd.engine.engine import EngineConfig, LLMEngine
+    from aurora_amd.engine.sampling_params import SamplingParams
+
+    eng = LLMEngine(EngineConfig(model="tiny-llama", device="cpu", seed=3,
+                                 kv_cache_pages=24, max_model_len=256,
+                                 max_num_seqs=8))
+    rnd = random.Random(42)
+    n_added = 0
+
+    def check(tag):
+        owned = []
+        if hasattr(eng, "_scratch_page"):
+            owned.append(eng._scratch_page)
+        for s in eng.seqs.values():
+            owned.extend(s.pages)
+        for s in eng.sessions.values():
+            owned.extend(s.pages)
+        owned.extend(eng.kv._free)
+        assert len(owned) == len(set(owned)), f"{tag}: double-owned page"
+        assert len(owned) == eng.kv.num_pages, (
+            f"{tag}: {eng.kv.num_pages - len(owned)} pages leaked")
+
+    for it in range(400):
+        op = rnd.random()
+        if op < 0.45 and len(eng.seqs) < 8:
+            n_added += 1
+            sess = f"s{rnd.randrange(6)}" if rnd.random() < 0.5 else None
+            try:
+                eng.add_request(
+                    f"r{n_added}",
+                    [rnd.randrange(1, 500) for _ in range(rnd.randrange(1, 120))],
+                    SamplingParams(max_tokens=rnd.randrange(1, 20),
+                                   ignore_eos=True),
+                    session_id=sess)
+            except MemoryError:
+                pass  # admission may reject under pressure; nothing leaks
+        elif op < 0.55 and eng.seqs:
+            eng.abort_request(rnd.choice(list(eng.seqs)))
+        elif eng.has_work():
+            eng.step()
+        check(f"iter {it}")
+    # drain everything; all pages return except scratch + session cache
+    while eng.has_work():
+        eng.step()
+        check("drain")
+    for rid in list(eng.seqs):
+        eng.abort_request(rid)
+    check("end")
