@@ -18,6 +18,7 @@ Design (MI355X-first):
 """
 from __future__ import annotations
 
+import codecs
 import dataclasses
 import itertools
 import time
@@ -83,6 +84,8 @@ class Sequence:
         self.status = WAITING
         self.finish_reason: Optional[str] = None
         self.out_bytes = bytearray()
+        # streaming detokenizer: holds partial UTF-8 sequences across tokens
+        self.text_dec = codecs.getincrementaldecoder("utf-8")(errors="replace")
         self.grammar = params.grammar.fork() if params.grammar is not None else None
         self.arrival = time.monotonic()
         self.first_token_time: Optional[float] = None
@@ -803,7 +806,10 @@ class LLMEngine:
             self.counters["generated_tokens"] += 1
             tb = self.tokenizer.token_bytes(tok)
             seq.out_bytes.extend(tb)
-            new_text = tb.decode("utf-8", errors="ignore")
+            # incremental UTF-8: a codepoint split across tokens (every
+            # non-ASCII char under the byte tokenizer) must surface once
+            # complete, not be dropped byte-by-byte
+            new_text = seq.text_dec.decode(bytes(tb))
             finish = None
             p = seq.params
             if tok == self.tokenizer.eos_id and not p.ignore_eos:
@@ -825,6 +831,7 @@ class LLMEngine:
                         finish = "stop"
                         break
             if finish:
+                new_text += seq.text_dec.decode(b"", True)  # flush partial char
                 self._finish(seq, finish)
             outputs.append(StepOutput(seq.req_id, [tok], new_text, bool(finish), finish))
         return outputs

@@ -137,3 +137,82 @@ def test_rejected_request_leaves_no_subscription():
         asyncio.run(go())
     finally:
         srv.shutdown()
+
+
+def test_streaming_text_preserves_multibyte_utf8():
+    """A codepoint split across tokens (every non-ASCII char under the
+    byte tokenizer) must appear in the streamed text once complete — the
+    old per-token errors='ignore' decode silently dropped it."""
+    from aurora_amd.engine.engine import EngineConfig, LLMEngine
+    from aurora_amd.engine.sampling_params import SamplingParams
+
+    LIT = "latency → 97 ms café".encode("utf-8")
+
+    class LiteralGrammar:
+        """Forces the exact byte string LIT, then completes."""
+
+        def __init__(self, pos=0):
+            self.pos = pos
+
+        def fork(self):
+            return LiteralGrammar(self.pos)
+
+        def is_complete(self):
+            return self.pos >= len(LIT)
+
+        def advance(self, data: bytes) -> bool:
+            if LIT[self.pos:self.pos + len(data)] != bytes(data) or not data:
+                return False
+            self.pos += len(data)
+            return True
+
+        def allowed_next_bytes(self):
+            return b"" if self.is_complete() else LIT[self.pos:self.pos + 1]
+
+    eng = LLMEngine(EngineConfig(model="tiny-llama", device="cpu", seed=0))
+    eng.add_request("utf8", [3, 4, 5],
+                    SamplingParams(temperature=0.0, max_tokens=len(LIT) + 4,
+                                   grammar=LiteralGrammar()))
+    text = ""
+    while eng.has_work():
+        for out in eng.step():
+            text += out.new_text
+    assert text == LIT.decode("utf-8"), repr(text)
+
+
+def test_streaming_text_flushes_partial_char_at_cutoff():
+    """max_tokens cutting generation mid-codepoint must flush a
+    replacement char, not strand bytes in the decoder."""
+    from aurora_amd.engine.engine import EngineConfig, LLMEngine
+    from aurora_amd.engine.sampling_params import SamplingParams
+
+    LIT = "aé".encode("utf-8")  # 3 bytes; cut after 2
+
+    class LiteralGrammar:
+        def __init__(self, pos=0):
+            self.pos = pos
+
+        def fork(self):
+            return LiteralGrammar(self.pos)
+
+        def is_complete(self):
+            return self.pos >= len(LIT)
+
+        def advance(self, data: bytes) -> bool:
+            if LIT[self.pos:self.pos + len(data)] != bytes(data) or not data:
+                return False
+            self.pos += len(data)
+            return True
+
+        def allowed_next_bytes(self):
+            return b"" if self.is_complete() else LIT[self.pos:self.pos + 1]
+
+    eng = LLMEngine(EngineConfig(model="tiny-llama", device="cpu", seed=0))
+    eng.add_request("cut", [3, 4, 5],
+                    SamplingParams(temperature=0.0, max_tokens=2,
+                                   grammar=LiteralGrammar()))
+    text = ""
+    while eng.has_work():
+        for out in eng.step():
+            text += out.new_text
+    assert text == "a�", repr(text)
