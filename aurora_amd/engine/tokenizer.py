@@ -84,10 +84,34 @@ class HFTokenizer(Tokenizer):
         from tokenizers import Tokenizer as _HT  # offline wheelhouse package
 
         self._t = _HT.from_file(path)
-        self.bos_id = self._t.token_to_id("<|begin_of_text|>") or self._t.token_to_id("<s>") or 0
-        self.eos_id = self._t.token_to_id("<|end_of_text|>") or self._t.token_to_id("</s>") or 1
+
+        def first_id(names: Sequence[str], default: int) -> int:
+            # explicit None checks: `or` would skip a special at id 0
+            for n in names:
+                i = self._t.token_to_id(n)
+                if i is not None:
+                    return i
+            return default
+
+        self.bos_id = first_id(("<|begin_of_text|>", "<s>", "<bos>"), 0)
+        self.eos_id = first_id(
+            ("<|end_of_text|>", "<|eot_id|>", "</s>", "<eos>",
+             "<|endoftext|>", "<|im_end|>"), 1)
         self.pad_id = self.eos_id
         self._byte_cache: dict[int, bytes] = {}
+        # ids whose tokens are control/special markers (no text contribution)
+        self._special_ids = set()
+        added = getattr(self._t, "get_added_tokens_decoder", None)
+        if added is not None:
+            for i, at in added().items():
+                if getattr(at, "special", True):
+                    self._special_ids.add(i)
+        else:  # older tokenizers: fall back to the declared specials
+            for n in ("<s>", "</s>", "<unk>", "<pad>", "<|begin_of_text|>",
+                      "<|end_of_text|>", "<|eot_id|>", "<|endoftext|>"):
+                i = self._t.token_to_id(n)
+                if i is not None:
+                    self._special_ids.add(i)
 
     @property
     def vocab_size(self) -> int:
@@ -100,10 +124,43 @@ class HFTokenizer(Tokenizer):
     def decode(self, ids: Sequence[int]) -> str:
         return self._t.decode(list(ids), skip_special_tokens=True)
 
+    @staticmethod
+    def _bytelevel_table() -> dict:
+        """GPT-2/ByteLevel char -> raw byte map (the printable-unicode
+        encoding byte-level BPE vocabs store their merges in)."""
+        bs = (list(range(33, 127)) + list(range(161, 173))
+              + list(range(174, 256)))
+        cs = list(bs)
+        n = 0
+        for b in range(256):
+            if b not in bs:
+                bs.append(b)
+                cs.append(256 + n)
+                n += 1
+        return {chr(c): b for b, c in zip(bs, cs)}
+
+    _BL_TABLE: Optional[dict] = None
+
     def token_bytes(self, token_id: int) -> bytes:
+        """Raw output bytes of one token, derived from the VOCAB string —
+        decoding a lone token would mangle codepoints whose UTF-8 bytes
+        are split across neighbouring tokens."""
         b = self._byte_cache.get(token_id)
         if b is None:
-            b = self._t.decode([token_id], skip_special_tokens=True).encode("utf-8")
+            s = self._t.id_to_token(token_id)
+            if s is None or token_id in self._special_ids:
+                b = b""  # special / control token contributes no text
+            elif s.startswith("<0x") and s.endswith(">") and len(s) == 6:
+                b = bytes([int(s[3:5], 16)])  # sentencepiece byte token
+            else:
+                if HFTokenizer._BL_TABLE is None:
+                    HFTokenizer._BL_TABLE = self._bytelevel_table()
+                tbl = HFTokenizer._BL_TABLE
+                if all(ch in tbl for ch in s):
+                    b = bytes(tbl[ch] for ch in s)  # byte-level BPE vocab
+                else:
+                    # plain-text vocab (sentencepiece "▁word" style)
+                    b = s.replace("▁", " ").encode("utf-8")
             self._byte_cache[token_id] = b
         return b
 

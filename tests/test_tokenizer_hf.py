@@ -1,0 +1,99 @@
+"""HFTokenizer (real BPE) tests: byte derivation from the vocab, buckets,
+and grammar-constrained decoding with multi-byte tokens.
+
+The BPE is trained offline in-test (`tokenizers` wheel, no network) —
+byte-level like Llama-3/Qwen2 production tokenizers (docstring contract:
+aurora_amd/engine/tokenizer.py HFTokenizer)."""
+import json
+
+import pytest
+
+from aurora_amd.engine.tokenizer import HFTokenizer
+
+
+@pytest.fixture(scope="module")
+def bpe_path(tmp_path_factory):
+    from tokenizers import Tokenizer, decoders, models, pre_tokenizers, trainers
+    tok = Tokenizer(models.BPE())
+    tok.pre_tokenizer = pre_tokenizers.ByteLevel(add_prefix_space=False)
+    tok.decoder = decoders.ByteLevel()
+    trainer = trainers.BpeTrainer(
+        vocab_size=320, special_tokens=["<s>", "</s>"],
+        initial_alphabet=pre_tokenizers.ByteLevel.alphabet())
+    corpus = ['{"tool": "kubectl", "args": {"command": "get pods"}}',
+              '{"tool": "final_answer", "args": {"text": "ok"}}',
+              "latency spike p99 postgres pool exhausted café →"] * 50
+    tok.train_from_iterator(corpus, trainer)
+    p = str(tmp_path_factory.mktemp("bpe") / "tokenizer.json")
+    tok.save(p)
+    return p
+
+
+def test_roundtrip_and_byte_concat(bpe_path):
+    t = HFTokenizer(bpe_path)
+    s = '{"tool": "kubectl"} café → p99'
+    ids = t.encode(s)
+    assert t.decode(ids) == s
+    # token_bytes must reconstruct the exact UTF-8 stream even when a
+    # codepoint's bytes are split across BPE tokens
+    assert b"".join(t.token_bytes(i) for i in ids).decode("utf-8") == s
+
+
+def test_specials_contribute_no_bytes(bpe_path):
+    t = HFTokenizer(bpe_path)
+    assert t.token_bytes(t.bos_id) == b""
+    assert t.token_bytes(t.eos_id) == b""
+    ids = t.encode("hi", add_bos=True)
+    assert ids[0] == t.bos_id
+
+
+def test_first_byte_buckets_cover_vocab(bpe_path):
+    t = HFTokenizer(bpe_path)
+    bk = t.first_byte_buckets()
+    assert len(bk) == 256  # byte-level alphabet covers every first byte
+    n = sum(len(v) for v in bk.values())
+    specials = sum(1 for i in range(t.vocab_size) if not t.token_bytes(i))
+    assert n == t.vocab_size - specials
+    for by, lst in bk.items():
+        for i in lst[:3]:
+            assert t.token_bytes(i)[0] == by
+
+
+def test_sentencepiece_byte_token_form(tmp_path):
+    """<0xNN> byte-fallback tokens decode to their raw byte."""
+    from tokenizers import Tokenizer, models
+    vocab = {"<s>": 0, "</s>": 1, "<0x0A>": 2, "hello": 3, "a": 4, "b": 5,
+             "ab": 6}
+    tok = Tokenizer(models.BPE(vocab=vocab, merges=[("a", "b")]))
+    p = str(tmp_path / "sp.json")
+    tok.save(p)
+    t = HFTokenizer(p)
+    assert t.token_bytes(2) == b"\n"
+    assert t.token_bytes(3) == b"hello"
+
+
+def test_engine_grammar_with_bpe_tokenizer(bpe_path):
+    """Grammar-constrained tool-call decoding with a multi-byte-token
+    tokenizer: every emitted token's FULL byte string is validated by the
+    FSM (engine._grammar_pick), so the output parses."""
+    from aurora_amd.engine.engine import EngineConfig, LLMEngine
+    from aurora_amd.engine.grammar import ToolCallGrammar
+    from aurora_amd.engine.sampling_params import SamplingParams
+
+    t = HFTokenizer(bpe_path)
+    eng = LLMEngine(EngineConfig(model="tiny-llama", device="cpu", seed=2),
+                    tokenizer=t)
+    g = ToolCallGrammar(["kubectl", "final_answer"], soft_close_bytes=60)
+    outs = eng.generate(
+        [t.encode("investigate the pod crash"),
+         t.encode("summarize status")],
+        SamplingParams(temperature=0.7, max_tokens=90, grammar=g))
+    for toks in outs:
+        # the FSM constrains JSON structure at the BYTE level; inside
+        # strings any bytes are legal, so decode like the engine streams
+        # (replacement chars are valid JSON string content)
+        text = b"".join(t.token_bytes(i) for i in toks).decode(
+            "utf-8", errors="replace")
+        doc = json.loads(text)
+        assert doc["tool"] in ("kubectl", "final_answer")
+        assert isinstance(doc["args"], dict)
