@@ -73,6 +73,8 @@ class Runtime:
                 kv_memory_fraction=self.settings.kv_memory_fraction,
                 max_num_seqs=self.settings.max_num_seqs,
                 tp_size=self.settings.tp_size,
+                checkpoint_path=self.settings.checkpoint_path or None,
+                tokenizer_path=self.settings.tokenizer_path or None,
             )
             engine_server = AsyncEngineServer(ecfg)
         self.engine = engine_server

@@ -36,6 +36,10 @@ class Settings:
     kv_memory_fraction: float = field(default_factory=lambda: float(os.environ.get("KV_MEMORY_FRACTION", "0.55")))
     max_num_seqs: int = field(default_factory=lambda: _i("ENGINE_MAX_SEQS", 64))
     tp_size: int = field(default_factory=lambda: _i("ENGINE_TP", 1))
+    # production weights: HF-layout safetensors dir + tokenizer.json
+    # (empty = random init / byte tokenizer, this environment's default)
+    checkpoint_path: str = field(default_factory=lambda: os.environ.get("MODEL_CHECKPOINT", ""))
+    tokenizer_path: str = field(default_factory=lambda: os.environ.get("MODEL_TOKENIZER", ""))
     # agent loop (reference: .env.example:111,160-161)
     agent_recursion_limit: int = field(default_factory=lambda: _i("AGENT_RECURSION_LIMIT", 240))
     max_input_tokens: int = field(default_factory=lambda: _i("MAX_INPUT_TOKENS", 20000))

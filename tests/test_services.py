@@ -436,3 +436,50 @@ def test_db_concurrent_writers():
     [t.join(timeout=60) for t in ts]
     assert not errs, errs
     assert len(store.list_incidents(limit=1000)) == 240
+
+
+def test_runtime_serves_checkpoint_and_tokenizer(tmp_path):
+    """MODEL_CHECKPOINT + MODEL_TOKENIZER settings reach the engine: the
+    service engine generates exactly what the exported model does and
+    tokenizes with the checkpoint's BPE (the production weights path)."""
+    import torch
+    from aurora_amd.engine.sampling_params import SamplingParams
+    from aurora_amd.models.config import get_spec
+    from aurora_amd.models.llama import LlamaForCausalLM
+    from aurora_amd.models.loader import save_llama_safetensors
+    from aurora_amd.service.runtime import Runtime
+    from aurora_amd.utils.config import Settings
+    from tokenizers import Tokenizer, decoders, models, pre_tokenizers, trainers
+
+    src = LlamaForCausalLM(get_spec("tiny-llama"), dtype=torch.float32)
+    src.init_random(19)
+    ck = str(tmp_path / "ck")
+    save_llama_safetensors(src, ck)
+    tok = Tokenizer(models.BPE())
+    tok.pre_tokenizer = pre_tokenizers.ByteLevel(add_prefix_space=False)
+    tok.decoder = decoders.ByteLevel()
+    tok.train_from_iterator(
+        ["investigate the incident"] * 30,
+        trainers.BpeTrainer(vocab_size=300, special_tokens=["<s>", "</s>"],
+                            initial_alphabet=pre_tokenizers.ByteLevel.alphabet()))
+    tok_path = str(tmp_path / "tokenizer.json")
+    tok.save(tok_path)
+
+    cfg = Settings(db_path=":memory:",
+                   secrets_path=str(tmp_path / "sec.json"))
+    cfg.engine_device = "cpu"
+    cfg.checkpoint_path = ck
+    cfg.tokenizer_path = tok_path
+    rt = Runtime(cfg)
+    try:
+        eng = rt.engine.engine
+        from aurora_amd.engine.tokenizer import HFTokenizer
+        assert isinstance(eng.tokenizer, HFTokenizer)
+        p = SamplingParams(temperature=0.0, max_tokens=5, ignore_eos=True)
+        got = eng.generate([[4, 5, 6]], p)
+        from aurora_amd.engine.engine import EngineConfig, LLMEngine
+        ref = LLMEngine(EngineConfig(model="tiny-llama", device="cpu"),
+                        model=src, tokenizer=eng.tokenizer)
+        assert got == ref.generate([[4, 5, 6]], p)
+    finally:
+        rt.engine.shutdown()
