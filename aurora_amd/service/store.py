@@ -45,6 +45,7 @@ CREATE INDEX IF NOT EXISTS idx_incidents_org ON incidents(org_id, created_at);
 CREATE TABLE IF NOT EXISTS alerts (
   id TEXT PRIMARY KEY, org_id TEXT NOT NULL, incident_id TEXT,
   source TEXT NOT NULL, fingerprint TEXT, payload TEXT, created_at REAL NOT NULL);
+CREATE INDEX IF NOT EXISTS idx_alerts_org_created ON alerts(org_id, created_at);
 CREATE TABLE IF NOT EXISTS chat_sessions (
   id TEXT PRIMARY KEY, org_id TEXT NOT NULL, user_id TEXT,
   messages TEXT NOT NULL DEFAULT '[]', is_background INTEGER NOT NULL DEFAULT 0,
@@ -55,6 +56,8 @@ CREATE TABLE IF NOT EXISTS llm_context_history (
 CREATE TABLE IF NOT EXISTS incident_thoughts (
   id INTEGER PRIMARY KEY AUTOINCREMENT, org_id TEXT NOT NULL,
   incident_id TEXT NOT NULL, content TEXT NOT NULL, ts REAL NOT NULL);
+CREATE INDEX IF NOT EXISTS idx_thoughts_org_incident
+  ON incident_thoughts(org_id, incident_id);
 CREATE TABLE IF NOT EXISTS execution_steps (
   id TEXT PRIMARY KEY, org_id TEXT NOT NULL, incident_id TEXT,
   session_id TEXT, tool TEXT NOT NULL, args TEXT, output_excerpt TEXT,
@@ -85,6 +88,8 @@ CREATE TABLE IF NOT EXISTS actions (
 CREATE TABLE IF NOT EXISTS lifecycle_events (
   id INTEGER PRIMARY KEY AUTOINCREMENT, org_id TEXT NOT NULL,
   incident_id TEXT NOT NULL, event TEXT NOT NULL, data TEXT, ts REAL NOT NULL);
+CREATE INDEX IF NOT EXISTS idx_lifecycle_org_incident
+  ON lifecycle_events(org_id, incident_id);
 CREATE TABLE IF NOT EXISTS llm_usage (
   id INTEGER PRIMARY KEY AUTOINCREMENT, org_id TEXT NOT NULL,
   model TEXT NOT NULL, prompt_tokens INTEGER NOT NULL,
