@@ -232,3 +232,58 @@ def test_chaos_engine_faults_requeue_until_complete():
         srv.shutdown()
 
     asyncio.run(main())
+
+
+def test_chaos_fault_in_post_rca_tail_still_completes():
+    """An engine that dies AFTER the workflow (during the overlapped
+    summary/recommendations tail, worker.py) must degrade to the
+    alert-only summary — the incident still reaches 'completed', with no
+    stranded asyncio task from the concurrent summary pass."""
+    import asyncio
+
+    from aurora_amd.engine.engine import EngineConfig
+    from aurora_amd.engine.server import AsyncEngineServer
+    from aurora_amd.service.runtime import Runtime
+    from aurora_amd.service.webhooks import process_webhook
+    from aurora_amd.utils.config import Settings
+
+    class TailFaultServer(AsyncEngineServer):
+        """Healthy for the workflow's stream(); every complete_text call
+        (the post-RCA passes) faults like a dead engine."""
+
+        async def complete_text(self, prompt, params, session_id=None):
+            raise RuntimeError("engine step failed: injected tail fault")
+
+    async def main():
+        srv = TailFaultServer(EngineConfig(model="tiny-llama", device="cpu",
+                                           seed=6))
+        cfg = Settings(db_path=":memory:",
+                       secrets_path="/tmp/chaos_tail_sec.json")
+        cfg.worker_concurrency = 2
+        rt = Runtime(cfg, engine_server=srv)
+        oid = rt.admin.create_org("chaos-tail")
+        rt.start()
+        iid, created = process_webhook(
+            rt, oid, "pagerduty",
+            {"event": {"data": {"id": "t1", "title": "tail fault alert",
+                                "service": {"summary": "svc"}}}},
+            rca_countdown_s=0.0,
+            rca_args={"max_turns": 2,
+                      "tool_subset": ["list_incidents", "final_answer"]})
+        assert created
+        store = rt.org(oid)
+        for _ in range(600):
+            inc = store.get_incident(iid)
+            if inc["aurora_status"] in ("completed", "failed"):
+                break
+            await asyncio.sleep(0.2)
+        inc = store.get_incident(iid)
+        assert inc["aurora_status"] == "completed", inc["aurora_status"]
+        # degraded artifacts: alert-only summary, no suggestions
+        assert "tail fault alert" in (inc["summary"] or "")
+        assert store.suggestions(iid) == []
+        assert store.get_postmortem(iid) is not None
+        await rt.stop()
+        srv.shutdown()
+
+    asyncio.run(main())
