@@ -158,3 +158,45 @@ def test_kv_page_accounting_invariant_fuzz():
     for rid in list(eng.seqs):
         eng.abort_request(rid)
     check("end")
+
+
+def test_session_reuse_equivalence_fuzz():
+    """Randomized multi-turn sessions — extending, partially diverging,
+    and fully replacing the cached prefix — must generate exactly what a
+    cold engine does for every request (the keep_pages common-prefix
+    carve in engine.add_request is the code under test)."""
+    import random
+
+    from aurora_amd.engine.engine import EngineConfig, LLMEngine
+    from aurora_amd.engine.sampling_params import SamplingParams
+
+    rnd = random.Random(7)
+    cfg = EngineConfig(model="tiny-llama", device="cpu", seed=5,
+                       kv_cache_pages=96, max_model_len=512)
+    eng = LLMEngine(cfg)
+    p = SamplingParams(temperature=0.0, max_tokens=4, ignore_eos=True)
+    history = {}  # session -> last prompt
+    n = 0
+    for it in range(30):
+        sess = f"s{rnd.randrange(4)}"
+        prev = history.get(sess)
+        if prev is None or rnd.random() < 0.25:
+            prompt = [rnd.randrange(2, 500) for _ in range(rnd.randrange(4, 60))]
+        elif rnd.random() < 0.6:
+            # extend the cached turn (the common fast path)
+            prompt = prev + [rnd.randrange(2, 500)
+                             for _ in range(rnd.randrange(1, 40))]
+        else:
+            # diverge mid-prefix: only part of the cache is reusable
+            keep = rnd.randrange(0, len(prev) + 1)
+            prompt = prev[:keep] + [rnd.randrange(2, 500)
+                                    for _ in range(rnd.randrange(1, 30))]
+        history[sess] = prompt
+        n += 1
+        eng.add_request(f"r{n}", prompt, p, session_id=sess)
+        got = []
+        while eng.has_work():
+            for o in eng.step():
+                got.extend(o.new_token_ids)
+        fresh = LLMEngine(cfg).generate([prompt], p)[0]
+        assert got == fresh, (it, sess, len(prompt))
