@@ -97,3 +97,45 @@ def test_engine_grammar_with_bpe_tokenizer(bpe_path):
         doc = json.loads(text)
         assert doc["tool"] in ("kubectl", "final_answer")
         assert isinstance(doc["args"], dict)
+
+
+def test_full_rca_pipeline_with_bpe_tokenizer(bpe_path, tmp_path):
+    """webhook -> queue -> guardrailed workflow (fused ReAct grammar) ->
+    summary/recommendations, all through an engine running a real BPE
+    tokenizer (MODEL_TOKENIZER) — the production tokenizer path."""
+    import asyncio
+
+    from aurora_amd.service.runtime import Runtime
+    from aurora_amd.service.webhooks import process_webhook
+    from aurora_amd.utils.config import Settings
+
+    async def main():
+        cfg = Settings(db_path=":memory:",
+                       secrets_path=str(tmp_path / "sec.json"))
+        cfg.engine_device = "cpu"
+        cfg.tokenizer_path = bpe_path
+        cfg.worker_concurrency = 2
+        rt = Runtime(cfg)
+        from aurora_amd.engine.tokenizer import HFTokenizer
+        assert isinstance(rt.engine.engine.tokenizer, HFTokenizer)
+        oid = rt.admin.create_org("bpe")
+        rt.start()
+        iid, created = process_webhook(
+            rt, oid, "pagerduty",
+            {"event": {"data": {"id": "b1", "title": "bpe pipeline alert",
+                                "service": {"summary": "svc"}}}},
+            rca_countdown_s=0.0,
+            rca_args={"max_turns": 2,
+                      "tool_subset": ["list_incidents", "final_answer"]})
+        assert created
+        store = rt.org(oid)
+        for _ in range(600):
+            inc = store.get_incident(iid)
+            if inc["aurora_status"] in ("completed", "failed"):
+                break
+            await asyncio.sleep(0.2)
+        assert store.get_incident(iid)["aurora_status"] == "completed"
+        await rt.stop()
+        rt.engine.shutdown()
+
+    asyncio.run(main())
