@@ -6,7 +6,7 @@ from __future__ import annotations
 
 import json
 import re
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 from .base import (fail_closed,
                    ProviderAPIError, Request, Transport, truncate_items,

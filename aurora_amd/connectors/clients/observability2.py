@@ -10,7 +10,7 @@ through the pluggable transport.
 from __future__ import annotations
 
 import json
-from typing import List, Optional
+from typing import Optional
 
 from .base import (fail_closed,
                    ProviderAPIError, Request, Transport, iso, series_stats,

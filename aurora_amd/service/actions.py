@@ -5,7 +5,7 @@ from __future__ import annotations
 
 import json
 import time
-from typing import Any, Dict
+from typing import Dict
 
 
 async def run_incident_actions(rt, org_id: str, incident_id: str) -> dict:

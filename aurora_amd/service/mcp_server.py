@@ -14,7 +14,7 @@ import asyncio
 import json
 import sys
 from dataclasses import dataclass
-from typing import Any, Callable, Dict, List, Optional
+from typing import Callable, Dict, List, Optional
 
 
 @dataclass
