@@ -44,6 +44,32 @@ class AlertCorrelator:
         self.rag = rag  # reuse the RAG encoder for similarity when present
         self.window_s = window_s
         self.threshold = threshold
+        self._inc_embs: dict = {}  # incident_id -> cached normalized vec
+
+    def _embed_one(self, text: str):
+        """Normalized embedding of one alert text via the shared encoder
+        (reference: similarity strategy over text2vec embeddings,
+        embedding_client.py:20-82); None when no encoder is attached."""
+        if self.rag is None:
+            return None
+        try:
+            v = self.rag._embed([text])
+        except Exception:  # noqa: BLE001 - similarity degrades to lexical
+            return None
+        if v is None:
+            return None
+        import torch.nn.functional as F
+        return F.normalize(v[0].float().cpu(), dim=0)
+
+    def _inc_vec(self, incident_id: str, text: str):
+        cached = self._inc_embs.get(incident_id)
+        if cached is None:
+            cached = self._embed_one(text)
+            if cached is not None:
+                if len(self._inc_embs) > 512:
+                    self._inc_embs.clear()
+                self._inc_embs[incident_id] = cached
+        return cached
 
     def correlate(self, title: str, payload: dict,
                   fingerprint: str = "") -> CorrelationResult:
@@ -59,12 +85,21 @@ class AlertCorrelator:
                 if alert["fingerprint"] == fingerprint and alert["incident_id"]:
                     return CorrelationResult(alert["incident_id"], 1.0, "fingerprint")
         text = _alert_text(title, payload)
+        qv = self._embed_one(text)  # [H] or None (no encoder / CPU)
         best: Optional[CorrelationResult] = None
         svc = payload.get("service") or payload.get("alert", {}).get("service")
         for inc in candidates:
             ipay = json.loads(inc.get("payload") or "{}")
-            sim = _jaccard(text, _alert_text(inc["title"], ipay))
+            itext = _alert_text(inc["title"], ipay)
+            sim = _jaccard(text, itext)
             strategy = "similarity"
+            if qv is not None:
+                iv = self._inc_vec(inc["id"], itext)
+                if iv is not None:
+                    cos = float((qv * iv).sum())
+                    if cos > sim:
+                        sim = cos
+                        strategy = "embedding"
             # topology bonus: services adjacent in the dependency graph
             if self.graph is not None and svc:
                 isvc = ipay.get("service") or ipay.get("alert", {}).get("service")

@@ -483,3 +483,39 @@ def test_runtime_serves_checkpoint_and_tokenizer(tmp_path):
         assert got == ref.generate([[4, 5, 6]], p)
     finally:
         rt.engine.shutdown()
+
+
+def test_correlator_embedding_strategy():
+    """With an encoder attached (AURORA_CORRELATOR_EMBED path), alerts
+    whose words differ but whose embeddings align correlate via the
+    'embedding' strategy; without one the same pair stays separate
+    (reference: similarity strategy over text2vec embeddings)."""
+    import torch
+    from aurora_amd.service.correlation import AlertCorrelator
+    from aurora_amd.service.store import AdminStore, Database
+
+    db = Database(":memory:")
+    admin = AdminStore(db)
+    store = admin.org_store(admin.create_org("t"))
+
+    class FakeRag:
+        def _embed(self, texts):
+            # same direction for anything mentioning checkout; orthogonal else
+            out = []
+            for t in texts:
+                v = torch.zeros(8)
+                v[0 if "checkout" in t else 3] = 1.0
+                out.append(v)
+            return torch.stack(out)
+
+    # lexical-only correlator: the two texts share no 3+-char words
+    lex = AlertCorrelator(store, rag=None)
+    iid1, created = lex.handle_alert(
+        "pagerduty", "checkout 5xx errors rising", {"description": "api"})
+    assert created
+    emb = AlertCorrelator(store, rag=FakeRag())
+    res = emb.correlate("payment flow broken for checkout", {})
+    assert res.incident_id == iid1 and res.strategy == "embedding", res
+    # orthogonal embedding + no shared words -> new incident
+    res2 = emb.correlate("disk full on build runner", {})
+    assert res2.incident_id is None
