@@ -55,6 +55,11 @@ class EngineConfig:
     seed: int = 0
     grammar_topk: int = 32  # candidates tried per step under a grammar
     use_hip_graphs: bool = True  # capture the decode forward per batch bucket
+    # cross-request prefix caching: identical full prompt pages (the shared
+    # system+skills prefix of concurrent investigations — measured 92% of a
+    # first-turn RCA prompt) are computed once and refcount-shared.
+    # AURORA_PREFIX_CACHE=0 is the kill switch.
+    prefix_cache: bool = True
 
 
 @dataclass
@@ -67,6 +72,16 @@ class StepOutput:
 
 
 _SEQ_SERIAL = itertools.count()
+
+
+def _chain_hash(prev: bytes, tokens: List[int]) -> bytes:
+    """Rolling hash of one full page of prompt tokens, chained so a page's
+    identity includes everything before it (vLLM-style block hashing)."""
+    import hashlib
+    from array import array
+    h = hashlib.blake2b(prev, digest_size=16)
+    h.update(array("q", tokens).tobytes())
+    return h.digest()
 
 
 class Sequence:
@@ -89,6 +104,9 @@ class Sequence:
         self.grammar = params.grammar.fork() if params.grammar is not None else None
         self.arrival = time.monotonic()
         self.first_token_time: Optional[float] = None
+        # prefix-cache registration progress (engine._register_prefix)
+        self.reg_pages = 0
+        self.prefix_hash = b""
         # bumped whenever this seq's pages are freed (preemption/session evict)
         # so graph block-table staging keyed on (req_id, gen, npages) can never
         # reuse a stale row after the free list hands back different page ids
@@ -170,6 +188,15 @@ class LLMEngine:
         self.seqs: Dict[str, Sequence] = {}
         # session KV reuse: session_id -> finished Sequence kept warm in cache
         self.sessions: "OrderedDict[str, Sequence]" = OrderedDict()
+        # cross-request prefix cache: chain-hash of full prompt pages ->
+        # resident page id (refcount-shared across sequences). Entries die
+        # with the page (kv.on_final_free).
+        self.prefix_cache = (config.prefix_cache
+                             and _os.environ.get("AURORA_PREFIX_CACHE", "1") == "1")
+        self._prefix_table: Dict[bytes, int] = {}
+        self._page_hash: Dict[int, bytes] = {}
+        self.kv.on_final_free = self._retain_page
+        self.kv.on_cache_evict = self._drop_page_hash
         self.gen = torch.Generator(device="cpu").manual_seed(config.seed)
         self._graph_cache: Dict[int, tuple] = {}
         self._graph_pool = None
@@ -190,7 +217,7 @@ class LLMEngine:
         self.counters = {
             "prompt_tokens": 0, "generated_tokens": 0, "prefill_steps": 0,
             "decode_steps": 0, "preemptions": 0, "grammar_fallbacks": 0,
-            "grammar_bucket_picks": 0,
+            "grammar_bucket_picks": 0, "prefix_cache_pages": 0,
         }
         # phase timers (seconds, cumulative) for perf triage
         self.timers = {"schedule": 0.0, "build": 0.0, "forward": 0.0,
@@ -231,8 +258,66 @@ class LLMEngine:
                 seq.num_cached = keep_pages * self.kv.page_size
             else:
                 self.kv.free(cached.pages)
+        if self.prefix_cache and seq.num_cached == 0:
+            self._attach_prefix(seq)
         self.seqs[request_id] = seq
         self.waiting.append(seq)
+
+    # ------------------------------------------------- prefix cache (global)
+
+    def _attach_prefix(self, seq: Sequence) -> None:
+        """Adopt already-resident pages for the longest matching full-page
+        prompt prefix. Never the page holding the final prompt token (at
+        least one token must run to produce logits, and the seq must only
+        ever write K/V into pages it owns exclusively)."""
+        P = self.kv.page_size
+        usable = (seq.num_prompt - 1) // P
+        h = b""
+        attached: List[int] = []
+        for k in range(usable):
+            h2 = _chain_hash(h, seq.token_ids[k * P:(k + 1) * P])
+            pid = self._prefix_table.get(h2)
+            if pid is None:
+                break
+            attached.append(pid)
+            h = h2  # hash of exactly the attached pages
+        if attached:
+            self.kv.incref(attached)
+            seq.pages = list(attached)
+            seq.num_cached = len(attached) * P
+            seq.reg_pages = len(attached)
+            seq.prefix_hash = h
+            self.counters["prefix_cache_pages"] += len(attached)
+
+    def _register_prefix(self, seq: Sequence) -> None:
+        """Publish this seq's freshly-prefilled full prompt pages so later
+        requests with the same prefix can share them."""
+        if not self.prefix_cache:
+            return
+        P = self.kv.page_size
+        complete = min(seq.num_cached, seq.num_prompt) // P
+        while seq.reg_pages < complete:
+            k = seq.reg_pages
+            seq.prefix_hash = _chain_hash(seq.prefix_hash,
+                                          seq.token_ids[k * P:(k + 1) * P])
+            pid = seq.pages[k]
+            if (seq.prefix_hash not in self._prefix_table
+                    and pid not in self._page_hash):
+                self._prefix_table[seq.prefix_hash] = pid
+                self._page_hash[pid] = seq.prefix_hash
+            seq.reg_pages += 1
+
+    def _retain_page(self, page: int) -> bool:
+        """Last ref dropped: keep registered prefix pages warm (LRU tier,
+        still counted free) so later identical prompts attach; others
+        recycle immediately."""
+        return page in self._page_hash
+
+    def _drop_page_hash(self, page: int) -> None:
+        """A retained page is being reclaimed for new content."""
+        h = self._page_hash.pop(page, None)
+        if h is not None and self._prefix_table.get(h) == page:
+            del self._prefix_table[h]
 
     def abort_request(self, request_id: str) -> None:
         seq = self.seqs.pop(request_id, None)
@@ -282,6 +367,8 @@ class LLMEngine:
             self.kv.free(victim.pages)
             victim.pages = []
             victim.num_cached = 0
+            victim.reg_pages = 0
+            victim.prefix_hash = b""
             victim.page_gen += 1
             victim.status = WAITING
             self.waiting.appendleft(victim)
@@ -406,6 +493,7 @@ class LLMEngine:
             for i, (seq, chunk) in enumerate(zip(seqs, chunks)):
                 had_output = seq.num_output > 0
                 seq.num_cached += chunk
+                self._register_prefix(seq)
                 self.counters["prompt_tokens"] += chunk
                 if seq.num_cached >= seq.prefill_target:
                     seq.status = RUNNING
@@ -503,6 +591,7 @@ class LLMEngine:
         for i, (seq, chunk) in enumerate(zip(prefill, chunks)):
             had_output = seq.num_output > 0
             seq.num_cached += chunk
+            self._register_prefix(seq)
             self.counters["prompt_tokens"] += chunk
             if seq.num_cached >= seq.prefill_target:
                 seq.status = RUNNING

@@ -12,7 +12,7 @@ Replaces nothing in the reference (it rents context windows; SURVEY.md
 """
 from __future__ import annotations
 
-from typing import List, Tuple
+from typing import Dict, List, Tuple
 
 import torch
 
@@ -42,6 +42,20 @@ class PagedKVCache:
             self.k_caches.append(torch.zeros(shape, device=device, dtype=dtype))
             self.v_caches.append(torch.zeros(shape, device=device, dtype=dtype))
         self._free: List[int] = list(range(num_pages - 1, -1, -1))
+        # page refcounts: cross-request prefix sharing gives one page many
+        # owners (each seq/session holding it counts one ref)
+        self._rc: Dict[int, int] = {}
+        # rc==0 pages whose content is retained for future prefix reuse
+        # (LRU; reclaimed by alloc() on demand — they still count as free)
+        from collections import OrderedDict as _OD
+        self._cached: "_OD[int, None]" = _OD()
+        #: called when a page's LAST ref drops; return True to RETAIN the
+        #: page content in the cached tier (the engine keeps prefix-table
+        #: entries alive for retained pages)
+        self.on_final_free = None
+        #: called when a retained page is reclaimed by alloc() — the engine
+        #: must drop its prefix-table entry before the page is rewritten
+        self.on_cache_evict = None
 
     @staticmethod
     def pages_for_memory(
@@ -61,15 +75,46 @@ class PagedKVCache:
     # -- page allocation ---------------------------------------------------
 
     def num_free(self) -> int:
-        return len(self._free)
+        return len(self._free) + len(self._cached)  # cached is reclaimable
 
     def alloc(self, n: int) -> List[int]:
-        if n > len(self._free):
-            raise MemoryError(f"KV cache exhausted: want {n} pages, have {len(self._free)}")
-        return [self._free.pop() for _ in range(n)]
+        if n > self.num_free():
+            raise MemoryError(f"KV cache exhausted: want {n} pages, have {self.num_free()}")
+        pages: List[int] = []
+        for _ in range(n):
+            if self._free:
+                pages.append(self._free.pop())
+            else:  # reclaim the least-recently-retained prefix page
+                p, _ = self._cached.popitem(last=False)
+                if self.on_cache_evict is not None:
+                    self.on_cache_evict(p)
+                pages.append(p)
+        for p in pages:
+            self._rc[p] = 1
+        return pages
+
+    def incref(self, pages: List[int]) -> None:
+        """Add one reference per page; revives cached (rc==0) pages."""
+        for p in pages:
+            if p in self._cached:
+                del self._cached[p]
+                self._rc[p] = 1
+            else:
+                self._rc[p] += 1
 
     def free(self, pages: List[int]) -> None:
-        self._free.extend(pages)
+        """Drop one reference per page; at refcount zero the page is
+        recycled, or retained in the cached tier when the owner asks."""
+        for p in pages:
+            rc = self._rc[p] - 1
+            if rc:
+                self._rc[p] = rc
+                continue
+            del self._rc[p]
+            if self.on_final_free is not None and self.on_final_free(p):
+                self._cached[p] = None  # retained, LRU tail
+            else:
+                self._free.append(p)
 
     def pages_needed(self, num_tokens: int) -> int:
         return (num_tokens + self.page_size - 1) // self.page_size

@@ -58,10 +58,11 @@ def test_random_interleavings_preserve_invariants(seed):
                 live.discard(out.request_id)
     assert not eng.has_work()
     assert not live, live
-    # all pages are either free or held by warm session caches
-    held = sum(len(s.pages) for s in eng.sessions.values())
-    assert eng.kv.num_free() + held == total_pages, (
-        eng.kv.num_free(), held, total_pages)
+    # all pages are free/retained or held by warm session caches (prefix
+    # sharing: a page held by several sessions counts once)
+    held = {p for s in eng.sessions.values() for p in s.pages}
+    assert eng.kv.num_free() + len(held) == total_pages, (
+        eng.kv.num_free(), len(held), total_pages)
 
 
 def test_staging_key_uniquely_identifies_page_assignment():
@@ -120,27 +121,43 @@ def test_kv_page_accounting_invariant_fuzz():
     n_added = 0
 
     def check(tag):
-        owned = []
+        from collections import Counter
+        owners = Counter()
         if hasattr(eng, "_scratch_page"):
-            owned.append(eng._scratch_page)
+            owners[eng._scratch_page] += 1
         for s in eng.seqs.values():
-            owned.extend(s.pages)
+            owners.update(s.pages)
         for s in eng.sessions.values():
-            owned.extend(s.pages)
-        owned.extend(eng.kv._free)
-        assert len(owned) == len(set(owned)), f"{tag}: double-owned page"
-        assert len(owned) == eng.kv.num_pages, (
-            f"{tag}: {eng.kv.num_pages - len(owned)} pages leaked")
+            owners.update(s.pages)
+        free = set(eng.kv._free)
+        cached = set(eng.kv._cached)
+        assert len(free) == len(eng.kv._free), f"{tag}: dup page in free list"
+        assert not (set(owners) & free), f"{tag}: page both owned and free"
+        assert not (set(owners) & cached), f"{tag}: page owned and cached"
+        assert not (free & cached), f"{tag}: page free and cached"
+        assert set(owners) | free | cached == set(range(eng.kv.num_pages)), (
+            f"{tag}: pages leaked")
+        # refcounts mirror ownership exactly (prefix sharing: rc = #owners)
+        assert dict(owners) == eng.kv._rc, f"{tag}: refcount drift"
+        # every retained page still has a live prefix-table identity
+        for pg in cached:
+            assert pg in eng._page_hash, f"{tag}: cached page without hash"
+
+    # fixed pool so many prompts share full 16-token pages -> the global
+    # prefix cache actually fires and the refcount invariant is exercised
+    common = [rnd.randrange(1, 500) for _ in range(96)]
 
     for it in range(400):
         op = rnd.random()
         if op < 0.45 and len(eng.seqs) < 8:
             n_added += 1
             sess = f"s{rnd.randrange(6)}" if rnd.random() < 0.5 else None
+            tail = [rnd.randrange(1, 500) for _ in range(rnd.randrange(1, 40))]
+            prompt = (common[:16 * rnd.randrange(0, 7)] + tail
+                      if rnd.random() < 0.6 else tail)
             try:
                 eng.add_request(
-                    f"r{n_added}",
-                    [rnd.randrange(1, 500) for _ in range(rnd.randrange(1, 120))],
+                    f"r{n_added}", prompt,
                     SamplingParams(max_tokens=rnd.randrange(1, 20),
                                    ignore_eos=True),
                     session_id=sess)
