@@ -258,8 +258,6 @@ class LLMEngine:
                 seq.num_cached = keep_pages * self.kv.page_size
             else:
                 self.kv.free(cached.pages)
-        if self.prefix_cache and seq.num_cached == 0:
-            self._attach_prefix(seq)
         self.seqs[request_id] = seq
         self.waiting.append(seq)
 
@@ -384,6 +382,12 @@ class LLMEngine:
         chunks: List[int] = []
         while self.waiting and len(self.running) + len(prefill) < self.config.max_num_seqs:
             seq = self.waiting.popleft()
+            # attach at SCHEDULE time, not admission: a burst of identical-
+            # prefix requests all arrives before any prefill has registered
+            # pages — by the time the 2nd..Nth are scheduled, the 1st has
+            # published the shared prefix
+            if self.prefix_cache and seq.num_cached == 0 and not seq.pages:
+                self._attach_prefix(seq)
             remaining = seq.prefill_target - seq.num_cached
             chunk = min(remaining, budget)
             if chunk <= 0:
@@ -392,6 +396,16 @@ class LLMEngine:
             # Admission never preempts running seqs (that thrashes); decode
             # growth below does the preempting. If nothing is running at all
             # and the seq still does not fit, it never will.
+            # Decode headroom: admitting a seq that leaves no room for the
+            # running batch to grow causes an immediate preemption storm
+            # (prefix sharing makes admission look cheap on tiny caches) —
+            # defer instead, unless nothing is running (progress guarantee).
+            if self.running or prefill:
+                need = self.kv.pages_needed(seq.num_cached + chunk) - len(seq.pages)
+                headroom = len(self.running) + len(prefill) + 1
+                if self.kv.num_free() < max(need, 0) + headroom:
+                    self.waiting.appendleft(seq)
+                    break
             reserved = self._try_reserve(seq, seq.num_cached + chunk)
             if not reserved:
                 self.waiting.appendleft(seq)

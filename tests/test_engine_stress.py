@@ -82,8 +82,10 @@ def test_staging_key_uniquely_identifies_page_assignment():
     for step in range(400):
         if live < 5 and rnd.random() < 0.5:
             rid = f"pk-{step}"
-            eng.add_request(rid, [5] * rnd.randint(3, 40),
-                            SamplingParams(max_tokens=rnd.randint(1, 30),
+            # short prompts admit cheaply; long decodes grow the running
+            # batch past the page budget so decode-growth preemption fires
+            eng.add_request(rid, [5] * rnd.randint(3, 10),
+                            SamplingParams(max_tokens=rnd.randint(30, 60),
                                            ignore_eos=True),
                             session_id=f"s-{step % 3}")
             live += 1

@@ -86,7 +86,10 @@ def test_session_cache_eviction_under_pressure():
                                  kv_cache_pages=10))
     p = SamplingParams(temperature=0.0, max_tokens=4, ignore_eos=True)
     for i in range(4):  # each session holds ~3 pages; 4 sessions > 12 pages
-        eng.add_request(f"r{i}", list(range(1, 40)), p, session_id=f"s{i}")
+        # distinct prompts: identical ones would legitimately all fit via
+        # cross-request prefix sharing (test_prefix_cache covers that)
+        eng.add_request(f"r{i}", list(range(1 + i * 7, 40 + i * 7)), p,
+                        session_id=f"s{i}")
         while eng.has_work():
             eng.step()
     assert len(eng.sessions) < 4  # oldest sessions were evicted
