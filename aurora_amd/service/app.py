@@ -731,6 +731,8 @@ def create_app(rt: Runtime) -> FastAPI:
             f"aurora_engine_steps_total{{kind=\"decode\"}} {eng.get('decode_steps', 0)}",
             "# TYPE aurora_engine_preemptions_total counter",
             f"aurora_engine_preemptions_total {eng.get('preemptions', 0)}",
+            "# TYPE aurora_prefix_cache_pages_total counter",
+            f"aurora_prefix_cache_pages_total {eng.get('prefix_cache_pages', 0)}",
             "# TYPE aurora_queue_depth gauge",
             f"aurora_queue_depth {rt.queue.pending_count()}",
             "# TYPE aurora_engine_healthy gauge",
