@@ -145,3 +145,49 @@ def test_prefix_cache_equivalence_fuzz(seed):
         assert (_run(eng, f"g{i}", prompt, max_tokens=mt)
                 == _run(ref, f"g{i}", prompt, max_tokens=mt)), (i, cut)
     assert eng.counters["prefix_cache_pages"] > 0  # sharing actually fired
+
+
+def test_abort_waiting_seq_with_attached_pages():
+    """A seq that attached shared pages but was never scheduled to run
+    (aborted from the waiting queue) must release its refs cleanly."""
+    eng = _mk(pages=32)
+    _run(eng, "warm", COMMON[:64] + [3])
+    p = SamplingParams(temperature=0.0, max_tokens=2, ignore_eos=True)
+    # long decode keeps the engine busy so "late" stays waiting after one step
+    eng.add_request("busy", COMMON[:64] + [5],
+                    SamplingParams(temperature=0.0, max_tokens=40,
+                                   ignore_eos=True))
+    eng.step()  # busy prefills (attaches); nothing else admitted yet
+    eng.add_request("late", COMMON[:64] + [6], p)
+    eng.step()  # late may attach at schedule time
+    eng.abort_request("late")
+    while eng.has_work():
+        eng.step()
+    # every page is free/retained or session-held; refcounts consistent
+    from collections import Counter
+    owners = Counter()
+    for s in eng.sessions.values():
+        owners.update(s.pages)
+    assert dict(owners) == eng.kv._rc
+    assert eng.kv.num_free() + len(set(owners)) == eng.kv.num_pages
+
+
+def test_preempted_seq_reattaches_cheaply():
+    """After preemption, a seq whose prompt pages are retained re-attaches
+    instead of recomputing the whole prefix (resume costs pages, not
+    prefill FLOPs)."""
+    eng = _mk(pages=14)
+    p_long = SamplingParams(temperature=0.0, max_tokens=60, ignore_eos=True)
+    eng.add_request("a", COMMON[:80] + [3], p_long)
+    eng.add_request("b", COMMON[:80] + [4], p_long)
+    outs = {}
+    while eng.has_work():
+        for o in eng.step():
+            outs.setdefault(o.request_id, []).extend(o.new_token_ids)
+    assert len(outs["a"]) == 60 and len(outs["b"]) == 60
+    # b shared a's prefix; any preemption churn stayed correct
+    ref = _mk(prefix_cache=False, pages=14)
+    ra = _run(ref, "a", COMMON[:80] + [3], max_tokens=60)
+    ref2 = _mk(prefix_cache=False, pages=14)
+    rb = _run(ref2, "b", COMMON[:80] + [4], max_tokens=60)
+    assert outs["a"] == ra and outs["b"] == rb
