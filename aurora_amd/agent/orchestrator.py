@@ -199,8 +199,10 @@ class Orchestrator:
         table (findings_writer.py)."""
         brief = ROLES[role]["brief"]
         subset = ROLES[role]["tools"] + ["final_answer"]
+        # brief lives in the system override only — duplicating it in the
+        # question would re-prefill it per turn as history
         sub_state = State(
-            question=f"[role: {role}] {brief}\nIncident: {state.question}",
+            question=f"[role: {role}] Incident: {state.question}",
             user_id=state.user_id, org_id=state.org_id,
             session_id=None,  # sub-agents keep their own scratch context
             incident_id=state.incident_id, is_background=True)
