@@ -363,6 +363,17 @@ class ReActTurnGrammar:
                 self.tool.nbytes += 1  # keep its soft-close budget live
             return ok
         # thought phase: every byte is legal; track the transition literal
+        if self.nbytes >= self.thought_bytes:
+            # over budget the steering is STRICT: a multi-byte token whose
+            # tail diverges from the literal must be rejected outright —
+            # under a BPE tokenizer a permissive mismatch would keep
+            # resetting the match and the turn would never transition
+            if b == self.LIT[self.lit_pos]:
+                self.lit_pos += 1
+                if self.lit_pos == len(self.LIT):
+                    self.phase = 1
+                return True
+            return False
         if b == self.LIT[self.lit_pos]:
             self.lit_pos += 1
             if self.lit_pos == len(self.LIT):

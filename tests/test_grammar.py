@@ -175,11 +175,26 @@ def test_react_turn_grammar_shapes():
 def test_react_turn_grammar_soft_transition():
     from aurora_amd.engine.grammar import ReActTurnGrammar
     g = ReActTurnGrammar(["kubectl"], thought_bytes=10)
-    assert g.advance(b"0123456789extra")  # over budget, still legal
+    assert g.advance(b"0123456789")  # thought up to the budget
     allowed = g.allowed_next_bytes()
     assert allowed == b"\n"  # steered onto the literal
+    # past the budget the steering is STRICT: diverging bytes are rejected
+    # (a BPE token's tail would otherwise reset the literal match forever)
+    assert not g.fork().advance(b"extra")
+    assert not g.fork().advance(b"\nAx")  # partial literal then divergence
     assert g.advance(b"\n")
     assert g.allowed_next_bytes() == b"A"
+    assert g.advance(b"Action:")
+    assert g.phase == 1
+
+
+def test_react_turn_budget_crossing_mid_token():
+    """A multi-byte token that starts under the budget and crosses it is
+    legal while it keeps matching the literal."""
+    from aurora_amd.engine.grammar import ReActTurnGrammar
+    g = ReActTurnGrammar(["kubectl"], thought_bytes=6)
+    assert g.advance(b"1234\nAction:")  # crosses the budget inside the literal
+    assert g.phase == 1
 
 
 def test_react_turn_grammar_engine_end_to_end():

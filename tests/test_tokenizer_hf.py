@@ -139,3 +139,29 @@ def test_full_rca_pipeline_with_bpe_tokenizer(bpe_path, tmp_path):
         rt.engine.shutdown()
 
     asyncio.run(main())
+
+
+def test_engine_fused_react_turn_with_bpe(bpe_path):
+    """The fused thought+action grammar under a multi-byte tokenizer:
+    every output must split at '\\nAction:' and parse as a tool call."""
+    from aurora_amd.engine.engine import EngineConfig, LLMEngine
+    from aurora_amd.engine.grammar import ReActTurnGrammar
+    from aurora_amd.engine.sampling_params import SamplingParams
+
+    t = HFTokenizer(bpe_path)
+    eng = LLMEngine(EngineConfig(model="tiny-llama", device="cpu", seed=7),
+                    tokenizer=t)
+    outs = eng.generate(
+        [t.encode("investigate latency spike"), t.encode("check pods")],
+        SamplingParams(temperature=0.6, max_tokens=200,
+                       grammar=ReActTurnGrammar(
+                           ["kubectl", "final_answer"], thought_bytes=60,
+                           tool_soft_close=50)))
+    for toks in outs:
+        text = b"".join(t.token_bytes(i) for i in toks).decode(
+            "utf-8", errors="replace")
+        head, sep, tail = text.rpartition("\nAction:")
+        assert sep, repr(text[-120:])
+        doc = json.loads(tail.strip())
+        assert doc["tool"] in ("kubectl", "final_answer")
+    assert eng.counters["grammar_fallbacks"] == 0
