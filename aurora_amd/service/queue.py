@@ -11,6 +11,7 @@ from __future__ import annotations
 
 import asyncio
 import json
+import sqlite3
 import time
 import traceback
 import uuid
@@ -66,8 +67,10 @@ class TaskQueue:
                 "max_attempts,created_at,updated_at) VALUES (?,?,?,?,?,?,?,?,?,?)",
                 [tid, name, org_id, json.dumps(args), "pending", dedup_key,
                  t + countdown_s, max_attempts, t, t])
-        except Exception:
+        except sqlite3.IntegrityError:
             return None  # dedup hit (UNIQUE dedup_key) — reference task.py:490
+            # (other DB errors must RAISE: swallowing them would silently
+            # drop the task)
         self._wake.set()
         return tid
 
