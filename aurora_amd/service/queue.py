@@ -115,10 +115,15 @@ class TaskQueue:
                             "updated_at=? WHERE id=?",
                             [f"no handler for {name}", t, row["id"]])
             return
+        args = json.loads(row["args"] or "{}")
+        # handlers that re-raise transient faults for retry can see whether
+        # this is the LAST attempt and fail terminally instead of stranding
+        # their domain state (e.g. an incident left 'pending' forever)
+        args["_task_meta"] = {"attempt": row["attempts"],
+                              "max_attempts": row["max_attempts"]}
         try:
             result = await asyncio.wait_for(
-                fn(json.loads(row["args"] or "{}")),
-                timeout=self.default_time_limit)
+                fn(args), timeout=self.default_time_limit)
             self.db.execute(
                 "UPDATE task_queue SET status='done', result=?, updated_at=? WHERE id=?",
                 [json.dumps(result)[:4000] if result is not None else None,

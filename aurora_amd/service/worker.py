@@ -117,15 +117,23 @@ async def run_background_chat(rt, args: Dict[str, Any]) -> dict:
     except Exception as e:  # noqa: BLE001
         msg = str(e)
         if "engine step failed" in msg or "engine shutdown" in msg:
-            # transient engine fault: re-raise so the queue retries with
-            # backoff (semantic failures below stay terminal). Status must
-            # leave 'running' FIRST or the retry dedups against our own
-            # crashed attempt and the incident strands (found by the
-            # fault-injection test; Celery acks-late redelivers fresh).
-            store.update_incident(incident_id, aurora_status="pending")
-            store.lifecycle(incident_id, "rca_engine_fault", {"error": msg})
-            raise
-        error = f"workflow failure: {e}"
+            meta = args.get("_task_meta") or {}
+            if meta.get("attempt", 1) < meta.get("max_attempts", 3):
+                # transient engine fault: re-raise so the queue retries with
+                # backoff (semantic failures below stay terminal). Status
+                # must leave 'running' FIRST or the retry dedups against our
+                # own crashed attempt and the incident strands (found by the
+                # fault-injection test; Celery acks-late redelivers fresh).
+                store.update_incident(incident_id, aurora_status="pending")
+                store.lifecycle(incident_id, "rca_engine_fault", {"error": msg})
+                raise
+            # LAST attempt: fail terminally so the incident is never left
+            # 'pending' with no retry coming (the stale-investigation
+            # reaper is the backstop, not the primary path)
+            store.lifecycle(incident_id, "rca_engine_fault_final", {"error": msg})
+            error = f"engine fault (retries exhausted): {e}"
+        else:
+            error = f"workflow failure: {e}"
 
     # ---- post-RCA stage ----
     if error and not final:

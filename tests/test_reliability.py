@@ -251,10 +251,14 @@ def test_chaos_fault_in_post_rca_tail_still_completes():
     from aurora_amd.utils.config import Settings
 
     class TailFaultServer(AsyncEngineServer):
-        """Healthy for the workflow's stream(); every complete_text call
-        (the post-RCA passes) faults like a dead engine."""
+        """Healthy for the workflow (stream + its forced-final/compression
+        complete_text calls, which carry the investigation prompt); every
+        post-RCA pass faults like a dead engine."""
 
         async def complete_text(self, prompt, params, session_id=None):
+            if "# Investigation so far" in prompt:  # workflow-shaped
+                return await super().complete_text(prompt, params,
+                                                   session_id=session_id)
             raise RuntimeError("engine step failed: injected tail fault")
 
     async def main():
@@ -286,6 +290,58 @@ def test_chaos_fault_in_post_rca_tail_still_completes():
         assert "tail fault alert" in (inc["summary"] or "")
         assert store.suggestions(iid) == []
         assert store.get_postmortem(iid) is not None
+        await rt.stop()
+        srv.shutdown()
+
+    asyncio.run(main())
+
+
+def test_chaos_persistent_engine_fault_terminates_as_failed():
+    """An engine that NEVER recovers: after the queue's retries are
+    exhausted the incident must end 'failed' — not strand 'pending'
+    waiting for a retry that will never come (worker.py _task_meta)."""
+    import asyncio
+
+    from aurora_amd.engine.engine import EngineConfig, StepOutput
+    from aurora_amd.engine.server import AsyncEngineServer
+    from aurora_amd.service.runtime import Runtime
+    from aurora_amd.service.webhooks import process_webhook
+    from aurora_amd.utils.config import Settings
+
+    class DeadServer(AsyncEngineServer):
+        async def stream(self, ids, params, session_id=None, request_id=None):
+            yield StepOutput(request_id or "x", [], "", True,
+                             "engine step failed: permanently dead")
+
+    async def main():
+        srv = DeadServer(EngineConfig(model="tiny-llama", device="cpu",
+                                      seed=1))
+        cfg = Settings(db_path=":memory:",
+                       secrets_path="/tmp/chaos_dead_sec.json")
+        cfg.worker_concurrency = 2
+        rt = Runtime(cfg, engine_server=srv)
+        oid = rt.admin.create_org("dead")
+        rt.start()
+        iid, created = process_webhook(
+            rt, oid, "pagerduty",
+            {"event": {"data": {"id": "d1", "title": "dead engine alert",
+                                "service": {"summary": "svc"}}}},
+            rca_countdown_s=0.0,
+            rca_args={"max_turns": 2,
+                      "tool_subset": ["list_incidents", "final_answer"]})
+        assert created
+        store = rt.org(oid)
+        for _ in range(600):
+            inc = store.get_incident(iid)
+            if inc["aurora_status"] in ("completed", "failed"):
+                break
+            await asyncio.sleep(0.1)
+        inc = store.get_incident(iid)
+        assert inc["aurora_status"] == "failed", inc["aurora_status"]
+        events = [e["event"] for e in store.lifecycle_events(iid)]
+        assert "rca_engine_fault" in events          # retried
+        assert "rca_engine_fault_final" in events    # then terminal
+        assert "dead engine alert" in (inc["summary"] or "")
         await rt.stop()
         srv.shutdown()
 
