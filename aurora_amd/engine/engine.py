@@ -244,22 +244,31 @@ class LLMEngine:
         if params.max_tokens > room:
             params = dataclasses.replace(params, max_tokens=max(room, 1))
         seq = Sequence(request_id, prompt_token_ids, params, session_id)
-        if session_id is not None and session_id in self.sessions:
-            cached = self.sessions.pop(session_id)
-            match = 0
-            limit = min(cached.num_cached, seq.num_prompt - 1)
-            while match < limit and cached.token_ids[match] == seq.token_ids[match]:
-                match += 1
-            # adopt whole pages covered by the matching prefix
-            keep_pages = match // self.kv.page_size
-            if keep_pages > 0:
-                seq.pages = cached.pages[:keep_pages]
-                self.kv.free(cached.pages[keep_pages:])
-                seq.num_cached = keep_pages * self.kv.page_size
-            else:
-                self.kv.free(cached.pages)
         self.seqs[request_id] = seq
         self.waiting.append(seq)
+
+    def _adopt_session(self, seq: Sequence) -> None:
+        """Adopt the warm session KV at SCHEDULE time, not admission: a
+        waiting seq must never hold pages (it can't be preempted, so
+        admission-time adoption let queued turn-N+1 requests hoard the
+        cache and livelock the running seq — pages stay in the evictable
+        session tier until this seq actually runs)."""
+        sid = seq.session_id
+        if sid is None or sid not in self.sessions:
+            return
+        cached = self.sessions.pop(sid)
+        match = 0
+        limit = min(cached.num_cached, seq.num_prompt - 1)
+        while match < limit and cached.token_ids[match] == seq.token_ids[match]:
+            match += 1
+        # adopt whole pages covered by the matching prefix
+        keep_pages = match // self.kv.page_size
+        if keep_pages > 0:
+            seq.pages = cached.pages[:keep_pages]
+            self.kv.free(cached.pages[keep_pages:])
+            seq.num_cached = keep_pages * self.kv.page_size
+        else:
+            self.kv.free(cached.pages)
 
     # ------------------------------------------------- prefix cache (global)
 
@@ -382,12 +391,15 @@ class LLMEngine:
         chunks: List[int] = []
         while self.waiting and len(self.running) + len(prefill) < self.config.max_num_seqs:
             seq = self.waiting.popleft()
-            # attach at SCHEDULE time, not admission: a burst of identical-
-            # prefix requests all arrives before any prefill has registered
-            # pages — by the time the 2nd..Nth are scheduled, the 1st has
-            # published the shared prefix
-            if self.prefix_cache and seq.num_cached == 0 and not seq.pages:
-                self._attach_prefix(seq)
+            # session adoption + prefix attach at SCHEDULE time, not
+            # admission: waiting seqs must hold no pages, and a burst of
+            # identical-prefix requests all arrives before any prefill has
+            # registered pages — by the time the 2nd..Nth are scheduled,
+            # the 1st has published the shared prefix
+            if seq.num_cached == 0 and not seq.pages:
+                self._adopt_session(seq)
+                if self.prefix_cache and seq.num_cached == 0 and not seq.pages:
+                    self._attach_prefix(seq)
             remaining = seq.prefill_target - seq.num_cached
             chunk = min(remaining, budget)
             if chunk <= 0:

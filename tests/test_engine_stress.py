@@ -219,3 +219,43 @@ def test_session_reuse_equivalence_fuzz():
                 got.extend(o.new_token_ids)
         fresh = LLMEngine(cfg).generate([prompt], p)[0]
         assert got == fresh, (it, sess, len(prompt))
+
+
+def test_turn_burst_oversubscribed_sessions_no_livelock():
+    """Turn-2 burst: 8 sessions each re-submitting an extended prompt on a
+    cache that fits only a few at once. Admission-time session adoption
+    let the WAITING seqs hoard pages (they are never preemptable), so the
+    one running seq could never grow and the engine preempt-thrashed
+    forever. Schedule-time adoption keeps the pages in the evictable
+    session tier; every request must finish."""
+    import time
+
+    import torch
+
+    from aurora_amd.engine.engine import EngineConfig, LLMEngine
+    from aurora_amd.engine.sampling_params import SamplingParams
+
+    eng = LLMEngine(EngineConfig(model="tiny-llama", device="cpu",
+                                 max_num_seqs=10, max_batched_tokens=16384))
+    rng = torch.Generator().manual_seed(1)
+    p = SamplingParams(temperature=0.7, top_p=0.95, max_tokens=128,
+                       ignore_eos=True)
+    ctx = {}
+    outs = {}
+    for i in range(8):
+        ctx[i] = torch.randint(0, 250, (768,), generator=rng).tolist()
+        eng.add_request(f"t1-{i}", ctx[i], p, session_id=f"inc-{i}")
+    while eng.has_work():
+        for o in eng.step():
+            outs.setdefault(o.request_id, []).extend(o.new_token_ids)
+    for i in range(8):  # extended turn-2 prompts, all queued at once
+        ctx[i].extend(outs[f"t1-{i}"][-128:])
+        ctx[i].extend(torch.randint(0, 250, (512,), generator=rng).tolist())
+        eng.add_request(f"t2-{i}", ctx[i], p, session_id=f"inc-{i}")
+    done = 0
+    t0 = time.time()
+    while eng.has_work():
+        assert time.time() - t0 < 120, (
+            f"livelock: {done}/8 done, {eng.counters['preemptions']} preemptions")
+        done += sum(1 for o in eng.step() if o.finished)
+    assert done == 8
