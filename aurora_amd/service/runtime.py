@@ -86,6 +86,7 @@ class Runtime:
         self._encoder = None  # None = unprobed, False = unavailable (CPU)
         self._embedding_rail = None
         self._graph: Dict[str, InfraGraph] = {}
+        self._correlators: Dict[str, Any] = {}
         self._rca_rate: Dict[str, float] = {}  # user -> last background RCA ts
         self.kubectl_agents: Dict[str, Any] = {}  # cluster_id -> AgentConn
         from ..connectors.oauth import StateCache
@@ -154,6 +155,22 @@ class Runtime:
         if org_id not in self._graph:
             self._graph[org_id] = InfraGraph(self.org(org_id))
         return self._graph[org_id]
+
+    def correlator(self, org_id: str):
+        """Per-org alert correlator (holds the per-incident embedding
+        cache across webhooks). Embedding similarity rides the shared GPU
+        encoder when opted in (AURORA_CORRELATOR_EMBED=1); default stays
+        lexical+topology so the webhook hot path never blocks on encoder
+        work."""
+        if org_id not in self._correlators:
+            import os as _os
+            from .correlation import AlertCorrelator
+            rag = (self.rag(org_id)
+                   if _os.environ.get("AURORA_CORRELATOR_EMBED", "0") == "1"
+                   else None)
+            self._correlators[org_id] = AlertCorrelator(
+                self.org(org_id), graph=self.graph(org_id), rag=rag)
+        return self._correlators[org_id]
 
     def embedding_rail(self):
         """Shared prompt-injection embedding classifier (lazy singleton;

@@ -242,13 +242,7 @@ def process_webhook(rt, org_id: str, source: str, payload: dict,
     if norm.get("external_id") is not None:
         norm["external_id"] = str(norm["external_id"])[:200]
     store = rt.org(org_id)
-    # embedding similarity rides the shared GPU encoder when opted in
-    # (AURORA_CORRELATOR_EMBED=1); default stays lexical+topology so the
-    # webhook hot path never blocks on encoder work
-    import os as _os
-    rag = (rt.rag(org_id)
-           if _os.environ.get("AURORA_CORRELATOR_EMBED", "0") == "1" else None)
-    corr = AlertCorrelator(store, graph=rt.graph(org_id), rag=rag)
+    corr = rt.correlator(org_id)
     enriched = dict(payload or {}, **{k: v for k, v in norm.items() if v})
     iid, created = corr.handle_alert(source, norm["title"], enriched,
                                      fingerprint=norm.get("fingerprint", ""),

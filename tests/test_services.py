@@ -476,11 +476,14 @@ def test_runtime_serves_checkpoint_and_tokenizer(tmp_path):
         from aurora_amd.engine.tokenizer import HFTokenizer
         assert isinstance(eng.tokenizer, HFTokenizer)
         p = SamplingParams(temperature=0.0, max_tokens=5, ignore_eos=True)
-        got = eng.generate([[4, 5, 6]], p)
+        # generate through the SERVER api — the engine is owned by the
+        # server's stepping thread; stepping it directly would race
+        import asyncio as _aio
+        res = _aio.run(rt.engine.generate([4, 5, 6], p))
         from aurora_amd.engine.engine import EngineConfig, LLMEngine
         ref = LLMEngine(EngineConfig(model="tiny-llama", device="cpu"),
                         model=src, tokenizer=eng.tokenizer)
-        assert got == ref.generate([[4, 5, 6]], p)
+        assert res.token_ids == ref.generate([[4, 5, 6]], p)[0]
     finally:
         rt.engine.shutdown()
 
