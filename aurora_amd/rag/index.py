@@ -30,36 +30,40 @@ def _tokens(text: str) -> List[str]:
 
 
 class BM25:
+    """Inverted-index BM25: per-term postings make a query cost the sum of
+    its terms' posting lists, not O(corpus) per term."""
+
     def __init__(self, k1: float = 1.2, b: float = 0.75):
         self.k1, self.b = k1, b
-        self.df: Counter = Counter()
-        self.doc_terms: List[Counter] = []
+        self.postings: Dict[str, Dict[int, int]] = defaultdict(dict)
         self.doc_len: List[int] = []
+        self._len_sum = 0
 
     def add(self, text: str) -> None:
+        i = len(self.doc_len)
         terms = Counter(_tokens(text))
-        self.doc_terms.append(terms)
-        self.doc_len.append(sum(terms.values()))
-        for t in terms:
-            self.df[t] += 1
+        n = sum(terms.values())
+        self.doc_len.append(n)
+        self._len_sum += n
+        for t, f in terms.items():
+            self.postings[t][i] = f
 
     def scores(self, query: str) -> Dict[int, float]:
         q = _tokens(query)
-        N = len(self.doc_terms)
+        N = len(self.doc_len)
         if N == 0:
             return {}
-        avg = sum(self.doc_len) / N
+        avg = self._len_sum / N
         out: Dict[int, float] = defaultdict(float)
         for term in q:
-            n = self.df.get(term, 0)
-            if n == 0:
+            plist = self.postings.get(term)
+            if not plist:
                 continue
+            n = len(plist)
             idf = math.log(1 + (N - n + 0.5) / (n + 0.5))
-            for i, terms in enumerate(self.doc_terms):
-                f = terms.get(term, 0)
-                if f:
-                    out[i] += idf * f * (self.k1 + 1) / (
-                        f + self.k1 * (1 - self.b + self.b * self.doc_len[i] / avg))
+            for i, f in plist.items():
+                out[i] += idf * f * (self.k1 + 1) / (
+                    f + self.k1 * (1 - self.b + self.b * self.doc_len[i] / avg))
         return out
 
 
