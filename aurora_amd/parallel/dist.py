@@ -34,8 +34,12 @@ def env_local_rank() -> int:
 
 
 def init_distributed(tp_size: int = 1, backend: Optional[str] = None,
-                     timeout_s: float = 600.0) -> None:
+                     timeout_s: float = 1800.0) -> None:
     """Initialise the default process group and the TP subgroup.
+
+    timeout 1800 s: the e2e bench's DP barrier is only reached after a
+    full multi-minute step per rank; rank skew must not trip the
+    collective watchdog at N=8.
 
     World layout: ranks are grouped into contiguous TP groups of size
     ``tp_size``; ranks in different groups are DP replicas.  xGMI note: TP
