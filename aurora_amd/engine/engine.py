@@ -248,11 +248,12 @@ class LLMEngine:
         self.waiting.append(seq)
 
     def _adopt_session(self, seq: Sequence) -> None:
-        """Adopt the warm session KV at SCHEDULE time, not admission: a
-        waiting seq must never hold pages (it can't be preempted, so
-        admission-time adoption let queued turn-N+1 requests hoard the
-        cache and livelock the running seq — pages stay in the evictable
-        session tier until this seq actually runs)."""
+        """Adopt the warm session KV at SCHEDULE time, not admission:
+        admission-time adoption let every queued turn-N+1 request hoard
+        unpreemptable pages and livelock the running seq. Pages stay in
+        the evictable session tier until a seq is actually considered;
+        at most ONE deferred seq holds adopted pages at a time (the
+        headroom guard breaks the admission loop right after it)."""
         sid = seq.session_id
         if sid is None or sid not in self.sessions:
             return
