@@ -167,3 +167,41 @@ def test_every_provider_tool_is_connection_gated():
     for name in provider_tools:
         t = GLOBAL_REGISTRY.get(name)
         assert t.requires_connection in known, (name, t.requires_connection)
+
+
+def test_every_skill_and_role_file_parses():
+    """All skills/roles markdown must carry valid YAML frontmatter and a
+    non-empty body — a malformed file silently vanishing from the prompt
+    would be invisible at runtime."""
+    import os
+
+    import yaml
+
+    import aurora_amd
+    base = os.path.dirname(aurora_amd.__file__)
+    n = 0
+    for root in (os.path.join(base, "agent", "skills", "library"),
+                 os.path.join(base, "agent", "roles")):
+        for dirpath, _, files in os.walk(root):
+            for fn in files:
+                if not fn.endswith(".md"):
+                    continue
+                raw = open(os.path.join(dirpath, fn)).read()
+                assert raw.startswith("---"), fn
+                _, fm, body = raw.split("---", 2)
+                meta = yaml.safe_load(fm)
+                assert meta.get("name"), fn
+                assert len(body.strip()) > 40, fn
+                n += 1
+    assert n >= 41  # 35 skills + 6 roles
+
+    from aurora_amd.agent.skills.registry import get_skill_registry
+    reg = get_skill_registry()
+    assert len(reg.skills) >= 35
+    # connection-gated skills must be resolvable against the catalog
+    from aurora_amd.connectors.registry import connector_catalog
+    providers = {c["name"] for c in connector_catalog()}
+    for sk in reg.skills.values():
+        if sk.requires_connection:
+            assert sk.requires_connection in providers, (
+                sk.name, sk.requires_connection)
