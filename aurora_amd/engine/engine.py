@@ -236,11 +236,20 @@ class LLMEngine:
             # fail at the API boundary, not as an embedding IndexError deep
             # inside the stepping thread
             raise ValueError(f"prompt token id outside vocab [0, {v})")
-        if len(prompt_token_ids) >= self.max_model_len:
-            prompt_token_ids = prompt_token_ids[-(self.max_model_len - params.max_tokens - 1):]
-        # clamp so prompt + output can never exceed max_model_len (the
+        # hard length ceiling: the model's positions AND the cache itself —
+        # a seq whose prompt+max_tokens exceeds total KV capacity can never
+        # complete and would preempt-cycle forever (found by the seed sweep)
+        reserved = 1 if self._graphs_enabled() else 0  # scratch page
+        capacity = max((self.kv.num_pages - reserved) * self.kv.page_size,
+                       self.kv.page_size)
+        hard = min(self.max_model_len, capacity)
+        if params.max_tokens >= hard:
+            params = dataclasses.replace(params, max_tokens=max(hard - 1, 1))
+        if len(prompt_token_ids) >= hard:
+            prompt_token_ids = prompt_token_ids[-max(hard - params.max_tokens - 1, 1):]
+        # clamp so prompt + output can never exceed the ceiling (the
         # per-step max_len check is the backstop; this keeps max_tokens honest)
-        room = self.max_model_len - len(prompt_token_ids)
+        room = hard - len(prompt_token_ids)
         if params.max_tokens > room:
             params = dataclasses.replace(params, max_tokens=max(room, 1))
         seq = Sequence(request_id, prompt_token_ids, params, session_id)
@@ -354,11 +363,28 @@ class LLMEngine:
             self.kv.free(old.pages)
             old.pages = []
 
+    def _release_waiting_pages(self, keep: Sequence) -> None:
+        """Strip adopted/attached pages from WAITING seqs (they re-adopt or
+        re-attach when scheduled). A deferred holder's pages must never cap
+        a running seq below its need — that is a livelock (a waiting seq is
+        not preemptable, so nothing else can reclaim them)."""
+        for w in self.waiting:
+            if w is keep or not w.pages:
+                continue
+            self.kv.free(w.pages)
+            w.pages = []
+            w.num_cached = 0
+            w.reg_pages = 0
+            w.prefix_hash = b""
+            w.page_gen += 1
+
     def _try_reserve(self, seq: Sequence, up_to_tokens: int) -> bool:
         need = self.kv.pages_needed(up_to_tokens) - len(seq.pages)
         if need <= 0:
             return True
         self._evict_sessions(need)
+        if self.kv.num_free() < need:
+            self._release_waiting_pages(keep=seq)
         if self.kv.num_free() < need:
             return False
         seq.pages.extend(self.kv.alloc(need))

@@ -259,3 +259,29 @@ def test_turn_burst_oversubscribed_sessions_no_livelock():
             f"livelock: {done}/8 done, {eng.counters['preemptions']} preemptions")
         done += sum(1 for o in eng.step() if o.finished)
     assert done == 8
+
+
+def test_request_exceeding_cache_capacity_is_clamped_and_completes():
+    """A request whose prompt+max_tokens exceeds TOTAL KV capacity can
+    never finish as asked — it used to preempt-cycle forever (found by a
+    seed sweep). Admission must clamp to cache capacity so the request
+    completes with what the cache can actually hold."""
+    from aurora_amd.engine.engine import EngineConfig, LLMEngine
+    from aurora_amd.engine.sampling_params import SamplingParams
+
+    eng = LLMEngine(EngineConfig(model="tiny-llama", device="cpu", seed=2,
+                                 kv_cache_pages=10, max_model_len=256))
+    cap = 10 * eng.kv.page_size  # 160 token slots total
+    # prompt 144 + max_tokens 24 = 168 > 160: the old engine livelocked
+    eng.add_request("big", [5] * 144,
+                    SamplingParams(max_tokens=24, ignore_eos=True))
+    seq = eng.seqs["big"]
+    assert seq.num_prompt + seq.params.max_tokens <= cap
+    out = []
+    for _ in range(500):
+        for o in eng.step():
+            out.extend(o.new_token_ids)
+        if not eng.has_work():
+            break
+    assert not eng.has_work()
+    assert len(out) == seq.params.max_tokens > 0
