@@ -218,6 +218,7 @@ class LLMEngine:
             "prompt_tokens": 0, "generated_tokens": 0, "prefill_steps": 0,
             "decode_steps": 0, "preemptions": 0, "grammar_fallbacks": 0,
             "grammar_bucket_picks": 0, "prefix_cache_pages": 0,
+            "steps_since_finish": 0,
         }
         # phase timers (seconds, cumulative) for perf triage
         self.timers = {"schedule": 0.0, "build": 0.0, "forward": 0.0,
@@ -506,6 +507,10 @@ class LLMEngine:
         self.timers["schedule"] += time.monotonic() - _t0
         if not seqs:
             return []
+        # thrash indicator: steps since ANY request finished. Steady growth
+        # alongside climbing preemptions is the signature of the livelock
+        # class fixed this round — surfaced so ops can see it, never acted on
+        self.counters["steps_since_finish"] += 1
         dev = self.device
         if mode == "mixed":
             self._dec_staged_ids = None
@@ -979,6 +984,7 @@ class LLMEngine:
         return outputs
 
     def _finish(self, seq: Sequence, reason: str) -> None:
+        self.counters["steps_since_finish"] = 0
         seq.status = FINISHED
         seq.finish_reason = reason
         if seq in self.running:
